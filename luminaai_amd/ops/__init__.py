@@ -1,0 +1,20 @@
+"""Fused op layer: hand-written CDNA4 (gfx950) HIP kernels with autograd,
+plus pure-PyTorch CPU reference implementations."""
+
+from . import reference
+from .interface import (
+    adamw_step,
+    fused_cross_entropy,
+    get_ext,
+    has_ext,
+    l2norm_sq,
+    rmsnorm,
+    rope,
+    swiglu,
+)
+from .reference import rope_cache
+
+__all__ = [
+    "adamw_step", "fused_cross_entropy", "get_ext", "has_ext", "l2norm_sq",
+    "reference", "rmsnorm", "rope", "rope_cache", "swiglu",
+]

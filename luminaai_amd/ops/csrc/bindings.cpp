@@ -1,0 +1,272 @@
+// PyTorch-ROCm bindings for the LuminaAI-AMD HIP kernels (gfx950).
+// Host-side code only; kernels live in the *.hip translation units and are
+// reached through extern "C" launchers taking hipStream_t.
+#include <torch/extension.h>
+#include <c10/hip/HIPStream.h>
+#include <hip/hip_runtime.h>
+
+#include <optional>
+#include <tuple>
+
+namespace {
+
+hipStream_t cur_stream() {
+  return c10::hip::getCurrentHIPStream().stream();
+}
+
+void check_hip(hipError_t e, const char* what) {
+  TORCH_CHECK(e == hipSuccess, what, " failed: ", hipGetErrorString(e));
+}
+
+bool is_bf16(const at::Tensor& t) { return t.scalar_type() == at::kBFloat16; }
+
+void check_dtype(const at::Tensor& t, const char* name) {
+  TORCH_CHECK(t.scalar_type() == at::kBFloat16 || t.scalar_type() == at::kFloat,
+              name, " must be bf16 or fp32, got ", t.scalar_type());
+}
+
+}  // namespace
+
+// ---- launchers from the .hip TUs ----------------------------------------
+extern "C" {
+hipError_t lumina_rmsnorm_fwd_bf16(const void*, const void*, void*, float*, int64_t, int, float, hipStream_t);
+hipError_t lumina_rmsnorm_fwd_f32(const void*, const void*, void*, float*, int64_t, int, float, hipStream_t);
+hipError_t lumina_rmsnorm_bwd_bf16(const void*, const void*, const void*, const float*, void*, float*, int64_t, int, hipStream_t);
+hipError_t lumina_rmsnorm_bwd_f32(const void*, const void*, const void*, const float*, void*, float*, int64_t, int, hipStream_t);
+hipError_t lumina_rope_bf16(const void*, const void*, void*, void*, const float*, const float*, const int*, int64_t, int, int, int, int, int, int, hipStream_t);
+hipError_t lumina_rope_f32(const void*, const void*, void*, void*, const float*, const float*, const int*, int64_t, int, int, int, int, int, int, hipStream_t);
+hipError_t lumina_swiglu_fwd_bf16(const void*, const void*, void*, int64_t, int, int64_t, int64_t, hipStream_t);
+hipError_t lumina_swiglu_fwd_f32(const void*, const void*, void*, int64_t, int, int64_t, int64_t, hipStream_t);
+hipError_t lumina_swiglu_bwd_bf16(const void*, const void*, const void*, void*, void*, int64_t, int, int64_t, int64_t, hipStream_t);
+hipError_t lumina_swiglu_bwd_f32(const void*, const void*, const void*, void*, void*, int64_t, int, int64_t, int64_t, hipStream_t);
+hipError_t lumina_ce_fwd_bf16(const void*, const int32_t*, const float*, float*, float*, int64_t, int, int, hipStream_t);
+hipError_t lumina_ce_fwd_f32(const void*, const int32_t*, const float*, float*, float*, int64_t, int, int, hipStream_t);
+hipError_t lumina_ce_bwd_bf16(const void*, const int32_t*, const float*, const float*, const float*, const float*, void*, int64_t, int, int, hipStream_t);
+hipError_t lumina_ce_bwd_f32(const void*, const int32_t*, const float*, const float*, const float*, const float*, void*, int64_t, int, int, hipStream_t);
+hipError_t lumina_l2norm_sq_bf16(const void*, int64_t, float*, hipStream_t);
+hipError_t lumina_l2norm_sq_f32(const void*, int64_t, float*, hipStream_t);
+hipError_t lumina_adamw_step(float*, const void*, int, float*, float*, void*, int, int64_t, float, float, float, float, float, float, float, const float*, float, float, hipStream_t);
+}
+
+// ---- RMSNorm -------------------------------------------------------------
+std::tuple<at::Tensor, at::Tensor> rmsnorm_fwd(const at::Tensor& x,
+                                               const at::Tensor& w,
+                                               double eps, bool save_inv) {
+  TORCH_CHECK(x.is_contiguous() && w.is_contiguous());
+  check_dtype(x, "x");
+  TORCH_CHECK(x.scalar_type() == w.scalar_type(), "x/w dtype mismatch");
+  const int H = (int)x.size(-1);
+  const int64_t N = x.numel() / H;
+  auto y = at::empty_like(x);
+  auto inv = save_inv
+      ? at::empty({N}, x.options().dtype(at::kFloat))
+      : at::Tensor();
+  float* invp = save_inv ? inv.data_ptr<float>() : nullptr;
+  if (is_bf16(x))
+    check_hip(lumina_rmsnorm_fwd_bf16(x.data_ptr(), w.data_ptr(), y.data_ptr(),
+                                      invp, N, H, (float)eps, cur_stream()),
+              "rmsnorm_fwd_bf16");
+  else
+    check_hip(lumina_rmsnorm_fwd_f32(x.data_ptr(), w.data_ptr(), y.data_ptr(),
+                                     invp, N, H, (float)eps, cur_stream()),
+              "rmsnorm_fwd_f32");
+  return {y, inv};
+}
+
+std::tuple<at::Tensor, at::Tensor> rmsnorm_bwd(const at::Tensor& gy,
+                                               const at::Tensor& x,
+                                               const at::Tensor& w,
+                                               const at::Tensor& inv) {
+  TORCH_CHECK(gy.is_contiguous() && x.is_contiguous() && w.is_contiguous());
+  const int H = (int)x.size(-1);
+  const int64_t N = x.numel() / H;
+  auto dx = at::empty_like(x);
+  auto dw = at::zeros({H}, x.options().dtype(at::kFloat));
+  if (is_bf16(x))
+    check_hip(lumina_rmsnorm_bwd_bf16(gy.data_ptr(), x.data_ptr(), w.data_ptr(),
+                                      inv.data_ptr<float>(), dx.data_ptr(),
+                                      dw.data_ptr<float>(), N, H, cur_stream()),
+              "rmsnorm_bwd_bf16");
+  else
+    check_hip(lumina_rmsnorm_bwd_f32(gy.data_ptr(), x.data_ptr(), w.data_ptr(),
+                                     inv.data_ptr<float>(), dx.data_ptr(),
+                                     dw.data_ptr<float>(), N, H, cur_stream()),
+              "rmsnorm_bwd_f32");
+  return {dx, dw};
+}
+
+// ---- RoPE ----------------------------------------------------------------
+// q: [B, S, Hq, D], k: [B, S, Hk, D] contiguous; cos/sin: [S_cache, D/2] fp32.
+std::tuple<at::Tensor, at::Tensor> rope_fwd(const at::Tensor& q,
+                                            const at::Tensor& k,
+                                            const at::Tensor& cos_t,
+                                            const at::Tensor& sin_t,
+                                            const std::optional<at::Tensor>& pos,
+                                            int64_t pos_offset, bool conj) {
+  TORCH_CHECK(q.is_contiguous() && k.is_contiguous());
+  TORCH_CHECK(cos_t.is_contiguous() && sin_t.is_contiguous());
+  TORCH_CHECK(cos_t.scalar_type() == at::kFloat && sin_t.scalar_type() == at::kFloat);
+  check_dtype(q, "q");
+  TORCH_CHECK(q.dim() == 4 && k.dim() == 4);
+  const int64_t B = q.size(0);
+  const int S = (int)q.size(1), Hq = (int)q.size(2), Hk = (int)k.size(2);
+  const int D = (int)q.size(3);
+  TORCH_CHECK(k.size(0) == B && k.size(1) == S && k.size(3) == D);
+  TORCH_CHECK(cos_t.size(1) == D / 2);
+  const int* posp = nullptr;
+  if (pos.has_value()) {
+    TORCH_CHECK(pos->scalar_type() == at::kInt && pos->is_contiguous());
+    TORCH_CHECK(pos->numel() == B * S);
+    posp = pos->data_ptr<int>();
+  }
+  auto oq = at::empty_like(q);
+  auto ok = at::empty_like(k);
+  if (is_bf16(q))
+    check_hip(lumina_rope_bf16(q.data_ptr(), k.data_ptr(), oq.data_ptr(),
+                               ok.data_ptr(), cos_t.data_ptr<float>(),
+                               sin_t.data_ptr<float>(), posp, B, S, Hq, Hk, D,
+                               (int)pos_offset, conj ? 1 : 0, cur_stream()),
+              "rope_bf16");
+  else
+    check_hip(lumina_rope_f32(q.data_ptr(), k.data_ptr(), oq.data_ptr(),
+                              ok.data_ptr(), cos_t.data_ptr<float>(),
+                              sin_t.data_ptr<float>(), posp, B, S, Hq, Hk, D,
+                              (int)pos_offset, conj ? 1 : 0, cur_stream()),
+              "rope_f32");
+  return {oq, ok};
+}
+
+// ---- SwiGLU ---------------------------------------------------------------
+// g, u: [rows, I] views into the fused gate_up output (row stride may exceed I).
+at::Tensor swiglu_fwd(const at::Tensor& g, const at::Tensor& u) {
+  check_dtype(g, "gate");
+  TORCH_CHECK(g.dim() == 2 && u.dim() == 2 && g.sizes() == u.sizes());
+  TORCH_CHECK(g.stride(1) == 1 && u.stride(1) == 1, "inner dim must be contiguous");
+  const int64_t rows = g.size(0);
+  const int I = (int)g.size(1);
+  auto y = at::empty({rows, I}, g.options());
+  auto fn = is_bf16(g) ? lumina_swiglu_fwd_bf16 : lumina_swiglu_fwd_f32;
+  check_hip(fn(g.data_ptr(), u.data_ptr(), y.data_ptr(), rows, I,
+               g.stride(0), u.stride(0), cur_stream()), "swiglu_fwd");
+  return y;
+}
+
+std::tuple<at::Tensor, at::Tensor> swiglu_bwd(const at::Tensor& dy,
+                                              const at::Tensor& g,
+                                              const at::Tensor& u) {
+  TORCH_CHECK(dy.is_contiguous());
+  TORCH_CHECK(g.stride(1) == 1 && u.stride(1) == 1);
+  const int64_t rows = g.size(0);
+  const int I = (int)g.size(1);
+  // dg/du are allocated as one fused [rows, 2I] buffer when g/u share storage
+  // layout, so the caller can feed the downstream GEMM one contiguous tensor.
+  at::Tensor dg, du;
+  bool fused = (g.stride(0) == 2 * I && u.stride(0) == 2 * I &&
+                u.data_ptr() == (char*)g.data_ptr() + I * g.element_size());
+  if (fused) {
+    auto dgu = at::empty({rows, 2 * I}, g.options());
+    dg = dgu.narrow(1, 0, I);
+    du = dgu.narrow(1, I, I);
+  } else {
+    dg = at::empty({rows, I}, g.options());
+    du = at::empty({rows, I}, u.options());
+  }
+  auto fn = is_bf16(g) ? lumina_swiglu_bwd_bf16 : lumina_swiglu_bwd_f32;
+  check_hip(fn(dy.data_ptr(), g.data_ptr(), u.data_ptr(), dg.data_ptr(),
+               du.data_ptr(), rows, I,
+               fused ? 2 * I : I, fused ? 2 * I : I, cur_stream()),
+            "swiglu_bwd");
+  return {dg, du};
+}
+
+// ---- fused CE ------------------------------------------------------------
+std::tuple<at::Tensor, at::Tensor> ce_fwd(const at::Tensor& logits,
+                                          const at::Tensor& labels,
+                                          const std::optional<at::Tensor>& weights,
+                                          int64_t ignore_index) {
+  TORCH_CHECK(logits.is_contiguous() && logits.dim() == 2);
+  TORCH_CHECK(labels.scalar_type() == at::kInt && labels.is_contiguous());
+  check_dtype(logits, "logits");
+  const int64_t N = logits.size(0);
+  const int V = (int)logits.size(1);
+  const float* wp = nullptr;
+  if (weights.has_value()) {
+    TORCH_CHECK(weights->scalar_type() == at::kFloat && weights->is_contiguous());
+    wp = weights->data_ptr<float>();
+  }
+  auto lse = at::empty({N}, logits.options().dtype(at::kFloat));
+  auto stats = at::zeros({4}, logits.options().dtype(at::kFloat));
+  auto fn = is_bf16(logits) ? lumina_ce_fwd_bf16 : lumina_ce_fwd_f32;
+  check_hip(fn(logits.data_ptr(), labels.data_ptr<int>(), wp,
+               lse.data_ptr<float>(), stats.data_ptr<float>(), N, V,
+               (int)ignore_index, cur_stream()), "ce_fwd");
+  return {lse, stats};
+}
+
+at::Tensor ce_bwd(const at::Tensor& logits, const at::Tensor& labels,
+                  const std::optional<at::Tensor>& weights,
+                  const at::Tensor& lse, const at::Tensor& stats,
+                  const at::Tensor& gscale, int64_t ignore_index) {
+  const int64_t N = logits.size(0);
+  const int V = (int)logits.size(1);
+  const float* wp = weights.has_value() ? weights->data_ptr<float>() : nullptr;
+  auto dlogits = at::empty_like(logits);
+  auto fn = is_bf16(logits) ? lumina_ce_bwd_bf16 : lumina_ce_bwd_f32;
+  check_hip(fn(logits.data_ptr(), labels.data_ptr<int>(), wp,
+               lse.data_ptr<float>(), stats.data_ptr<float>(),
+               gscale.data_ptr<float>(), dlogits.data_ptr(), N, V,
+               (int)ignore_index, cur_stream()), "ce_bwd");
+  return dlogits;
+}
+
+// ---- optimizer -----------------------------------------------------------
+at::Tensor l2norm_sq(const at::Tensor& x) {
+  TORCH_CHECK(x.is_contiguous());
+  check_dtype(x, "x");
+  auto out = at::zeros({1}, x.options().dtype(at::kFloat));
+  auto fn = is_bf16(x) ? lumina_l2norm_sq_bf16 : lumina_l2norm_sq_f32;
+  check_hip(fn(x.data_ptr(), x.numel(), out.data_ptr<float>(), cur_stream()),
+            "l2norm_sq");
+  return out;
+}
+
+void adamw_step(at::Tensor& master, const at::Tensor& grad, at::Tensor& m,
+                at::Tensor& v, const std::optional<at::Tensor>& w_out,
+                double lr, double beta1, double beta2, double eps, double wd,
+                int64_t step, const std::optional<at::Tensor>& gnorm_sq,
+                double max_norm, double grad_scale) {
+  TORCH_CHECK(master.scalar_type() == at::kFloat && master.is_contiguous());
+  TORCH_CHECK(m.scalar_type() == at::kFloat && v.scalar_type() == at::kFloat);
+  check_dtype(grad, "grad");
+  const int64_t n = master.numel();
+  TORCH_CHECK(grad.numel() == n && m.numel() == n && v.numel() == n);
+  void* wp = nullptr;
+  int w_bf16 = 1;
+  if (w_out.has_value()) {
+    TORCH_CHECK(w_out->numel() == n && w_out->is_contiguous());
+    wp = w_out->data_ptr();
+    w_bf16 = is_bf16(*w_out) ? 1 : 0;
+  }
+  const float* np = gnorm_sq.has_value() ? gnorm_sq->data_ptr<float>() : nullptr;
+  const float bias1 = 1.0f / (1.0f - (float)std::pow(beta1, (double)step));
+  const float bias2 = 1.0f / (1.0f - (float)std::pow(beta2, (double)step));
+  check_hip(lumina_adamw_step(master.data_ptr<float>(), grad.data_ptr(),
+                              is_bf16(grad) ? 1 : 0, m.data_ptr<float>(),
+                              v.data_ptr<float>(), wp, w_bf16, n, (float)lr,
+                              (float)beta1, (float)beta2, (float)eps, (float)wd,
+                              bias1, bias2, np, (float)max_norm,
+                              (float)grad_scale, cur_stream()),
+            "adamw_step");
+}
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
+  mod.def("rmsnorm_fwd", &rmsnorm_fwd, "RMSNorm forward (gfx950)");
+  mod.def("rmsnorm_bwd", &rmsnorm_bwd, "RMSNorm backward (gfx950)");
+  mod.def("rope_fwd", &rope_fwd, "RoPE q/k rotation (gfx950)");
+  mod.def("swiglu_fwd", &swiglu_fwd, "SwiGLU forward (gfx950)");
+  mod.def("swiglu_bwd", &swiglu_bwd, "SwiGLU backward (gfx950)");
+  mod.def("ce_fwd", &ce_fwd, "fused CE+accuracy forward (gfx950)");
+  mod.def("ce_bwd", &ce_bwd, "fused CE backward (gfx950)");
+  mod.def("l2norm_sq", &l2norm_sq, "flat L2 norm squared (gfx950)");
+  mod.def("adamw_step", &adamw_step, "fused clip+AdamW on flat buffers (gfx950)");
+}

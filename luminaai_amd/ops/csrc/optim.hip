@@ -1,0 +1,139 @@
+// Flat-buffer fused optimizer kernels for CDNA4 (gfx950).
+//
+// MI355X-native replacement for the reference's multi-tensor machinery
+// (/root/reference/Src/Main_Scripts/training/fused_grad_clip.cu:26-175 and the
+// vendored ColossalAI multi_tensor_adam.cu / multi_tensor_l2norm_kernel.cu):
+// instead of chunked pointer tables, the trainer keeps params / grads /
+// optimizer state in ONE flat buffer each (sized for 288 GB HBM3E), so every
+// optimizer-path op is a single contiguous streaming kernel:
+//
+//   l2norm_sq:   ||g||^2 of the flat grad          (one pass, atomic combine)
+//   adamw_step:  grad-clip + AdamW + bf16 weight materialisation in ONE pass,
+//                reading the norm from device memory => zero host syncs.
+//
+// Memory per param/step: read g(2B) + m,v,master(12B), write m,v,master(12B)
+// + bf16 weight(2B) = 28 B — at ~6.3 TB/s achievable HBM BW an 8B-param
+// update costs ~35 ms; the flat layout keeps it at exactly that roofline.
+#include "common.h"
+
+template <typename E, int BLOCK>
+__global__ void l2norm_sq_kernel(const typename E::storage* __restrict__ x,
+                                 int64_t n, float* __restrict__ out) {
+  __shared__ float red[16];
+  float acc = 0.f;
+  if (sizeof(typename E::storage) == 2) {
+    const ushortx8* xv = reinterpret_cast<const ushortx8*>(x);
+    const int64_t nvec = n / 8;
+    for (int64_t i = (int64_t)blockIdx.x * BLOCK + threadIdx.x; i < nvec;
+         i += (int64_t)gridDim.x * BLOCK) {
+      ushortx8 v = xv[i];
+      #pragma unroll
+      for (int j = 0; j < 8; ++j) { float f = bf16_to_f32(v[j]); acc += f * f; }
+    }
+    for (int64_t i = nvec * 8 + (int64_t)blockIdx.x * BLOCK + threadIdx.x; i < n;
+         i += (int64_t)gridDim.x * BLOCK) {
+      float f = E::load(x + i); acc += f * f;
+    }
+  } else {
+    const floatx4* xv = reinterpret_cast<const floatx4*>(x);
+    const int64_t nvec = n / 4;
+    for (int64_t i = (int64_t)blockIdx.x * BLOCK + threadIdx.x; i < nvec;
+         i += (int64_t)gridDim.x * BLOCK) {
+      floatx4 v = xv[i];
+      #pragma unroll
+      for (int j = 0; j < 4; ++j) acc += v[j] * v[j];
+    }
+    for (int64_t i = nvec * 4 + (int64_t)blockIdx.x * BLOCK + threadIdx.x; i < n;
+         i += (int64_t)gridDim.x * BLOCK) {
+      float f = E::load(x + i); acc += f * f;
+    }
+  }
+  acc = block_reduce_sum(acc, red);
+  if (threadIdx.x == 0 && acc != 0.f) atomicAdd(out, acc);
+}
+
+// One-pass grad-clip + AdamW. The clip scale is computed per-thread from the
+// device-resident ||g||^2 (written by l2norm_sq on the same stream), so the
+// whole optimizer step runs without a host round trip:
+//   gnorm = sqrt(*gnorm_sq) * grad_scale
+//   clip  = max_norm > 0 && gnorm > max_norm ? max_norm / (gnorm + 1e-6) : 1
+//   g~    = g * grad_scale * clip
+//   m     = b1*m + (1-b1)*g~ ;  v = b2*v + (1-b2)*g~^2
+//   master -= lr * (m*bias1 / (sqrt(v*bias2) + eps) + wd*master)
+//   w_out  = bf16(master)
+template <typename G, typename W>
+__global__ void adamw_step_kernel(float* __restrict__ master,
+                                  const typename G::storage* __restrict__ grad,
+                                  float* __restrict__ m, float* __restrict__ v,
+                                  typename W::storage* __restrict__ w_out,
+                                  int64_t n, float lr, float beta1, float beta2,
+                                  float eps, float wd, float bias1, float bias2,
+                                  const float* __restrict__ gnorm_sq,
+                                  float max_norm, float grad_scale) {
+  float clip = 1.0f;
+  if (max_norm > 0.f && gnorm_sq) {
+    const float gn = sqrtf(*gnorm_sq) * grad_scale;
+    if (gn > max_norm) clip = max_norm / (gn + 1e-6f);
+  }
+  const float gs = grad_scale * clip;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += gstride()) {
+    const float g = G::load(grad + i) * gs;
+    float mi = m[i] = beta1 * m[i] + (1.f - beta1) * g;
+    float vi = v[i] = beta2 * v[i] + (1.f - beta2) * g * g;
+    float p = master[i];
+    p -= lr * (mi * bias1 / (sqrtf(vi * bias2) + eps) + wd * p);
+    master[i] = p;
+    if (w_out) W::store(w_out + i, p);
+  }
+}
+
+extern "C" {
+
+hipError_t lumina_l2norm_sq_bf16(const void* x, int64_t n, float* out,
+                                 hipStream_t s) {
+  constexpr int B = 256;
+  int grid = elementwise_grid(n, B, 16);
+  l2norm_sq_kernel<BF16Elem, B><<<grid, B, 0, s>>>((const uint16_t*)x, n, out);
+  HIP_CHECK_LAST();
+  return hipSuccess;
+}
+
+hipError_t lumina_l2norm_sq_f32(const void* x, int64_t n, float* out,
+                                hipStream_t s) {
+  constexpr int B = 256;
+  int grid = elementwise_grid(n, B, 16);
+  l2norm_sq_kernel<F32Elem, B><<<grid, B, 0, s>>>((const float*)x, n, out);
+  HIP_CHECK_LAST();
+  return hipSuccess;
+}
+
+hipError_t lumina_adamw_step(float* master, const void* grad, int grad_is_bf16,
+                             float* m, float* v, void* w_out, int wout_is_bf16,
+                             int64_t n, float lr, float beta1, float beta2,
+                             float eps, float wd, float bias1, float bias2,
+                             const float* gnorm_sq, float max_norm,
+                             float grad_scale, hipStream_t s) {
+  const int block = 256;
+  const int grid = elementwise_grid(n, block, 4);
+  if (grad_is_bf16 && wout_is_bf16)
+    adamw_step_kernel<BF16Elem, BF16Elem><<<grid, block, 0, s>>>(
+        master, (const uint16_t*)grad, m, v, (uint16_t*)w_out, n, lr, beta1,
+        beta2, eps, wd, bias1, bias2, gnorm_sq, max_norm, grad_scale);
+  else if (grad_is_bf16)
+    adamw_step_kernel<BF16Elem, F32Elem><<<grid, block, 0, s>>>(
+        master, (const uint16_t*)grad, m, v, (float*)w_out, n, lr, beta1,
+        beta2, eps, wd, bias1, bias2, gnorm_sq, max_norm, grad_scale);
+  else if (wout_is_bf16)
+    adamw_step_kernel<F32Elem, BF16Elem><<<grid, block, 0, s>>>(
+        master, (const float*)grad, m, v, (uint16_t*)w_out, n, lr, beta1,
+        beta2, eps, wd, bias1, bias2, gnorm_sq, max_norm, grad_scale);
+  else
+    adamw_step_kernel<F32Elem, F32Elem><<<grid, block, 0, s>>>(
+        master, (const float*)grad, m, v, (float*)w_out, n, lr, beta1,
+        beta2, eps, wd, bias1, bias2, gnorm_sq, max_norm, grad_scale);
+  HIP_CHECK_LAST();
+  return hipSuccess;
+}
+
+}  // extern "C"

@@ -1,0 +1,253 @@
+"""Autograd interface over the gfx950 HIP kernels with a CPU reference path.
+
+Dispatch policy:
+- CUDA (= ROCm/HIP) tensors -> the _lumina_hip extension. If the extension is
+  missing on a GPU machine this raises loudly (no silent eager fallback) unless
+  LUMINA_ALLOW_FALLBACK=1 is set.
+- CPU tensors -> the pure-PyTorch reference implementations (ops/reference.py).
+
+Every Function here has a hand-written backward on both paths; numerics tests
+(tests/test_ops_gpu.py) compare the HIP path against fp32 references.
+"""
+
+from __future__ import annotations
+
+import os
+from typing import Optional
+
+import torch
+
+from . import reference as ref
+
+_EXT = None
+_TRIED = False
+
+
+def _load_ext():
+    global _EXT, _TRIED
+    if _TRIED:
+        return _EXT
+    _TRIED = True
+    from . import build_ext
+    _EXT = build_ext.load_prebuilt()
+    return _EXT
+
+
+def get_ext():
+    return _load_ext()
+
+
+def has_ext() -> bool:
+    return _load_ext() is not None
+
+
+def _ext_or_raise():
+    e = _load_ext()
+    if e is None:
+        if os.environ.get("LUMINA_ALLOW_FALLBACK") == "1":
+            return None
+        raise RuntimeError(
+            "luminaai_amd: tensor is on GPU but the _lumina_hip extension is not "
+            "built. Run `python -m luminaai_amd.ops.build_ext` (or __graft_entry__"
+            ".build()). Set LUMINA_ALLOW_FALLBACK=1 to allow the slow eager path.")
+    return e
+
+
+def use_hip(t: torch.Tensor) -> bool:
+    if not t.is_cuda:
+        return False
+    return _ext_or_raise() is not None
+
+
+# ------------------------------------------------------------------ RMSNorm
+class RMSNormFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, weight, eps):
+        if use_hip(x):
+            ext = get_ext()
+            y, inv = ext.rmsnorm_fwd(x.contiguous(), weight.contiguous(), eps, True)
+        else:
+            y, inv = ref.rmsnorm_fwd_train(x, weight, eps)
+        ctx.save_for_backward(x, weight, inv)
+        return y
+
+    @staticmethod
+    def backward(ctx, gy):
+        x, weight, inv = ctx.saved_tensors
+        if use_hip(x):
+            dx, dw = get_ext().rmsnorm_bwd(gy.contiguous(), x.contiguous(),
+                                           weight.contiguous(), inv)
+        else:
+            dx, dw = ref.rmsnorm_bwd(gy, x, weight, inv)
+        return dx, dw.to(weight.dtype), None
+
+
+def rmsnorm(x: torch.Tensor, weight: torch.Tensor, eps: float = 1e-6) -> torch.Tensor:
+    if torch.is_grad_enabled() and (x.requires_grad or weight.requires_grad):
+        return RMSNormFn.apply(x, weight, eps)
+    if use_hip(x):
+        y, _ = get_ext().rmsnorm_fwd(x.contiguous(), weight.contiguous(), eps, False)
+        return y
+    return ref.rmsnorm_fwd(x, weight, eps)
+
+
+# -------------------------------------------------------------------- RoPE
+class RoPEFn(torch.autograd.Function):
+    """Rotate q and k ([B, S, H, D] layout) in one launch. Backward is the
+    inverse rotation (conj=True)."""
+
+    @staticmethod
+    def forward(ctx, q, k, cos, sin, pos, pos_offset):
+        ctx.pos_offset = pos_offset
+        if use_hip(q):
+            oq, ok = get_ext().rope_fwd(q.contiguous(), k.contiguous(), cos, sin,
+                                        pos, pos_offset, False)
+            ctx.save_for_backward(cos, sin, pos if pos is not None else torch.empty(0))
+            ctx.has_pos = pos is not None
+            return oq, ok
+        ctx.save_for_backward(cos, sin, pos if pos is not None else torch.empty(0))
+        ctx.has_pos = pos is not None
+        return _rope_ref_bshd(q, k, cos, sin, pos, pos_offset, conj=False)
+
+    @staticmethod
+    def backward(ctx, gq, gk):
+        cos, sin, pos = ctx.saved_tensors
+        pos = pos if ctx.has_pos else None
+        if use_hip(gq):
+            dq, dk = get_ext().rope_fwd(gq.contiguous(), gk.contiguous(), cos, sin,
+                                        pos, ctx.pos_offset, True)
+        else:
+            dq, dk = _rope_ref_bshd(gq, gk, cos, sin, pos, ctx.pos_offset, conj=True)
+        return dq, dk, None, None, None, None
+
+
+def _rope_ref_bshd(q, k, cos, sin, pos, pos_offset, conj):
+    """Reference rotation on [B, S, H, D] tensors (fp32 math)."""
+    B, S, _, D = q.shape
+    half = D // 2
+    if pos is not None:
+        c = cos[pos.view(B, S).long()]  # [B, S, half]
+        s = sin[pos.view(B, S).long()]
+    else:
+        c = cos[pos_offset:pos_offset + S].unsqueeze(0).expand(B, S, half)
+        s = sin[pos_offset:pos_offset + S].unsqueeze(0).expand(B, S, half)
+    if conj:
+        s = -s
+    c = c.unsqueeze(2)  # [B, S, 1, half]
+    s = s.unsqueeze(2)
+
+    def rot(x):
+        x32 = x.float()
+        x1, x2 = x32.chunk(2, dim=-1)
+        return torch.cat([x1 * c - x2 * s, x2 * c + x1 * s], dim=-1).to(x.dtype)
+
+    return rot(q), rot(k)
+
+
+def rope(q, k, cos, sin, pos: Optional[torch.Tensor] = None, pos_offset: int = 0):
+    return RoPEFn.apply(q, k, cos, sin, pos, pos_offset)
+
+
+# ------------------------------------------------------------------ SwiGLU
+class SwiGLUFn(torch.autograd.Function):
+    """y = silu(gate) * up on 2-D views (gate/up may be the two halves of a
+    fused gate_up projection, sharing storage with row stride 2I)."""
+
+    @staticmethod
+    def forward(ctx, gate, up):
+        ctx.save_for_backward(gate, up)
+        if use_hip(gate):
+            return get_ext().swiglu_fwd(gate, up)
+        return ref.swiglu_fwd(gate, up)
+
+    @staticmethod
+    def backward(ctx, gy):
+        gate, up = ctx.saved_tensors
+        if use_hip(gate):
+            dg, du = get_ext().swiglu_bwd(gy.contiguous(), gate, up)
+        else:
+            dg, du = ref.swiglu_bwd(gy, gate, up)
+        return dg, du
+
+
+def swiglu(gate: torch.Tensor, up: torch.Tensor) -> torch.Tensor:
+    shape = gate.shape
+    g2 = gate.reshape(-1, shape[-1]) if gate.dim() != 2 else gate
+    u2 = up.reshape(-1, shape[-1]) if up.dim() != 2 else up
+    y = SwiGLUFn.apply(g2, u2)
+    return y.view(shape)
+
+
+# ------------------------------------------------- fused CE + accuracy loss
+class FusedCEFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, logits, labels_i32, weights, ignore_index):
+        ext = get_ext()
+        lse, stats = ext.ce_fwd(logits, labels_i32, weights, ignore_index)
+        ctx.save_for_backward(logits, labels_i32,
+                              weights if weights is not None else torch.empty(0),
+                              lse, stats)
+        ctx.has_w = weights is not None
+        ctx.ignore_index = ignore_index
+        loss = stats[0] / stats[1].clamp_min(1e-8)
+        ctx.mark_non_differentiable(stats)
+        return loss, stats
+
+    @staticmethod
+    def backward(ctx, gloss, _gstats):
+        logits, labels, weights, lse, stats = ctx.saved_tensors
+        weights = weights if ctx.has_w else None
+        dlogits = get_ext().ce_bwd(logits, labels, weights, lse, stats,
+                                   gloss.reshape(1).float().contiguous(),
+                                   ctx.ignore_index)
+        return dlogits, None, None, None
+
+
+def fused_cross_entropy(logits: torch.Tensor, labels: torch.Tensor,
+                        loss_weights: Optional[torch.Tensor] = None,
+                        ignore_index: int = -100):
+    """Returns (loss, accuracy, n_valid). GPU: single-pass HIP kernel;
+    CPU: reference. `labels` may be int64; cast happens here."""
+    logits2 = logits.reshape(-1, logits.shape[-1])
+    labels1 = labels.reshape(-1)
+    w = loss_weights.reshape(-1).float() if loss_weights is not None else None
+    if use_hip(logits2):
+        loss, stats = FusedCEFn.apply(logits2.contiguous(),
+                                      labels1.to(torch.int32).contiguous(),
+                                      w.contiguous() if w is not None else None,
+                                      ignore_index)
+        acc = stats[2] / stats[3].clamp_min(1.0)
+        return loss, acc, stats[3]
+    return ref.fused_cross_entropy(logits2, labels1, w, ignore_index)
+
+
+# ----------------------------------------------------------- optimizer path
+def l2norm_sq(flat: torch.Tensor) -> torch.Tensor:
+    if use_hip(flat):
+        return get_ext().l2norm_sq(flat)
+    return flat.float().pow(2).sum().reshape(1)
+
+
+def adamw_step(master: torch.Tensor, grad: torch.Tensor, m: torch.Tensor,
+               v: torch.Tensor, w_out: Optional[torch.Tensor], lr: float,
+               beta1: float, beta2: float, eps: float, wd: float, step: int,
+               gnorm_sq: Optional[torch.Tensor], max_norm: float,
+               grad_scale: float = 1.0):
+    """Fused clip+AdamW on flat buffers. CPU path mirrors the kernel math."""
+    if use_hip(master):
+        get_ext().adamw_step(master, grad, m, v, w_out, lr, beta1, beta2, eps,
+                             wd, step, gnorm_sq, max_norm, grad_scale)
+        return
+    g = grad.float() * grad_scale
+    if max_norm > 0 and gnorm_sq is not None:
+        gn = gnorm_sq.sum().sqrt().item() * grad_scale
+        if gn > max_norm:
+            g = g * (max_norm / (gn + 1e-6))
+    m.mul_(beta1).add_(g, alpha=1 - beta1)
+    v.mul_(beta2).addcmul_(g, g, value=1 - beta2)
+    bias1 = 1.0 / (1.0 - beta1 ** step)
+    bias2 = 1.0 / (1.0 - beta2 ** step)
+    denom = (v * bias2).sqrt().add_(eps)
+    master.add_(-lr * ((m * bias1) / denom + wd * master))
+    if w_out is not None:
+        w_out.copy_(master.to(w_out.dtype))
