@@ -1,0 +1,153 @@
+#!/usr/bin/env python3
+"""Headline benchmark: whole-node training tokens/sec.
+
+Flagship config (BASELINE.json): 8-expert top-2 MoE, ~1.3B active / ~8B total
+(b1 architecture: 1908 hidden, 31 layers, GQA 12:4, seq 2048), bf16, synthetic
+data, random-init weights.
+
+Single GPU:   python bench.py --gpus 1 --steps 20 --warmup 5
+Multi-GPU:    python -m torch.distributed.run --nnodes=1 --nproc-per-node N \
+                  --master-addr 127.0.0.1 bench.py --gpus N --steps K --warmup W
+
+One JSON line is printed by rank 0 (driver contract).
+"""
+
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import sys
+import time
+
+import torch
+
+sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+
+from luminaai_amd.config import ConfigPresets  # noqa: E402
+from luminaai_amd.data.dataset import SyntheticDataset  # noqa: E402
+from luminaai_amd.data.tokenizer import ConversationTokenizer  # noqa: E402
+from luminaai_amd.models import DeepSeekTransformer, config_to_deepseek_config  # noqa: E402
+from luminaai_amd.parallel import comm  # noqa: E402
+from luminaai_amd.training import Trainer  # noqa: E402
+
+BASELINE_TOKENS_PER_SEC = 73000.0  # reference b1 MoE 8E top-2 on A100 40GB
+
+
+def build_config(args):
+    cfg = ConfigPresets.b1_moe()
+    if args.preset:
+        cfg = ConfigPresets.get(args.preset)
+    cfg.num_workers = 0
+    cfg.compile = False
+    cfg.micro_batch_size = args.micro_batch
+    cfg.gradient_accumulation_steps = args.accum
+    cfg.seq_length = args.seq_len or cfg.seq_length
+    cfg.gradient_checkpointing = not args.no_checkpointing
+    cfg.zero_stage = args.zero if args.zero is not None else \
+        (2 if comm.env_world_size() > 1 else 0)
+    cfg.eval_every_n_batches = 0
+    cfg.save_every_n_batches = 0
+    return cfg
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=10)
+    ap.add_argument("--warmup", type=int, default=3)
+    ap.add_argument("--micro-batch", type=int, default=4)
+    ap.add_argument("--accum", type=int, default=1)
+    ap.add_argument("--seq-len", type=int, default=None)
+    ap.add_argument("--preset", type=str, default=None)
+    ap.add_argument("--zero", type=int, default=None)
+    ap.add_argument("--no-checkpointing", action="store_true")
+    args = ap.parse_args()
+
+    distributed = comm.init_distributed()
+    rank = comm.get_rank()
+    world = comm.get_world_size()
+    device = torch.device("cuda", comm.env_local_rank()) \
+        if torch.cuda.is_available() else torch.device("cpu")
+
+    cfg = build_config(args)
+    torch.manual_seed(cfg.seed + rank)
+    tok = ConversationTokenizer(max_length=cfg.seq_length)
+    model_cfg = config_to_deepseek_config(cfg)
+    model = DeepSeekTransformer(model_cfg)
+    trainer = Trainer(model, tok, cfg)
+    trainer._setup_scheduler(args.steps + args.warmup + 10)
+
+    micro = cfg.micro_batch_size or 1
+    steps_total = args.warmup + args.steps
+    ds = SyntheticDataset(cfg.vocab_size, cfg.seq_length,
+                          micro * cfg.gradient_accumulation_steps * steps_total,
+                          seed=cfg.seed + rank)
+    batches = [ds[i] for i in range(len(ds))]
+
+    def run_step(step_idx):
+        base = step_idx * cfg.gradient_accumulation_steps * micro
+        for a in range(cfg.gradient_accumulation_steps):
+            boundary = a == cfg.gradient_accumulation_steps - 1
+            trainer.engine.set_sync(boundary)
+            rows = [batches[base + a * micro + j] for j in range(micro)]
+            batch = {k: torch.stack([r[k] for r in rows]) for k in rows[0]}
+            trainer.train_step(batch)
+        trainer.optimizer_step()
+
+    # ---- warmup
+    for i in range(args.warmup):
+        run_step(i)
+
+    # ---- timed region
+    if distributed:
+        comm.barrier()
+    if device.type == "cuda":
+        torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for i in range(args.warmup, steps_total):
+        run_step(i)
+    if device.type == "cuda":
+        torch.cuda.synchronize()
+    if distributed:
+        comm.barrier()
+    t1 = time.perf_counter()
+
+    elapsed = t1 - t0
+    elapsed = comm.all_reduce_scalar(elapsed, op="max")  # MAX over ranks
+    tokens_per_step = micro * cfg.gradient_accumulation_steps * cfg.seq_length * world
+    tokens_per_sec = tokens_per_step * args.steps / elapsed
+    ms_per_step = elapsed / args.steps * 1000.0
+
+    if rank == 0:
+        loss = trainer._metric_floats().get("ce_loss")
+        result = {
+            "metric": "training_tokens_per_sec",
+            "value": round(tokens_per_sec, 1),
+            "unit": "tokens/s",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(ms_per_step, 2),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": round(tokens_per_sec / BASELINE_TOKENS_PER_SEC, 3),
+            "dtype": trainer.precision.spec.name,
+            "data": "synthetic",
+            "config": {
+                "model": args.preset or "b1_moe_8e_top2_8B",
+                "global_batch": micro * cfg.gradient_accumulation_steps * world,
+                "seq_len": cfg.seq_length,
+                "parallelism": f"dp{world}" + (f"_zero{cfg.zero_stage}"
+                                               if cfg.zero_stage else ""),
+                "micro_batch": micro,
+                "grad_accum": cfg.gradient_accumulation_steps,
+                "final_loss": loss,
+            },
+        }
+        print(json.dumps(result))
+    comm.cleanup()
+
+
+if __name__ == "__main__":
+    main()

@@ -71,9 +71,10 @@ __global__ void adamw_step_kernel(float* __restrict__ master,
                                   const float* __restrict__ gnorm_sq,
                                   float max_norm, float grad_scale) {
   float clip = 1.0f;
-  if (max_norm > 0.f && gnorm_sq) {
+  if (gnorm_sq) {
     const float gn = sqrtf(*gnorm_sq) * grad_scale;
-    if (gn > max_norm) clip = max_norm / (gn + 1e-6f);
+    if (!isfinite(gn)) return;  // NaN/Inf grads: skip the whole step
+    if (max_norm > 0.f && gn > max_norm) clip = max_norm / (gn + 1e-6f);
   }
   const float gs = grad_scale * clip;
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;

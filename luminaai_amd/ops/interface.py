@@ -12,6 +12,7 @@ Every Function here has a hand-written backward on both paths; numerics tests
 
 from __future__ import annotations
 
+import math
 import os
 from typing import Optional
 
@@ -239,9 +240,11 @@ def adamw_step(master: torch.Tensor, grad: torch.Tensor, m: torch.Tensor,
                              wd, step, gnorm_sq, max_norm, grad_scale)
         return
     g = grad.float() * grad_scale
-    if max_norm > 0 and gnorm_sq is not None:
+    if gnorm_sq is not None:
         gn = gnorm_sq.sum().sqrt().item() * grad_scale
-        if gn > max_norm:
+        if not math.isfinite(gn):
+            return  # NaN/Inf grads: skip the whole step (matches HIP kernel)
+        if max_norm > 0 and gn > max_norm:
             g = g * (max_norm / (gn + 1e-6))
     m.mul_(beta1).add_(g, alpha=1 - beta1)
     v.mul_(beta2).addcmul_(g, g, value=1 - beta2)

@@ -1,0 +1,288 @@
+"""FlatAdamW: flat-buffer fused AdamW for MI355X, ZeRO-shardable.
+
+MI355X-native replacement for the reference optimizer path (torch AdamW +
+FusedGradClip ctypes kernels, /root/reference/Src/Main_Scripts/training/
+cuda_kernels.py:253-402, fused_grad_clip.cu; DeepSpeed ZeRO stages via
+backend_deepspeed.py:129-165). Design:
+
+- every parameter's storage is re-pointed into ONE flat bf16 buffer per
+  parameter group (weights), padded to a multiple of 256*world so ZeRO
+  shards are equal-size and 16-byte aligned;
+- every `p.grad` is a view into ONE flat bf16 grad buffer, so RCCL
+  all-reduce / reduce-scatter over xGMI and the grad-norm kernel operate on
+  contiguous memory with zero gather/scatter;
+- the optimizer step is two HIP kernels per group (l2norm_sq + fused
+  clip/AdamW/weight-materialise) with NO host synchronisation — the clip
+  scale (and NaN skip) is resolved from device memory inside the kernel;
+- with shard_world > 1 (ZeRO-1/2), master/m/v are allocated SHARD-SIZE only
+  (memory 6+12/N bytes/param instead of 18) and `step()` updates just this
+  rank's shard; the engine (parallel/zero.py) all-gathers the bf16 weights.
+
+In fp32 mode (CPU tests, shard_world == 1) parameters alias the master.
+"""
+
+from __future__ import annotations
+
+import math
+from typing import Dict, Iterable, List, Optional, Tuple
+
+import torch
+import torch.nn as nn
+
+from ..ops import interface as K
+
+NO_DECAY_KEYWORDS = ("bias", "norm", "gate.weight", "mod_router")
+ALIGN = 256
+
+
+def split_decay_groups(model: nn.Module) -> List[Dict]:
+    """Two param groups: weight-decayed matrices vs norms/biases/router gates
+    (reference backend_fsdp.py:220-252 decay grouping)."""
+    decay, no_decay = [], []
+    for name, p in model.named_parameters():
+        if not p.requires_grad:
+            continue
+        if any(k in name for k in NO_DECAY_KEYWORDS) or p.dim() <= 1:
+            no_decay.append(p)
+        else:
+            decay.append(p)
+    groups = []
+    if decay:
+        groups.append({"params": decay, "weight_decay": None})  # default wd
+    if no_decay:
+        groups.append({"params": no_decay, "weight_decay": 0.0})
+    return groups
+
+
+class _FlatGroup:
+    """One parameter group flattened into contiguous (padded) buffers."""
+
+    def __init__(self, params: List[torch.Tensor], lr: float, weight_decay: float,
+                 shard_rank: int = 0, shard_world: int = 1):
+        self.params = params
+        self.lr = lr
+        self.weight_decay = weight_decay
+        self.shard_rank = shard_rank
+        self.shard_world = shard_world
+        self.numel = sum(p.numel() for p in params)
+        quantum = ALIGN * shard_world
+        self.padded = (self.numel + quantum - 1) // quantum * quantum
+        self.shard_size = self.padded // shard_world
+        self.shard_lo = shard_rank * self.shard_size
+        self.shard_hi = self.shard_lo + self.shard_size
+
+        device = params[0].device
+        self.dtype = params[0].dtype
+        self.device = device
+        self.bf16 = self.dtype == torch.bfloat16
+        sharded = shard_world > 1
+
+        if self.bf16:
+            self.flat_w = torch.zeros(self.padded, device=device,
+                                      dtype=torch.bfloat16)
+        else:
+            assert not sharded or self.dtype == torch.float32
+            self.flat_w = (torch.zeros(self.padded, device=device,
+                                       dtype=torch.float32) if sharded else None)
+        self.flat_g = torch.zeros(self.padded, device=device, dtype=self.dtype)
+
+        # fill flat weights & re-point params
+        off = 0
+        self.offsets: List[Tuple[int, int]] = []
+        w_master_full = (self.flat_w is None)
+        if w_master_full:
+            self.master = torch.zeros(self.padded, device=device,
+                                      dtype=torch.float32)
+        for p in params:
+            n = p.numel()
+            if w_master_full:
+                self.master[off:off + n].copy_(p.data.reshape(-1).float())
+                p.data = self.master[off:off + n].view(p.shape)
+            else:
+                self.flat_w[off:off + n].copy_(p.data.reshape(-1).to(self.dtype))
+                p.data = self.flat_w[off:off + n].view(p.shape)
+            p.grad = self.flat_g[off:off + n].view(p.shape)
+            self.offsets.append((off, n))
+            off += n
+
+        # optimizer state: shard-size (== padded when world==1)
+        if not w_master_full:
+            self.master = self.flat_w[self.shard_lo:self.shard_hi].float()
+        state_size = self.padded if w_master_full else self.shard_size
+        self.m = torch.zeros(state_size, device=device, dtype=torch.float32)
+        self.v = torch.zeros_like(self.m)
+        self._master_is_params = w_master_full
+
+    # ---- views used by step() -------------------------------------------
+    def update_grad(self) -> torch.Tensor:
+        """The grad slice this rank's update consumes."""
+        if self._master_is_params:
+            return self.flat_g
+        return self.flat_g[self.shard_lo:self.shard_hi]
+
+    def update_weight_out(self) -> Optional[torch.Tensor]:
+        if self._master_is_params:
+            return None  # params alias master
+        return self.flat_w[self.shard_lo:self.shard_hi]
+
+    def weight_view(self) -> torch.Tensor:
+        return self.master if self._master_is_params else self.flat_w
+
+    def grad_view(self) -> torch.Tensor:
+        return self.flat_g
+
+
+class FlatAdamW:
+    """torch-optimizer-like interface over flat fused AdamW (ZeRO-aware)."""
+
+    def __init__(self, model_or_groups, lr: float = 1e-4,
+                 betas: Tuple[float, float] = (0.9, 0.95), eps: float = 1e-8,
+                 weight_decay: float = 0.01, max_grad_norm: float = 1.0,
+                 shard_rank: int = 0, shard_world: int = 1):
+        if isinstance(model_or_groups, nn.Module):
+            groups = split_decay_groups(model_or_groups)
+        else:
+            groups = list(model_or_groups)
+        self.defaults = dict(lr=lr, betas=betas, eps=eps, weight_decay=weight_decay)
+        self.betas = betas
+        self.eps = eps
+        self.max_grad_norm = max_grad_norm
+        self.shard_rank = shard_rank
+        self.shard_world = shard_world
+        self.step_count = 0
+        self.groups: List[_FlatGroup] = [
+            _FlatGroup(list(g["params"]), lr=g.get("lr", lr),
+                       weight_decay=(weight_decay if g.get("weight_decay") is None
+                                     else g["weight_decay"]),
+                       shard_rank=shard_rank, shard_world=shard_world)
+            for g in groups
+        ]
+        self._last_norm_sq: Optional[torch.Tensor] = None
+
+    # ---- torch-compat surface ---------------------------------------------
+    @property
+    def param_groups(self):
+        return [_GroupProxy(g) for g in self.groups]
+
+    def zero_grad(self, set_to_none: bool = False):
+        for g in self.groups:
+            g.flat_g.zero_()
+
+    def local_grad_norm_sq(self, shard_only: bool = False) -> torch.Tensor:
+        total = None
+        for g in self.groups:
+            t = g.update_grad() if shard_only else g.flat_g
+            ns = K.l2norm_sq(t)
+            total = ns if total is None else total + ns
+        return total
+
+    @torch.no_grad()
+    def step(self, grad_scale: float = 1.0, closure=None,
+             norm_sq: Optional[torch.Tensor] = None,
+             shard_only: bool = False):
+        """Fused clip + AdamW.
+
+        grad_scale multiplies stored grads (1/accum, DP averaging, ...).
+        norm_sq: externally supplied global ||g||^2 (ZeRO engines pass the
+        all-reduced value); default computes it locally.
+        shard_only: grads already reduced into this rank's shard (ZeRO-2).
+        """
+        self.step_count += 1
+        if norm_sq is None and self.max_grad_norm > 0:
+            norm_sq = self.local_grad_norm_sq(shard_only=shard_only)
+        self._last_norm_sq = norm_sq
+        for g in self.groups:
+            K.adamw_step(
+                g.master, g.update_grad(), g.m, g.v, g.update_weight_out(),
+                g.lr, self.betas[0], self.betas[1], self.eps, g.weight_decay,
+                self.step_count, norm_sq, self.max_grad_norm, grad_scale)
+
+    def last_grad_norm(self) -> float:
+        """Host-visible grad norm of the LAST step (synchronises)."""
+        if self._last_norm_sq is None:
+            return 0.0
+        val = float(self._last_norm_sq.sum())
+        return math.sqrt(val) if math.isfinite(val) and val >= 0 else float("nan")
+
+    # ---- state dict --------------------------------------------------------
+    def state_dict(self) -> Dict:
+        return {
+            "step_count": self.step_count,
+            "defaults": self.defaults,
+            "shard_rank": self.shard_rank,
+            "shard_world": self.shard_world,
+            "groups": [
+                {"master": g.master.cpu(), "m": g.m.cpu(), "v": g.v.cpu(),
+                 "lr": g.lr, "weight_decay": g.weight_decay, "numel": g.numel}
+                for g in self.groups
+            ],
+        }
+
+    def load_state_dict(self, sd: Dict):
+        self.step_count = sd.get("step_count", 0)
+        if sd.get("shard_world", 1) != self.shard_world:
+            raise ValueError("FlatAdamW: resharding checkpoints not supported yet "
+                             f"(saved world={sd.get('shard_world')}, "
+                             f"current={self.shard_world})")
+        for g, gs in zip(self.groups, sd["groups"]):
+            if g.numel != gs["numel"]:
+                raise ValueError(f"FlatAdamW state size mismatch ({g.numel} vs "
+                                 f"{gs['numel']})")
+            g.master.copy_(gs["master"].to(g.master.device))
+            g.m.copy_(gs["m"].to(g.m.device))
+            g.v.copy_(gs["v"].to(g.v.device))
+            g.lr = gs.get("lr", g.lr)
+            g.weight_decay = gs.get("weight_decay", g.weight_decay)
+            if not g._master_is_params:
+                g.flat_w[g.shard_lo:g.shard_hi].copy_(g.master.to(g.dtype))
+
+    # ---- dynamic architecture (expert add/prune) ---------------------------
+    def rebuild(self, model: nn.Module):
+        """Re-flatten after parameters changed (expert add/prune). State of
+        surviving parameters is preserved by identity match (single-rank)."""
+        old_state = {}
+        for g in self.groups:
+            if g.shard_world > 1:
+                old_state = {}
+                break
+            for p, (off, n) in zip(g.params, g.offsets):
+                old_state[id(p)] = (g.master[off:off + n].clone(),
+                                    g.m[off:off + n].clone(),
+                                    g.v[off:off + n].clone())
+        lr = self.groups[0].lr if self.groups else self.defaults["lr"]
+        groups = split_decay_groups(model)
+        new_groups = []
+        for gd in groups:
+            wd = gd.get("weight_decay")
+            fg = _FlatGroup(list(gd["params"]), lr=lr,
+                            weight_decay=(self.defaults["weight_decay"]
+                                          if wd is None else wd),
+                            shard_rank=self.shard_rank,
+                            shard_world=self.shard_world)
+            if fg.shard_world == 1:
+                for p, (off, n) in zip(fg.params, fg.offsets):
+                    st = old_state.get(id(p))
+                    if st is not None and st[0].numel() == n:
+                        fg.master[off:off + n].copy_(st[0])
+                        fg.m[off:off + n].copy_(st[1])
+                        fg.v[off:off + n].copy_(st[2])
+                        if not fg._master_is_params:
+                            fg.flat_w[off:off + n].copy_(st[0].to(fg.dtype))
+            new_groups.append(fg)
+        self.groups = new_groups
+
+
+class _GroupProxy(dict):
+    """dict-like view letting generic code mutate group lr / weight_decay."""
+
+    def __init__(self, group: _FlatGroup):
+        super().__init__(lr=group.lr, weight_decay=group.weight_decay,
+                         params=group.params)
+        self._g = group
+
+    def __setitem__(self, key, value):
+        super().__setitem__(key, value)
+        if key == "lr":
+            self._g.lr = value
+        elif key == "weight_decay":
+            self._g.weight_decay = value

@@ -1,0 +1,95 @@
+"""Tokenizer + dataset pipeline tests."""
+
+import torch
+
+from luminaai_amd.data import (
+    BaseTrainingDataset, ConversationDataset, ConversationTokenizer,
+    InterleavedDataset, SyntheticDataset, create_dataloader, setup_datasets,
+)
+
+
+def test_tokenizer_roundtrip(tokenizer):
+    text = "hello world, this is a test"
+    ids = tokenizer.encode(text)
+    assert tokenizer.decode(ids) == text
+
+
+def test_tokenizer_special_tokens(tokenizer):
+    assert len(tokenizer.special_tokens) == 13
+    assert tokenizer.vocab_size % 128 == 0
+    assert tokenizer.pad_token_id == 0
+    for tok, tid in tokenizer.special_tokens.items():
+        assert tid >= tokenizer.base_vocab_size
+
+
+def test_encode_conversation_weights(tokenizer):
+    conv = {"messages": [
+        {"role": "user", "content": "hi"},
+        {"role": "assistant", "content": "hello there"},
+    ]}
+    ids, w = tokenizer.encode_conversation(conv, return_loss_weights=True)
+    assert len(ids) == len(w)
+    # special tokens have zero weight; assistant content has elevated weight
+    assert w[0] == 0.0 and w[1] == 0.0
+    assert max(w) == tokenizer.assistant_loss_weight
+    # role structure round-trips
+    text = tokenizer.decode(ids)
+    assert "<|im_start|>" in text and "<|assistant|>" in text
+
+
+def test_base_dataset_chunks(tokenizer, sample_text, tiny_config):
+    ds = BaseTrainingDataset(sample_text, tokenizer, seq_length=32)
+    assert len(ds) > 0
+    row = ds[0]
+    assert row["input_ids"].shape == (32,)
+    assert torch.equal(row["labels"][:-1], row["input_ids"][1:])
+
+
+def test_conversation_dataset(tokenizer, sample_conversations):
+    ds = ConversationDataset(sample_conversations, tokenizer, seq_length=48)
+    assert len(ds) == 8
+    row = ds[0]
+    assert row["input_ids"].shape == (48,)
+    assert row["labels"].shape == (48,)
+    assert (row["labels"][row["loss_weights"] == 0] == -100).all()
+
+
+def test_setup_datasets_modes(tokenizer, sample_conversations, tiny_config):
+    tiny_config.train_data_path = sample_conversations
+    tiny_config.eval_data_path = ""
+    train, evalds = setup_datasets(tiny_config, tokenizer)
+    assert isinstance(train, ConversationDataset)
+    assert evalds is None
+
+
+def test_setup_datasets_synthetic_fallback(tokenizer, tiny_config):
+    tiny_config.train_data_path = "/nonexistent/file.jsonl"
+    train, _ = setup_datasets(tiny_config, tokenizer)
+    assert isinstance(train, SyntheticDataset)
+
+
+def test_dataloader(tokenizer, sample_conversations, tiny_config):
+    ds = ConversationDataset(sample_conversations, tokenizer,
+                             seq_length=tiny_config.seq_length)
+    dl = create_dataloader(ds, tiny_config, shuffle=True)
+    batch = next(iter(dl))
+    assert batch["input_ids"].shape[0] == tiny_config.micro_batch_size
+
+
+def test_interleaved(tokenizer, sample_conversations):
+    a = ConversationDataset(sample_conversations, tokenizer, 32)
+    b = SyntheticDataset(128, 32, 4)
+    ds = InterleavedDataset(a, b, 0.5)
+    assert len(ds) == len(a) + len(b)
+    _ = ds[0], ds[1], ds[len(ds) - 1]
+
+
+def test_streaming_dataset(tokenizer, sample_text):
+    from luminaai_amd.data import StreamingBaseTrainingDataset
+    ds = StreamingBaseTrainingDataset(sample_text, tokenizer, 32)
+    rows = []
+    for i, row in enumerate(ds):
+        rows.append(row)
+        if i >= 2:
+            break
+    assert rows and rows[0]["input_ids"].shape == (32,)

@@ -1,0 +1,154 @@
+"""CPU reference-op tests: semantics of every fused op vs plain torch autograd
+(the same references the HIP kernels are tested against on GPU)."""
+
+import math
+
+import pytest
+import torch
+import torch.nn.functional as F
+
+from luminaai_amd.ops import reference as ref
+from luminaai_amd.ops import interface as K
+
+
+def test_rmsnorm_fwd_matches_formula():
+    torch.manual_seed(0)
+    x = torch.randn(4, 33, dtype=torch.float64).float()
+    w = torch.randn(33)
+    y = ref.rmsnorm_fwd(x, w, 1e-6)
+    expect = x * torch.rsqrt(x.pow(2).mean(-1, keepdim=True) + 1e-6) * w
+    assert torch.allclose(y, expect, atol=1e-5)
+
+
+def test_rmsnorm_bwd_matches_autograd():
+    torch.manual_seed(1)
+    x = torch.randn(8, 64, requires_grad=True)
+    w = torch.randn(64, requires_grad=True)
+    y = K.rmsnorm(x, w, 1e-6)
+    g = torch.randn_like(y)
+    y.backward(g)
+    x2 = x.detach().clone().requires_grad_(True)
+    w2 = w.detach().clone().requires_grad_(True)
+    y2 = x2 * torch.rsqrt(x2.pow(2).mean(-1, keepdim=True) + 1e-6) * w2
+    y2.backward(g)
+    assert torch.allclose(x.grad, x2.grad, atol=1e-4)
+    assert torch.allclose(w.grad, w2.grad, atol=1e-4)
+
+
+def test_rope_rotation_preserves_norm():
+    cos, sin = ref.rope_cache(16, 8)
+    q = torch.randn(2, 16, 3, 8)
+    k = torch.randn(2, 16, 2, 8)
+    oq, ok = K.rope(q, k, cos, sin)
+    assert torch.allclose(oq.norm(dim=-1), q.norm(dim=-1), atol=1e-4)
+    assert oq.shape == q.shape and ok.shape == k.shape
+    # position 0 is identity
+    assert torch.allclose(oq[:, 0], q[:, 0], atol=1e-5)
+
+
+def test_rope_backward_is_inverse_rotation():
+    cos, sin = ref.rope_cache(8, 4)
+    q = torch.randn(1, 8, 2, 4, requires_grad=True)
+    k = torch.randn(1, 8, 2, 4, requires_grad=True)
+    oq, ok = K.rope(q, k, cos, sin)
+    (oq.sum() + ok.sum()).backward()
+    # gradient of a rotation is the transposed (inverse) rotation of ones
+    g = torch.ones_like(q)
+    gq, gk = K.RoPEFn.apply(g, torch.ones_like(k), cos, sin, None, 0)
+    # rotation by -theta of ones equals backward grad
+    inv_q, _ = ref.rope_apply(g.transpose(1, 2), g.transpose(1, 2), cos, -sin)
+    assert torch.allclose(q.grad, inv_q.transpose(1, 2), atol=1e-4)
+
+
+def test_swiglu_matches_autograd():
+    torch.manual_seed(2)
+    g = torch.randn(16, 32, requires_grad=True)
+    u = torch.randn(16, 32, requires_grad=True)
+    y = K.swiglu(g, u)
+    gy = torch.randn_like(y)
+    y.backward(gy)
+    g2 = g.detach().clone().requires_grad_(True)
+    u2 = u.detach().clone().requires_grad_(True)
+    (F.silu(g2) * u2).backward(gy)
+    assert torch.allclose(g.grad, g2.grad, atol=1e-5)
+    assert torch.allclose(u.grad, u2.grad, atol=1e-5)
+
+
+def test_fused_ce_matches_torch():
+    torch.manual_seed(3)
+    logits = torch.randn(10, 50)
+    labels = torch.randint(0, 50, (10,))
+    labels[3] = -100
+    loss, acc, n = ref.fused_cross_entropy(logits, labels)
+    expect = F.cross_entropy(logits, labels, ignore_index=-100)
+    assert torch.allclose(loss, expect, atol=1e-5)
+    assert n == 9
+
+
+def test_fused_ce_weighted():
+    torch.manual_seed(4)
+    logits = torch.randn(6, 20)
+    labels = torch.randint(0, 20, (6,))
+    w = torch.tensor([2.0, 1.0, 0.0, 1.0, 1.0, 3.0])
+    loss, acc, n = ref.fused_cross_entropy(logits, labels, w)
+    nll = F.cross_entropy(logits, labels, reduction="none")
+    expect = (nll * w).sum() / w.sum()
+    assert torch.allclose(loss, expect, atol=1e-5)
+
+
+def test_topk_gating_normalized():
+    logits = torch.randn(32, 8)
+    wts, idx, probs = ref.topk_gating(logits, 2)
+    assert torch.allclose(wts.sum(-1), torch.ones(32), atol=1e-5)
+    assert (idx >= 0).all() and (idx < 8).all()
+    assert torch.allclose(probs.sum(-1), torch.ones(32), atol=1e-5)
+
+
+def test_load_balancing_loss_uniform_is_one():
+    # perfectly uniform routing -> loss == 1.0 (E * E * (1/E) * (1/E))
+    probs = torch.full((64, 4), 0.25)
+    topi = torch.arange(64).remainder(4).unsqueeze(1)
+    loss = ref.load_balancing_loss(probs, topi, 4)
+    assert abs(float(loss) - 1.0) < 1e-5
+
+
+def test_adamw_step_cpu_matches_torch_adamw():
+    torch.manual_seed(5)
+    p0 = torch.randn(100)
+    g0 = torch.randn(100)
+    # our flat implementation
+    master = p0.clone()
+    m = torch.zeros(100)
+    v = torch.zeros(100)
+    K.adamw_step(master, g0.clone(), m, v, None, lr=1e-2, beta1=0.9,
+                 beta2=0.95, eps=1e-8, wd=0.0, step=1, gnorm_sq=None,
+                 max_norm=0.0)
+    # torch reference
+    p = p0.clone().requires_grad_(True)
+    opt = torch.optim.AdamW([p], lr=1e-2, betas=(0.9, 0.95), eps=1e-8,
+                            weight_decay=0.0)
+    p.grad = g0.clone()
+    opt.step()
+    assert torch.allclose(master, p.detach(), atol=1e-6)
+
+
+def test_adamw_nan_skip():
+    master = torch.ones(10)
+    g = torch.full((10,), float("nan"))
+    m = torch.zeros(10)
+    v = torch.zeros(10)
+    ns = torch.tensor([float("nan")])
+    K.adamw_step(master, g, m, v, None, 1e-2, 0.9, 0.95, 1e-8, 0.0, 1, ns, 1.0)
+    assert torch.allclose(master, torch.ones(10))  # step skipped
+
+
+def test_adamw_clip():
+    master = torch.zeros(4)
+    g = torch.full((4,), 10.0)
+    m = torch.zeros(4)
+    v = torch.zeros(4)
+    ns = (g.pow(2).sum()).reshape(1)
+    K.adamw_step(master, g, m, v, None, lr=1.0, beta1=0.0, beta2=0.0,
+                 eps=1e-8, wd=0.0, step=1, gnorm_sq=ns, max_norm=1.0)
+    # after clip, g_i = 10 * (1/20) = 0.5 each; m = g; v = g^2; update = -lr*m/sqrt(v) ~ -1
+    assert torch.allclose(master, torch.full((4,), -1.0), atol=1e-3)
