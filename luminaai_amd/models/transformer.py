@@ -162,8 +162,10 @@ class RotaryEmbedding(nn.Module):
         self.head_dim = head_dim
         self.theta = theta
         self.max_seq = 0
-        self.register_buffer("cos_cached", torch.empty(0), persistent=False)
-        self.register_buffer("sin_cached", torch.empty(0), persistent=False)
+        # plain attributes, NOT buffers: the tables must stay fp32 even when
+        # the model is cast to bf16 (module.to(dtype) converts buffers).
+        self.cos_cached = torch.empty(0)
+        self.sin_cached = torch.empty(0)
         self._build(max_seq, torch.device("cpu"))
 
     def _build(self, seq_len: int, device):

@@ -16,6 +16,10 @@
 struct MS { float m; float s; };  // online max / sum-exp state
 
 DEV_INLINE MS ms_combine(MS a, MS b) {
+  // a lane that saw no elements carries {-inf, 0}; exp(-inf - -inf) = NaN,
+  // so empty states must pass through untouched.
+  if (!(a.m > -INFINITY)) return b;
+  if (!(b.m > -INFINITY)) return a;
   MS r;
   r.m = fmaxf(a.m, b.m);
   r.s = a.s * __expf(a.m - r.m) + b.s * __expf(b.m - r.m);
