@@ -352,11 +352,12 @@ class MoEFFNLayer(nn.Module):
         bufv = buf[:E * C].view(E, C, h)
 
         # --- grouped expert GEMMs (hipBLASLt strided-batched)
-        gu = torch.bmm(bufv, self.w_gate_up.to(x.dtype))        # [E, C, 2I]
+        gu = ops.interface.expert_bmm(bufv, self.w_gate_up.to(x.dtype))  # [E, C, 2I]
         I = self.intermediate_size
         gu2 = gu.view(E * C, 2 * I)
         act = ops.swiglu(gu2.narrow(1, 0, I), gu2.narrow(1, I, I))
-        y = torch.bmm(act.view(E, C, I), self.w_down.to(x.dtype))  # [E, C, h]
+        y = ops.interface.expert_bmm(act.view(E, C, I),
+                                     self.w_down.to(x.dtype))   # [E, C, h]
 
         # --- weighted combine back to token order
         yf = y.reshape(E * C, h)

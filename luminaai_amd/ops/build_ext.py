@@ -54,6 +54,11 @@ def build(verbose: bool = False):
 
     objs = compile_kernels(verbose=verbose)
     BUILD_DIR.mkdir(exist_ok=True)
+    # ninja only tracks bindings.cpp; force a relink when any kernel .o is
+    # newer than the linked extension.
+    so = BUILD_DIR / "_lumina_hip.so"
+    if so.exists() and any(_newer(Path(o), so) for o in objs):
+        so.unlink()
     module = cpp_extension.load(
         name="_lumina_hip",
         sources=[str(CSRC / "bindings.cpp")],

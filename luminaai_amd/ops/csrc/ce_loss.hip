@@ -44,7 +44,8 @@ __global__ void ce_fwd_kernel(const typename E::storage* __restrict__ logits,
     MS st = {-INFINITY, 0.f};
     float amax_v = -INFINITY;
     int amax_i = 0;
-    if (sizeof(typename E::storage) == 2) {
+    if (sizeof(typename E::storage) == 2 && (V % 8) == 0) {
+      // row base stays 16B-aligned only when V is a multiple of 8
       const ushortx8* xv = reinterpret_cast<const ushortx8*>(x);
       const int nvec = V / 8;
       for (int i = threadIdx.x; i < nvec; i += BLOCK) {

@@ -50,6 +50,10 @@ __global__ void rope_kernel(const typename E::storage* __restrict__ q,
     float x2 = E::load(xin + base + half);
     E::store(xout + base, x1 * c - x2 * sv);
     E::store(xout + base + half, x2 * c + x1 * sv);
+    // odd head_dim (e.g. 1908/12 = 159): the last element passes through
+    if ((D & 1) && d == 0)
+      E::store(xout + row * (int64_t)D + (D - 1),
+               E::load(xin + row * (int64_t)D + (D - 1)));
   }
 }
 

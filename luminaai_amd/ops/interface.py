@@ -137,10 +137,17 @@ def _rope_ref_bshd(q, k, cos, sin, pos, pos_offset, conj):
     c = c.unsqueeze(2)  # [B, S, 1, half]
     s = s.unsqueeze(2)
 
+    D = q.shape[-1]
+    half = D // 2
+
     def rot(x):
         x32 = x.float()
-        x1, x2 = x32.chunk(2, dim=-1)
-        return torch.cat([x1 * c - x2 * s, x2 * c + x1 * s], dim=-1).to(x.dtype)
+        x1 = x32[..., :half]
+        x2 = x32[..., half:2 * half]
+        parts = [x1 * c - x2 * s, x2 * c + x1 * s]
+        if D % 2:  # odd head_dim: last element passes through
+            parts.append(x32[..., 2 * half:])
+        return torch.cat(parts, dim=-1).to(x.dtype)
 
     return rot(q), rot(k)
 
@@ -220,6 +227,40 @@ def fused_cross_entropy(logits: torch.Tensor, labels: torch.Tensor,
         acc = stats[2] / stats[3].clamp_min(1.0)
         return loss, acc, stats[3]
     return ref.fused_cross_entropy(logits2, labels1, w, ignore_index)
+
+
+# ------------------------------------------------------- batched expert GEMM
+class BatchedLinearFn(torch.autograd.Function):
+    """bmm with a safe backward for ROCm.
+
+    torch's built-in BmmBackward computes grad_x = grad @ w.transpose(1, 2)
+    as a strided batched GEMM with a transposed-B operand — that pattern
+    memory-faults in this ROCm hipBLASLt/rocBLAS build for large bf16 batches
+    (verified on MI355X: any K/N, batched-only, transposed-B only). Here the
+    transposed operand is materialised contiguous first (sub-ms copy), and
+    grad_w uses the transposed-A form, which is fine.
+    """
+
+    @staticmethod
+    def forward(ctx, x, w):
+        ctx.save_for_backward(x, w)
+        return torch.bmm(x, w)
+
+    @staticmethod
+    def backward(ctx, go):
+        x, w = ctx.saved_tensors
+        go = go.contiguous()
+        gx = gw = None
+        if ctx.needs_input_grad[0]:
+            gx = torch.bmm(go, w.transpose(1, 2).contiguous())
+        if ctx.needs_input_grad[1]:
+            gw = torch.bmm(x.transpose(1, 2), go)
+        return gx, gw
+
+
+def expert_bmm(x: torch.Tensor, w: torch.Tensor) -> torch.Tensor:
+    """Grouped expert GEMM [E, C, K] x [E, K, N] (hipBLASLt strided-batched)."""
+    return BatchedLinearFn.apply(x, w)
 
 
 # ----------------------------------------------------------- optimizer path

@@ -79,9 +79,16 @@ def rope_apply(q: torch.Tensor, k: torch.Tensor, cos: torch.Tensor, sin: torch.T
     c = cos[pos_offset:pos_offset + S].to(q.dtype)  # [S, D/2]
     s = sin[pos_offset:pos_offset + S].to(q.dtype)
 
+    D = q.shape[-1]
+    half = D // 2
+
     def rot(x):
-        x1, x2 = x.chunk(2, dim=-1)
-        return torch.cat([x1 * c - x2 * s, x2 * c + x1 * s], dim=-1)
+        x1 = x[..., :half]
+        x2 = x[..., half:2 * half]
+        parts = [x1 * c - x2 * s, x2 * c + x1 * s]
+        if D % 2:
+            parts.append(x[..., 2 * half:])
+        return torch.cat(parts, dim=-1)
 
     return rot(q), rot(k)
 
