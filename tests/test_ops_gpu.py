@@ -185,7 +185,8 @@ def test_adamw_step_matches_torch():
         p.grad = g0.float()
         opt.step()
     assert torch.allclose(master, p.detach(), atol=1e-4, rtol=1e-3)
-    assert torch.allclose(w_out.float(), master, atol=1e-2)
+    # w_out is the bf16 rounding of master: rel err up to 2^-8
+    assert torch.allclose(w_out.float(), master, atol=1e-2, rtol=8e-3)
 
 
 def test_adamw_nan_skip_gpu():
