@@ -1,0 +1,135 @@
+"""Incremental-decoding generation engine.
+
+Rebuild of the reference GenerationEngine (/root/reference/Src/Main_Scripts/
+Chat.py:355-465) with the O(L²)-per-token flaw fixed: the reference re-ran the
+full forward for every generated token even though its attention supported a
+KV cache (Chat.py:381, model.py:694-702). Here the prompt is prefilled once
+and each new token does a single-position forward against per-layer KV caches.
+Sampling: repetition penalty over the last `rep_window` tokens, temperature,
+top-k, top-p nucleus, greedy; 4 named sampling modes (Chat.py:60-85)."""
+
+from __future__ import annotations
+
+import time
+from dataclasses import dataclass, field
+from typing import Callable, Dict, List, Optional
+
+import torch
+
+SAMPLING_MODES: Dict[str, Dict] = {
+    "greedy": {"temperature": 0.0, "top_k": 0, "top_p": 1.0},
+    "standard": {"temperature": 0.8, "top_k": 50, "top_p": 0.9},
+    "creative": {"temperature": 1.1, "top_k": 100, "top_p": 0.95},
+    "precise": {"temperature": 0.3, "top_k": 20, "top_p": 0.85},
+}
+
+
+@dataclass
+class GenerationConfig:
+    max_new_tokens: int = 256
+    temperature: float = 0.8
+    top_k: int = 50
+    top_p: float = 0.9
+    repetition_penalty: float = 1.1
+    rep_window: int = 50
+    stop_token_ids: List[int] = field(default_factory=list)
+    max_context: int = 2048
+
+    @classmethod
+    def from_mode(cls, mode: str, **overrides) -> "GenerationConfig":
+        kw = dict(SAMPLING_MODES.get(mode, SAMPLING_MODES["standard"]))
+        kw.update(overrides)
+        return cls(**kw)
+
+
+class GenerationEngine:
+    def __init__(self, model, tokenizer, device: Optional[torch.device] = None):
+        self.model = model
+        self.tokenizer = tokenizer
+        self.device = device or next(model.parameters()).device
+        self.stats = {"tokens_generated": 0, "time_in_generate": 0.0,
+                      "prefill_tokens": 0}
+
+    @torch.no_grad()
+    def generate(self, prompt_ids: List[int],
+                 config: Optional[GenerationConfig] = None,
+                 stream_callback: Optional[Callable[[int], None]] = None) -> List[int]:
+        """Returns the generated token ids (not including the prompt)."""
+        cfg = config or GenerationConfig()
+        t0 = time.perf_counter()
+        self.model.eval()
+        stop = set(cfg.stop_token_ids) | {self.tokenizer.eos_token_id}
+
+        ids = prompt_ids[-cfg.max_context:]
+        x = torch.tensor([ids], dtype=torch.long, device=self.device)
+        caches = self.model.make_kv_caches()
+
+        # ---- prefill: one forward over the whole prompt
+        logits, _, _ = self.model(x, kv_caches=caches)
+        self.stats["prefill_tokens"] += len(ids)
+        generated: List[int] = []
+        recent: List[int] = list(ids)
+
+        for _ in range(cfg.max_new_tokens):
+            next_id = self._sample(logits[0, -1].float(), cfg,
+                                   recent[-cfg.rep_window:])
+            if next_id in stop:
+                break
+            generated.append(next_id)
+            recent.append(next_id)
+            if stream_callback is not None:
+                stream_callback(next_id)
+            if caches[0].seq_len >= cfg.max_context:
+                break
+            # ---- single-token incremental step
+            step = torch.tensor([[next_id]], dtype=torch.long, device=self.device)
+            logits, _, _ = self.model(step, kv_caches=caches)
+
+        self.stats["tokens_generated"] += len(generated)
+        self.stats["time_in_generate"] += time.perf_counter() - t0
+        return generated
+
+    def generate_text(self, prompt: str, config: Optional[GenerationConfig] = None,
+                      stream: bool = False) -> str:
+        ids = self.tokenizer.encode(prompt)
+        out: List[str] = []
+
+        def cb(tok_id):
+            piece = self.tokenizer.decode([tok_id])
+            out.append(piece)
+            if stream:
+                print(piece, end="", flush=True)
+
+        self.generate(ids, config, stream_callback=cb)
+        return "".join(out)
+
+    # ------------------------------------------------------------------
+    def _sample(self, logits: torch.Tensor, cfg: GenerationConfig,
+                recent: List[int]) -> int:
+        # repetition penalty (reference Chat.py:392-398)
+        if cfg.repetition_penalty != 1.0 and recent:
+            idx = torch.tensor(sorted(set(recent)), device=logits.device)
+            sel = logits[idx]
+            logits[idx] = torch.where(sel > 0, sel / cfg.repetition_penalty,
+                                      sel * cfg.repetition_penalty)
+        if cfg.temperature <= 0.0:
+            return int(logits.argmax())
+        logits = logits / cfg.temperature
+        if cfg.top_k and cfg.top_k > 0:
+            kth = torch.topk(logits, min(cfg.top_k, logits.numel())).values[-1]
+            logits = logits.masked_fill(logits < kth, float("-inf"))
+        if cfg.top_p < 1.0:
+            sorted_logits, sorted_idx = torch.sort(logits, descending=True)
+            probs = torch.softmax(sorted_logits, dim=-1)
+            cum = torch.cumsum(probs, dim=-1)
+            cut = cum - probs > cfg.top_p       # keep first token above p
+            sorted_logits = sorted_logits.masked_fill(cut, float("-inf"))
+            logits = torch.full_like(logits, float("-inf")).scatter(
+                0, sorted_idx, sorted_logits)
+        probs = torch.softmax(logits, dim=-1)
+        return int(torch.multinomial(probs, 1))
+
+    def get_stats(self) -> Dict:
+        t = self.stats["time_in_generate"]
+        return dict(self.stats,
+                    tokens_per_sec=self.stats["tokens_generated"] / t if t else 0.0)

@@ -1,0 +1,142 @@
+"""Checkpoint discovery and smart loading for inference.
+
+Rebuild of the reference loaders (/root/reference/Src/Main_Scripts/Chat.py:
+load_checkpoint_smart :132 (prefix stripping), load_zero_shards :163
+(ZeRO-shard merge), infer_config_from_state_dict :219 (architecture from
+tensor shapes), find_latest_checkpoint :301)."""
+
+from __future__ import annotations
+
+import glob
+import os
+import re
+from typing import Dict, List, Optional
+
+import torch
+
+from ..models.transformer import DeepSeekConfig
+
+_STRIP_PREFIXES = ("module.", "_orig_mod.", "model.")
+
+
+def _strip_prefixes(sd: Dict[str, torch.Tensor]) -> Dict[str, torch.Tensor]:
+    out = {}
+    for k, v in sd.items():
+        for p in _STRIP_PREFIXES:
+            if k.startswith(p):
+                k = k[len(p):]
+        out[k] = v
+    return out
+
+
+def load_checkpoint_smart(path: str, map_location="cpu") -> Dict:
+    """Load a checkpoint file; returns {"model_state_dict": ..., ...} with
+    wrapper prefixes stripped. Accepts raw state dicts too."""
+    payload = torch.load(path, map_location=map_location, weights_only=False)
+    if isinstance(payload, dict) and "model_state_dict" in payload:
+        payload["model_state_dict"] = _strip_prefixes(payload["model_state_dict"])
+        return payload
+    if isinstance(payload, dict) and all(
+            torch.is_tensor(v) for v in payload.values()):
+        return {"model_state_dict": _strip_prefixes(payload)}
+    raise ValueError(f"unrecognised checkpoint format in {path}")
+
+
+def load_zero_shards(shard_dir: str, map_location="cpu") -> Dict:
+    """Merge per-rank ZeRO checkpoint shards (rank{N}.pt / *_rank_{N}.pt)
+    into one full state dict. Shards carry disjoint flat-optimizer partitions;
+    the model_state_dict is identical across ranks (stage 1/2) so the merge
+    takes rank0's model and concatenates optimizer shards per group."""
+    patterns = ["*rank*[0-9].pt", "*shard*[0-9].pt"]
+    files: List[str] = []
+    for p in patterns:
+        files.extend(glob.glob(os.path.join(shard_dir, p)))
+    files = sorted(set(files),
+                   key=lambda f: int(re.findall(r"(\d+)", os.path.basename(f))[-1]))
+    if not files:
+        raise FileNotFoundError(f"no shard files in {shard_dir}")
+    shards = [torch.load(f, map_location=map_location, weights_only=False)
+              for f in files]
+    merged = dict(shards[0])
+    merged["model_state_dict"] = _strip_prefixes(shards[0]["model_state_dict"])
+    opt_shards = [s.get("optimizer_state_dict") for s in shards]
+    if all(o is not None for o in opt_shards) and \
+            all("groups" in o for o in opt_shards):
+        groups = []
+        for gi in range(len(opt_shards[0]["groups"])):
+            g0 = dict(opt_shards[0]["groups"][gi])
+            for key in ("exp_avg", "exp_avg_sq", "master"):
+                parts = [o["groups"][gi][key] for o in opt_shards
+                         if o["groups"][gi].get(key) is not None]
+                if parts:
+                    g0[key] = torch.cat(parts)
+            groups.append(g0)
+        merged["optimizer_state_dict"] = dict(opt_shards[0], groups=groups)
+    return merged
+
+
+def infer_config_from_state_dict(sd: Dict[str, torch.Tensor]) -> DeepSeekConfig:
+    """Reconstruct a DeepSeekConfig from tensor shapes
+    (reference Chat.py:219-300)."""
+    vocab_size, hidden = sd["embed_tokens.weight"].shape
+    layer_ids = {int(m.group(1)) for k in sd
+                 if (m := re.match(r"layers\.(\d+)\.", k))}
+    num_layers = max(layer_ids) + 1 if layer_ids else 0
+
+    qkv = sd["layers.0.attention.qkv_proj.weight"]      # [q + 2kv, h]
+    o = sd["layers.0.attention.o_proj.weight"]          # [h, q]
+    q_size = o.shape[1]
+    kv_size = (qkv.shape[0] - q_size) // 2
+
+    # head_dim from the RoPE-compatible assumption head_dim = hidden/num_heads;
+    # num_heads is the largest divisor consistent with q_size == hidden.
+    # q_size == hidden always holds for this family, so take gcd-style probe:
+    num_heads = None
+    for h in (64, 48, 40, 32, 24, 20, 16, 12, 8, 6, 4, 2, 1):
+        if hidden % h == 0 and q_size % h == 0 and kv_size % (hidden // h) == 0:
+            num_heads = h
+            break
+    num_heads = num_heads or 1
+    head_dim = hidden // num_heads
+    num_kv_heads = max(1, kv_size // head_dim)
+
+    use_moe = any(".w_gate_up" in k for k in sd)
+    num_experts = 8
+    moe_inter = None
+    if use_moe:
+        for k, v in sd.items():
+            if k.endswith(".w_gate_up"):
+                num_experts, _, i2 = v.shape
+                moe_inter = i2 // 2
+                break
+    inter = moe_inter
+    for k, v in sd.items():
+        if k.endswith("ffn.gate_up_proj.weight"):
+            inter = v.shape[0] // 2
+            break
+    use_mod = any("mod_router" in k for k in sd)
+    tie = "lm_head.weight" not in sd or \
+        sd["lm_head.weight"].data_ptr() == sd["embed_tokens.weight"].data_ptr()
+
+    return DeepSeekConfig(
+        vocab_size=vocab_size, hidden_size=hidden, num_layers=num_layers,
+        num_heads=num_heads, num_kv_heads=num_kv_heads,
+        intermediate_size=inter, use_moe=use_moe, num_experts=num_experts,
+        use_mod=use_mod, tie_word_embeddings=tie,
+        moe_pattern="all" if use_moe and all(
+            f"layers.{i}.ffn.w_gate_up" in sd for i in range(num_layers))
+        else ("every_2nd" if use_moe else "all"),
+    )
+
+
+def find_latest_checkpoint(search_dirs: Optional[List[str]] = None) -> Optional[str]:
+    """Newest .pt under checkpoints/ or experiments/ (reference Chat.py:301)."""
+    search_dirs = search_dirs or ["checkpoints", "experiments", "."]
+    candidates: List[str] = []
+    for d in search_dirs:
+        candidates.extend(glob.glob(os.path.join(d, "**", "*.pt"),
+                                    recursive=True))
+    candidates = [c for c in candidates if not c.endswith(".tmp")]
+    if not candidates:
+        return None
+    return max(candidates, key=os.path.getmtime)
