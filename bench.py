@@ -43,7 +43,8 @@ def build_config(args):
     cfg.micro_batch_size = args.micro_batch
     cfg.gradient_accumulation_steps = args.accum
     cfg.seq_length = args.seq_len or cfg.seq_length
-    cfg.gradient_checkpointing = not args.no_checkpointing
+    # 288 GB HBM3E: the b1 flagship never needs activation recompute
+    cfg.gradient_checkpointing = args.checkpointing
     cfg.zero_stage = args.zero if args.zero is not None else \
         (2 if comm.env_world_size() > 1 else 0)
     cfg.eval_every_n_batches = 0
@@ -56,12 +57,16 @@ def main():
     ap.add_argument("--gpus", type=int, default=1)
     ap.add_argument("--steps", type=int, default=10)
     ap.add_argument("--warmup", type=int, default=3)
-    ap.add_argument("--micro-batch", type=int, default=4)
+    ap.add_argument("--micro-batch", type=int, default=8)
     ap.add_argument("--accum", type=int, default=1)
     ap.add_argument("--seq-len", type=int, default=None)
     ap.add_argument("--preset", type=str, default=None)
     ap.add_argument("--zero", type=int, default=None)
-    ap.add_argument("--no-checkpointing", action="store_true")
+    ap.add_argument("--ep", type=int, default=None,
+                    help="expert-parallel degree (default: world if it "
+                         "divides num_experts)")
+    ap.add_argument("--checkpointing", action="store_true",
+                    help="enable activation recompute (off by default)")
     args = ap.parse_args()
 
     distributed = comm.init_distributed()
@@ -71,6 +76,11 @@ def main():
         if torch.cuda.is_available() else torch.device("cpu")
 
     cfg = build_config(args)
+    ep = args.ep if args.ep is not None else \
+        (world if cfg.use_moe and world > 1
+         and cfg.num_experts % world == 0 else 1)
+    from luminaai_amd.parallel.mesh import init_mesh
+    init_mesh(ep if world > 1 else 1)
     torch.manual_seed(cfg.seed + rank)
     tok = ConversationTokenizer(max_length=cfg.seq_length)
     model_cfg = config_to_deepseek_config(cfg)
@@ -138,8 +148,9 @@ def main():
                 "model": args.preset or "b1_moe_8e_top2_8B",
                 "global_batch": micro * cfg.gradient_accumulation_steps * world,
                 "seq_len": cfg.seq_length,
-                "parallelism": f"dp{world}" + (f"_zero{cfg.zero_stage}"
-                                               if cfg.zero_stage else ""),
+                "parallelism": f"dp{world}"
+                + (f"_zero{cfg.zero_stage}" if cfg.zero_stage else "")
+                + (f"_ep{ep}" if ep > 1 else ""),
                 "micro_batch": micro,
                 "grad_accum": cfg.gradient_accumulation_steps,
                 "final_loss": loss,

@@ -300,10 +300,25 @@ class MoEFFNLayer(nn.Module):
         self.routing_noise_std = config.routing_noise_std
         self.expert_dropout = 0.0
 
+        # expert parallelism: local shard of the expert weights, token
+        # all-to-all over the EP group at forward (parallel/expert_parallel.py)
+        from ..parallel.mesh import get_mesh
+        mesh = get_mesh()
+        if mesh is not None and mesh.ep_size > 1 and \
+                config.num_experts % mesh.ep_size == 0:
+            self.ep_size = mesh.ep_size
+            self.ep_group = mesh.ep_group
+        else:
+            self.ep_size = 1
+            self.ep_group = None
+        self.num_local_experts = config.num_experts // self.ep_size
+
         self.gate = nn.Linear(h, config.num_experts, bias=False)
         I = config.intermediate_size
-        self.w_gate_up = nn.Parameter(torch.empty(config.num_experts, h, 2 * I))
-        self.w_down = nn.Parameter(torch.empty(config.num_experts, I, h))
+        self.w_gate_up = nn.Parameter(
+            torch.empty(self.num_local_experts, h, 2 * I))
+        self.w_down = nn.Parameter(
+            torch.empty(self.num_local_experts, I, h))
 
         # routing stats (device tensors; read via get_routing_stats)
         self.register_buffer("_usage_counts",
@@ -351,13 +366,29 @@ class MoEFFNLayer(nn.Module):
         buf = torch.index_put(buf, (dest,), xf[tok])
         bufv = buf[:E * C].view(E, C, h)
 
+        # --- EP token exchange: [E, C, h] -> peers owning each expert shard.
+        # Shapes are static (capacity-bucketed), so the all-to-all needs no
+        # length metadata and maps 1:1 onto the 7 xGMI links.
+        EL = self.num_local_experts
+        if self.ep_size > 1:
+            from ..parallel.expert_parallel import all_to_all
+            bufv = all_to_all(bufv, self.ep_group)      # [ep*EL, C, h]
+            bufv = bufv.view(self.ep_size, EL, C, h).transpose(0, 1) \
+                .reshape(EL, self.ep_size * C, h)       # tokens per local expert
+
         # --- grouped expert GEMMs (hipBLASLt strided-batched)
-        gu = ops.interface.expert_bmm(bufv, self.w_gate_up.to(x.dtype))  # [E, C, 2I]
+        gu = ops.interface.expert_bmm(bufv, self.w_gate_up.to(x.dtype))
         I = self.intermediate_size
-        gu2 = gu.view(E * C, 2 * I)
+        gu2 = gu.reshape(-1, 2 * I)
         act = ops.swiglu(gu2.narrow(1, 0, I), gu2.narrow(1, I, I))
-        y = ops.interface.expert_bmm(act.view(E, C, I),
-                                     self.w_down.to(x.dtype))   # [E, C, h]
+        y = ops.interface.expert_bmm(act.view(EL, -1, I),
+                                     self.w_down.to(x.dtype))
+
+        if self.ep_size > 1:
+            from ..parallel.expert_parallel import all_to_all
+            y = y.view(EL, self.ep_size, C, h).transpose(0, 1) \
+                .reshape(self.ep_size * EL, C, h)       # back to [E, C, h] order
+            y = all_to_all(y, self.ep_group)
 
         # --- weighted combine back to token order
         yf = y.reshape(E * C, h)
@@ -398,6 +429,11 @@ class MoEFFNLayer(nn.Module):
     def add_expert(self, noise_std: float = 0.01):
         """Append one expert initialised to the mean of existing experts + noise
         (reference trainer.py:1337-1376) and grow the gate."""
+        if self.ep_size > 1:
+            raise RuntimeError("dynamic expert add/prune is not supported "
+                               "under expert parallelism (EP shards are "
+                               "fixed-size); run with ep_size=1 to evolve "
+                               "the architecture")
         E, h, I2 = self.w_gate_up.shape
         new_gu = self.w_gate_up.mean(0, keepdim=True) + \
             torch.randn(1, h, I2, device=self.w_gate_up.device,
@@ -413,11 +449,15 @@ class MoEFFNLayer(nn.Module):
         self.gate.weight.data[:E] = old_gate.weight.data
         self.gate.weight.data[E] = old_gate.weight.data.mean(0)
         self.num_experts = E + 1
+        self.num_local_experts = E + 1
         self._usage_counts = torch.zeros(E + 1, device=self._usage_counts.device)
 
     @torch.no_grad()
     def prune_expert(self, idx: int):
         """Remove expert `idx` and its gate row (reference trainer.py:1378-1448)."""
+        if self.ep_size > 1:
+            raise RuntimeError("dynamic expert add/prune is not supported "
+                               "under expert parallelism")
         E = self.num_experts
         assert E > 1 and 0 <= idx < E
         keep = [i for i in range(E) if i != idx]
@@ -431,6 +471,7 @@ class MoEFFNLayer(nn.Module):
                               dtype=old_gate.weight.dtype)
         self.gate.weight.data.copy_(old_gate.weight.data[kt])
         self.num_experts = E - 1
+        self.num_local_experts = E - 1
         self.top_k = min(self.top_k, self.num_experts)
         self._usage_counts = torch.zeros(E - 1, device=self._usage_counts.device)
 

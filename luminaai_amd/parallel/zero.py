@@ -43,13 +43,14 @@ class ZeroEngine:
     def __init__(self, optimizer: "FlatAdamW", stage: int = 0,
                  bucket_bytes: int = 50_000_000,
                  overlap_comm: bool = True,
-                 process_group=None):
+                 process_group=None, mesh=None):
         assert stage in (0, 1, 2), "ZeRO-3 not implemented yet"
         self.opt = optimizer
         self.stage = stage
         self.bucket_bytes = bucket_bytes
         self.overlap = overlap_comm
         self.pg = process_group
+        self.mesh = mesh                    # ParallelMesh when EP is active
         self.world = comm.get_world_size()
         self.rank = comm.get_rank()
         self._hooks = []
@@ -61,6 +62,12 @@ class ZeroEngine:
         if self.world > 1 and self.overlap and stage in (0, 1):
             self._install_hooks()
 
+    def _dp_groups(self):
+        return [g for g in self.opt.groups if g.comm == "dp"]
+
+    def _expert_groups(self):
+        return [g for g in self.opt.groups if g.comm == "expert"]
+
     # ------------------------------------------------------------- hooks
     def _install_hooks(self):
         """Bucketed all-reduce launched as grads become ready during backward.
@@ -68,7 +75,7 @@ class ZeroEngine:
         when every param mapping into it has accumulated its grad."""
         self.sync_enabled = True
         plan = []  # (group, lo, hi, param_ids)
-        for g in self.opt.groups:
+        for g in self._dp_groups():
             lo = 0
             cur_ids = []
             cur_hi = 0
@@ -100,7 +107,7 @@ class ZeroEngine:
         for bi, (g, lo, hi, ids) in enumerate(plan):
             for pid in ids:
                 param_to_bucket[pid] = bi
-        for g in self.opt.groups:
+        for g in self._dp_groups():
             for p in g.params:
                 h = p.register_post_accumulate_grad_hook(
                     make_hook(param_to_bucket[id(p)]))
@@ -129,10 +136,10 @@ class ZeroEngine:
                 self._works.clear()
                 self.reset_bucket_state()
             else:
-                for g in self.opt.groups:
+                for g in self._dp_groups():
                     dist.all_reduce(g.flat_g, group=self.pg)
         else:  # stage 2: reduce-scatter into this rank's shard
-            for g in self.opt.groups:
+            for g in self._dp_groups():
                 shard = g.flat_g[g.shard_lo:g.shard_hi]
                 if self._backend() == "gloo":
                     # gloo lacks reduce_scatter; tests-only fallback
@@ -141,16 +148,37 @@ class ZeroEngine:
                     tmp = torch.empty_like(shard)
                     dist.reduce_scatter_tensor(tmp, g.flat_g, group=self.pg)
                     shard.copy_(tmp)
+        # expert grads: replicas of the same expert shard live across the
+        # expert-dp group (no-op when every replica set has one member)
+        if self.mesh is not None and self.mesh.expert_dp_size > 1:
+            for g in self._expert_groups():
+                dist.all_reduce(g.flat_g, group=self.mesh.expert_dp_group)
 
     def global_grad_norm_sq(self) -> Optional[torch.Tensor]:
         if self.opt.max_grad_norm <= 0:
             return None
+        from ..ops import interface as K
         shard_only = self.stage == 2
-        ns = self.opt.local_grad_norm_sq(shard_only=shard_only)
-        if self.world > 1 and shard_only:
+        ns = None
+        for g in self._dp_groups():
+            t = g.update_grad() if shard_only else g.flat_g
+            n = K.l2norm_sq(t)
+            ns = n if ns is None else ns + n
+        if ns is not None and self.world > 1 and shard_only:
             dist.all_reduce(ns, group=self.pg)
-        # stages 0/1: grads are already globally reduced (summed); the norm of
-        # the summed grad is what clipping applies to (after grad_scale).
+        # stages 0/1: dp grads are already globally reduced (summed); the norm
+        # of the summed grad is what clipping applies to (after grad_scale).
+        exp = self._expert_groups()
+        if exp:
+            ns_e = None
+            for g in exp:
+                n = K.l2norm_sq(g.flat_g)
+                ns_e = n if ns_e is None else ns_e + n
+            if self.mesh is not None and self.mesh.ep_size > 1:
+                # each rank holds E/ep experts (post expert-dp reduce) ->
+                # summing over ONE ep group covers every expert exactly once
+                dist.all_reduce(ns_e, group=self.mesh.ep_group)
+            ns = ns_e if ns is None else ns + ns_e
         return ns
 
     def step(self, grad_scale: float = 1.0):
@@ -160,7 +188,7 @@ class ZeroEngine:
         self.opt.step(grad_scale=grad_scale, norm_sq=norm_sq,
                       shard_only=self.stage == 2)
         if self.stage in (1, 2) and self.world > 1:
-            for g in self.opt.groups:
+            for g in self._dp_groups():
                 shard = g.update_weight_out()
                 if self._backend() == "gloo":
                     chunks = list(g.weight_view().chunk(self.world))
@@ -182,14 +210,23 @@ class ZeroEngine:
 
     # ------------------------------------------------------------- misc
     def broadcast_parameters(self):
-        """Rank-0 weights to all (initial sync)."""
+        """Initial weight sync: replicated (dp) params from rank 0 to all;
+        EP-sharded expert params from the first rank of each expert-dp
+        replica set (they differ across EP ranks by design)."""
         if self.world <= 1:
             return
-        for g in self.opt.groups:
+        for g in self._dp_groups():
             dist.broadcast(g.weight_view(), src=0, group=self.pg)
             if not g._master_is_params:
                 g.master.copy_(
                     g.weight_view()[g.shard_lo:g.shard_hi].float())
+        if self.mesh is not None and self.mesh.expert_dp_size > 1:
+            pg = self.mesh.expert_dp_group
+            src = dist.get_process_group_ranks(pg)[0]
+            for g in self._expert_groups():
+                dist.broadcast(g.weight_view(), src=src, group=pg)
+                if not g._master_is_params:
+                    g.master.copy_(g.weight_view().float())
 
     def remove_hooks(self):
         for h in self._hooks:

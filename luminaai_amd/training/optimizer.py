@@ -35,14 +35,20 @@ NO_DECAY_KEYWORDS = ("bias", "norm", "gate.weight", "mod_router")
 ALIGN = 256
 
 
-def split_decay_groups(model: nn.Module) -> List[Dict]:
-    """Two param groups: weight-decayed matrices vs norms/biases/router gates
-    (reference backend_fsdp.py:220-252 decay grouping)."""
-    decay, no_decay = [], []
+def split_decay_groups(model: nn.Module, ep_active: bool = False) -> List[Dict]:
+    """Param groups: weight-decayed matrices vs norms/biases/router gates
+    (reference backend_fsdp.py:220-252 decay grouping). When expert
+    parallelism is active, EP-sharded expert weights form their own group
+    (comm='expert'): their grads are NOT replicated across the EP group and
+    must not join the global DP all-reduce."""
+    from ..parallel.expert_parallel import is_expert_param
+    decay, no_decay, expert = [], [], []
     for name, p in model.named_parameters():
         if not p.requires_grad:
             continue
-        if any(k in name for k in NO_DECAY_KEYWORDS) or p.dim() <= 1:
+        if ep_active and is_expert_param(name):
+            expert.append(p)
+        elif any(k in name for k in NO_DECAY_KEYWORDS) or p.dim() <= 1:
             no_decay.append(p)
         else:
             decay.append(p)
@@ -51,6 +57,9 @@ def split_decay_groups(model: nn.Module) -> List[Dict]:
         groups.append({"params": decay, "weight_decay": None})  # default wd
     if no_decay:
         groups.append({"params": no_decay, "weight_decay": 0.0})
+    if expert:
+        groups.append({"params": expert, "weight_decay": None,
+                       "comm": "expert"})
     return groups
 
 
@@ -58,10 +67,12 @@ class _FlatGroup:
     """One parameter group flattened into contiguous (padded) buffers."""
 
     def __init__(self, params: List[torch.Tensor], lr: float, weight_decay: float,
-                 shard_rank: int = 0, shard_world: int = 1):
+                 shard_rank: int = 0, shard_world: int = 1,
+                 comm: str = "dp"):
         self.params = params
         self.lr = lr
         self.weight_decay = weight_decay
+        self.comm = comm                  # "dp" (replicated) | "expert" (EP-sharded)
         self.shard_rank = shard_rank
         self.shard_world = shard_world
         self.numel = sum(p.numel() for p in params)
@@ -138,9 +149,10 @@ class FlatAdamW:
     def __init__(self, model_or_groups, lr: float = 1e-4,
                  betas: Tuple[float, float] = (0.9, 0.95), eps: float = 1e-8,
                  weight_decay: float = 0.01, max_grad_norm: float = 1.0,
-                 shard_rank: int = 0, shard_world: int = 1):
+                 shard_rank: int = 0, shard_world: int = 1,
+                 ep_active: bool = False):
         if isinstance(model_or_groups, nn.Module):
-            groups = split_decay_groups(model_or_groups)
+            groups = split_decay_groups(model_or_groups, ep_active=ep_active)
         else:
             groups = list(model_or_groups)
         self.defaults = dict(lr=lr, betas=betas, eps=eps, weight_decay=weight_decay)
@@ -150,11 +162,14 @@ class FlatAdamW:
         self.shard_rank = shard_rank
         self.shard_world = shard_world
         self.step_count = 0
+        # expert groups are EP-sharded by construction -> no ZeRO shard on top
         self.groups: List[_FlatGroup] = [
             _FlatGroup(list(g["params"]), lr=g.get("lr", lr),
                        weight_decay=(weight_decay if g.get("weight_decay") is None
                                      else g["weight_decay"]),
-                       shard_rank=shard_rank, shard_world=shard_world)
+                       shard_rank=0 if g.get("comm") == "expert" else shard_rank,
+                       shard_world=1 if g.get("comm") == "expert" else shard_world,
+                       comm=g.get("comm", "dp"))
             for g in groups
         ]
         self._last_norm_sq: Optional[torch.Tensor] = None

@@ -64,16 +64,21 @@ class Trainer:
             if torch.cuda.is_available() else torch.device("cpu")
         self.model = self.precision.cast_model(model).to(self.device)
 
+        from ..parallel.mesh import get_mesh
+        mesh = get_mesh()
+        ep_active = mesh is not None and mesh.ep_size > 1
+        self.mesh = mesh
         self.optimizer = FlatAdamW(
             self.model, lr=config.learning_rate,
             betas=(0.9, 0.95), eps=1e-8,
             weight_decay=config.weight_decay, max_grad_norm=1.0,
             shard_rank=comm.get_rank() if config.zero_stage in (1, 2) else 0,
-            shard_world=comm.get_world_size() if config.zero_stage in (1, 2) else 1)
+            shard_world=comm.get_world_size() if config.zero_stage in (1, 2) else 1,
+            ep_active=ep_active)
         self.engine = engine if engine is not None else ZeroEngine(
             self.optimizer, stage=min(config.zero_stage, 2),
             bucket_bytes=config.reduce_bucket_size,
-            overlap_comm=config.overlap_comm)
+            overlap_comm=config.overlap_comm, mesh=mesh)
         if comm.is_distributed():
             self.engine.broadcast_parameters()
 
@@ -254,7 +259,14 @@ class Trainer:
         self.scheduler = create_scheduler(self.optimizer, self.config, total_steps)
 
     def save_checkpoint(self, is_best: bool = False, tag: Optional[str] = None) -> str:
-        if comm.get_rank() != 0:
+        if self.mesh is not None and self.mesh.ep_size > 1:
+            # every EP rank holds distinct experts: dp_rank 0 of each EP
+            # slot writes its own shard file (merge via inference.loader)
+            if self.mesh.dp_rank != 0:
+                return ""
+            tag = (tag or f"checkpoint_step_{self.global_step}") + \
+                f"_ep_rank_{self.mesh.ep_rank}"
+        elif comm.get_rank() != 0:
             return ""
         path = self.checkpoints.save_checkpoint(
             self.model, self.optimizer, self.scheduler,
