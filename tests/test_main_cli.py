@@ -1,0 +1,58 @@
+"""End-to-end CLI test: tiny synthetic training run through main()."""
+
+import json
+import os
+
+import pytest
+
+
+def test_main_synthetic_run(tmp_path, monkeypatch):
+    monkeypatch.chdir(tmp_path)
+    from luminaai_amd.main import main
+    result = main([
+        "--preset", "debug", "--synthetic-steps", "3",
+        "--experiment-name", "cli_test", "--precision", "fp32",
+        "--micro-batch", "2", "--accum", "1", "--seq-len", "32",
+        "--set", "hidden_size=64", "--set", "num_layers=2",
+        "--set", "num_heads=4", "--set", "num_kv_heads=2",
+        "--set", "vocab_size=512", "--set", "intermediate_size=128",
+        "--set", "num_workers=0", "--set", "use_moe=false",
+        "--set", "use_mod=false", "--set", "gradient_checkpointing=false",
+        "--set", "eval_every_n_batches=0", "--set", "save_every_n_batches=0",
+    ])
+    assert result["global_step"] >= 3
+    exp = tmp_path / "experiments" / "cli_test"
+    assert (exp / "config.yaml").exists()
+    assert (exp / "training_summary.json").exists()
+    summary = json.loads((exp / "training_summary.json").read_text())
+    assert summary["global_step"] >= 3
+    assert (exp / "training_report.html").exists()
+
+
+def test_main_data_run(tmp_path, monkeypatch):
+    monkeypatch.chdir(tmp_path)
+    from luminaai_amd.utils import generate_sample_data
+    data = str(tmp_path / "train.jsonl")
+    generate_sample_data(data, n=8)
+    from luminaai_amd.main import main
+    result = main([
+        "--preset", "debug", "--train-data", data,
+        "--experiment-name", "cli_data_test", "--precision", "fp32",
+        "--epochs", "1", "--micro-batch", "2", "--accum", "1",
+        "--seq-len", "32",
+        "--set", "hidden_size=64", "--set", "num_layers=2",
+        "--set", "num_heads=4", "--set", "num_kv_heads=2",
+        "--set", "vocab_size=512", "--set", "intermediate_size=128",
+        "--set", "num_workers=0", "--set", "use_moe=false",
+        "--set", "use_mod=false", "--set", "gradient_checkpointing=false",
+        "--set", "eval_every_n_batches=0", "--set", "save_every_n_batches=0",
+        "--set", "eval_data_path=",
+    ])
+    assert result["global_step"] > 0
+
+
+def test_unknown_config_field_rejected():
+    from luminaai_amd.main import main
+    with pytest.raises(SystemExit):
+        main(["--preset", "debug", "--set", "not_a_field=1",
+              "--synthetic-steps", "1"])
