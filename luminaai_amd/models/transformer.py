@@ -349,22 +349,12 @@ class MoEFFNLayer(nn.Module):
 
         aux = ref_ops.load_balancing_loss(probs, topi, E) * self.load_balancing_weight
 
-        # --- capacity-bucketed dispatch (no host sync)
+        # --- capacity-bucketed dispatch: gather-only plan (ops/interface.py
+        # MoERoutingPlan — no atomics, no host sync, hipGraph-capturable)
         C = max(1, int(math.ceil(N * k / E * self.capacity_factor)))
-        flat_e = topi.reshape(-1)                       # [Nk]
-        order = torch.argsort(flat_e, stable=True)      # [Nk]
-        tok = torch.div(order, k, rounding_mode="floor")
-        counts = torch.bincount(flat_e, minlength=E)    # [E]
-        offs = torch.cumsum(counts, 0) - counts
-        sorted_e = flat_e[order]
-        pos = torch.arange(N * k, device=x.device) - offs[sorted_e]
-        valid = pos < C
-        dest = torch.where(valid, sorted_e * C + pos,
-                           torch.full_like(pos, E * C))
-
-        buf = x.new_zeros(E * C + 1, h)
-        buf = torch.index_put(buf, (dest,), xf[tok])
-        bufv = buf[:E * C].view(E, C, h)
+        plan = ops.interface.moe_routing_plan(topi, E, C)
+        counts = plan.counts
+        bufv = ops.interface.moe_dispatch(xf, plan).view(E, C, h)
 
         # --- EP token exchange: [E, C, h] -> peers owning each expert shard.
         # Shapes are static (capacity-bucketed), so the all-to-all needs no
@@ -390,11 +380,9 @@ class MoEFFNLayer(nn.Module):
                 .reshape(self.ep_size * EL, C, h)       # back to [E, C, h] order
             y = all_to_all(y, self.ep_group)
 
-        # --- weighted combine back to token order
+        # --- weighted combine back to token order (gather + k-reduce)
         yf = y.reshape(E * C, h)
-        gathered = yf[dest.clamp_max(E * C - 1)]        # [Nk, h]
-        w_sorted = (topw.reshape(-1)[order] * valid.float()).to(x.dtype)
-        out = x.new_zeros(N, h).index_add(0, tok, gathered * w_sorted.unsqueeze(1))
+        out = ops.interface.moe_combine(yf, topw.reshape(-1), plan)
 
         # routing stats (device-side, no sync)
         if not torch.jit.is_scripting():
