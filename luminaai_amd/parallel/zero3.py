@@ -1,0 +1,513 @@
+"""ZeRO-3: parameter + gradient + optimizer-state sharding on RCCL/xGMI.
+
+Replaces the reference's DeepSpeed ZeRO-3 / FSDP FULL_SHARD / ColossalAI
+Gemini paths (reference backend_deepspeed.py:129-165, backend_fsdp.py:151-198,
+vendored zero/gemini) with one engine built for this framework's flat-buffer
+model:
+
+- sharding unit = one TransformerBlock (plus a root unit: embeddings /
+  final norm / untied head) — each unit's params live as views into a
+  RESIZABLE flat bf16 buffer; only the rank's 1/world shard (bf16 weight +
+  fp32 master/m/v + bf16 grad accumulator) persists.
+- forward: a pre-hook all-gathers the unit's flat buffer from the bf16 weight
+  shards (prefetching the next unit on a side stream), a post-hook frees the
+  full buffer (storage resize-to-0, so autograd-saved views stay valid).
+- backward: weights re-materialise either via the module's
+  full_backward_pre_hook (plain path) or via the forward pre-hook firing
+  inside activation-recompute (checkpointed path — recompute is detected with
+  torch._C._current_graph_task_id(), which is only >= 0 inside a backward
+  pass, so the free-after-forward hook knows not to free mid-recompute).
+- after a unit's last param grad accumulates, the full grad buffer is
+  reduce-scattered into the persistent shard accumulator and freed; gradient
+  accumulation over micro-batches sums in the shard (same bf16-accumulate
+  semantics as the ZeRO-1/2 engine).
+- optimizer: the fused HIP clip+AdamW kernel updates each shard in place
+  (master fp32 -> bf16 weight shard); there is NO post-step weight
+  all-gather — the next forward's unit gathers are the re-materialisation.
+
+xGMI sizing: unit buffers for the b-series blocks are 30-300 MB — each
+all-gather is one fully-connected exchange over the 7 links; the per-unit
+pipeline (prefetch next while computing current) keeps links and CUs busy.
+"""
+
+from __future__ import annotations
+
+import math
+from typing import Dict, List, Optional
+
+import torch
+import torch.distributed as dist
+import torch.nn as nn
+
+from ..ops import interface as K
+from . import comm
+
+ALIGN = 256
+
+
+def _in_backward() -> bool:
+    """True inside a backward pass (including checkpoint recompute)."""
+    try:
+        return torch._C._current_graph_task_id() != -1
+    except AttributeError:
+        return False
+
+
+def _decay_split(named_params):
+    from ..training.optimizer import NO_DECAY_KEYWORDS
+    decay, no_decay = [], []
+    for name, p in named_params:
+        if not p.requires_grad:
+            continue
+        if any(k in name for k in NO_DECAY_KEYWORDS) or p.dim() <= 1:
+            no_decay.append(p)
+        else:
+            decay.append(p)
+    return decay, no_decay
+
+
+class _Zero3Segment:
+    """One (params, weight_decay) flat segment of a sharding unit."""
+
+    def __init__(self, params: List[torch.Tensor], lr: float,
+                 weight_decay: float, rank: int, world: int):
+        self.params = params
+        self.lr = lr
+        self.weight_decay = weight_decay
+        self.rank = rank
+        self.world = world
+        self.numel = sum(p.numel() for p in params)
+        q = ALIGN * world
+        self.padded = (self.numel + q - 1) // q * q
+        self.shard_size = self.padded // world
+        lo = rank * self.shard_size
+        hi = lo + self.shard_size
+
+        device = params[0].device
+        self.dtype = params[0].dtype
+        self.flat_w = torch.zeros(self.padded, device=device, dtype=self.dtype)
+        self.flat_g = torch.zeros(self.padded, device=device, dtype=self.dtype)
+        self.offsets = []
+        off = 0
+        for p in params:
+            n = p.numel()
+            self.flat_w[off:off + n].copy_(p.data.reshape(-1).to(self.dtype))
+            p.data = self.flat_w[off:off + n].view(p.shape)
+            p.grad = self.flat_g[off:off + n].view(p.shape)
+            self.offsets.append((off, n))
+            off += n
+
+        # persistent shard state
+        self.w_shard = self.flat_w[lo:hi].clone()
+        self.master = self.w_shard.float().clone()
+        self.g_shard = torch.zeros_like(self.w_shard)
+        self.m = torch.zeros(self.shard_size, device=device, dtype=torch.float32)
+        self.v = torch.zeros_like(self.m)
+
+    # -------------------------------------------------- storage management
+    def weights_alloc(self) -> bool:
+        return self.flat_w.untyped_storage().size() > 0
+
+    def free_weights(self):
+        self.flat_w.untyped_storage().resize_(0)
+
+    def free_grads(self):
+        self.flat_g.untyped_storage().resize_(0)
+
+    def alloc_grads(self):
+        st = self.flat_g.untyped_storage()
+        if st.size() == 0:
+            st.resize_(self.padded * self.flat_g.element_size())
+        self.flat_g.zero_()
+
+    def gather(self, pg=None):
+        st = self.flat_w.untyped_storage()
+        if st.size() > 0:
+            return
+        st.resize_(self.padded * self.flat_w.element_size())
+        if self.world == 1:
+            self.flat_w.copy_(self.w_shard)
+        elif dist.get_backend(pg) == "gloo":
+            chunks = list(self.flat_w.chunk(self.world))
+            dist.all_gather(chunks, self.w_shard.contiguous(), group=pg)
+        else:
+            dist.all_gather_into_tensor(self.flat_w, self.w_shard, group=pg)
+
+    def reduce_into_shard(self, pg=None):
+        """flat_g -> += g_shard (sum over ranks), then free flat_g."""
+        lo = self.rank * self.shard_size
+        hi = lo + self.shard_size
+        if self.world == 1:
+            self.g_shard.add_(self.flat_g[lo:hi])
+        elif dist.get_backend(pg) == "gloo":
+            dist.all_reduce(self.flat_g, group=pg)
+            self.g_shard.add_(self.flat_g[lo:hi])
+        else:
+            tmp = torch.empty_like(self.g_shard)
+            dist.reduce_scatter_tensor(tmp, self.flat_g, group=pg)
+            self.g_shard.add_(tmp)
+        self.free_grads()
+
+    def step(self, step_count: int, norm_sq, max_norm: float,
+             grad_scale: float):
+        K.adamw_step(self.master, self.g_shard, self.m, self.v, self.w_shard,
+                     self.lr, 0.9, 0.95, 1e-8, self.weight_decay, step_count,
+                     norm_sq, max_norm, grad_scale)
+
+    def refresh_from_shard(self, pg=None):
+        """w_shard -> flat_w (persistent units after an optimizer step)."""
+        if self.flat_w.untyped_storage().size() == 0:
+            return
+        if self.world == 1:
+            lo = self.rank * self.shard_size
+            self.flat_w[lo:lo + self.shard_size].copy_(self.w_shard)
+        elif dist.get_backend(pg) == "gloo":
+            chunks = list(self.flat_w.chunk(self.world))
+            dist.all_gather(chunks, self.w_shard.contiguous(), group=pg)
+        else:
+            dist.all_gather_into_tensor(self.flat_w, self.w_shard, group=pg)
+
+    def sync_shard_from_full(self):
+        """flat_w -> w_shard/master (after load_state_dict wrote params)."""
+        lo = self.rank * self.shard_size
+        self.w_shard.copy_(self.flat_w[lo:lo + self.shard_size])
+        self.master.copy_(self.w_shard.float())
+
+
+class _Zero3Unit:
+    """A module whose parameters gather/free together."""
+
+    def __init__(self, name: str, module: nn.Module, lr: float,
+                 weight_decay: float, rank: int, world: int,
+                 persistent: bool = False):
+        self.name = name
+        self.module = module
+        self.persistent = persistent
+        decay, no_decay = _decay_split(module.named_parameters())
+        self.segments: List[_Zero3Segment] = []
+        if decay:
+            self.segments.append(_Zero3Segment(decay, lr, weight_decay,
+                                               rank, world))
+        if no_decay:
+            self.segments.append(_Zero3Segment(no_decay, lr, 0.0, rank, world))
+        self.n_params = sum(len(s.params) for s in self.segments)
+        self._grads_pending = self.n_params
+
+    def gather(self, pg=None):
+        for s in self.segments:
+            s.gather(pg)
+
+    def free_weights(self):
+        if self.persistent:
+            return
+        for s in self.segments:
+            s.free_weights()
+
+    def alloc_grads(self):
+        for s in self.segments:
+            if s.flat_g.untyped_storage().size() == 0:
+                s.alloc_grads()
+
+    def grads_allocated(self) -> bool:
+        return all(s.flat_g.untyped_storage().size() > 0
+                   for s in self.segments)
+
+
+class Zero3Optimizer:
+    """FlatAdamW-compatible surface over the shard segments (the Trainer,
+    scheduler and checkpointing code only touch this interface)."""
+
+    def __init__(self, engine: "Zero3Engine", max_grad_norm: float = 1.0):
+        self.engine = engine
+        self.max_grad_norm = max_grad_norm
+        self.step_count = 0
+        self.shard_rank = engine.rank
+        self.shard_world = engine.world
+        self._last_norm_sq: Optional[torch.Tensor] = None
+
+    @property
+    def groups(self):
+        return [s for u in self.engine.units for s in u.segments]
+
+    @property
+    def param_groups(self):
+        return self.groups
+
+    def zero_grad(self, set_to_none: bool = False):
+        for g in self.groups:
+            g.g_shard.zero_()
+
+    def local_grad_norm_sq(self, shard_only: bool = True) -> torch.Tensor:
+        total = None
+        for g in self.groups:
+            ns = K.l2norm_sq(g.g_shard)
+            total = ns if total is None else total + ns
+        return total
+
+    @torch.no_grad()
+    def step(self, grad_scale: float = 1.0, closure=None,
+             norm_sq: Optional[torch.Tensor] = None, shard_only: bool = True):
+        self.step_count += 1
+        if norm_sq is None and self.max_grad_norm > 0:
+            norm_sq = self.local_grad_norm_sq()
+        self._last_norm_sq = norm_sq
+        for g in self.groups:
+            g.step(self.step_count, norm_sq, self.max_grad_norm, grad_scale)
+
+    def last_grad_norm(self) -> float:
+        if self._last_norm_sq is None:
+            return 0.0
+        val = float(self._last_norm_sq.sum())
+        return math.sqrt(val) if math.isfinite(val) and val >= 0 else float("nan")
+
+    def state_dict(self) -> Dict:
+        return {
+            "step_count": self.step_count,
+            "zero_stage": 3,
+            "shard_rank": self.shard_rank,
+            "shard_world": self.shard_world,
+            "groups": [
+                {"master": g.master.cpu(), "m": g.m.cpu(), "v": g.v.cpu(),
+                 "lr": g.lr, "weight_decay": g.weight_decay, "numel": g.numel}
+                for g in self.groups
+            ],
+        }
+
+    def load_state_dict(self, sd: Dict):
+        if sd.get("shard_world", 1) != self.shard_world:
+            raise ValueError("ZeRO-3 resharding of checkpoints not supported "
+                             f"(saved world={sd.get('shard_world')})")
+        self.step_count = sd.get("step_count", 0)
+        for g, gs in zip(self.groups, sd["groups"]):
+            g.master.copy_(gs["master"].to(g.master.device))
+            g.m.copy_(gs["m"].to(g.m.device))
+            g.v.copy_(gs["v"].to(g.v.device))
+            g.lr = gs.get("lr", g.lr)
+            g.weight_decay = gs.get("weight_decay", g.weight_decay)
+            g.w_shard.copy_(g.master.to(g.w_shard.dtype))
+
+    def rebuild(self, model):
+        raise RuntimeError("dynamic expert add/prune is not supported under "
+                           "ZeRO-3 (param shards are fixed); use ZeRO-0/1/2")
+
+
+class Zero3Engine:
+    """ZeroEngine-compatible engine: set_sync / reduce_gradients / step /
+    zero_grad / broadcast_parameters, plus the unit hook machinery."""
+
+    stage = 3
+
+    def __init__(self, model: nn.Module, config, process_group=None,
+                 mesh=None):
+        if mesh is not None and mesh.ep_size > 1:
+            raise RuntimeError("ZeRO-3 + expert parallelism not supported "
+                               "yet; use ZeRO-2 with EP")
+        self.model = model
+        self.pg = process_group
+        self.mesh = mesh
+        self.world = comm.get_world_size()
+        self.rank = comm.get_rank()
+        self.bucket_bytes = getattr(config, "reduce_bucket_size", 50_000_000)
+        self.overlap = getattr(config, "overlap_comm", True)
+        lr = config.learning_rate
+        wd = config.weight_decay
+
+        blocks = list(getattr(model, "layers", []))
+        # root unit: everything not inside a block (embed/final-norm/head)
+        self.units: List[_Zero3Unit] = []
+        root_mod = _RootShell(model, blocks)
+        self.units.append(_Zero3Unit("root", root_mod, lr, wd,
+                                     self.rank, self.world, persistent=True))
+        for i, b in enumerate(blocks):
+            self.units.append(_Zero3Unit(f"block{i}", b, lr, wd,
+                                         self.rank, self.world))
+        self.optimizer = Zero3Optimizer(self)
+        self._unit_of_param: Dict[int, _Zero3Unit] = {}
+        self._hooks = []
+        self._install_hooks()
+        self.sync_enabled = True
+        self._prefetch_stream = (torch.cuda.Stream()
+                                 if torch.cuda.is_available() else None)
+        if self.world == 1:
+            self._finalize_init()
+        self._finalized = self.world == 1
+
+    # ---------------------------------------------------------------- init
+    def _finalize_init(self):
+        """Free non-persistent unit weights (first gather re-materialises)."""
+        for u in self.units:
+            u.free_weights()
+        self._finalized = True
+
+    def broadcast_parameters(self):
+        """Rank-0 initial weights to all, then shard + free."""
+        if self.world > 1:
+            for u in self.units:
+                for s in u.segments:
+                    s.gather(self.pg)  # no-op pre-finalize (still allocated)
+                    dist.broadcast(s.flat_w, src=0, group=self.pg)
+                    lo = s.rank * s.shard_size
+                    s.w_shard.copy_(s.flat_w[lo:lo + s.shard_size])
+                    s.master.copy_(s.w_shard.float())
+        self._finalize_init()
+
+    # ---------------------------------------------------------------- hooks
+    def _install_hooks(self):
+        for ui, unit in enumerate(self.units):
+            if unit.persistent:
+                for s in unit.segments:
+                    for p in s.params:
+                        self._unit_of_param[id(p)] = unit
+                        self._hooks.append(p.register_post_accumulate_grad_hook(
+                            self._make_grad_hook(unit)))
+                continue
+            self._hooks.append(unit.module.register_forward_pre_hook(
+                self._make_fwd_pre(ui)))
+            self._hooks.append(unit.module.register_forward_hook(
+                self._make_fwd_post(ui)))
+            self._hooks.append(unit.module.register_full_backward_pre_hook(
+                self._make_bwd_pre(ui)))
+            for s in unit.segments:
+                for p in s.params:
+                    self._unit_of_param[id(p)] = unit
+                    self._hooks.append(p.register_post_accumulate_grad_hook(
+                        self._make_grad_hook(unit)))
+
+    def _make_fwd_pre(self, ui):
+        def hook(module, args):
+            unit = self.units[ui]
+            if not self._finalized:
+                return
+            unit.gather(self.pg)
+            if _in_backward():
+                # checkpoint recompute: backward follows immediately
+                unit.alloc_grads()
+            elif self._prefetch_stream is not None and ui + 1 < len(self.units):
+                nxt = self.units[ui + 1]
+                with torch.cuda.stream(self._prefetch_stream):
+                    nxt.gather(self.pg)
+        return hook
+
+    def _make_fwd_post(self, ui):
+        def hook(module, args, output):
+            if not self._finalized or _in_backward():
+                return  # inside recompute: the unit backward still needs them
+            unit = self.units[ui]
+            if module.training:
+                unit.free_weights()
+            # eval/generation: keep gathered for reuse across decode steps
+        return hook
+
+    def _make_bwd_pre(self, ui):
+        def hook(module, grad_output):
+            unit = self.units[ui]
+            unit.gather(self.pg)
+            unit.alloc_grads()
+        return hook
+
+    def _make_grad_hook(self, unit):
+        def hook(param):
+            unit._grads_pending -= 1
+            if unit._grads_pending == 0:
+                unit._grads_pending = unit.n_params
+                for s in unit.segments:
+                    s.reduce_into_shard(self.pg)
+                    if unit.persistent:
+                        # persistent units have no backward-pre hook to
+                        # re-allocate grads: keep the buffer live (zeroed)
+                        s.alloc_grads()
+                unit.free_weights()
+        return hook
+
+    # ------------------------------------------------------------- engine API
+    def set_sync(self, enabled: bool):
+        # ZeRO-3 must reduce every micro-batch (full grads cannot persist)
+        self.sync_enabled = True
+
+    def reduce_gradients(self):
+        pass  # handled per-unit by the grad hooks
+
+    def global_grad_norm_sq(self) -> Optional[torch.Tensor]:
+        if self.optimizer.max_grad_norm <= 0:
+            return None
+        ns = self.optimizer.local_grad_norm_sq()
+        if self.world > 1:
+            dist.all_reduce(ns, group=self.pg)
+        return ns
+
+    def step(self, grad_scale: float = 1.0):
+        # flush units whose backward did not touch every param (e.g. a gate
+        # with zero routed tokens): their grad buffers are still allocated
+        for u in self.units:
+            partial = u._grads_pending != u.n_params
+            for s in u.segments:
+                if partial or (not u.persistent
+                               and s.flat_g.untyped_storage().size() > 0):
+                    s.reduce_into_shard(self.pg)
+                    if u.persistent:
+                        s.alloc_grads()
+            u._grads_pending = u.n_params
+            u.free_weights()
+        norm_sq = self.global_grad_norm_sq()
+        self.optimizer.step(grad_scale=grad_scale, norm_sq=norm_sq)
+        for u in self.units:
+            if u.persistent:
+                for s in u.segments:
+                    s.refresh_from_shard(self.pg)
+
+    def sync_shards_from_full(self):
+        """After load_state_dict materialised full weights into flat buffers."""
+        for u in self.units:
+            for s in u.segments:
+                s.sync_shard_from_full()
+
+    def zero_grad(self):
+        self.optimizer.zero_grad()
+
+    def remove_hooks(self):
+        for h in self._hooks:
+            h.remove()
+        self._hooks.clear()
+
+    # ------------------------------------------------------------- checkpoint
+    class _GatherAll:
+        def __init__(self, engine):
+            self.engine = engine
+
+        def __enter__(self):
+            for u in self.engine.units:
+                u.gather(self.engine.pg)
+            return self
+
+        def __exit__(self, *exc):
+            if self.engine.model.training:
+                for u in self.engine.units:
+                    u.free_weights()
+
+    def gathered_weights(self):
+        """Context manager materialising ALL params (state_dict/checkpoint)."""
+        return Zero3Engine._GatherAll(self)
+
+
+class _RootShell(nn.Module):
+    """Wraps the model's non-block parameters as one pseudo-module so the
+    root unit can flatten them (embeddings, final norm, untied lm_head)."""
+
+    def __init__(self, model: nn.Module, blocks):
+        super().__init__()
+        in_blocks = set()
+        for b in blocks:
+            for p in b.parameters():
+                in_blocks.add(id(p))
+        self._names = []
+        self._params = []
+        for n, p in model.named_parameters():
+            if id(p) not in in_blocks:
+                self._names.append(n)
+                self._params.append(p)
+
+    def named_parameters(self, *a, **kw):
+        return list(zip(self._names, self._params))
+
+    def parameters(self, *a, **kw):
+        return list(self._params)

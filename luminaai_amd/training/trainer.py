@@ -68,17 +68,26 @@ class Trainer:
         mesh = get_mesh()
         ep_active = mesh is not None and mesh.ep_size > 1
         self.mesh = mesh
-        self.optimizer = FlatAdamW(
-            self.model, lr=config.learning_rate,
-            betas=(0.9, 0.95), eps=1e-8,
-            weight_decay=config.weight_decay, max_grad_norm=1.0,
-            shard_rank=comm.get_rank() if config.zero_stage in (1, 2) else 0,
-            shard_world=comm.get_world_size() if config.zero_stage in (1, 2) else 1,
-            ep_active=ep_active)
-        self.engine = engine if engine is not None else ZeroEngine(
-            self.optimizer, stage=min(config.zero_stage, 2),
-            bucket_bytes=config.reduce_bucket_size,
-            overlap_comm=config.overlap_comm, mesh=mesh)
+        if engine is not None:
+            self.engine = engine
+            self.optimizer = engine.opt if hasattr(engine, "opt") \
+                else engine.optimizer
+        elif config.zero_stage >= 3:
+            from ..parallel.zero3 import Zero3Engine
+            self.engine = Zero3Engine(self.model, config, mesh=mesh)
+            self.optimizer = self.engine.optimizer
+        else:
+            self.optimizer = FlatAdamW(
+                self.model, lr=config.learning_rate,
+                betas=(0.9, 0.95), eps=1e-8,
+                weight_decay=config.weight_decay, max_grad_norm=1.0,
+                shard_rank=comm.get_rank() if config.zero_stage in (1, 2) else 0,
+                shard_world=comm.get_world_size() if config.zero_stage in (1, 2) else 1,
+                ep_active=ep_active)
+            self.engine = ZeroEngine(
+                self.optimizer, stage=min(config.zero_stage, 2),
+                bucket_bytes=config.reduce_bucket_size,
+                overlap_comm=config.overlap_comm, mesh=mesh)
         if comm.is_distributed():
             self.engine.broadcast_parameters()
 
@@ -268,21 +277,30 @@ class Trainer:
                 f"_ep_rank_{self.mesh.ep_rank}"
         elif comm.get_rank() != 0:
             return ""
-        path = self.checkpoints.save_checkpoint(
-            self.model, self.optimizer, self.scheduler,
-            global_step=self.global_step, epoch=self.epoch,
-            config=self.config, model_config=self.model.config,
-            metrics=self._metric_floats(), is_best=is_best, tag=tag)
+        ctx = self.engine.gathered_weights() if self.engine.stage >= 3 \
+            else contextlib.nullcontext()
+        with ctx:
+            path = self.checkpoints.save_checkpoint(
+                self.model, self.optimizer, self.scheduler,
+                global_step=self.global_step, epoch=self.epoch,
+                config=self.config, model_config=self.model.config,
+                metrics=self._metric_floats(), is_best=is_best, tag=tag)
         self.checkpoint_history.append(path)
         return path
 
     def load_checkpoint(self, which: str = "latest", load_optimizer: bool = True):
         payload = self.checkpoints.load_checkpoint(which, map_location=self.device)
-        self.model.load_state_dict(payload["model_state_dict"])
-        # re-materialise the flat bf16 weights from the loaded params
-        for g in self.optimizer.groups:
-            for p, (off, n) in zip(g.params, g.offsets):
-                pass  # params ARE views into flat buffers; load_state_dict wrote in place
+        if self.engine.stage >= 3:
+            with self.engine.gathered_weights():
+                self.model.load_state_dict(payload["model_state_dict"])
+                self.engine.sync_shards_from_full()
+        else:
+            # params ARE views into the flat buffers: in-place load
+            self.model.load_state_dict(payload["model_state_dict"])
+            for g in self.optimizer.groups:
+                if hasattr(g, "_master_is_params") and not g._master_is_params:
+                    g.master.copy_(
+                        g.weight_view()[g.shard_lo:g.shard_hi].float())
         if load_optimizer and payload.get("optimizer_state_dict"):
             self.optimizer.load_state_dict(payload["optimizer_state_dict"])
         if load_optimizer and payload.get("scheduler_state_dict") and self.scheduler:
@@ -389,7 +407,7 @@ class Trainer:
 
     def add_expert(self) -> bool:
         layers = self._moe_layers()
-        if not layers:
+        if not layers or self.engine.stage >= 3:
             return False
         for l in layers:
             l.add_expert()
@@ -401,7 +419,7 @@ class Trainer:
 
     def prune_expert(self, expert_idx: Optional[int] = None) -> bool:
         layers = self._moe_layers()
-        if not layers or layers[0].num_experts <= 2:
+        if not layers or layers[0].num_experts <= 2 or self.engine.stage >= 3:
             return False
         for l in layers:
             idx = expert_idx

@@ -1,0 +1,177 @@
+"""ZeRO-3 engine tests: single-process equivalence with the ZeRO-0 path
+(fp32, CPU) and multi-process (gloo x2) sync + DDP equivalence."""
+
+import os
+import socket
+
+import pytest
+import torch
+import torch.distributed as dist
+import torch.multiprocessing as mp
+
+WORLD = 2
+
+
+def _cfg(zero_stage, rank=0, ckpt=False):
+    from luminaai_amd.config import Config
+    return Config(vocab_size=512, hidden_size=64, num_layers=2, num_heads=4,
+                  num_kv_heads=2, seq_length=32, intermediate_size=128,
+                  micro_batch_size=2, gradient_accumulation_steps=1,
+                  num_workers=0, use_moe=False, use_mod=False,
+                  zero_stage=zero_stage, precision="fp32",
+                  gradient_checkpointing=ckpt,
+                  experiment_name=f"z3_test_{zero_stage}_r{rank}",
+                  eval_every_n_batches=0, save_every_n_batches=0)
+
+
+def _train(cfg, steps=3, seed=500):
+    from luminaai_amd.data.tokenizer import ConversationTokenizer
+    from luminaai_amd.models import DeepSeekTransformer, config_to_deepseek_config
+    from luminaai_amd.training import Trainer
+    torch.manual_seed(1234)
+    model = DeepSeekTransformer(config_to_deepseek_config(cfg))
+    t = Trainer(model, ConversationTokenizer(), cfg)
+    t._setup_scheduler(10)
+    torch.manual_seed(seed)
+    for _ in range(steps):
+        ids = torch.randint(1, cfg.vocab_size, (2, cfg.seq_length + 1))
+        t.engine.set_sync(True)
+        t.train_step({"input_ids": ids[:, :-1], "labels": ids[:, 1:]})
+        t.optimizer_step()
+    return t
+
+
+def _full_weights(t):
+    """Materialise and snapshot all params."""
+    if t.engine.stage >= 3:
+        with t.engine.gathered_weights():
+            return {k: v.detach().clone()
+                    for k, v in t.model.state_dict().items()}
+    return {k: v.detach().clone() for k, v in t.model.state_dict().items()}
+
+
+def test_zero3_matches_zero0_single_proc():
+    t0 = _train(_cfg(0))
+    t3 = _train(_cfg(3))
+    w0 = _full_weights(t0)
+    w3 = _full_weights(t3)
+    for k in w0:
+        torch.testing.assert_close(w3[k], w0[k], rtol=1e-5, atol=1e-6,
+                                   msg=f"mismatch in {k}")
+    assert t3.optimizer.last_grad_norm() == pytest.approx(
+        t0.optimizer.last_grad_norm(), rel=1e-5)
+
+
+def test_zero3_with_activation_checkpointing():
+    t_plain = _train(_cfg(3, ckpt=False))
+    t_ckpt = _train(_cfg(3, ckpt=True))
+    wp = _full_weights(t_plain)
+    wc = _full_weights(t_ckpt)
+    for k in wp:
+        torch.testing.assert_close(wc[k], wp[k], rtol=1e-5, atol=1e-6,
+                                   msg=f"mismatch in {k}")
+
+
+def test_zero3_grad_accumulation():
+    cfg = _cfg(3)
+    cfg.gradient_accumulation_steps = 2
+    from luminaai_amd.data.tokenizer import ConversationTokenizer
+    from luminaai_amd.models import DeepSeekTransformer, config_to_deepseek_config
+    from luminaai_amd.training import Trainer
+    torch.manual_seed(1234)
+    model = DeepSeekTransformer(config_to_deepseek_config(cfg))
+    t = Trainer(model, ConversationTokenizer(), cfg)
+    t._setup_scheduler(10)
+    torch.manual_seed(42)
+    for _ in range(2):  # two micro-batches
+        ids = torch.randint(1, cfg.vocab_size, (2, cfg.seq_length + 1))
+        t.train_step({"input_ids": ids[:, :-1], "labels": ids[:, 1:]})
+    t.optimizer_step()
+    assert t.global_step == 1
+    assert t.optimizer.last_grad_norm() > 0
+
+
+def test_zero3_checkpoint_roundtrip(tmp_path, monkeypatch):
+    monkeypatch.chdir(tmp_path)
+    t = _train(_cfg(3), steps=2)
+    path = t.save_checkpoint()
+    assert path and os.path.exists(path)
+    w_before = _full_weights(t)
+    # train further, then roll back
+    torch.manual_seed(99)
+    ids = torch.randint(1, 512, (2, 33))
+    t.engine.set_sync(True)
+    t.train_step({"input_ids": ids[:, :-1], "labels": ids[:, 1:]})
+    t.optimizer_step()
+    t.load_checkpoint(path)
+    w_after = _full_weights(t)
+    for k in w_before:
+        torch.testing.assert_close(w_after[k], w_before[k],
+                                   msg=f"mismatch in {k}")
+
+
+# ---------------------------------------------------------------- multi-proc
+def _free_port():
+    with socket.socket() as s:
+        s.bind(("127.0.0.1", 0))
+        return s.getsockname()[1]
+
+
+def _run(rank, world, port, fn_name, q):
+    os.environ.update({
+        "MASTER_ADDR": "127.0.0.1", "MASTER_PORT": str(port),
+        "RANK": str(rank), "LOCAL_RANK": str(rank), "WORLD_SIZE": str(world),
+    })
+    dist.init_process_group("gloo", init_method="env://", rank=rank,
+                            world_size=world)
+    try:
+        import test_zero3
+        result = getattr(test_zero3, fn_name)(rank, world)
+        q.put((rank, "ok", result))
+    except Exception:  # noqa: BLE001
+        import traceback
+        q.put((rank, "err", traceback.format_exc()))
+    finally:
+        dist.destroy_process_group()
+
+
+def _spawn(fn_name):
+    ctx = mp.get_context("spawn")
+    q = ctx.SimpleQueue()
+    port = _free_port()
+    procs = [ctx.Process(target=_run, args=(r, WORLD, port, fn_name, q))
+             for r in range(WORLD)]
+    for p in procs:
+        p.start()
+    results = {}
+    for _ in range(WORLD):
+        rank, status, payload = q.get()
+        assert status == "ok", f"rank {rank} failed:\n{payload}"
+        results[rank] = payload
+    for p in procs:
+        p.join(timeout=60)
+    return results
+
+
+def zero3_worker(rank, world):
+    t = _train(_cfg(3, rank=rank), seed=500 + rank)
+    w = _full_weights(t)
+    key = "embed_tokens.weight"
+    return {"checksum": float(w[key].sum()), "norm": float(w[key].norm()),
+            "grad_norm": t.optimizer.last_grad_norm()}
+
+
+def ddp_ref_worker(rank, world):
+    t = _train(_cfg(0, rank=rank), seed=500 + rank)
+    w = _full_weights(t)
+    key = "embed_tokens.weight"
+    return {"checksum": float(w[key].sum()), "norm": float(w[key].norm()),
+            "grad_norm": t.optimizer.last_grad_norm()}
+
+
+def test_zero3_ranks_in_sync_and_match_ddp():
+    z3 = _spawn("zero3_worker")
+    assert z3[0]["checksum"] == pytest.approx(z3[1]["checksum"], abs=1e-4)
+    ddp = _spawn("ddp_ref_worker")
+    assert z3[0]["checksum"] == pytest.approx(ddp[0]["checksum"], rel=1e-4)
+    assert z3[0]["grad_norm"] == pytest.approx(ddp[0]["grad_norm"], rel=1e-4)
