@@ -46,6 +46,7 @@ hipError_t lumina_ce_bwd_f32(const void*, const int32_t*, const float*, const fl
 hipError_t lumina_l2norm_sq_bf16(const void*, int64_t, float*, hipStream_t);
 hipError_t lumina_l2norm_sq_f32(const void*, int64_t, float*, hipStream_t);
 hipError_t lumina_adamw_step(float*, const void*, int, float*, float*, void*, int, int64_t, float, float, float, float, float, float, float, const float*, float, float, hipStream_t);
+void launch_grouped_gemm_nt(const void*, const void*, void*, int, int, int, int, int64_t, int64_t, int64_t, hipStream_t);
 }
 
 // ---- RMSNorm -------------------------------------------------------------
@@ -259,7 +260,27 @@ void adamw_step(at::Tensor& master, const at::Tensor& grad, at::Tensor& m,
             "adamw_step");
 }
 
+// ---- grouped NT GEMM -----------------------------------------------------
+at::Tensor grouped_gemm_nt(const at::Tensor& A, const at::Tensor& B) {
+  // out[e] = A[e] @ B[e]^T : A [E,M,K], B [E,N,K] -> out [E,M,N]
+  TORCH_CHECK(A.is_contiguous() && B.is_contiguous(),
+              "grouped_gemm_nt needs contiguous operands");
+  TORCH_CHECK(is_bf16(A) && is_bf16(B), "grouped_gemm_nt is bf16-only");
+  TORCH_CHECK(A.dim() == 3 && B.dim() == 3 && A.size(0) == B.size(0)
+              && A.size(2) == B.size(2), "shape mismatch");
+  const int E = (int)A.size(0), M = (int)A.size(1);
+  const int K = (int)A.size(2), N = (int)B.size(1);
+  auto O = at::empty({E, M, N}, A.options());
+  launch_grouped_gemm_nt(A.data_ptr(), B.data_ptr(), O.data_ptr(),
+                         E, M, N, K, (int64_t)M * K, (int64_t)N * K,
+                         (int64_t)M * N, cur_stream());
+  check_hip(hipGetLastError(), "grouped_gemm_nt");
+  return O;
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
+  mod.def("grouped_gemm_nt", &grouped_gemm_nt,
+          "grouped expert GEMM out[e]=A[e]@B[e]^T, bf16 MFMA (gfx950)");
   mod.def("rmsnorm_fwd", &rmsnorm_fwd, "RMSNorm forward (gfx950)");
   mod.def("rmsnorm_bwd", &rmsnorm_bwd, "RMSNorm backward (gfx950)");
   mod.def("rope_fwd", &rope_fwd, "RoPE q/k rotation (gfx950)");

@@ -231,14 +231,16 @@ def fused_cross_entropy(logits: torch.Tensor, labels: torch.Tensor,
 
 # ------------------------------------------------------- batched expert GEMM
 class BatchedLinearFn(torch.autograd.Function):
-    """bmm with a safe backward for ROCm.
+    """bmm with an MI355X-native backward.
 
     torch's built-in BmmBackward computes grad_x = grad @ w.transpose(1, 2)
     as a strided batched GEMM with a transposed-B operand — that pattern
     memory-faults in this ROCm hipBLASLt/rocBLAS build for large bf16 batches
-    (verified on MI355X: any K/N, batched-only, transposed-B only). Here the
-    transposed operand is materialised contiguous first (sub-ms copy), and
-    grad_w uses the transposed-A form, which is fine.
+    (verified on MI355X: any K/N, batched-only, transposed-B only). grad_x
+    therefore runs on the hand-written MFMA grouped NT-GEMM kernel
+    (csrc/grouped_gemm.hip) — both operands row-major, K-contiguous
+    fragments, no transpose copy. grad_w uses the transposed-A hipBLASLt
+    form, which is fine.
     """
 
     @staticmethod
@@ -252,15 +254,25 @@ class BatchedLinearFn(torch.autograd.Function):
         go = go.contiguous()
         gx = gw = None
         if ctx.needs_input_grad[0]:
-            gx = torch.bmm(go, w.transpose(1, 2).contiguous())
+            gx = grouped_gemm_nt(go, w)
         if ctx.needs_input_grad[1]:
             gw = torch.bmm(x.transpose(1, 2), go)
         return gx, gw
 
 
 def expert_bmm(x: torch.Tensor, w: torch.Tensor) -> torch.Tensor:
-    """Grouped expert GEMM [E, C, K] x [E, K, N] (hipBLASLt strided-batched)."""
+    """Grouped expert GEMM [E, C, K] x [E, K, N] (hipBLASLt strided-batched
+    forward; hand-written MFMA NT kernel for grad_x)."""
     return BatchedLinearFn.apply(x, w)
+
+
+def grouped_gemm_nt(a: torch.Tensor, b: torch.Tensor) -> torch.Tensor:
+    """out[e] = a[e] @ b[e]^T with a [E,M,K], b [E,N,K] both row-major
+    (contraction over the trailing dim). HIP MFMA kernel on GPU bf16;
+    fp32-math fallback elsewhere."""
+    if use_hip(a) and a.dtype == torch.bfloat16 and b.dtype == torch.bfloat16:
+        return get_ext().grouped_gemm_nt(a.contiguous(), b.contiguous())
+    return torch.matmul(a, b.transpose(1, 2).contiguous())
 
 
 # ----------------------------------------------------------- optimizer path
