@@ -217,6 +217,16 @@ class GroupedQueryAttention(nn.Module):
                                   self.q_size + 2 * self.kv_size, bias=False)
         self.o_proj = nn.Linear(self.q_size, config.hidden_size, bias=False)
         self.dropout = config.dropout
+        # Ulysses sequence parallelism (parallel/sequence_parallel.py):
+        # heads scattered / sequence gathered around the SDPA core.
+        from ..parallel.mesh import get_mesh
+        mesh = get_mesh()
+        self.sp_size = mesh.sp_size if mesh is not None else 1
+        self.sp_group = mesh.sp_group if mesh is not None else None
+        if self.sp_size > 1:
+            assert self.num_heads % self.sp_size == 0 and \
+                self.num_kv_heads % self.sp_size == 0, \
+                "num_heads and num_kv_heads must divide sp_size"
 
     def forward(self, x, rope_cs, pos: Optional[torch.Tensor] = None,
                 pos_offset: int = 0, kv_cache: Optional[KVCache] = None,
@@ -229,10 +239,21 @@ class GroupedQueryAttention(nn.Module):
         v = v.view(B, S, self.num_kv_heads, self.head_dim)
 
         cos, sin = rope_cs
+        # RoPE first, on local tokens with their GLOBAL positions — the
+        # Ulysses exchange below then needs no position bookkeeping.
         q, k = ops.rope(q, k, cos, sin, pos, pos_offset)
 
         if kv_cache is not None:
             k, v = kv_cache.append(k, v)
+
+        run_sp = self.sp_size > 1 and kv_cache is None and S > 1 \
+            and attn_mask is None
+        if run_sp:
+            from ..parallel.sequence_parallel import (
+                scatter_heads_gather_seq, scatter_seq_gather_heads)
+            q = scatter_heads_gather_seq(q, self.sp_size, self.sp_group)
+            k = scatter_heads_gather_seq(k, self.sp_size, self.sp_group)
+            v = scatter_heads_gather_seq(v, self.sp_size, self.sp_group)
 
         # SDPA wants [B, H, S, D]
         qt = q.transpose(1, 2)
@@ -246,7 +267,12 @@ class GroupedQueryAttention(nn.Module):
             is_causal=is_causal,
             enable_gqa=self.num_kv_heads != self.num_heads,
         )
-        out = out.transpose(1, 2).reshape(B, S, self.q_size)
+        out = out.transpose(1, 2)
+        if run_sp:
+            from ..parallel.sequence_parallel import scatter_seq_gather_heads
+            out = scatter_seq_gather_heads(out.contiguous(), self.sp_size,
+                                           self.sp_group)
+        out = out.reshape(B, S, self.q_size)
         return self.o_proj(out)
 
 
@@ -637,6 +663,11 @@ class DeepSeekTransformer(nn.Module):
         x = self.embed_tokens(input_ids) * self.embed_scale
         if kv_caches is not None and kv_caches[0].seq_len > 0:
             pos_offset = kv_caches[0].seq_len
+        # Ulysses SP: this rank holds sequence slice [sp_rank*S, (sp_rank+1)*S)
+        from ..parallel.mesh import get_mesh
+        mesh = get_mesh()
+        if mesh is not None and mesh.sp_size > 1 and kv_caches is None and S > 1:
+            pos_offset = pos_offset + mesh.sp_rank * S
         rope_cs = self.rotary.get(pos_offset + S, x.device)
 
         attn_mask = None

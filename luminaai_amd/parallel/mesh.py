@@ -28,22 +28,29 @@ _MESH: Optional["ParallelMesh"] = None
 
 
 class ParallelMesh:
-    def __init__(self, ep_size: int = 1):
+    def __init__(self, ep_size: int = 1, sp_size: int = 1):
         world = comm.get_world_size()
         rank = comm.get_rank()
-        assert world % ep_size == 0, \
+        assert world % max(ep_size, 1) == 0, \
             f"world {world} not divisible by ep_size {ep_size}"
+        assert world % max(sp_size, 1) == 0, \
+            f"world {world} not divisible by sp_size {sp_size}"
+        assert ep_size <= 1 or sp_size <= 1, \
+            "EP and Ulysses-SP composition is not supported yet"
         self.world = world
         self.rank = rank
         self.ep_size = ep_size
-        self.dp_size = world // ep_size
-        self.ep_rank = rank % ep_size
+        self.sp_size = sp_size
+        self.dp_size = world // (ep_size * max(sp_size, 1))
+        self.ep_rank = rank % ep_size if ep_size > 1 else 0
         self.dp_rank = rank // ep_size
+        self.sp_rank = rank % sp_size if sp_size > 1 else 0
         self.ep_group = None
         self.expert_dp_group = None
+        self.sp_group = None
         if world > 1 and ep_size > 1:
             # build ALL groups on every rank (dist.new_group is collective)
-            for d in range(self.dp_size):
+            for d in range(world // ep_size):
                 ranks = list(range(d * ep_size, (d + 1) * ep_size))
                 g = dist.new_group(ranks)
                 if rank in ranks:
@@ -53,15 +60,21 @@ class ParallelMesh:
                 g = dist.new_group(ranks)
                 if rank in ranks:
                     self.expert_dp_group = g
+        if world > 1 and sp_size > 1:
+            for d in range(world // sp_size):
+                ranks = list(range(d * sp_size, (d + 1) * sp_size))
+                g = dist.new_group(ranks)
+                if rank in ranks:
+                    self.sp_group = g
 
     @property
     def expert_dp_size(self) -> int:
-        return self.dp_size
+        return self.world // self.ep_size if self.ep_size > 1 else self.world
 
 
-def init_mesh(ep_size: int = 1) -> ParallelMesh:
+def init_mesh(ep_size: int = 1, sp_size: int = 1) -> ParallelMesh:
     global _MESH
-    _MESH = ParallelMesh(ep_size)
+    _MESH = ParallelMesh(ep_size, sp_size)
     return _MESH
 
 
