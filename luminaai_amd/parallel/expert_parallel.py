@@ -25,14 +25,18 @@ class _AllToAllFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x: torch.Tensor, group):
         ctx.group = group
-        out = torch.empty_like(x)
-        dist.all_to_all_single(out, x.contiguous(), group=group)
+        x = x.contiguous()
+        # NOTE empty_like would PRESERVE non-contiguous strides and the
+        # collective writes assuming dense layout — always allocate dense.
+        out = torch.empty(x.shape, dtype=x.dtype, device=x.device)
+        dist.all_to_all_single(out, x, group=group)
         return out
 
     @staticmethod
     def backward(ctx, gy: torch.Tensor):
-        gx = torch.empty_like(gy)
-        dist.all_to_all_single(gx, gy.contiguous(), group=ctx.group)
+        gy = gy.contiguous()
+        gx = torch.empty(gy.shape, dtype=gy.dtype, device=gy.device)
+        dist.all_to_all_single(gx, gy, group=ctx.group)
         return gx, None
 
 
