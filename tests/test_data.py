@@ -93,3 +93,43 @@ def test_streaming_dataset(tokenizer, sample_text):
         if i >= 2:
             break
     assert rows and rows[0]["input_ids"].shape == (32,)
+
+
+# ---------------------------------------------------------------- acquisition
+def test_oasst_prepare(tmp_path):
+    import json
+    from luminaai_amd.data.acquisition import prepare_oasst
+    trees = [
+        {"prompt": {"role": "user", "text": "What is the weather like today in general terms?",
+                    "replies": [{"role": "assistant",
+                                 "text": "Weather varies by region; check a local forecast for specifics.",
+                                 "replies": []}]}},
+        {"prompt": {"role": "user", "text": "hi", "replies": []}},  # too short
+    ]
+    p = tmp_path / "trees.jsonl"
+    p.write_text("\n".join(json.dumps(t) for t in trees))
+    stats = prepare_oasst(str(p), str(tmp_path / "out"))
+    assert stats["kept"] == 1
+    assert len(stats["shards"]) == 1
+    row = json.loads(open(stats["shards"][0]).read())
+    assert row["messages"][1]["role"] == "assistant"
+
+
+def test_text_corpus_prepare(tmp_path):
+    from luminaai_amd.data.acquisition import prepare_text_corpus
+    raw = tmp_path / "raw.txt"
+    raw.write_text("== See also ==\nshort\n" + ("solid sentence content. " * 20) + "\n")
+    stats = prepare_text_corpus(str(raw), str(tmp_path / "clean.txt"),
+                                source="wikipedia", min_chars=100)
+    assert stats["kept"] == 1
+
+
+def test_conversation_quality():
+    from luminaai_amd.data.acquisition import conversation_quality
+    good = {"messages": [{"role": "user", "content": "Explain entropy in thermodynamics please."},
+                         {"role": "assistant", "content": "Entropy measures the dispersal of energy among microstates of a system."}]}
+    assert conversation_quality(good) > 0.5
+    assert conversation_quality({"messages": []}) == 0.0
+    degenerate = {"messages": [{"role": "user", "content": "a" * 500},
+                               {"role": "assistant", "content": "b" * 500}]}
+    assert conversation_quality(degenerate) < 0.5
