@@ -1,0 +1,90 @@
+"""fp8 (OCP e4m3fn) GEMM path for gfx950.
+
+The reference's fp8 support was registry-only and never reached a kernel
+(reference trainer.py:157-356). Here fp8 is real: gfx950 MFMA runs OCP
+e4m3fn at ~2x the bf16 rate (MI355X_MICROARCH.md §Matrix cores; NOT the
+MI300X fnuz variants), reached through hipBLASLt via torch._scaled_mm.
+
+Scheme: bf16 master weights; per-tensor dynamic scaling (amax / 448 for
+e4m3fn) on both activations and weights; fp32 accumulate; bf16 output.
+The backward runs in bf16 (standard "fp8 forward, bf16 backward" training
+recipe — wgrad/dgrad keep full bf16 fidelity).
+"""
+
+from __future__ import annotations
+
+import torch
+import torch.nn as nn
+
+E4M3_MAX = 448.0
+
+
+def _amax_scale(t: torch.Tensor) -> torch.Tensor:
+    amax = t.abs().amax().float().clamp_min(1e-12)
+    return (E4M3_MAX / amax).clamp(max=1e12)
+
+
+def quantize_e4m3(t: torch.Tensor):
+    """Returns (fp8 tensor, inverse scale fp32 scalar tensor)."""
+    s = _amax_scale(t)
+    q = (t.float() * s).clamp(-E4M3_MAX, E4M3_MAX).to(torch.float8_e4m3fn)
+    return q, (1.0 / s)
+
+
+class _FP8MatmulFn(torch.autograd.Function):
+    """y = x @ w.T in fp8 forward; bf16 backward."""
+
+    @staticmethod
+    def forward(ctx, x, w):
+        ctx.save_for_backward(x, w)
+        shp = x.shape[:-1]
+        x2 = x.reshape(-1, x.shape[-1])
+        xq, xs = quantize_e4m3(x2)
+        wq, ws = quantize_e4m3(w)
+        y = torch._scaled_mm(xq, wq.t(), scale_a=xs, scale_b=ws,
+                             out_dtype=torch.bfloat16)
+        return y.reshape(*shp, w.shape[0])
+
+    @staticmethod
+    def backward(ctx, gy):
+        x, w = ctx.saved_tensors
+        gy2 = gy.reshape(-1, gy.shape[-1])
+        gx = gw = None
+        if ctx.needs_input_grad[0]:
+            gx = (gy2 @ w.to(gy.dtype)).reshape_as(x)
+        if ctx.needs_input_grad[1]:
+            gw = gy2.t() @ x.reshape(-1, x.shape[-1]).to(gy.dtype)
+        return gx, gw
+
+
+class FP8Linear(nn.Linear):
+    """Drop-in nn.Linear whose forward GEMM runs on fp8 MFMA.
+    Falls back to bf16 matmul off-GPU or for tiny shapes."""
+
+    fp8_min_dim = 64  # _scaled_mm needs %16 shapes; skip tiny layers
+
+    def forward(self, x):
+        w = self.weight
+        if (x.is_cuda and self.bias is None
+                and x.shape[-1] % 16 == 0 and w.shape[0] % 16 == 0
+                and min(w.shape) >= self.fp8_min_dim):
+            return _FP8MatmulFn.apply(x, w)
+        return super().forward(x)
+
+
+def convert_linears_to_fp8(model: nn.Module, min_dim: int = 64) -> int:
+    """Swap eligible nn.Linear modules for FP8Linear in place
+    (cf. the reference QuantizationManager's bnb Linear8bitLt swap,
+    trainer.py:658-679). Returns the number converted."""
+    n = 0
+    for mod in model.modules():
+        for name, child in list(mod.named_children()):
+            if type(child) is nn.Linear and child.bias is None \
+                    and min(child.weight.shape) >= min_dim:
+                new = FP8Linear(child.in_features, child.out_features,
+                                bias=False, device=child.weight.device,
+                                dtype=child.weight.dtype)
+                new.weight = child.weight
+                setattr(mod, name, new)
+                n += 1
+    return n

@@ -63,6 +63,11 @@ class Trainer:
         self.device = torch.device("cuda", comm.env_local_rank()) \
             if torch.cuda.is_available() else torch.device("cpu")
         self.model = self.precision.cast_model(model).to(self.device)
+        if self.precision.spec.name == "fp8" and self.device.type == "cuda":
+            from ..ops.fp8 import convert_linears_to_fp8
+            n_fp8 = convert_linears_to_fp8(self.model)
+            if logger:
+                logger.info(f"fp8: {n_fp8} Linear layers on e4m3fn MFMA")
 
         from ..parallel.mesh import get_mesh
         mesh = get_mesh()

@@ -269,3 +269,48 @@ def test_expert_bmm_backward_uses_nt_kernel():
     torch.matmul(x32, w32).backward(g.float())
     torch.testing.assert_close(x.grad.float(), x32.grad, rtol=8e-2, atol=8e-2)
     torch.testing.assert_close(w.grad.float(), w32.grad, rtol=8e-2, atol=8e-2)
+
+
+# ---------------------------------------------------------------- fp8 GEMM
+def test_fp8_linear_matches_bf16():
+    from luminaai_amd.ops.fp8 import FP8Linear
+    torch.manual_seed(0)
+    lin = FP8Linear(256, 512, bias=False, device=_dev(), dtype=torch.bfloat16)
+    x = torch.randn(4, 128, 256, device=_dev(), dtype=torch.bfloat16,
+                    requires_grad=True)
+    y = lin(x)
+    ref = torch.nn.functional.linear(x.float(), lin.weight.float())
+    # e4m3 has ~2 mantissa bits: generous tolerance, relative to magnitude
+    err = (y.float() - ref).abs().max() / ref.abs().max()
+    assert float(err) < 0.1, f"fp8 relative error {float(err)}"
+    y.sum().backward()
+    assert x.grad is not None and lin.weight.grad is not None
+    assert torch.isfinite(x.grad.float()).all()
+
+
+def test_fp8_convert_and_train_step():
+    from luminaai_amd.config import ConfigPresets
+    from luminaai_amd.data.tokenizer import ConversationTokenizer
+    from luminaai_amd.models import DeepSeekTransformer, config_to_deepseek_config
+    from luminaai_amd.training import Trainer
+    cfg = ConfigPresets.debug()
+    cfg.precision = "fp8"
+    cfg.use_moe = False
+    cfg.use_mod = False
+    cfg.num_workers = 0
+    cfg.seq_length = 128
+    cfg.micro_batch_size = 2
+    cfg.gradient_accumulation_steps = 1
+    cfg.gradient_checkpointing = False
+    torch.manual_seed(0)
+    model = DeepSeekTransformer(config_to_deepseek_config(cfg))
+    t = Trainer(model, ConversationTokenizer(), cfg)
+    t._setup_scheduler(10)
+    from luminaai_amd.ops.fp8 import FP8Linear
+    n_fp8 = sum(1 for m in t.model.modules() if isinstance(m, FP8Linear))
+    assert n_fp8 > 0, "no Linear layers converted to fp8"
+    ids = torch.randint(1, cfg.vocab_size, (2, cfg.seq_length + 1))
+    out = t.train_step({"input_ids": ids[:, :-1], "labels": ids[:, 1:]})
+    t.optimizer_step()
+    loss = float(out["ce_loss"].detach())
+    assert loss == loss and loss > 0
