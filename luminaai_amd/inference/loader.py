@@ -75,6 +75,34 @@ def load_zero_shards(shard_dir: str, map_location="cpu") -> Dict:
     return merged
 
 
+def merge_ep_checkpoints(paths: List[str], map_location="cpu") -> Dict:
+    """Merge per-EP-rank checkpoint files (written by Trainer.save_checkpoint
+    under expert parallelism as *_ep_rank_{r}.pt) into one full-expert
+    payload: expert weights (w_gate_up / w_down) are concatenated along the
+    expert dim in ep-rank order; everything else is taken from rank 0."""
+    def _ep_rank(p):
+        m = re.search(r"_ep_rank_(\d+)", os.path.basename(p))
+        return int(m.group(1)) if m else 0
+
+    paths = sorted(paths, key=_ep_rank)
+    payloads = [torch.load(p, map_location=map_location, weights_only=False)
+                for p in paths]
+    sds = [_strip_prefixes(p["model_state_dict"]) for p in payloads]
+    merged_sd = dict(sds[0])
+    for key in sds[0]:
+        if ".w_gate_up" in key or ".w_down" in key:
+            merged_sd[key] = torch.cat([sd[key] for sd in sds], dim=0)
+    out = dict(payloads[0])
+    out["model_state_dict"] = merged_sd
+    return out
+
+
+def find_ep_shards(any_shard_path: str) -> List[str]:
+    """All sibling _ep_rank_* files of one EP checkpoint shard."""
+    base = re.sub(r"_ep_rank_\d+", "_ep_rank_*", any_shard_path)
+    return sorted(glob.glob(base))
+
+
 def infer_config_from_state_dict(sd: Dict[str, torch.Tensor]) -> DeepSeekConfig:
     """Reconstruct a DeepSeekConfig from tensor shapes
     (reference Chat.py:219-300)."""

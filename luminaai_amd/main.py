@@ -30,7 +30,7 @@ from .config import Config, ConfigPresets
 from .data.dataset import SyntheticDataset, setup_datasets
 from .data.tokenizer import ConversationTokenizer
 from .models import DeepSeekTransformer, config_to_deepseek_config
-from .monitoring import ProductionLogger, TrainingHealthMonitor
+from .monitoring import ProductionLogger, TrainingHealthMonitor, WandbLogger
 from .parallel import comm
 from .parallel.mesh import init_mesh
 from .training import (AdaptiveTrainingOrchestrator, EnhancedChinchillaScaler)
@@ -192,10 +192,16 @@ def main(argv: Optional[list] = None) -> dict:
         signal.signal(signal.SIGTERM, _emergency)
 
     health = TrainingHealthMonitor(check_every=cfg.health_check_interval)
-    trainer.set_metrics_hook(lambda m: (
-        orch._enqueue_metrics(m),
-        health.log_step({k: v for k, v in m.as_dict().items()
-                         if isinstance(v, (int, float))}, m.step)))
+    wb = WandbLogger(cfg, enabled=cfg.enable_wandb and rank == 0)
+
+    def _hook(m):
+        orch._enqueue_metrics(m)
+        floats = {k: v for k, v in m.as_dict().items()
+                  if isinstance(v, (int, float))}
+        health.log_step(floats, m.step)
+        wb.log(floats, m.step)
+
+    trainer.set_metrics_hook(_hook)
 
     t0 = time.time()
     try:
@@ -203,6 +209,7 @@ def main(argv: Optional[list] = None) -> dict:
                                          train_ds, eval_ds)
     finally:
         orch.cleanup()
+        wb.finish()
         comm.cleanup()
 
     if rank == 0:

@@ -1,3 +1,5 @@
 from .logger import MetricsCollector, TrainingHealthMonitor, ProductionLogger
+from .wandb_compat import WandbLogger
 
-__all__ = ["MetricsCollector", "TrainingHealthMonitor", "ProductionLogger"]
+__all__ = ["MetricsCollector", "ProductionLogger", "TrainingHealthMonitor",
+           "WandbLogger"]

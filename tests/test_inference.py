@@ -129,3 +129,39 @@ def test_chat_save_session(small_model, tokenizer, tmp_path):
     assert chat.handle_command(f"/save {p}").startswith("saved")
     data = json.loads(open(p).read())
     assert len(data["history"]) == 2
+
+
+def test_merge_ep_checkpoints(tiny_moe_config, tmp_path):
+    """Two EP shard files with disjoint experts merge to the full model."""
+    from luminaai_amd.inference.loader import find_ep_shards, merge_ep_checkpoints
+    torch.manual_seed(0)
+    full = DeepSeekTransformer(config_to_deepseek_config(tiny_moe_config))
+    sd = full.state_dict()
+    E = tiny_moe_config.num_experts
+    for r in range(2):
+        shard = dict(sd)
+        for k, v in sd.items():
+            if ".w_gate_up" in k or ".w_down" in k:
+                EL = E // 2
+                shard[k] = v[r * EL:(r + 1) * EL].clone()
+        torch.save({"model_state_dict": shard, "global_step": 5},
+                   tmp_path / f"ck_step_5_ep_rank_{r}.pt")
+    shards = find_ep_shards(str(tmp_path / "ck_step_5_ep_rank_0.pt"))
+    assert len(shards) == 2
+    merged = merge_ep_checkpoints(shards)
+    for k, v in sd.items():
+        torch.testing.assert_close(merged["model_state_dict"][k], v)
+    m2 = DeepSeekTransformer(config_to_deepseek_config(tiny_moe_config))
+    m2.load_state_dict(merged["model_state_dict"])
+
+
+def test_wandb_fallback_logger(tmp_path, monkeypatch, tiny_config):
+    monkeypatch.chdir(tmp_path)
+    from luminaai_amd.monitoring import WandbLogger
+    tiny_config.enable_wandb = True
+    wb = WandbLogger(tiny_config, enabled=True)
+    wb.log({"loss": 1.5}, step=1)
+    wb.finish()
+    import glob as g
+    files = g.glob("experiments/*/wandb_fallback.jsonl")
+    assert files, "fallback jsonl not written"
