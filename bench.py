@@ -57,7 +57,7 @@ def main():
     ap.add_argument("--gpus", type=int, default=1)
     ap.add_argument("--steps", type=int, default=10)
     ap.add_argument("--warmup", type=int, default=3)
-    ap.add_argument("--micro-batch", type=int, default=8)
+    ap.add_argument("--micro-batch", type=int, default=16)
     ap.add_argument("--accum", type=int, default=1)
     ap.add_argument("--seq-len", type=int, default=None)
     ap.add_argument("--preset", type=str, default=None)
