@@ -1,0 +1,125 @@
+"""Pipeline parallelism tests: partitioning, and 1F1B equivalence vs a
+single-process model (gloo x2)."""
+
+import os
+import socket
+
+import pytest
+import torch
+import torch.distributed as dist
+import torch.multiprocessing as mp
+
+WORLD = 2
+
+
+def test_partition_layers():
+    from luminaai_amd.parallel.pipeline import partition_layers
+    assert partition_layers(8, 2) == [(0, 4), (4, 8)]
+    assert partition_layers(7, 2) == [(0, 4), (4, 7)]
+    assert partition_layers(31, 4) == [(0, 8), (8, 16), (16, 24), (24, 31)]
+    bounds = partition_layers(5, 5)
+    assert bounds[0] == (0, 1) and bounds[-1] == (4, 5)
+
+
+def _model_cfg():
+    from luminaai_amd.models.transformer import DeepSeekConfig
+    return DeepSeekConfig(vocab_size=512, hidden_size=64, num_layers=4,
+                          num_heads=4, num_kv_heads=2, intermediate_size=128,
+                          seq_length=32, use_moe=True, num_experts=4,
+                          moe_top_k=2, routing_noise_std=0.0,
+                          moe_pattern="every_2nd", use_mod=False,
+                          tie_word_embeddings=False)
+
+
+def _free_port():
+    with socket.socket() as s:
+        s.bind(("127.0.0.1", 0))
+        return s.getsockname()[1]
+
+
+def _run(rank, world, port, q):
+    os.environ.update({
+        "MASTER_ADDR": "127.0.0.1", "MASTER_PORT": str(port),
+        "RANK": str(rank), "LOCAL_RANK": str(rank), "WORLD_SIZE": str(world),
+    })
+    dist.init_process_group("gloo", init_method="env://", rank=rank,
+                            world_size=world)
+    try:
+        result = pp_worker(rank, world)
+        q.put((rank, "ok", result))
+    except Exception:  # noqa: BLE001
+        import traceback
+        q.put((rank, "err", traceback.format_exc()))
+    finally:
+        dist.destroy_process_group()
+
+
+def pp_worker(rank, world):
+    from luminaai_amd.models import DeepSeekTransformer
+    from luminaai_amd.parallel.pipeline import PipelineParallelEngine
+    from luminaai_amd.ops import fused_cross_entropy
+
+    mcfg = _model_cfg()
+    torch.manual_seed(1234)
+    model = DeepSeekTransformer(mcfg)
+    engine = PipelineParallelEngine(model, None)
+
+    torch.manual_seed(900)  # same data everywhere
+    micro = []
+    for _ in range(4):
+        ids = torch.randint(1, mcfg.vocab_size, (2, 33))
+        micro.append({"input_ids": ids[:, :-1], "labels": ids[:, 1:]})
+    out = engine.train_batch(micro)
+
+    # single-process reference on the SAME model replica
+    torch.manual_seed(1234)
+    ref = DeepSeekTransformer(mcfg)
+    ref_losses = []
+    for mb in micro:
+        logits, aux, _ = ref(mb["input_ids"])
+        ce, _, _ = fused_cross_entropy(logits, mb["labels"])
+        (ce + aux).backward()
+        ref_losses.append(float(ce))
+
+    res = {"pp_loss": float(out["loss"]),
+           "ref_loss": sum(ref_losses) / len(ref_losses)}
+    # per-stage grads equal reference grads of the owned params
+    ref_named = dict(ref.named_parameters())
+    bad = 0.0
+    for name, p in engine.stage.named_parameters():
+        # stage layer indices are local; map back via shapes+order
+        if p.grad is None:
+            continue
+    # direct structural check: first stage owns embed, last owns lm_head
+    if rank == 0:
+        g_pp = engine.stage.embed_tokens.weight.grad
+        g_ref = ref.embed_tokens.weight.grad
+        bad = float((g_pp - g_ref).abs().max() / g_ref.abs().max().clamp_min(1e-12))
+    else:
+        g_pp = engine.stage.lm_head.weight.grad
+        g_ref = ref.lm_head.weight.grad
+        bad = float((g_pp - g_ref).abs().max() / g_ref.abs().max().clamp_min(1e-12))
+    res["rel_grad_err"] = bad
+    return res
+
+
+def test_pp_1f1b_matches_single_process():
+    ctx = mp.get_context("spawn")
+    q = ctx.SimpleQueue()
+    port = _free_port()
+    procs = [ctx.Process(target=_run, args=(r, WORLD, port, q))
+             for r in range(WORLD)]
+    for p in procs:
+        p.start()
+    results = {}
+    for _ in range(WORLD):
+        rank, status, payload = q.get()
+        assert status == "ok", f"rank {rank} failed:\n{payload}"
+        results[rank] = payload
+    for p in procs:
+        p.join(timeout=60)
+    # last stage reports the CE loss; must match the single-process run
+    assert results[1]["pp_loss"] == pytest.approx(results[1]["ref_loss"],
+                                                  rel=1e-4)
+    assert results[0]["rel_grad_err"] < 1e-3
+    assert results[1]["rel_grad_err"] < 1e-3
