@@ -12,10 +12,14 @@
 //
 // Structure (cdna_hip_programming.md §5): 128x128 tile, BK=64, 256 threads
 // (4 waves as 2x2, 64x64 output per wave), v_mfma_f32_16x16x32_bf16,
-// single-LDS-image register-staged pipeline (prefetch tile t+1 into VGPRs
-// while computing tile t; ds_write after the barrier — T14), st_16x32 XOR
-// swizzle on the LDS image so fragment ds_read_b128 is bank-spread.
-// Arbitrary M/N/K (predicated edge loads/stores; zero-filled K tail).
+// st_16x32 XOR-swizzled LDS image. Two staging paths:
+//  - GLDS=true (K % 64 == 0): global_load_lds dwordx4 into a double-buffered
+//    LDS image, next K-tile in flight during the MFMA block. glds writes
+//    lane-linear, so the swizzle is applied to the per-lane GLOBAL source
+//    address (guide: "pre-swizzled global src + swizzled ds_read addr").
+//    Ragged M/N rows clamp to a valid row; the epilogue masks them.
+//  - GLDS=false: predicated register staging + ds_write, handles any K
+//    (zero-filled tail) — the fallback for K % 64 != 0.
 
 #include "common.h"
 
@@ -28,9 +32,10 @@ typedef __attribute__((ext_vector_type(4))) float  f32x4;
 #define THREADS 256
 // per-thread staging chunks: tile bytes (128*64*2) / (256 threads * 16B) = 4
 #define CHUNKS 4
+#define TILE_ELEMS (BM * BK)
 
 // LDS byte offset for element (row, k) of a [128][64] bf16 tile image with
-// the st_16x32 swizzle: flip byte-bit-5 with byte-bit-9 (= row bit 2).
+// the st_16x32 swizzle: flip byte-bit-5 with row bit 2.
 DEV_INLINE int swz(int row, int kbyte) {
   return row * (BK * 2) + (kbyte ^ (((row >> 2) & 1) << 5));
 }
@@ -49,6 +54,10 @@ DEV_INLINE ushortx8 load8_guard(const uint16_t* __restrict__ p, bool row_ok,
   return v;
 }
 
+typedef const __attribute__((address_space(1))) void* gas_ptr;
+typedef __attribute__((address_space(3))) void* las_ptr;
+
+template <bool GLDS>
 __global__ __launch_bounds__(THREADS, 2)
 void grouped_gemm_nt_kernel(const uint16_t* __restrict__ Aall,
                             const uint16_t* __restrict__ Ball,
@@ -56,9 +65,8 @@ void grouped_gemm_nt_kernel(const uint16_t* __restrict__ Aall,
                             int M, int N, int K,
                             int64_t strideA, int64_t strideB,
                             int64_t strideO) {
-  __shared__ uint16_t lds[2 * BM * BK];      // [A tile | B tile], swizzled
-  uint16_t* As = lds;
-  uint16_t* Bs = lds + BM * BK;
+  // [A | B] per buffer; glds path double-buffers (64 KiB total)
+  __shared__ uint16_t lds[(GLDS ? 4 : 2) * TILE_ELEMS];
 
   const int e = blockIdx.z;
   const uint16_t* A = Aall + e * strideA;
@@ -72,90 +80,152 @@ void grouped_gemm_nt_kernel(const uint16_t* __restrict__ Aall,
   const int wave = t >> 6;
   const int wm = wave >> 1;                  // wave row (0..1) -> 64 rows
   const int wn = wave & 1;                   // wave col (0..1) -> 64 cols
-
-  // staging map: chunk c (0..1023) -> row = c>>3, 16B k-group = c&7
-  int s_row[CHUNKS], s_koff[CHUNKS];
-  #pragma unroll
-  for (int i = 0; i < CHUNKS; ++i) {
-    int c = t + i * THREADS;
-    s_row[i] = c >> 3;
-    s_koff[i] = (c & 7) * 8;                 // in elements
-  }
-
   const int KT = (K + BK - 1) / BK;
-  ushortx8 ra[CHUNKS], rb[CHUNKS];
-
-  // ---- load tile 0 into registers
-  #pragma unroll
-  for (int i = 0; i < CHUNKS; ++i) {
-    int gm = tileM + s_row[i];
-    int gn = tileN + s_row[i];
-    int gk = s_koff[i];
-    ra[i] = load8_guard(A + (int64_t)gm * K + gk, gm < M, gk, K);
-    rb[i] = load8_guard(B + (int64_t)gn * K + gk, gn < N, gk, K);
-  }
-  // ---- write tile 0 to LDS
-  #pragma unroll
-  for (int i = 0; i < CHUNKS; ++i) {
-    *reinterpret_cast<ushortx8*>(
-        reinterpret_cast<char*>(As) + swz(s_row[i], s_koff[i] * 2)) = ra[i];
-    *reinterpret_cast<ushortx8*>(
-        reinterpret_cast<char*>(Bs) + swz(s_row[i], s_koff[i] * 2)) = rb[i];
-  }
-  __syncthreads();
+  const int fr = lane & 15;                  // fragment row/col index
+  const int fg = lane >> 4;                  // k-subgroup 0..3 (8 elems)
 
   f32x4 acc[4][4] = {};
+  int buf = 0;
 
-  const int fr = lane & 15;                  // fragment row/col index
-  const int fg = lane >> 4;                  // k-subgroup 0..3 (8 elems each)
-
-  for (int kt = 0; kt < KT; ++kt) {
-    // prefetch next K-tile into registers (overlaps the MFMA block below)
-    if (kt + 1 < KT) {
-      const int k0 = (kt + 1) * BK;
+  if constexpr (GLDS) {
+    // ---- glds staging: wave w, piece i covers chunks c = w*64 + i*256 + l
+    const int maxA = M - 1, maxB = N - 1;
+    auto issue_tile = [&](int which, int kt) {
+      const int k0 = kt * BK;
+      uint16_t* base = lds + which * 2 * TILE_ELEMS;
       #pragma unroll
       for (int i = 0; i < CHUNKS; ++i) {
-        int gm = tileM + s_row[i];
-        int gn = tileN + s_row[i];
-        int gk = k0 + s_koff[i];
-        ra[i] = load8_guard(A + (int64_t)gm * K + gk, gm < M, gk, K);
-        rb[i] = load8_guard(B + (int64_t)gn * K + gk, gn < N, gk, K);
+        const int c = wave * 64 + i * 256 + lane;
+        const int row = c >> 3;
+        const int kbB = ((c & 7) * 16) ^ (((row >> 2) & 1) << 5);
+        // A piece
+        {
+          const int r = row > maxA - tileM ? (maxA - tileM < 0 ? 0 : maxA - tileM) : row;
+          const char* gp = reinterpret_cast<const char*>(
+              A + (int64_t)(tileM + r) * K + k0) + kbB;
+          las_ptr lp = (las_ptr)(reinterpret_cast<char*>(base)
+                                 + (wave * 64 + i * 256) * 16);
+          __builtin_amdgcn_global_load_lds((gas_ptr)gp, lp, 16, 0, 0);
+        }
+        // B piece
+        {
+          const int r = row > maxB - tileN ? (maxB - tileN < 0 ? 0 : maxB - tileN) : row;
+          const char* gp = reinterpret_cast<const char*>(
+              B + (int64_t)(tileN + r) * K + k0) + kbB;
+          las_ptr lp = (las_ptr)(reinterpret_cast<char*>(base + TILE_ELEMS)
+                                 + (wave * 64 + i * 256) * 16);
+          __builtin_amdgcn_global_load_lds((gas_ptr)gp, lp, 16, 0, 0);
+        }
       }
-    }
+    };
 
-    // compute on the staged tile: 2 x (8 ds_read_b128 + 16 MFMA)
-    #pragma unroll
-    for (int kk = 0; kk < BK; kk += 32) {
-      bf16x8 af[4], bf[4];
-      const int kb = (kk + fg * 8) * 2;      // byte offset of this lane's k8
+    issue_tile(0, 0);
+    asm volatile("s_waitcnt vmcnt(0)");
+    __syncthreads();
+
+    for (int kt = 0; kt < KT; ++kt) {
+      if (kt + 1 < KT)
+        issue_tile(buf ^ 1, kt + 1);        // in flight under the MFMAs
+      const uint16_t* As = lds + buf * 2 * TILE_ELEMS;
+      const uint16_t* Bs = As + TILE_ELEMS;
       #pragma unroll
-      for (int m = 0; m < 4; ++m)
-        af[m] = *reinterpret_cast<const bf16x8*>(
-            reinterpret_cast<const char*>(As)
-            + swz(wm * 64 + m * 16 + fr, kb));
-      #pragma unroll
-      for (int n = 0; n < 4; ++n)
-        bf[n] = *reinterpret_cast<const bf16x8*>(
-            reinterpret_cast<const char*>(Bs)
-            + swz(wn * 64 + n * 16 + fr, kb));
-      #pragma unroll
-      for (int m = 0; m < 4; ++m)
+      for (int kk = 0; kk < BK; kk += 32) {
+        bf16x8 af[4], bf[4];
+        const int kb = (kk + fg * 8) * 2;
+        #pragma unroll
+        for (int m = 0; m < 4; ++m)
+          af[m] = *reinterpret_cast<const bf16x8*>(
+              reinterpret_cast<const char*>(As)
+              + swz(wm * 64 + m * 16 + fr, kb));
         #pragma unroll
         for (int n = 0; n < 4; ++n)
-          acc[m][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              af[m], bf[n], acc[m][n], 0, 0, 0);
+          bf[n] = *reinterpret_cast<const bf16x8*>(
+              reinterpret_cast<const char*>(Bs)
+              + swz(wn * 64 + n * 16 + fr, kb));
+        #pragma unroll
+        for (int m = 0; m < 4; ++m)
+          #pragma unroll
+          for (int n = 0; n < 4; ++n)
+            acc[m][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                af[m], bf[n], acc[m][n], 0, 0, 0);
+      }
+      __syncthreads();                      // drains the in-flight glds too
+      buf ^= 1;
     }
+  } else {
+    // ---- predicated register staging (any K)
+    uint16_t* As = lds;
+    uint16_t* Bs = lds + TILE_ELEMS;
+    int s_row[CHUNKS], s_koff[CHUNKS];
+    #pragma unroll
+    for (int i = 0; i < CHUNKS; ++i) {
+      int c = t + i * THREADS;
+      s_row[i] = c >> 3;
+      s_koff[i] = (c & 7) * 8;               // in elements
+    }
+    ushortx8 ra[CHUNKS], rb[CHUNKS];
 
+    #pragma unroll
+    for (int i = 0; i < CHUNKS; ++i) {
+      int gm = tileM + s_row[i];
+      int gn = tileN + s_row[i];
+      int gk = s_koff[i];
+      ra[i] = load8_guard(A + (int64_t)gm * K + gk, gm < M, gk, K);
+      rb[i] = load8_guard(B + (int64_t)gn * K + gk, gn < N, gk, K);
+    }
+    #pragma unroll
+    for (int i = 0; i < CHUNKS; ++i) {
+      *reinterpret_cast<ushortx8*>(
+          reinterpret_cast<char*>(As) + swz(s_row[i], s_koff[i] * 2)) = ra[i];
+      *reinterpret_cast<ushortx8*>(
+          reinterpret_cast<char*>(Bs) + swz(s_row[i], s_koff[i] * 2)) = rb[i];
+    }
     __syncthreads();
-    if (kt + 1 < KT) {
+
+    for (int kt = 0; kt < KT; ++kt) {
+      if (kt + 1 < KT) {
+        const int k0 = (kt + 1) * BK;
+        #pragma unroll
+        for (int i = 0; i < CHUNKS; ++i) {
+          int gm = tileM + s_row[i];
+          int gn = tileN + s_row[i];
+          int gk = k0 + s_koff[i];
+          ra[i] = load8_guard(A + (int64_t)gm * K + gk, gm < M, gk, K);
+          rb[i] = load8_guard(B + (int64_t)gn * K + gk, gn < N, gk, K);
+        }
+      }
       #pragma unroll
-      for (int i = 0; i < CHUNKS; ++i) {
-        *reinterpret_cast<ushortx8*>(
-            reinterpret_cast<char*>(As) + swz(s_row[i], s_koff[i] * 2)) = ra[i];
-        *reinterpret_cast<ushortx8*>(
-            reinterpret_cast<char*>(Bs) + swz(s_row[i], s_koff[i] * 2)) = rb[i];
+      for (int kk = 0; kk < BK; kk += 32) {
+        bf16x8 af[4], bf[4];
+        const int kb = (kk + fg * 8) * 2;
+        #pragma unroll
+        for (int m = 0; m < 4; ++m)
+          af[m] = *reinterpret_cast<const bf16x8*>(
+              reinterpret_cast<const char*>(As)
+              + swz(wm * 64 + m * 16 + fr, kb));
+        #pragma unroll
+        for (int n = 0; n < 4; ++n)
+          bf[n] = *reinterpret_cast<const bf16x8*>(
+              reinterpret_cast<const char*>(Bs)
+              + swz(wn * 64 + n * 16 + fr, kb));
+        #pragma unroll
+        for (int m = 0; m < 4; ++m)
+          #pragma unroll
+          for (int n = 0; n < 4; ++n)
+            acc[m][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                af[m], bf[n], acc[m][n], 0, 0, 0);
       }
       __syncthreads();
+      if (kt + 1 < KT) {
+        #pragma unroll
+        for (int i = 0; i < CHUNKS; ++i) {
+          *reinterpret_cast<ushortx8*>(
+              reinterpret_cast<char*>(As) + swz(s_row[i], s_koff[i] * 2)) = ra[i];
+          *reinterpret_cast<ushortx8*>(
+              reinterpret_cast<char*>(Bs) + swz(s_row[i], s_koff[i] * 2)) = rb[i];
+        }
+        __syncthreads();
+      }
     }
   }
 
@@ -180,7 +250,16 @@ extern "C" void launch_grouped_gemm_nt(const void* A, const void* B, void* O,
                                        int64_t strideA, int64_t strideB,
                                        int64_t strideO, hipStream_t stream) {
   dim3 grid((M + BM - 1) / BM, (N + BN - 1) / BN, E);
-  hipLaunchKernelGGL(grouped_gemm_nt_kernel, grid, dim3(THREADS), 0, stream,
-                     (const uint16_t*)A, (const uint16_t*)B, (uint16_t*)O,
-                     M, N, K, strideA, strideB, strideO);
+  const bool aligned = (K % BK == 0)
+      && ((reinterpret_cast<uintptr_t>(A) & 15) == 0)
+      && ((reinterpret_cast<uintptr_t>(B) & 15) == 0);
+  if (aligned) {
+    hipLaunchKernelGGL(grouped_gemm_nt_kernel<true>, grid, dim3(THREADS), 0,
+                       stream, (const uint16_t*)A, (const uint16_t*)B,
+                       (uint16_t*)O, M, N, K, strideA, strideB, strideO);
+  } else {
+    hipLaunchKernelGGL(grouped_gemm_nt_kernel<false>, grid, dim3(THREADS), 0,
+                       stream, (const uint16_t*)A, (const uint16_t*)B,
+                       (uint16_t*)O, M, N, K, strideA, strideB, strideO);
+  }
 }
