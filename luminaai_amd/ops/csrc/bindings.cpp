@@ -47,6 +47,11 @@ hipError_t lumina_l2norm_sq_bf16(const void*, int64_t, float*, hipStream_t);
 hipError_t lumina_l2norm_sq_f32(const void*, int64_t, float*, hipStream_t);
 hipError_t lumina_adamw_step(float*, const void*, int, float*, float*, void*, int, int64_t, float, float, float, float, float, float, float, const float*, float, float, hipStream_t);
 void launch_grouped_gemm_nt(const void*, const void*, void*, int, int, int, int, int64_t, int64_t, int64_t, hipStream_t);
+hipError_t lumina_moe_gather_rows(const void*, const int64_t*, const bool*, void*, int64_t, int, int, hipStream_t);
+hipError_t lumina_moe_dispatch_bwd(const void*, const int64_t*, void*, int64_t, int, int, int64_t, int, hipStream_t);
+hipError_t lumina_moe_combine_fwd(const void*, const float*, const int64_t*, void*, int64_t, int, int, int64_t, int, hipStream_t);
+hipError_t lumina_moe_combine_bwd_y(const void*, const float*, const int64_t*, const int64_t*, const bool*, void*, int64_t, int64_t, int, int, hipStream_t);
+hipError_t lumina_moe_combine_bwd_w(const void*, const void*, const int64_t*, float*, int64_t, int, int, int64_t, int, hipStream_t);
 }
 
 // ---- RMSNorm -------------------------------------------------------------
@@ -260,6 +265,78 @@ void adamw_step(at::Tensor& master, const at::Tensor& grad, at::Tensor& m,
             "adamw_step");
 }
 
+// ---- fused MoE dispatch/combine gathers ----------------------------------
+at::Tensor moe_gather_rows(const at::Tensor& x, const at::Tensor& src_tok,
+                           const at::Tensor& fill) {
+  TORCH_CHECK(x.is_contiguous() && x.dim() == 2);
+  const int64_t n_slots = src_tok.numel();
+  auto buf = at::empty({n_slots, x.size(1)}, x.options());
+  check_hip(lumina_moe_gather_rows(x.data_ptr(), src_tok.data_ptr<int64_t>(),
+                                   fill.data_ptr<bool>(), buf.data_ptr(),
+                                   n_slots, (int)x.size(1),
+                                   is_bf16(x) ? 1 : 0, cur_stream()),
+            "moe_gather_rows");
+  return buf;
+}
+
+at::Tensor moe_dispatch_bwd(const at::Tensor& gbuf, const at::Tensor& slot_tm,
+                            int64_t k) {
+  TORCH_CHECK(gbuf.is_contiguous() && gbuf.dim() == 2);
+  const int64_t n_tok = slot_tm.numel() / k;
+  auto gx = at::empty({n_tok, gbuf.size(1)}, gbuf.options());
+  check_hip(lumina_moe_dispatch_bwd(gbuf.data_ptr(),
+                                    slot_tm.data_ptr<int64_t>(),
+                                    gx.data_ptr(), n_tok, (int)k,
+                                    (int)gbuf.size(1), gbuf.size(0),
+                                    is_bf16(gbuf) ? 1 : 0, cur_stream()),
+            "moe_dispatch_bwd");
+  return gx;
+}
+
+at::Tensor moe_combine_fwd(const at::Tensor& y, const at::Tensor& w_tm,
+                           const at::Tensor& slot_tm, int64_t k) {
+  TORCH_CHECK(y.is_contiguous() && y.dim() == 2);
+  TORCH_CHECK(w_tm.scalar_type() == at::kFloat);
+  const int64_t n_tok = slot_tm.numel() / k;
+  auto out = at::empty({n_tok, y.size(1)}, y.options());
+  check_hip(lumina_moe_combine_fwd(y.data_ptr(), w_tm.data_ptr<float>(),
+                                   slot_tm.data_ptr<int64_t>(),
+                                   out.data_ptr(), n_tok, (int)k,
+                                   (int)y.size(1), y.size(0),
+                                   is_bf16(y) ? 1 : 0, cur_stream()),
+            "moe_combine_fwd");
+  return out;
+}
+
+at::Tensor moe_combine_bwd_y(const at::Tensor& gout, const at::Tensor& w_tm,
+                             const at::Tensor& inv, const at::Tensor& src_tok,
+                             const at::Tensor& fill) {
+  TORCH_CHECK(gout.is_contiguous() && gout.dim() == 2);
+  const int64_t n_slots = inv.numel();
+  auto gy = at::empty({n_slots, gout.size(1)}, gout.options());
+  check_hip(lumina_moe_combine_bwd_y(
+                gout.data_ptr(), w_tm.data_ptr<float>(),
+                inv.data_ptr<int64_t>(), src_tok.data_ptr<int64_t>(),
+                fill.data_ptr<bool>(), gy.data_ptr(), n_slots,
+                w_tm.numel(), (int)gout.size(1), is_bf16(gout) ? 1 : 0,
+                cur_stream()),
+            "moe_combine_bwd_y");
+  return gy;
+}
+
+at::Tensor moe_combine_bwd_w(const at::Tensor& gout, const at::Tensor& y,
+                             const at::Tensor& slot_tm, int64_t k) {
+  TORCH_CHECK(gout.is_contiguous() && y.is_contiguous());
+  const int64_t n_flat = slot_tm.numel();
+  auto gw = at::empty({n_flat}, gout.options().dtype(at::kFloat));
+  check_hip(lumina_moe_combine_bwd_w(
+                gout.data_ptr(), y.data_ptr(), slot_tm.data_ptr<int64_t>(),
+                gw.data_ptr<float>(), n_flat, (int)k, (int)gout.size(1),
+                y.size(0), is_bf16(gout) ? 1 : 0, cur_stream()),
+            "moe_combine_bwd_w");
+  return gw;
+}
+
 // ---- grouped NT GEMM -----------------------------------------------------
 at::Tensor grouped_gemm_nt(const at::Tensor& A, const at::Tensor& B) {
   // out[e] = A[e] @ B[e]^T : A [E,M,K], B [E,N,K] -> out [E,M,N]
@@ -281,6 +358,11 @@ at::Tensor grouped_gemm_nt(const at::Tensor& A, const at::Tensor& B) {
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("grouped_gemm_nt", &grouped_gemm_nt,
           "grouped expert GEMM out[e]=A[e]@B[e]^T, bf16 MFMA (gfx950)");
+  mod.def("moe_gather_rows", &moe_gather_rows, "MoE dispatch gather (gfx950)");
+  mod.def("moe_dispatch_bwd", &moe_dispatch_bwd, "MoE dispatch backward");
+  mod.def("moe_combine_fwd", &moe_combine_fwd, "MoE weighted combine");
+  mod.def("moe_combine_bwd_y", &moe_combine_bwd_y, "MoE combine backward (y)");
+  mod.def("moe_combine_bwd_w", &moe_combine_bwd_w, "MoE combine backward (w)");
   mod.def("rmsnorm_fwd", &rmsnorm_fwd, "RMSNorm forward (gfx950)");
   mod.def("rmsnorm_bwd", &rmsnorm_bwd, "RMSNorm backward (gfx950)");
   mod.def("rope_fwd", &rope_fwd, "RoPE q/k rotation (gfx950)");

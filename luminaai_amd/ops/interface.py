@@ -339,11 +339,17 @@ def moe_routing_plan(topi: torch.Tensor, num_experts: int,
 
 class MoEDispatchFn(torch.autograd.Function):
     """buf[s] = xf[src_tok[s]] * fill[s]; backward gathers grad rows back
-    per (token, choice) and reduces over k — no scatter."""
+    per (token, choice) and reduces over k — no scatter. On GPU both
+    directions run as single fused HIP kernels (csrc/moe.hip)."""
 
     @staticmethod
     def forward(ctx, xf, src_tok, fill_mask, slot_tm, k):
-        buf = xf.index_select(0, src_tok) * fill_mask.unsqueeze(1).to(xf.dtype)
+        if use_hip(xf):
+            buf = get_ext().moe_gather_rows(xf.contiguous(), src_tok,
+                                            fill_mask)
+        else:
+            buf = xf.index_select(0, src_tok) \
+                * fill_mask.unsqueeze(1).to(xf.dtype)
         ctx.save_for_backward(slot_tm)
         ctx.k = k
         ctx.EC = buf.shape[0]
@@ -352,6 +358,10 @@ class MoEDispatchFn(torch.autograd.Function):
     @staticmethod
     def backward(ctx, gbuf):
         (slot_tm,) = ctx.saved_tensors
+        if use_hip(gbuf):
+            return (get_ext().moe_dispatch_bwd(gbuf.contiguous(), slot_tm,
+                                               ctx.k),
+                    None, None, None, None)
         keep = (slot_tm < ctx.EC).unsqueeze(1).to(gbuf.dtype)
         g = gbuf.index_select(0, slot_tm.clamp_max(ctx.EC - 1)) * keep
         N = slot_tm.numel() // ctx.k
@@ -365,12 +375,17 @@ class MoECombineFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, y, w_tm, slot_tm, inv, src_tok, fill_mask, k):
         EC = y.shape[0]
-        keep = slot_tm < EC
-        slot_c = slot_tm.clamp_max(EC - 1)
-        wk = (w_tm * keep.to(w_tm.dtype)).to(y.dtype)
-        y_sel = y.index_select(0, slot_c)
         N = slot_tm.numel() // k
-        out = (y_sel * wk.unsqueeze(1)).view(N, k, -1).sum(1)
+        if use_hip(y):
+            out = get_ext().moe_combine_fwd(y.contiguous(),
+                                            w_tm.float().contiguous(),
+                                            slot_tm, k)
+        else:
+            keep = slot_tm < EC
+            slot_c = slot_tm.clamp_max(EC - 1)
+            wk = (w_tm * keep.to(w_tm.dtype)).to(y.dtype)
+            y_sel = y.index_select(0, slot_c)
+            out = (y_sel * wk.unsqueeze(1)).view(N, k, -1).sum(1)
         ctx.save_for_backward(y, w_tm, slot_tm, inv, src_tok, fill_mask)
         ctx.k = k
         return out
@@ -381,6 +396,13 @@ class MoECombineFn(torch.autograd.Function):
         EC = y.shape[0]
         k = ctx.k
         Nk = slot_tm.numel()
+        if use_hip(y):
+            gout = gout.contiguous()
+            w32 = w_tm.float().contiguous()
+            grad_y = get_ext().moe_combine_bwd_y(gout, w32, inv, src_tok,
+                                                 fill_mask)
+            gw = get_ext().moe_combine_bwd_w(gout, y.contiguous(), slot_tm, k)
+            return (grad_y, gw.to(w_tm.dtype), None, None, None, None, None)
         # grad_y[s] = gout[src_tok[s]] * w_tm[inv[s]] * fill[s]
         w_slot = w_tm[inv.clamp_max(Nk - 1)] * fill_mask.to(w_tm.dtype)
         grad_y = gout.index_select(0, src_tok) * w_slot.unsqueeze(1).to(gout.dtype)
