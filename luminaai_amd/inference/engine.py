@@ -62,7 +62,7 @@ class GenerationEngine:
 
         ids = prompt_ids[-cfg.max_context:]
         x = torch.tensor([ids], dtype=torch.long, device=self.device)
-        caches = self.model.make_kv_caches()
+        caches = self.model.make_kv_caches(max_len=cfg.max_context)
 
         # ---- prefill: one forward over the whole prompt
         logits, _, _ = self.model(x, kv_caches=caches)

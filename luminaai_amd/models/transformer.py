@@ -182,22 +182,42 @@ class RotaryEmbedding(nn.Module):
 
 
 class KVCache:
-    """Per-layer incremental KV cache for decoding ([B, S, Hkv, D] layout)."""
+    """Per-layer incremental KV cache for decoding ([B, S, Hkv, D] layout).
 
-    def __init__(self):
+    With max_len > 0 the buffers are preallocated ONCE and appends are
+    in-place slice writes — no per-token torch.cat reallocation/copy (the
+    decode hot loop stays allocation-free and hipGraph-friendly). Without
+    max_len it falls back to growing concat."""
+
+    def __init__(self, max_len: int = 0):
+        self.max_len = max_len
         self.k: Optional[torch.Tensor] = None
         self.v: Optional[torch.Tensor] = None
+        self._len = 0
 
     @property
     def seq_len(self) -> int:
-        return 0 if self.k is None else self.k.shape[1]
+        return self._len
 
     def append(self, k: torch.Tensor, v: torch.Tensor):
+        B, S, H, D = k.shape
+        if self.max_len > 0:
+            if self.k is None:
+                cap = max(self.max_len, S)
+                self.k = k.new_empty(B, cap, H, D)
+                self.v = v.new_empty(B, cap, H, D)
+            end = min(self._len + S, self.k.shape[1])
+            take = end - self._len
+            self.k[:, self._len:end] = k[:, :take]
+            self.v[:, self._len:end] = v[:, :take]
+            self._len = end
+            return self.k[:, :end], self.v[:, :end]
         if self.k is None:
             self.k, self.v = k, v
         else:
             self.k = torch.cat([self.k, k], dim=1)
             self.v = torch.cat([self.v, v], dim=1)
+        self._len = self.k.shape[1]
         return self.k, self.v
 
 
@@ -756,5 +776,6 @@ class DeepSeekTransformer(nn.Module):
     def get_moe_layers(self) -> List[MoEFFNLayer]:
         return [l.ffn for l in self.layers if l.is_moe]
 
-    def make_kv_caches(self) -> List[KVCache]:
-        return [KVCache() for _ in self.layers]
+    def make_kv_caches(self, max_len: int = 0) -> List[KVCache]:
+        """max_len > 0 preallocates static buffers (serving hot path)."""
+        return [KVCache(max_len) for _ in self.layers]

@@ -165,3 +165,19 @@ def test_wandb_fallback_logger(tmp_path, monkeypatch, tiny_config):
     import glob as g
     files = g.glob("experiments/*/wandb_fallback.jsonl")
     assert files, "fallback jsonl not written"
+
+
+def test_static_kv_cache_matches_dynamic(small_model):
+    m = small_model.eval()
+    ids = torch.randint(0, 500, (1, 10))
+    with torch.no_grad():
+        dyn = m.make_kv_caches()
+        a, _, _ = m(ids[:, :6], kv_caches=dyn)
+        sta = m.make_kv_caches(max_len=16)
+        b, _, _ = m(ids[:, :6], kv_caches=sta)
+        torch.testing.assert_close(a, b)
+        for t in range(6, 10):
+            a, _, _ = m(ids[:, t:t + 1], kv_caches=dyn)
+            b, _, _ = m(ids[:, t:t + 1], kv_caches=sta)
+            torch.testing.assert_close(a, b, rtol=1e-5, atol=1e-5)
+        assert sta[0].k.shape[1] == 16  # preallocated, not grown
