@@ -396,10 +396,15 @@ class MoEFFNLayer(nn.Module):
         aux = ref_ops.load_balancing_loss(probs, topi, E) * self.load_balancing_weight
 
         # --- capacity-bucketed dispatch: gather-only plan (ops/interface.py
-        # MoERoutingPlan — no atomics, no host sync, hipGraph-capturable)
+        # MoERoutingPlan — no atomics, no host sync, hipGraph-capturable).
+        # `placement` (load balancer, parallel/load_balance.py) maps expert
+        # id -> bucket slot so hot experts spread across EP ranks.
         C = max(1, int(math.ceil(N * k / E * self.capacity_factor)))
-        plan = ops.interface.moe_routing_plan(topi, E, C)
-        counts = plan.counts
+        placement = getattr(self, "placement", None)
+        route_idx = topi if placement is None else placement[topi]
+        plan = ops.interface.moe_routing_plan(route_idx, E, C)
+        counts = plan.counts if placement is None \
+            else plan.counts[placement]        # back to per-expert order
         bufv = ops.interface.moe_dispatch(xf, plan).view(E, C, h)
 
         # --- EP token exchange: [E, C, h] -> peers owning each expert shard.

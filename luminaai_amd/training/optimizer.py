@@ -161,6 +161,7 @@ class FlatAdamW:
         self.max_grad_norm = max_grad_norm
         self.shard_rank = shard_rank
         self.shard_world = shard_world
+        self.ep_active = ep_active
         self.step_count = 0
         # expert groups are EP-sharded by construction -> no ZeRO shard on top
         self.groups: List[_FlatGroup] = [
@@ -265,15 +266,17 @@ class FlatAdamW:
                                     g.m[off:off + n].clone(),
                                     g.v[off:off + n].clone())
         lr = self.groups[0].lr if self.groups else self.defaults["lr"]
-        groups = split_decay_groups(model)
+        groups = split_decay_groups(model, ep_active=self.ep_active)
         new_groups = []
         for gd in groups:
             wd = gd.get("weight_decay")
+            expert = gd.get("comm") == "expert"
             fg = _FlatGroup(list(gd["params"]), lr=lr,
                             weight_decay=(self.defaults["weight_decay"]
                                           if wd is None else wd),
-                            shard_rank=self.shard_rank,
-                            shard_world=self.shard_world)
+                            shard_rank=0 if expert else self.shard_rank,
+                            shard_world=1 if expert else self.shard_world,
+                            comm=gd.get("comm", "dp"))
             if fg.shard_world == 1:
                 for p, (off, n) in zip(fg.params, fg.offsets):
                     st = old_state.get(id(p))

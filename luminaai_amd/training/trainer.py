@@ -437,6 +437,47 @@ class Trainer:
                                  overlap_comm=self.engine.overlap)
         return True
 
+    def apply_expert_load_balance(self) -> bool:
+        """Re-place experts across EP ranks by observed load (LPT packing;
+        parallel/load_balance.py — ColossalAI LoadBalancer counterpart).
+        Rebuilds the optimizer (fresh moments for expert weights)."""
+        layers = self._moe_layers()
+        if not layers or self.engine.stage >= 3:
+            return False
+        from ..parallel.load_balance import (apply_placement, imbalance,
+                                             plan_placement)
+        changed = False
+        for l in layers:
+            counts = l._usage_counts.clone()
+            if comm.is_distributed():
+                import torch.distributed as dist
+                dist.all_reduce(counts)
+            total = float(counts.sum())
+            if total <= 0:
+                continue
+            loads = (counts / total).tolist()
+            ep = max(l.ep_size, 1)
+            if ep == 1:
+                continue                      # nothing to balance locally
+            order = plan_placement(loads, ep)
+            ident = list(range(l.num_experts))
+            if imbalance(loads, order, ep) < imbalance(loads, ident, ep) - 1e-6 \
+                    or getattr(l, "placement", None) is not None:
+                apply_placement(l, order)
+                changed = True
+        if changed:
+            # weights moved IN PLACE through the flat buffers; the Adam
+            # moments no longer describe the experts now in each slot —
+            # reset them and refresh the fp32 master from the moved weights
+            for g in self.optimizer.groups:
+                if getattr(g, "comm", "dp") == "expert":
+                    g.m.zero_()
+                    g.v.zero_()
+                    if not g._master_is_params:
+                        g.master.copy_(
+                            g.weight_view()[g.shard_lo:g.shard_hi].float())
+        return changed
+
     def adjust_capacity_factor(self, new_factor: float):
         for l in self._moe_layers():
             l.capacity_factor = new_factor

@@ -323,3 +323,50 @@ def test_zero2_matches_ddp():
     a = _spawn("ddp_worker")
     b = _spawn("zero2_worker")
     assert a[0]["checksum"] == pytest.approx(b[0]["checksum"], rel=1e-4)
+
+
+# ---- expert load balancing -------------------------------------------------
+def ep_load_balance_worker(rank, world):
+    """Applying a non-trivial placement must not change model outputs."""
+    from luminaai_amd.models import DeepSeekTransformer
+    from luminaai_amd.parallel.load_balance import apply_placement
+    from luminaai_amd.parallel.mesh import init_mesh, reset_mesh
+
+    mcfg = _moe_model_cfg()
+    init_mesh(world)
+    torch.manual_seed(1234)
+    ep = DeepSeekTransformer(mcfg).eval()
+    torch.manual_seed(42 + rank)
+    ids = torch.randint(1, mcfg.vocab_size, (2, mcfg.seq_length))
+    with torch.no_grad():
+        before, _, _ = ep(ids)
+        order = [1, 3, 0, 2]        # shuffle experts across both ranks
+        for l in ep.get_moe_layers():
+            apply_placement(l, order)
+        after, _, _ = ep(ids)
+        # and back to identity
+        for l in ep.get_moe_layers():
+            apply_placement(l, list(range(mcfg.num_experts)))
+        back, _, _ = ep(ids)
+    reset_mesh()
+    d1 = float((after - before).abs().max())
+    d2 = float((back - before).abs().max())
+    return {"d_placed": d1, "d_back": d2}
+
+
+def test_ep_load_balance_placement_transparent():
+    res = _spawn("ep_load_balance_worker")
+    for r in range(2):
+        assert res[r]["d_placed"] < 1e-4, res
+        assert res[r]["d_back"] < 1e-4, res
+
+
+def test_plan_placement_lpt():
+    from luminaai_amd.parallel.load_balance import imbalance, plan_placement
+    loads = [0.4, 0.3, 0.1, 0.05, 0.05, 0.04, 0.03, 0.03]
+    order = plan_placement(loads, 4)
+    assert sorted(order) == list(range(8))
+    assert imbalance(loads, order, 4) < imbalance(loads, list(range(8)), 4)
+    # the two heaviest experts must land on different ranks
+    pos = {e: i // 2 for i, e in enumerate(order)}
+    assert pos[0] != pos[1]
