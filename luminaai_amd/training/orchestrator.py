@@ -197,6 +197,10 @@ class ArchitectureEvolution:
             decision = AdaptiveDecision(
                 "prune_expert", None,
                 f"under-utilised experts (util={util:.2f})", 0.7)
+        elif imb > 2.0:
+            decision = AdaptiveDecision(
+                "load_balance", None,
+                f"EP load imbalance {imb:.1f}", 0.7)
         if decision is not None:
             self.last_change_step = step
         return decision
@@ -430,6 +434,8 @@ class AdaptiveTrainingOrchestrator:
             ok = t.add_expert()
         elif d.action == "prune_expert":
             ok = t.prune_expert()
+        elif d.action == "load_balance":
+            ok = t.apply_expert_load_balance()
         elif d.action == "batch_size":
             t.adjust_batch_size(int(d.value))
             ok = True
