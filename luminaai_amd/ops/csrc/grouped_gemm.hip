@@ -263,3 +263,140 @@ extern "C" void launch_grouped_gemm_nt(const void* A, const void* B, void* O,
                        (uint16_t*)O, M, N, K, strideA, strideB, strideO);
   }
 }
+
+// ---------------------------------------------------------------------------
+// v2: same 128x128/BK=64 glds structure on v_mfma_f32_32x32x16_bf16
+// (per-wave 64x64 = 2x2 fragments of 32x32; higher MFMA ceiling than the
+// 16x16 shape: 2382 vs 2075 TF ubench). Fragment maps (cdna4):
+//   A/B: lane l holds 8 contiguous k at row/col = l&31, k8 = (l>>5)
+//   C/D: col = lane&31, row = (reg&3) + 8*(reg>>2) + 4*(lane>>5)
+// The 32-lane fragment groups hit LDS banks differently from the 16-lane
+// pattern, so the swizzle is a template knob measured on hardware:
+//   SWZ 0: byte5 ^= row bit2 (st_16x32, as v1)
+//   SWZ 1: byte5:4 ^= row bits1:0 (spreads 32-row groups over 4 slots)
+typedef __attribute__((ext_vector_type(16))) float f32x16;
+
+template <int SWZ>
+DEV_INLINE int swz2(int row, int kbyte) {
+  if (SWZ == 0) return row * (BK * 2) + (kbyte ^ (((row >> 2) & 1) << 5));
+  return row * (BK * 2) + (kbyte ^ ((row & 3) << 4));
+}
+
+template <int SWZ>
+__global__ __launch_bounds__(THREADS, 2)
+void grouped_gemm_nt32_kernel(const uint16_t* __restrict__ Aall,
+                              const uint16_t* __restrict__ Ball,
+                              uint16_t* __restrict__ Oall,
+                              int M, int N, int K,
+                              int64_t strideA, int64_t strideB,
+                              int64_t strideO) {
+  __shared__ uint16_t lds[4 * TILE_ELEMS];
+  const int e = blockIdx.z;
+  const uint16_t* A = Aall + e * strideA;
+  const uint16_t* B = Ball + e * strideB;
+  uint16_t* O = Oall + e * strideO;
+  const int tileM = blockIdx.x * BM;
+  const int tileN = blockIdx.y * BN;
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int wm = wave >> 1;
+  const int wn = wave & 1;
+  const int KT = K / BK;                     // aligned-only kernel
+  const int fr = lane & 31;
+  const int fg = lane >> 5;                  // k8-group 0..1
+  const int maxA = M - 1, maxB = N - 1;
+
+  f32x16 acc[2][2] = {};
+  int buf = 0;
+
+  auto issue_tile = [&](int which, int kt) {
+    const int k0 = kt * BK;
+    uint16_t* base = lds + which * 2 * TILE_ELEMS;
+    #pragma unroll
+    for (int i = 0; i < CHUNKS; ++i) {
+      const int c = wave * 64 + i * 256 + lane;
+      const int row = c >> 3;
+      const int kbB = swz2<SWZ>(0, (c & 7) * 16);  // swizzle of kbyte only
+      const int kbBr = swz2<SWZ>(row, (c & 7) * 16) - row * (BK * 2);
+      {
+        const int r = row > maxA - tileM ? (maxA - tileM < 0 ? 0 : maxA - tileM) : row;
+        const char* gp = reinterpret_cast<const char*>(
+            A + (int64_t)(tileM + r) * K + k0) + kbBr;
+        las_ptr lp = (las_ptr)(reinterpret_cast<char*>(base)
+                               + (wave * 64 + i * 256) * 16);
+        __builtin_amdgcn_global_load_lds((gas_ptr)gp, lp, 16, 0, 0);
+      }
+      {
+        const int r = row > maxB - tileN ? (maxB - tileN < 0 ? 0 : maxB - tileN) : row;
+        const char* gp = reinterpret_cast<const char*>(
+            B + (int64_t)(tileN + r) * K + k0) + kbBr;
+        las_ptr lp = (las_ptr)(reinterpret_cast<char*>(base + TILE_ELEMS)
+                               + (wave * 64 + i * 256) * 16);
+        __builtin_amdgcn_global_load_lds((gas_ptr)gp, lp, 16, 0, 0);
+      }
+    }
+  };
+
+  issue_tile(0, 0);
+  asm volatile("s_waitcnt vmcnt(0)");
+  __syncthreads();
+
+  for (int kt = 0; kt < KT; ++kt) {
+    if (kt + 1 < KT) issue_tile(buf ^ 1, kt + 1);
+    const uint16_t* As = lds + buf * 2 * TILE_ELEMS;
+    const uint16_t* Bs = As + TILE_ELEMS;
+    #pragma unroll
+    for (int kk = 0; kk < BK; kk += 16) {
+      bf16x8 af[2], bf[2];
+      const int kb = (kk + fg * 8) * 2;
+      #pragma unroll
+      for (int m = 0; m < 2; ++m)
+        af[m] = *reinterpret_cast<const bf16x8*>(
+            reinterpret_cast<const char*>(As)
+            + swz2<SWZ>(wm * 64 + m * 32 + fr, kb));
+      #pragma unroll
+      for (int n = 0; n < 2; ++n)
+        bf[n] = *reinterpret_cast<const bf16x8*>(
+            reinterpret_cast<const char*>(Bs)
+            + swz2<SWZ>(wn * 64 + n * 32 + fr, kb));
+      #pragma unroll
+      for (int m = 0; m < 2; ++m)
+        #pragma unroll
+        for (int n = 0; n < 2; ++n)
+          acc[m][n] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+              af[m], bf[n], acc[m][n], 0, 0, 0);
+    }
+    __syncthreads();
+    buf ^= 1;
+  }
+
+  #pragma unroll
+  for (int m = 0; m < 2; ++m) {
+    #pragma unroll
+    for (int n = 0; n < 2; ++n) {
+      #pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        int row = tileM + wm * 64 + m * 32
+                  + (r & 3) + 8 * (r >> 2) + 4 * fg;
+        int col = tileN + wn * 64 + n * 32 + fr;
+        if (row < M && col < N)
+          O[(int64_t)row * N + col] = f32_to_bf16(acc[m][n][r]);
+      }
+    }
+  }
+}
+
+extern "C" void launch_grouped_gemm_nt_v2(const void* A, const void* B,
+                                          void* O, int E, int M, int N, int K,
+                                          int64_t sA, int64_t sB, int64_t sO,
+                                          int swz_mode, hipStream_t stream) {
+  dim3 grid((M + BM - 1) / BM, (N + BN - 1) / BN, E);
+  if (swz_mode == 0)
+    hipLaunchKernelGGL(grouped_gemm_nt32_kernel<0>, grid, dim3(THREADS), 0,
+                       stream, (const uint16_t*)A, (const uint16_t*)B,
+                       (uint16_t*)O, M, N, K, sA, sB, sO);
+  else
+    hipLaunchKernelGGL(grouped_gemm_nt32_kernel<1>, grid, dim3(THREADS), 0,
+                       stream, (const uint16_t*)A, (const uint16_t*)B,
+                       (uint16_t*)O, M, N, K, sA, sB, sO);
+}

@@ -412,8 +412,9 @@ class Trainer:
 
     def add_expert(self) -> bool:
         layers = self._moe_layers()
-        if not layers or self.engine.stage >= 3:
-            return False
+        if not layers or self.engine.stage >= 3 or \
+                (self.mesh is not None and self.mesh.ep_size > 1):
+            return False  # EP shards are fixed-size; use load balancing
         for l in layers:
             l.add_expert()
         self.optimizer.rebuild(self.model)
@@ -424,7 +425,9 @@ class Trainer:
 
     def prune_expert(self, expert_idx: Optional[int] = None) -> bool:
         layers = self._moe_layers()
-        if not layers or layers[0].num_experts <= 2 or self.engine.stage >= 3:
+        if not layers or layers[0].num_experts <= 2 or \
+                self.engine.stage >= 3 or \
+                (self.mesh is not None and self.mesh.ep_size > 1):
             return False
         for l in layers:
             idx = expert_idx
