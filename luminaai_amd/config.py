@@ -355,7 +355,9 @@ class ConfigPresets:
     @staticmethod
     def debug_300m() -> Config:
         return Config(
-            vocab_size=50304, hidden_size=786, num_layers=6, num_heads=4,
+            # 768, not the reference's 786 (config_manager.py:823) — 786 is
+            # not divisible by num_heads=4 and would break head_dim
+            vocab_size=50304, hidden_size=768, num_layers=6, num_heads=4,
             num_kv_heads=4, seq_length=512,
             batch_size=4, micro_batch_size=1, gradient_accumulation_steps=4,
             use_moe=True, use_mod=False, num_experts=8, moe_top_k=2,
