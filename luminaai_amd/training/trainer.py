@@ -273,6 +273,20 @@ class Trainer:
         self.scheduler = create_scheduler(self.optimizer, self.config, total_steps)
 
     def save_checkpoint(self, is_best: bool = False, tag: Optional[str] = None) -> str:
+        # ZeRO-1/2/3: every rank owns a distinct optimizer-state shard —
+        # write them as sibling files (merged by inference.loader's
+        # load_zero_shards, or re-loaded shard-wise on resume)
+        if comm.is_distributed() and self.config.zero_stage >= 1 \
+                and comm.get_rank() != 0:
+            # "optim_shard_" prefix keeps these out of the manager's
+            # "checkpoint_*.pt" discovery glob
+            stem = tag or f"checkpoint_step_{self.global_step}"
+            path = os.path.join(
+                str(self.checkpoints.dir),
+                f"optim_shard_{stem}_rank{comm.get_rank()}.pt")
+            os.makedirs(os.path.dirname(path), exist_ok=True)
+            torch.save({"optimizer_state_dict": self.optimizer.state_dict(),
+                        "global_step": self.global_step}, path)
         if self.mesh is not None and self.mesh.ep_size > 1:
             # every EP rank holds distinct experts: dp_rank 0 of each EP
             # slot writes its own shard file (merge via inference.loader)
@@ -306,8 +320,22 @@ class Trainer:
                 if hasattr(g, "_master_is_params") and not g._master_is_params:
                     g.master.copy_(
                         g.weight_view()[g.shard_lo:g.shard_hi].float())
-        if load_optimizer and payload.get("optimizer_state_dict"):
-            self.optimizer.load_state_dict(payload["optimizer_state_dict"])
+        opt_state = payload.get("optimizer_state_dict")
+        if load_optimizer and comm.is_distributed() \
+                and self.config.zero_stage >= 1 and comm.get_rank() != 0:
+            # this rank's optimizer shard lives in a sibling file
+            base = self.checkpoints.resolve(which)
+            if base is not None:
+                shard_path = os.path.join(
+                    os.path.dirname(str(base)),
+                    f"optim_shard_{os.path.basename(str(base))[:-3]}"
+                    f"_rank{comm.get_rank()}.pt")
+                if os.path.exists(shard_path):
+                    opt_state = torch.load(shard_path, map_location=self.device,
+                                           weights_only=False).get(
+                        "optimizer_state_dict")
+        if load_optimizer and opt_state:
+            self.optimizer.load_state_dict(opt_state)
         if load_optimizer and payload.get("scheduler_state_dict") and self.scheduler:
             self.scheduler.load_state_dict(payload["scheduler_state_dict"])
         self.global_step = payload.get("global_step", 0)
