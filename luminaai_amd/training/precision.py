@@ -36,7 +36,21 @@ _REGISTRY: Dict[str, PrecisionSpec] = {
     "fp16": PrecisionSpec("fp16", torch.float16, torch.float16, True),
     "mixed_fp16": PrecisionSpec("mixed_fp16", torch.float16, torch.float16, True),
     "fp8": PrecisionSpec("fp8", torch.bfloat16, torch.bfloat16, False,
-                         "bf16 weights; fp8 e4m3fn GEMM inputs where wired"),
+                         "bf16 weights; fp8 e4m3fn GEMM forward (ops/fp8.py)"),
+    "fp8_e5m2": PrecisionSpec("fp8_e5m2", torch.bfloat16, torch.bfloat16, False,
+                              "reserved: e5m2 variant of the fp8 path"),
+    "fp64": PrecisionSpec("fp64", torch.float64, torch.float64, False,
+                          "debugging only; no MFMA path"),
+    # registry aliases kept for reference Config compatibility
+    # (trainer.py:157-356 listed int8/int4/... — those were post-training
+    # quantization targets, served here by the inference loaders, not a
+    # training dtype): they map to bf16 training with a warning.
+    "int8": PrecisionSpec("int8", torch.bfloat16, torch.bfloat16, False,
+                          "alias->bf16 (PTQ is an inference concern)"),
+    "int4": PrecisionSpec("int4", torch.bfloat16, torch.bfloat16, False,
+                          "alias->bf16 (PTQ is an inference concern)"),
+    "mixed": PrecisionSpec("mixed", torch.bfloat16, torch.bfloat16, False,
+                           "alias of mixed_bf16"),
 }
 
 

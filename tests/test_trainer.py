@@ -224,3 +224,33 @@ def test_nan_batch_does_not_poison_weights(trainer, tiny_moe_config):
     trainer.engine.step(grad_scale=1.0)
     assert torch.isfinite(trainer.model.embed_tokens.weight.detach()).all()
     assert torch.allclose(trainer.model.embed_tokens.weight.detach(), w0)
+
+
+def test_loss_decreases_on_learnable_data(tiny_config, tokenizer):
+    """Loss-curve regression: a tiny model on repetitive data must learn
+    (the reference had no such test — SURVEY.md §4 gap list)."""
+    import torch
+    from luminaai_amd.data.dataset import ConversationDataset, create_dataloader
+    from luminaai_amd.models import DeepSeekTransformer, config_to_deepseek_config
+    from luminaai_amd.training import Trainer
+    from luminaai_amd.utils import generate_sample_data
+    import tempfile, os
+    with tempfile.TemporaryDirectory() as d:
+        p = os.path.join(d, "train.jsonl")
+        generate_sample_data(p, n=16, seed=0)
+        cfg = tiny_config
+        cfg.learning_rate = 2e-3
+        cfg.num_epochs = 4
+        torch.manual_seed(0)
+        model = DeepSeekTransformer(config_to_deepseek_config(cfg))
+        t = Trainer(model, tokenizer, cfg)
+        ds = ConversationDataset(p, tokenizer, cfg.seq_length)
+        dl = create_dataloader(ds, cfg, shuffle=False)
+        t._setup_scheduler(64)
+        first = None
+        for epoch in range(4):
+            stats = t.train_epoch(dl, epoch)
+            if first is None:
+                first = stats["mean_loss"]
+        assert stats["mean_loss"] < first * 0.8, \
+            f"no learning: {first} -> {stats['mean_loss']}"
