@@ -120,7 +120,7 @@ void grouped_gemm_nt_kernel(const uint16_t* __restrict__ Aall,
     };
 
     issue_tile(0, 0);
-    asm volatile("s_waitcnt vmcnt(0)");
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
     __syncthreads();
 
     for (int kt = 0; kt < KT; ++kt) {
@@ -338,7 +338,7 @@ void grouped_gemm_nt32_kernel(const uint16_t* __restrict__ Aall,
   };
 
   issue_tile(0, 0);
-  asm volatile("s_waitcnt vmcnt(0)");
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
   __syncthreads();
 
   for (int kt = 0; kt < KT; ++kt) {
@@ -399,4 +399,134 @@ extern "C" void launch_grouped_gemm_nt_v2(const void* A, const void* B,
     hipLaunchKernelGGL(grouped_gemm_nt32_kernel<1>, grid, dim3(THREADS), 0,
                        stream, (const uint16_t*)A, (const uint16_t*)B,
                        (uint16_t*)O, M, N, K, sA, sB, sO);
+}
+
+// ---------------------------------------------------------------------------
+// v3: 16x16x32 with THREE LDS buffers, counted vmcnt and raw s_barrier —
+// tiles stay in flight across barriers instead of draining (the 2-buffer
+// path's __syncthreads carries a vmcnt(0) while a glds is outstanding).
+// Per-wave glds queues are same-aged across waves (uniform staging order),
+// so one counted vmcnt at every wave + a barrier guarantees the tile all
+// waves are about to read has fully landed. 96 KiB LDS -> 1 block/CU.
+template <int DUMMY>
+__global__ __launch_bounds__(THREADS, 1)
+void grouped_gemm_nt3_kernel(const uint16_t* __restrict__ Aall,
+                             const uint16_t* __restrict__ Ball,
+                             uint16_t* __restrict__ Oall,
+                             int M, int N, int K,
+                             int64_t strideA, int64_t strideB,
+                             int64_t strideO) {
+  __shared__ uint16_t lds[6 * TILE_ELEMS];   // 3 buffers x [A|B]
+  const int e = blockIdx.z;
+  const uint16_t* A = Aall + e * strideA;
+  const uint16_t* B = Ball + e * strideB;
+  uint16_t* O = Oall + e * strideO;
+  const int tileM = blockIdx.x * BM;
+  const int tileN = blockIdx.y * BN;
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int wm = wave >> 1;
+  const int wn = wave & 1;
+  const int KT = K / BK;
+  const int fr = lane & 15;
+  const int fg = lane >> 4;
+  const int maxA = M - 1, maxB = N - 1;
+
+  f32x4 acc[4][4] = {};
+
+  auto issue_tile = [&](int which, int kt) {
+    const int k0 = kt * BK;
+    uint16_t* base = lds + which * 2 * TILE_ELEMS;
+    #pragma unroll
+    for (int i = 0; i < CHUNKS; ++i) {
+      const int c = wave * 64 + i * 256 + lane;
+      const int row = c >> 3;
+      const int kbB = ((c & 7) * 16) ^ (((row >> 2) & 1) << 5);
+      {
+        const int r = row > maxA - tileM ? (maxA - tileM < 0 ? 0 : maxA - tileM) : row;
+        const char* gp = reinterpret_cast<const char*>(
+            A + (int64_t)(tileM + r) * K + k0) + kbB;
+        las_ptr lp = (las_ptr)(reinterpret_cast<char*>(base)
+                               + (wave * 64 + i * 256) * 16);
+        __builtin_amdgcn_global_load_lds((gas_ptr)gp, lp, 16, 0, 0);
+      }
+      {
+        const int r = row > maxB - tileN ? (maxB - tileN < 0 ? 0 : maxB - tileN) : row;
+        const char* gp = reinterpret_cast<const char*>(
+            B + (int64_t)(tileN + r) * K + k0) + kbB;
+        las_ptr lp = (las_ptr)(reinterpret_cast<char*>(base + TILE_ELEMS)
+                               + (wave * 64 + i * 256) * 16);
+        __builtin_amdgcn_global_load_lds((gas_ptr)gp, lp, 16, 0, 0);
+      }
+    }
+  };
+
+  // prologue: t0 and t1 in flight; wait for t0 only (t1 keeps flying)
+  issue_tile(0, 0);
+  if (KT > 1) issue_tile(1, 1);
+  if (KT > 1)
+    asm volatile("s_waitcnt vmcnt(8)" ::: "memory");   // t1's 8 glds may stay outstanding
+  else
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __builtin_amdgcn_s_barrier();
+
+  for (int kt = 0; kt < KT; ++kt) {
+    if (kt + 2 < KT)
+      issue_tile((kt + 2) % 3, kt + 2);
+    const uint16_t* As = lds + (kt % 3) * 2 * TILE_ELEMS;
+    const uint16_t* Bs = As + TILE_ELEMS;
+    #pragma unroll
+    for (int kk = 0; kk < BK; kk += 32) {
+      bf16x8 af[4], bf[4];
+      const int kb = (kk + fg * 8) * 2;
+      #pragma unroll
+      for (int m = 0; m < 4; ++m)
+        af[m] = *reinterpret_cast<const bf16x8*>(
+            reinterpret_cast<const char*>(As)
+            + swz(wm * 64 + m * 16 + fr, kb));
+      #pragma unroll
+      for (int n = 0; n < 4; ++n)
+        bf[n] = *reinterpret_cast<const bf16x8*>(
+            reinterpret_cast<const char*>(Bs)
+            + swz(wn * 64 + n * 16 + fr, kb));
+      #pragma unroll
+      for (int m = 0; m < 4; ++m)
+        #pragma unroll
+        for (int n = 0; n < 4; ++n)
+          acc[m][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              af[m], bf[n], acc[m][n], 0, 0, 0);
+    }
+    // next iteration reads tile kt+1: wait until only the newest tile's
+    // glds can still be in flight, then rendezvous
+    if (kt + 2 < KT)
+      asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
+    else
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+  }
+
+  #pragma unroll
+  for (int m = 0; m < 4; ++m) {
+    #pragma unroll
+    for (int n = 0; n < 4; ++n) {
+      #pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int row = tileM + wm * 64 + m * 16 + fg * 4 + r;
+        int col = tileN + wn * 64 + n * 16 + fr;
+        if (row < M && col < N)
+          O[(int64_t)row * N + col] = f32_to_bf16(acc[m][n][r]);
+      }
+    }
+  }
+}
+
+extern "C" void launch_grouped_gemm_nt_v3(const void* A, const void* B,
+                                          void* O, int E, int M, int N, int K,
+                                          int64_t sA, int64_t sB, int64_t sO,
+                                          hipStream_t stream) {
+  dim3 grid((M + BM - 1) / BM, (N + BN - 1) / BN, E);
+  hipLaunchKernelGGL(grouped_gemm_nt3_kernel<0>, grid, dim3(THREADS), 0,
+                     stream, (const uint16_t*)A, (const uint16_t*)B,
+                     (uint16_t*)O, M, N, K, sA, sB, sO);
 }

@@ -31,3 +31,5 @@ r2 = torch.matmul(a2.float(), b2.float().transpose(1,2))
 for mode in (0,1):
     o2 = ext.grouped_gemm_nt_v2(a2, b2, mode)
     print("ragged mode", mode, "relerr", float((o2.float()-r2).abs().max()/r2.abs().max()))
+check(ext.grouped_gemm_nt_v3(a, b), "v3")
+bench(lambda: ext.grouped_gemm_nt_v3(a, b), "v3-3buf-vmcnt")
