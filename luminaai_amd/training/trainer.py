@@ -226,10 +226,11 @@ class Trainer:
             if eval_loader is not None:
                 stats["eval"] = self.evaluate(eval_loader)
             history.append(stats)
-            if comm.get_rank() == 0:
-                self.save_checkpoint(
-                    is_best=stats.get("eval", {}).get("loss", float("inf"))
-                    <= self.best_eval_loss)
+            # every rank participates (ZeRO shard writes + Z3 gather
+            # collectives); only the designated rank writes the main file
+            self.save_checkpoint(
+                is_best=stats.get("eval", {}).get("loss", float("inf"))
+                <= self.best_eval_loss)
             if self.should_stop:
                 break
         return {"epochs": history, "global_step": self.global_step,
@@ -273,32 +274,36 @@ class Trainer:
         self.scheduler = create_scheduler(self.optimizer, self.config, total_steps)
 
     def save_checkpoint(self, is_best: bool = False, tag: Optional[str] = None) -> str:
-        # ZeRO-1/2/3: every rank owns a distinct optimizer-state shard —
-        # write them as sibling files (merged by inference.loader's
-        # load_zero_shards, or re-loaded shard-wise on resume)
-        if comm.is_distributed() and self.config.zero_stage >= 1 \
-                and comm.get_rank() != 0:
-            # "optim_shard_" prefix keeps these out of the manager's
-            # "checkpoint_*.pt" discovery glob
-            stem = tag or f"checkpoint_step_{self.global_step}"
-            path = os.path.join(
-                str(self.checkpoints.dir),
-                f"optim_shard_{stem}_rank{comm.get_rank()}.pt")
-            os.makedirs(os.path.dirname(path), exist_ok=True)
-            torch.save({"optimizer_state_dict": self.optimizer.state_dict(),
-                        "global_step": self.global_step}, path)
-        if self.mesh is not None and self.mesh.ep_size > 1:
-            # every EP rank holds distinct experts: dp_rank 0 of each EP
-            # slot writes its own shard file (merge via inference.loader)
-            if self.mesh.dp_rank != 0:
-                return ""
-            tag = (tag or f"checkpoint_step_{self.global_step}") + \
-                f"_ep_rank_{self.mesh.ep_rank}"
-        elif comm.get_rank() != 0:
-            return ""
+        """CALL ON EVERY RANK when distributed: ZeRO-3 gathering is a
+        collective, and non-zero ranks write their own optimizer shards.
+        Only the designated rank writes the main checkpoint file."""
+        # Z3 weight gather involves all ranks — enter the context everywhere
         ctx = self.engine.gathered_weights() if self.engine.stage >= 3 \
             else contextlib.nullcontext()
         with ctx:
+            # ZeRO-1/2/3: every rank owns a distinct optimizer-state shard —
+            # write them as sibling files (merged by inference.loader's
+            # load_zero_shards, or re-loaded shard-wise on resume).
+            # "optim_shard_" prefix keeps them out of the manager's
+            # "checkpoint_*.pt" discovery glob.
+            if comm.is_distributed() and self.config.zero_stage >= 1 \
+                    and comm.get_rank() != 0:
+                stem = tag or f"checkpoint_step_{self.global_step}"
+                path = os.path.join(
+                    str(self.checkpoints.dir),
+                    f"optim_shard_{stem}_rank{comm.get_rank()}.pt")
+                os.makedirs(os.path.dirname(path), exist_ok=True)
+                torch.save({"optimizer_state_dict": self.optimizer.state_dict(),
+                            "global_step": self.global_step}, path)
+            if self.mesh is not None and self.mesh.ep_size > 1:
+                # every EP rank holds distinct experts: dp_rank 0 of each EP
+                # slot writes its own shard file (merge via inference.loader)
+                if self.mesh.dp_rank != 0:
+                    return ""
+                tag = (tag or f"checkpoint_step_{self.global_step}") + \
+                    f"_ep_rank_{self.mesh.ep_rank}"
+            elif comm.get_rank() != 0:
+                return ""
             path = self.checkpoints.save_checkpoint(
                 self.model, self.optimizer, self.scheduler,
                 global_step=self.global_step, epoch=self.epoch,
