@@ -175,3 +175,27 @@ def test_zero3_ranks_in_sync_and_match_ddp():
     ddp = _spawn("ddp_ref_worker")
     assert z3[0]["checksum"] == pytest.approx(ddp[0]["checksum"], rel=1e-4)
     assert z3[0]["grad_norm"] == pytest.approx(ddp[0]["grad_norm"], rel=1e-4)
+
+
+def zero3_save_worker(rank, world):
+    """Multi-rank ZeRO-3 save must not deadlock (gather is collective) and
+    must write rank-0 main file + rank-1 optimizer shard."""
+    from luminaai_amd.training import CheckpointManager
+    tmp = os.environ["Z3_SAVE_TMP"]
+    t = _train(_cfg(3, rank=rank), steps=1, seed=500 + rank)
+    t.checkpoints = CheckpointManager(os.path.join(tmp, "z3_ckpts"))
+    path = t.save_checkpoint()
+    dist.barrier()
+    files = sorted(os.listdir(os.path.join(tmp, "z3_ckpts")))
+    return {"path": path, "files": files}
+
+
+def test_zero3_multirank_save(tmp_path):
+    os.environ["Z3_SAVE_TMP"] = str(tmp_path)
+    res = _spawn("zero3_save_worker")
+    assert res[0]["path"].endswith(".pt")
+    assert res[1]["path"] == ""          # rank1 writes only its shard
+    names = res[0]["files"]
+    assert any(n.startswith("checkpoint_step_") for n in names), names
+    assert any(n.startswith("optim_shard_") and "rank1" in n
+               for n in names), names
