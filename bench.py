@@ -81,10 +81,15 @@ def main():
          and cfg.num_experts % world == 0 else 1)
     from luminaai_amd.parallel.mesh import init_mesh
     init_mesh(ep if world > 1 else 1)
+    if device.type == "cuda":
+        torch.cuda.set_device(device)
     torch.manual_seed(cfg.seed + rank)
     tok = ConversationTokenizer(max_length=cfg.seq_length)
     model_cfg = config_to_deepseek_config(cfg)
-    model = DeepSeekTransformer(model_cfg)
+    # build + random-init straight on the GPU (a ~10 GB fp32 CPU init per
+    # rank would serialize 8-rank startup)
+    with torch.device(device):
+        model = DeepSeekTransformer(model_cfg)
     trainer = Trainer(model, tok, cfg)
     trainer._setup_scheduler(args.steps + args.warmup + 10)
 
