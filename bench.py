@@ -45,8 +45,13 @@ def build_config(args):
     cfg.seq_length = args.seq_len or cfg.seq_length
     # 288 GB HBM3E: the b1 flagship never needs activation recompute
     cfg.gradient_checkpointing = args.checkpointing
+    # ZeRO-1 by default for multi-GPU: in this engine grads live in the full
+    # flat buffer under stage 1 AND 2 (p.grad views need it), so the stages
+    # have equal memory — but stage 1's bucketed all-reduce overlaps with
+    # backward (post-accumulate hooks) while stage 2's reduce-scatter is an
+    # exposed boundary collective. Override with --zero 2.
     cfg.zero_stage = args.zero if args.zero is not None else \
-        (2 if comm.env_world_size() > 1 else 0)
+        (1 if comm.env_world_size() > 1 else 0)
     cfg.eval_every_n_batches = 0
     cfg.save_every_n_batches = 0
     return cfg
