@@ -289,7 +289,6 @@ class GroupedQueryAttention(nn.Module):
         )
         out = out.transpose(1, 2)
         if run_sp:
-            from ..parallel.sequence_parallel import scatter_seq_gather_heads
             out = scatter_seq_gather_heads(out.contiguous(), self.sp_size,
                                            self.sp_group)
         out = out.reshape(B, S, self.q_size)
