@@ -49,6 +49,7 @@ hipError_t lumina_adamw_step(float*, const void*, int, float*, float*, void*, in
 void launch_grouped_gemm_nt(const void*, const void*, void*, int, int, int, int, int64_t, int64_t, int64_t, hipStream_t);
 void launch_grouped_gemm_nt_v2(const void*, const void*, void*, int, int, int, int, int64_t, int64_t, int64_t, int, hipStream_t);
 void launch_grouped_gemm_nt_v3(const void*, const void*, void*, int, int, int, int, int64_t, int64_t, int64_t, hipStream_t);
+void launch_grouped_gemm_nt_v4(const void*, const void*, void*, int, int, int, int, int64_t, int64_t, int64_t, hipStream_t);
 hipError_t lumina_moe_gather_rows(const void*, const int64_t*, const bool*, void*, int64_t, int, int, hipStream_t);
 hipError_t lumina_moe_dispatch_bwd(const void*, const int64_t*, void*, int64_t, int, int, int64_t, int, hipStream_t);
 hipError_t lumina_moe_combine_fwd(const void*, const float*, const int64_t*, void*, int64_t, int, int, int64_t, int, hipStream_t);
@@ -388,7 +389,24 @@ at::Tensor grouped_gemm_nt_v3(const at::Tensor& A, const at::Tensor& B) {
   return O;
 }
 
+at::Tensor grouped_gemm_nt_v4(const at::Tensor& A, const at::Tensor& B) {
+  TORCH_CHECK(A.is_contiguous() && B.is_contiguous());
+  TORCH_CHECK(is_bf16(A) && is_bf16(B));
+  TORCH_CHECK(A.dim() == 3 && B.dim() == 3 && A.size(2) == B.size(2));
+  TORCH_CHECK(A.size(2) % 64 == 0, "v4 kernel requires K % 64 == 0");
+  const int E = (int)A.size(0), M = (int)A.size(1);
+  const int K = (int)A.size(2), N = (int)B.size(1);
+  auto O = at::empty({E, M, N}, A.options());
+  launch_grouped_gemm_nt_v4(A.data_ptr(), B.data_ptr(), O.data_ptr(),
+                            E, M, N, K, (int64_t)M * K, (int64_t)N * K,
+                            (int64_t)M * N, cur_stream());
+  check_hip(hipGetLastError(), "grouped_gemm_nt_v4");
+  return O;
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
+  mod.def("grouped_gemm_nt_v4", &grouped_gemm_nt_v4,
+          "2-buffer raw-barrier counted-vmcnt variant (K%64==0)");
   mod.def("grouped_gemm_nt_v3", &grouped_gemm_nt_v3,
           "3-buffer counted-vmcnt variant (K%64==0)");
   mod.def("grouped_gemm_nt", &grouped_gemm_nt,

@@ -530,3 +530,123 @@ extern "C" void launch_grouped_gemm_nt_v3(const void* A, const void* B,
                      stream, (const uint16_t*)A, (const uint16_t*)B,
                      (uint16_t*)O, M, N, K, sA, sB, sO);
 }
+
+// ---------------------------------------------------------------------------
+// v4: v1's 2-buffer glds geometry, but raw s_barrier + issue-then-counted-
+// vmcnt ordering: each iteration first issues tile kt+1 (8 glds), then waits
+// vmcnt(8) — draining exactly the OLDER tile kt while kt+1 stays in flight —
+// computes, and crosses a raw barrier with the prefetch still outstanding
+// (same-aged per-wave queues make the counted wait a global guarantee).
+// Keeps 64 KiB LDS -> 2 blocks/CU, unlike the 3-buffer v3.
+template <int DUMMY>
+__global__ __launch_bounds__(THREADS, 2)
+void grouped_gemm_nt4_kernel(const uint16_t* __restrict__ Aall,
+                             const uint16_t* __restrict__ Ball,
+                             uint16_t* __restrict__ Oall,
+                             int M, int N, int K,
+                             int64_t strideA, int64_t strideB,
+                             int64_t strideO) {
+  __shared__ uint16_t lds[4 * TILE_ELEMS];
+  const int e = blockIdx.z;
+  const uint16_t* A = Aall + e * strideA;
+  const uint16_t* B = Ball + e * strideB;
+  uint16_t* O = Oall + e * strideO;
+  const int tileM = blockIdx.x * BM;
+  const int tileN = blockIdx.y * BN;
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int wm = wave >> 1;
+  const int wn = wave & 1;
+  const int KT = K / BK;
+  const int fr = lane & 15;
+  const int fg = lane >> 4;
+  const int maxA = M - 1, maxB = N - 1;
+  f32x4 acc[4][4] = {};
+
+  auto issue_tile = [&](int which, int kt) {
+    const int k0 = kt * BK;
+    uint16_t* base = lds + which * 2 * TILE_ELEMS;
+    #pragma unroll
+    for (int i = 0; i < CHUNKS; ++i) {
+      const int c = wave * 64 + i * 256 + lane;
+      const int row = c >> 3;
+      const int kbB = ((c & 7) * 16) ^ (((row >> 2) & 1) << 5);
+      {
+        const int r = row > maxA - tileM ? (maxA - tileM < 0 ? 0 : maxA - tileM) : row;
+        const char* gp = reinterpret_cast<const char*>(
+            A + (int64_t)(tileM + r) * K + k0) + kbB;
+        las_ptr lp = (las_ptr)(reinterpret_cast<char*>(base)
+                               + (wave * 64 + i * 256) * 16);
+        __builtin_amdgcn_global_load_lds((gas_ptr)gp, lp, 16, 0, 0);
+      }
+      {
+        const int r = row > maxB - tileN ? (maxB - tileN < 0 ? 0 : maxB - tileN) : row;
+        const char* gp = reinterpret_cast<const char*>(
+            B + (int64_t)(tileN + r) * K + k0) + kbB;
+        las_ptr lp = (las_ptr)(reinterpret_cast<char*>(base + TILE_ELEMS)
+                               + (wave * 64 + i * 256) * 16);
+        __builtin_amdgcn_global_load_lds((gas_ptr)gp, lp, 16, 0, 0);
+      }
+    }
+  };
+
+  issue_tile(0, 0);
+  int buf = 0;
+  for (int kt = 0; kt < KT; ++kt) {
+    if (kt + 1 < KT) {
+      issue_tile(buf ^ 1, kt + 1);
+      asm volatile("s_waitcnt vmcnt(8)" ::: "memory");   // tile kt landed
+    } else {
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    }
+    __builtin_amdgcn_s_barrier();      // all waves see tile kt in LDS
+    const uint16_t* As = lds + buf * 2 * TILE_ELEMS;
+    const uint16_t* Bs = As + TILE_ELEMS;
+    #pragma unroll
+    for (int kk = 0; kk < BK; kk += 32) {
+      bf16x8 af[4], bf[4];
+      const int kb = (kk + fg * 8) * 2;
+      #pragma unroll
+      for (int m = 0; m < 4; ++m)
+        af[m] = *reinterpret_cast<const bf16x8*>(
+            reinterpret_cast<const char*>(As)
+            + swz(wm * 64 + m * 16 + fr, kb));
+      #pragma unroll
+      for (int n = 0; n < 4; ++n)
+        bf[n] = *reinterpret_cast<const bf16x8*>(
+            reinterpret_cast<const char*>(Bs)
+            + swz(wn * 64 + n * 16 + fr, kb));
+      #pragma unroll
+      for (int m = 0; m < 4; ++m)
+        #pragma unroll
+        for (int n = 0; n < 4; ++n)
+          acc[m][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              af[m], bf[n], acc[m][n], 0, 0, 0);
+    }
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();      // done reading buf; next iter reuses it
+    buf ^= 1;
+  }
+
+  #pragma unroll
+  for (int m = 0; m < 4; ++m)
+    #pragma unroll
+    for (int n = 0; n < 4; ++n)
+      #pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int row = tileM + wm * 64 + m * 16 + fg * 4 + r;
+        int col = tileN + wn * 64 + n * 16 + fr;
+        if (row < M && col < N)
+          O[(int64_t)row * N + col] = f32_to_bf16(acc[m][n][r]);
+      }
+}
+
+extern "C" void launch_grouped_gemm_nt_v4(const void* A, const void* B,
+                                          void* O, int E, int M, int N, int K,
+                                          int64_t sA, int64_t sB, int64_t sO,
+                                          hipStream_t stream) {
+  dim3 grid((M + BM - 1) / BM, (N + BN - 1) / BN, E);
+  hipLaunchKernelGGL(grouped_gemm_nt4_kernel<0>, grid, dim3(THREADS), 0,
+                     stream, (const uint16_t*)A, (const uint16_t*)B,
+                     (uint16_t*)O, M, N, K, sA, sB, sO);
+}

@@ -33,3 +33,11 @@ for mode in (0,1):
     print("ragged mode", mode, "relerr", float((o2.float()-r2).abs().max()/r2.abs().max()))
 check(ext.grouped_gemm_nt_v3(a, b), "v3")
 bench(lambda: ext.grouped_gemm_nt_v3(a, b), "v3-3buf-vmcnt")
+check(ext.grouped_gemm_nt_v4(a, b), "v4")
+bench(lambda: ext.grouped_gemm_nt_v4(a, b), "v4-2buf-raw")
+# race screen: repeated runs must be deterministic
+o1 = ext.grouped_gemm_nt_v4(a, b)
+for _ in range(5):
+    d = (ext.grouped_gemm_nt_v4(a, b) - o1).abs().max()
+    assert float(d) == 0.0, f"nondeterminism {float(d)}"
+print("v4 race-screen ok")
