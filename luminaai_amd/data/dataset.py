@@ -195,7 +195,13 @@ class HybridDatasetManager:
 
 # ----------------------------------------------------------------------
 def create_dataloader(dataset: Dataset, config, shuffle: bool = True) -> DataLoader:
-    """Tuned DataLoader (reference create_fast_dataloader, dataset.py:807-845)."""
+    """Tuned DataLoader (reference create_fast_dataloader, dataset.py:807-845).
+
+    Under data parallelism each rank gets a disjoint 1/world slice via
+    DistributedSampler (the DP group for data purposes is the world minus
+    sequence parallelism: SP ranks of one replica must see the SAME batch
+    and slice it along the sequence)."""
+    import torch.distributed as dist
     is_iterable = isinstance(dataset, IterableDataset)
     kwargs = dict(
         batch_size=config.micro_batch_size or 1,
@@ -203,7 +209,18 @@ def create_dataloader(dataset: Dataset, config, shuffle: bool = True) -> DataLoa
         pin_memory=config.pin_memory and torch.cuda.is_available(),
         drop_last=True,
     )
-    if not is_iterable:
+    if not is_iterable and dist.is_available() and dist.is_initialized() \
+            and dist.get_world_size() > 1:
+        from ..parallel.mesh import get_mesh
+        mesh = get_mesh()
+        sp = mesh.sp_size if mesh is not None else 1
+        world = dist.get_world_size() // max(sp, 1)
+        rank = dist.get_rank() // max(sp, 1)
+        from torch.utils.data.distributed import DistributedSampler
+        kwargs["sampler"] = DistributedSampler(
+            dataset, num_replicas=world, rank=rank, shuffle=shuffle,
+            seed=config.seed, drop_last=True)
+    elif not is_iterable:
         kwargs["shuffle"] = shuffle
     if config.num_workers > 0:
         kwargs["prefetch_factor"] = config.prefetch_factor
