@@ -357,6 +357,13 @@ class Trainer:
         if self._metrics_hook is None:
             return
         m = self.get_current_metrics()
+        # Distributed: the orchestrators on every rank must reach IDENTICAL
+        # decisions from their independent monitor threads — average the
+        # loss across ranks here (training thread, so the collective is safe
+        # and aligned across ranks; grad_norm/lr/step are already global).
+        if comm.is_distributed() and m.loss is not None:
+            m.loss = comm.all_reduce_scalar(float(m.loss), op="sum") \
+                / comm.get_world_size()
         try:
             self._metrics_hook(m)
         except Exception:  # noqa: BLE001 — monitoring must never kill training
