@@ -178,3 +178,40 @@ def prepare_text_corpus(input_path: str, out_path: str, source: str = "generic",
                 n_out += 1
     return {"lines_in": n_in, "kept": n_out, "source": source,
             "out_path": out_path}
+
+
+def main(argv=None):
+    """CLI: prepare local corpora (reference Dataset_download.py /
+    multi_source_dataset.py CLIs, network-optional)."""
+    import argparse
+    ap = argparse.ArgumentParser(prog="luminaai-amd data",
+                                 description="dataset acquisition/preparation")
+    sub = ap.add_subparsers(dest="cmd", required=True)
+    p1 = sub.add_parser("oasst", help="prepare OASST message trees")
+    p1.add_argument("input", help="local trees .jsonl (or 'download')")
+    p1.add_argument("--out", default="data/oasst")
+    p1.add_argument("--min-quality", type=float, default=0.5)
+    p2 = sub.add_parser("text", help="clean a raw text/JSONL corpus")
+    p2.add_argument("input")
+    p2.add_argument("--out", default="data/corpus.txt")
+    p2.add_argument("--source", default="generic",
+                    choices=sorted(SOURCE_PROCESSORS) + ["generic"])
+    p3 = sub.add_parser("validate", help="validate a conversations JSONL")
+    p3.add_argument("input")
+    args = ap.parse_args(argv)
+    if args.cmd == "oasst":
+        stats = (download_oasst(args.out, min_quality=args.min_quality)
+                 if args.input == "download"
+                 else prepare_oasst(args.input, args.out,
+                                    min_quality=args.min_quality))
+    elif args.cmd == "text":
+        stats = prepare_text_corpus(args.input, args.out, source=args.source)
+    else:
+        from ..utils.data_processing import validate_jsonl
+        stats = validate_jsonl(args.input)
+    print(json.dumps(stats, indent=2, default=str))
+    return stats
+
+
+if __name__ == "__main__":
+    main()

@@ -599,6 +599,38 @@ class Trainer:
                         f"accum={self.accum_steps} (attempt {attempt + 1})")
         raise RuntimeError("training failed after 10 OOM retries")
 
+    def profile_training_loop_overhead(self, batch: Dict, iters: int = 5) -> Dict:
+        """Per-phase wall breakdown of one step (reference trainer.py:3821).
+        Returns ms for h2d / forward / backward / optimizer."""
+        import time as _time
+
+        def _sync():
+            if self.device.type == "cuda":
+                torch.cuda.synchronize()
+
+        res = {"h2d_ms": 0.0, "forward_ms": 0.0, "backward_ms": 0.0,
+               "optimizer_ms": 0.0}
+        self.model.train()
+        for _ in range(iters):
+            _sync(); t0 = _time.perf_counter()
+            ids = batch["input_ids"].to(self.device, non_blocking=False)
+            labels = batch["labels"].to(self.device, non_blocking=False)
+            _sync(); t1 = _time.perf_counter()
+            logits, aux, _ = self.model(ids)
+            out = self.compute_loss(logits, labels, None, aux)
+            _sync(); t2 = _time.perf_counter()
+            out["loss"].backward()
+            _sync(); t3 = _time.perf_counter()
+            self.engine.set_sync(True)
+            self._micro_in_cycle = 1
+            self.optimizer_step()
+            _sync(); t4 = _time.perf_counter()
+            res["h2d_ms"] += (t1 - t0) * 1e3
+            res["forward_ms"] += (t2 - t1) * 1e3
+            res["backward_ms"] += (t3 - t2) * 1e3
+            res["optimizer_ms"] += (t4 - t3) * 1e3
+        return {k: round(v / iters, 3) for k, v in res.items()}
+
     # ---- throughput --------------------------------------------------------
     def _calculate_throughput(self) -> float:
         now = time.perf_counter()

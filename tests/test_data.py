@@ -133,3 +133,13 @@ def test_conversation_quality():
     degenerate = {"messages": [{"role": "user", "content": "a" * 500},
                                {"role": "assistant", "content": "b" * 500}]}
     assert conversation_quality(degenerate) < 0.5
+
+
+def test_acquisition_cli(tmp_path, capsys):
+    import json as _json
+    from luminaai_amd.data.acquisition import main as acq_main
+    raw = tmp_path / "c.jsonl"
+    raw.write_text('{"messages": [{"role": "user", "content": "a question here"},'
+                   '{"role": "assistant", "content": "a fine answer here"}]}\n')
+    stats = acq_main(["validate", str(raw)])
+    assert stats["valid"] == 1

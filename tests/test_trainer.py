@@ -254,3 +254,15 @@ def test_loss_decreases_on_learnable_data(tiny_config, tokenizer):
                 first = stats["mean_loss"]
         assert stats["mean_loss"] < first * 0.8, \
             f"no learning: {first} -> {stats['mean_loss']}"
+
+
+def test_profile_training_loop_overhead(tiny_config, tokenizer, small_model):
+    from luminaai_amd.training import Trainer
+    import torch
+    t = Trainer(small_model, tokenizer, tiny_config)
+    t._setup_scheduler(10)
+    ids = torch.randint(1, 512, (2, 33))
+    res = t.profile_training_loop_overhead(
+        {"input_ids": ids[:, :-1], "labels": ids[:, 1:]}, iters=2)
+    assert set(res) == {"h2d_ms", "forward_ms", "backward_ms", "optimizer_ms"}
+    assert all(v >= 0 for v in res.values())
