@@ -388,6 +388,11 @@ class Trainer:
         d = self._metric_floats()
         mem = (torch.cuda.memory_allocated() / 1e9
                if torch.cuda.is_available() else 0.0)
+        # MoE routing stats walk every layer and sync device scalars —
+        # refresh them every few steps, not per emission
+        self._stats_tick = getattr(self, "_stats_tick", 0) + 1
+        if self._stats_tick % 10 == 1 or not hasattr(self, "_stats_cache"):
+            self._stats_cache = self._extract_moe_routing_stats()
         return TrainingMetrics(
             step=self.global_step, epoch=self.epoch,
             loss=d.get("loss"), aux_loss=d.get("aux_loss"),
@@ -396,7 +401,7 @@ class Trainer:
             accuracy=d.get("accuracy"),
             perplexity=math.exp(min(d.get("ce_loss", 20.0), 20.0))
             if "ce_loss" in d else None,
-            memory_gb=mem, expert_stats=self._extract_moe_routing_stats(),
+            memory_gb=mem, expert_stats=self._stats_cache,
             timestamp=time.time())
 
     def get_lr(self) -> float:
