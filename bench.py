@@ -54,6 +54,8 @@ def build_config(args):
         (1 if comm.env_world_size() > 1 else 0)
     cfg.eval_every_n_batches = 0
     cfg.save_every_n_batches = 0
+    if args.precision:
+        cfg.precision = args.precision
     return cfg
 
 
@@ -72,6 +74,9 @@ def main():
                          "divides num_experts)")
     ap.add_argument("--checkpointing", action="store_true",
                     help="enable activation recompute (off by default)")
+    ap.add_argument("--precision", default=None,
+                    help="override compute precision (e.g. fp8 for e4m3 "
+                         "MFMA GEMM forwards; headline stays bf16)")
     args = ap.parse_args()
 
     distributed = comm.init_distributed()

@@ -416,13 +416,17 @@ class MoEFFNLayer(nn.Module):
             bufv = bufv.view(self.ep_size, EL, C, h).transpose(0, 1) \
                 .reshape(EL, self.ep_size * C, h)       # tokens per local expert
 
-        # --- grouped expert GEMMs (hipBLASLt strided-batched)
-        gu = ops.interface.expert_bmm(bufv, self.w_gate_up.to(x.dtype))
+        # --- grouped expert GEMMs (hipBLASLt strided-batched; optional fp8
+        # e4m3 MFMA forward when the precision manager enables it)
+        if getattr(self, "use_fp8", False):
+            from ..ops.fp8 import expert_bmm_fp8 as _ebmm
+        else:
+            _ebmm = ops.interface.expert_bmm
+        gu = _ebmm(bufv, self.w_gate_up.to(x.dtype))
         I = self.intermediate_size
         gu2 = gu.reshape(-1, 2 * I)
         act = ops.swiglu(gu2.narrow(1, 0, I), gu2.narrow(1, I, I))
-        y = ops.interface.expert_bmm(act.view(EL, -1, I),
-                                     self.w_down.to(x.dtype))
+        y = _ebmm(act.view(EL, -1, I), self.w_down.to(x.dtype))
 
         if self.ep_size > 1:
             from ..parallel.expert_parallel import all_to_all

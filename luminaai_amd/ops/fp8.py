@@ -72,6 +72,47 @@ class FP8Linear(nn.Linear):
         return super().forward(x)
 
 
+class _FP8ExpertBmmFn(torch.autograd.Function):
+    """Grouped expert GEMM with fp8 e4m3 forward (per-expert 2D scaled_mm —
+    hipBLASLt has no batched fp8 entry; E launches amortise fine) and bf16
+    backward through the MFMA NT kernel + transA bmm."""
+
+    @staticmethod
+    def forward(ctx, x, w):
+        ctx.save_for_backward(x, w)
+        E, C, K = x.shape
+        N = w.shape[2]
+        out = x.new_empty(E, C, N)
+        for e in range(E):
+            xq, xs = quantize_e4m3(x[e])
+            wq, ws = quantize_e4m3(w[e].t().contiguous())   # [N, K]
+            out[e] = torch._scaled_mm(xq, wq.t(), scale_a=xs, scale_b=ws,
+                                      out_dtype=torch.bfloat16)
+        return out
+
+    @staticmethod
+    def backward(ctx, go):
+        from .interface import grouped_gemm_nt
+        x, w = ctx.saved_tensors
+        go = go.contiguous()
+        gx = gw = None
+        if ctx.needs_input_grad[0]:
+            gx = grouped_gemm_nt(go, w)
+        if ctx.needs_input_grad[1]:
+            gw = torch.bmm(x.transpose(1, 2), go)
+        return gx, gw
+
+
+def expert_bmm_fp8(x: torch.Tensor, w: torch.Tensor) -> torch.Tensor:
+    """fp8-forward grouped expert GEMM: x [E,C,K] @ w [E,K,N] -> [E,C,N].
+    Requires K,N % 16 == 0 and a GPU; falls back to bf16 bmm otherwise."""
+    if x.is_cuda and x.shape[-1] % 16 == 0 and w.shape[-1] % 16 == 0 \
+            and x.shape[1] % 16 == 0:
+        return _FP8ExpertBmmFn.apply(x, w)
+    from .interface import expert_bmm
+    return expert_bmm(x, w)
+
+
 def convert_linears_to_fp8(model: nn.Module, min_dim: int = 64) -> int:
     """Swap eligible nn.Linear modules for FP8Linear in place
     (cf. the reference QuantizationManager's bnb Linear8bitLt swap,

@@ -66,8 +66,13 @@ class Trainer:
         if self.precision.spec.name == "fp8" and self.device.type == "cuda":
             from ..ops.fp8 import convert_linears_to_fp8
             n_fp8 = convert_linears_to_fp8(self.model)
+            n_moe = 0
+            for l in self.model.get_moe_layers():
+                l.use_fp8 = True
+                n_moe += 1
             if logger:
-                logger.info(f"fp8: {n_fp8} Linear layers on e4m3fn MFMA")
+                logger.info(f"fp8: {n_fp8} Linear layers + {n_moe} MoE "
+                            "expert groups on e4m3fn MFMA")
 
         from ..parallel.mesh import get_mesh
         mesh = get_mesh()

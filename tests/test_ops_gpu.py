@@ -349,3 +349,20 @@ def test_moe_dispatch_combine_hip_vs_cpu(dtype):
                                rtol=1e-2, atol=1e-2)
     torch.testing.assert_close(w_g.grad.cpu(), w_cpu.grad,
                                rtol=1e-2, atol=1e-2)
+
+
+def test_fp8_expert_bmm_matches_bf16():
+    from luminaai_amd.ops.fp8 import expert_bmm_fp8
+    torch.manual_seed(0)
+    E, C, K, N = 4, 64, 128, 256
+    x = torch.randn(E, C, K, device=_dev(), dtype=torch.bfloat16,
+                    requires_grad=True)
+    w = torch.randn(E, K, N, device=_dev(), dtype=torch.bfloat16,
+                    requires_grad=True)
+    y = expert_bmm_fp8(x, w)
+    ref = torch.matmul(x.float(), w.float())
+    err = (y.float() - ref).abs().max() / ref.abs().max()
+    assert float(err) < 0.1, f"fp8 expert bmm relerr {float(err)}"
+    y.sum().backward()
+    assert torch.isfinite(x.grad.float()).all()
+    assert torch.isfinite(w.grad.float()).all()
