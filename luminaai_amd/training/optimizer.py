@@ -174,6 +174,9 @@ class FlatAdamW:
             for g in groups
         ]
         self._last_norm_sq: Optional[torch.Tensor] = None
+        self._norm_pinned: Optional[torch.Tensor] = None
+        self._norm_event = None
+        self._norm_cached = 0.0
 
     # ---- torch-compat surface ---------------------------------------------
     @property
@@ -207,6 +210,14 @@ class FlatAdamW:
         if norm_sq is None and self.max_grad_norm > 0:
             norm_sq = self.local_grad_norm_sq(shard_only=shard_only)
         self._last_norm_sq = norm_sq
+        if norm_sq is not None and norm_sq.is_cuda:
+            # async D2H so last_grad_norm() never syncs the step pipeline
+            if self._norm_pinned is None:
+                self._norm_pinned = torch.zeros(1, pin_memory=True)
+                self._norm_event = torch.cuda.Event()
+            self._norm_pinned.copy_(norm_sq.sum().reshape(1).float(),
+                                    non_blocking=True)
+            self._norm_event.record()
         for g in self.groups:
             K.adamw_step(
                 g.master, g.update_grad(), g.m, g.v, g.update_weight_out(),
@@ -214,9 +225,18 @@ class FlatAdamW:
                 self.step_count, norm_sq, self.max_grad_norm, grad_scale)
 
     def last_grad_norm(self) -> float:
-        """Host-visible grad norm of the LAST step (synchronises)."""
+        """Host-visible grad norm of the most recent COMPLETED step. On GPU
+        this is non-blocking (value lags one step if the copy is still in
+        flight); on CPU it reads directly."""
         if self._last_norm_sq is None:
             return 0.0
+        if self._last_norm_sq.is_cuda:
+            if self._norm_event is not None and self._norm_event.query():
+                val = float(self._norm_pinned[0])
+                self._norm_cached = (math.sqrt(val)
+                                     if math.isfinite(val) and val >= 0
+                                     else float("nan"))
+            return self._norm_cached
         val = float(self._last_norm_sq.sum())
         return math.sqrt(val) if math.isfinite(val) and val >= 0 else float("nan")
 
