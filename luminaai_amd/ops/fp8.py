@@ -31,6 +31,15 @@ def quantize_e4m3(t: torch.Tensor):
     return q, (1.0 / s)
 
 
+def _quantized_weight_2d(w: torch.Tensor):
+    ent = _WCACHE.get(id(w))
+    if ent is not None and ent[2] == _EPOCH:
+        return ent[0], ent[1]
+    q, s = quantize_e4m3(w)
+    _WCACHE[id(w)] = (q, s, _EPOCH)
+    return q, s
+
+
 class _FP8MatmulFn(torch.autograd.Function):
     """y = x @ w.T in fp8 forward; bf16 backward."""
 
@@ -40,7 +49,7 @@ class _FP8MatmulFn(torch.autograd.Function):
         shp = x.shape[:-1]
         x2 = x.reshape(-1, x.shape[-1])
         xq, xs = quantize_e4m3(x2)
-        wq, ws = quantize_e4m3(w)
+        wq, ws = _quantized_weight_2d(w)
         y = torch._scaled_mm(xq, wq.t(), scale_a=xs, scale_b=ws,
                              out_dtype=torch.bfloat16)
         return y.reshape(*shp, w.shape[0])
@@ -72,6 +81,32 @@ class FP8Linear(nn.Linear):
         return super().forward(x)
 
 
+# quantized-weight cache: expert weights only change at optimizer steps, so
+# the transpose+amax+cast of the (large) weight operand is done once per step
+# (Trainer.optimizer_step calls invalidate_weight_cache); activations are
+# quantized per call.
+_WCACHE = {}
+_EPOCH = 0
+
+
+def invalidate_weight_cache():
+    global _EPOCH
+    _EPOCH += 1
+
+
+def _quantized_weight(w: torch.Tensor):
+    ent = _WCACHE.get(id(w))
+    if ent is not None and ent[2] == _EPOCH:
+        return ent[0], ent[1]
+    qs, ss = [], []
+    for e in range(w.shape[0]):
+        q, s = quantize_e4m3(w[e].t().contiguous())     # [N, K] row-major
+        qs.append(q)
+        ss.append(s)
+    _WCACHE[id(w)] = (qs, ss, _EPOCH)
+    return qs, ss
+
+
 class _FP8ExpertBmmFn(torch.autograd.Function):
     """Grouped expert GEMM with fp8 e4m3 forward (per-expert 2D scaled_mm —
     hipBLASLt has no batched fp8 entry; E launches amortise fine) and bf16
@@ -83,10 +118,11 @@ class _FP8ExpertBmmFn(torch.autograd.Function):
         E, C, K = x.shape
         N = w.shape[2]
         out = x.new_empty(E, C, N)
+        wq, ws = _quantized_weight(w)
         for e in range(E):
             xq, xs = quantize_e4m3(x[e])
-            wq, ws = quantize_e4m3(w[e].t().contiguous())   # [N, K]
-            out[e] = torch._scaled_mm(xq, wq.t(), scale_a=xs, scale_b=ws,
+            out[e] = torch._scaled_mm(xq, wq[e].t(), scale_a=xs,
+                                      scale_b=ws[e],
                                       out_dtype=torch.bfloat16)
         return out
 

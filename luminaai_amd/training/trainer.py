@@ -165,6 +165,9 @@ class Trainer:
             if self._micro_in_cycle else 1.0
         self.engine.step(grad_scale=grad_scale)
         self.engine.zero_grad()
+        if self.precision.spec.name == "fp8":
+            from ..ops import fp8 as _fp8
+            _fp8.invalidate_weight_cache()
         self._micro_in_cycle = 0
         self.global_step += 1
         self._maybe_expire_lr_override()
