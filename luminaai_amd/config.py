@@ -130,6 +130,11 @@ class Config:
 
     # Monitoring and fault tolerance
     health_check_interval: int = 100
+    # orchestrator metric emission cadence (optimizer steps). Emission reads
+    # device scalars (loss) -> one sync per emission; at sub-10ms steps that
+    # plus the monitor thread's GIL share costs ~35%, at ~1s steps <0.5%.
+    # Raise for small/fast models.
+    metrics_emit_every: int = 1
     auto_resume: bool = True
     backup_every_n_hours: int = 6
     max_retries: int = 3

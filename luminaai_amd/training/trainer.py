@@ -191,7 +191,9 @@ class Trainer:
             if boundary:
                 self.optimizer_step()
                 self.engine.set_sync(False)
-                self._emit_metrics(out)
+                every = max(1, getattr(self.config, "metrics_emit_every", 1))
+                if self.global_step % every == 0:
+                    self._emit_metrics(out)
             losses.append(out["ce_loss"].detach())
             if self.config.eval_every_n_batches and \
                     (i + 1) % self.config.eval_every_n_batches == 0 and \
