@@ -53,8 +53,11 @@ class GenerationEngine:
     @torch.no_grad()
     def generate(self, prompt_ids: List[int],
                  config: Optional[GenerationConfig] = None,
-                 stream_callback: Optional[Callable[[int], None]] = None) -> List[int]:
-        """Returns the generated token ids (not including the prompt)."""
+                 stream_callback: Optional[Callable[[int], None]] = None,
+                 use_graph: Optional[bool] = None) -> List[int]:
+        """Returns the generated token ids (not including the prompt).
+        On GPU the single-token steps replay a captured hipGraph by default
+        (graph_decode.py); pass use_graph=False for the eager loop."""
         cfg = config or GenerationConfig()
         t0 = time.perf_counter()
         self.model.eval()
@@ -62,28 +65,47 @@ class GenerationEngine:
 
         ids = prompt_ids[-cfg.max_context:]
         x = torch.tensor([ids], dtype=torch.long, device=self.device)
-        caches = self.model.make_kv_caches(max_len=cfg.max_context)
+        if use_graph is None:
+            use_graph = self.device.type == "cuda"
 
-        # ---- prefill: one forward over the whole prompt
-        logits, _, _ = self.model(x, kv_caches=caches)
+        dec = None
+        if use_graph:
+            from .graph_decode import GraphedDecoder
+            if getattr(self, "_graph_dec", None) is None or \
+                    self._graph_dec.max_context != cfg.max_context:
+                self._graph_dec = GraphedDecoder(self.model, cfg.max_context)
+            dec = self._graph_dec
+            dec.reset()
+            last_logits = dec.prefill(x)[0].float()
+            seq_len = lambda: dec.seq_len  # noqa: E731
+        else:
+            caches = self.model.make_kv_caches(max_len=cfg.max_context)
+            logits, _, _ = self.model(x, kv_caches=caches)
+            last_logits = logits[0, -1].float()
+            seq_len = lambda: caches[0].seq_len  # noqa: E731
+
         self.stats["prefill_tokens"] += len(ids)
         generated: List[int] = []
         recent: List[int] = list(ids)
 
         for _ in range(cfg.max_new_tokens):
-            next_id = self._sample(logits[0, -1].float(), cfg,
-                                   recent[-cfg.rep_window:])
+            next_id = self._sample(last_logits, cfg, recent[-cfg.rep_window:])
             if next_id in stop:
                 break
             generated.append(next_id)
             recent.append(next_id)
             if stream_callback is not None:
                 stream_callback(next_id)
-            if caches[0].seq_len >= cfg.max_context:
+            if seq_len() >= cfg.max_context:
                 break
             # ---- single-token incremental step
-            step = torch.tensor([[next_id]], dtype=torch.long, device=self.device)
-            logits, _, _ = self.model(step, kv_caches=caches)
+            step = torch.tensor([[next_id]], dtype=torch.long,
+                                device=self.device)
+            if dec is not None:
+                last_logits = dec.step(step)[0]
+            else:
+                logits, _, _ = self.model(step, kv_caches=caches)
+                last_logits = logits[0, -1].float()
 
         self.stats["tokens_generated"] += len(generated)
         self.stats["time_in_generate"] += time.perf_counter() - t0

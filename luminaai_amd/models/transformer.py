@@ -189,23 +189,48 @@ class KVCache:
     decode hot loop stays allocation-free and hipGraph-friendly). Without
     max_len it falls back to growing concat."""
 
-    def __init__(self, max_len: int = 0):
+    def __init__(self, max_len: int = 0, graph_mode: bool = False):
         self.max_len = max_len
+        self.graph_mode = graph_mode and max_len > 0
         self.k: Optional[torch.Tensor] = None
         self.v: Optional[torch.Tensor] = None
         self._len = 0
+        self.pos_dev: Optional[torch.Tensor] = None  # device write cursor
 
     @property
     def seq_len(self) -> int:
         return self._len
 
+    def _alloc(self, k):
+        B, S, H, D = k.shape
+        cap = max(self.max_len, S)
+        self.k = k.new_zeros(B, cap, H, D)
+        self.v = k.new_zeros(B, cap, H, D)
+        if self.graph_mode:
+            self.pos_dev = torch.zeros(1, dtype=torch.int32, device=k.device)
+
     def append(self, k: torch.Tensor, v: torch.Tensor):
         B, S, H, D = k.shape
+        if self.graph_mode:
+            # hipGraph-replayable: the write position lives ON DEVICE and
+            # every returned shape is static (full buffer; the attention
+            # masks past the cursor). Single-token steps only post-capture.
+            if self.k is None:
+                self._alloc(k)
+            if S == 1:
+                idx = self.pos_dev.long()                      # [1]
+                self.k.index_copy_(1, idx, k)
+                self.v.index_copy_(1, idx, v)
+                self.pos_dev.add_(1)
+            else:  # prefill (eager, before capture)
+                self.k[:, self._len:self._len + S] = k
+                self.v[:, self._len:self._len + S] = v
+                self.pos_dev.add_(S)
+            self._len += S
+            return self.k, self.v                              # FULL buffers
         if self.max_len > 0:
             if self.k is None:
-                cap = max(self.max_len, S)
-                self.k = k.new_empty(B, cap, H, D)
-                self.v = v.new_empty(B, cap, H, D)
+                self._alloc(k)
             end = min(self._len + S, self.k.shape[1])
             take = end - self._len
             self.k[:, self._len:end] = k[:, :take]
@@ -265,6 +290,14 @@ class GroupedQueryAttention(nn.Module):
 
         if kv_cache is not None:
             k, v = kv_cache.append(k, v)
+            if kv_cache.graph_mode:
+                # full static buffers: mask every slot at/after the device
+                # write cursor (hipGraph-replayable — no host shapes change)
+                L = k.shape[1]
+                ar = torch.arange(L, device=x.device).view(1, 1, 1, L)
+                live = ar < kv_cache.pos_dev.view(1, 1, 1, 1)
+                attn_mask = torch.zeros(1, 1, S, L, device=x.device,
+                                        dtype=x.dtype).masked_fill(~live, -1e4)
 
         run_sp = self.sp_size > 1 and kv_cache is None and S > 1 \
             and attn_mask is None
@@ -689,8 +722,16 @@ class DeepSeekTransformer(nn.Module):
         """Returns (logits, total_aux_loss, aux_losses dict)."""
         B, S = input_ids.shape
         x = self.embed_tokens(input_ids) * self.embed_scale
+        pos = None
         if kv_caches is not None and kv_caches[0].seq_len > 0:
             pos_offset = kv_caches[0].seq_len
+        if kv_caches is not None and S == 1 \
+                and getattr(kv_caches[0], "graph_mode", False) \
+                and kv_caches[0].pos_dev is not None:
+            # device-side position (layer appends bump their own cursors, so
+            # snapshot layer 0's BEFORE any append)
+            pos = kv_caches[0].pos_dev.clone().view(1, 1) \
+                .expand(B, 1).contiguous()
         # Ulysses SP: this rank holds sequence slice [sp_rank*S, (sp_rank+1)*S)
         from ..parallel.mesh import get_mesh
         mesh = get_mesh()
@@ -715,10 +756,10 @@ class DeepSeekTransformer(nn.Module):
             cache = kv_caches[i] if kv_caches is not None else None
             if self.gradient_checkpointing and self.training and cache is None:
                 x, aux = torch.utils.checkpoint.checkpoint(
-                    layer, x, rope_cs, None, pos_offset, None, attn_mask,
+                    layer, x, rope_cs, pos, pos_offset, None, attn_mask,
                     use_reentrant=False)
             else:
-                x, aux = layer(x, rope_cs, None, pos_offset, cache, attn_mask)
+                x, aux = layer(x, rope_cs, pos, pos_offset, cache, attn_mask)
             if layer.is_moe:
                 total_aux = total_aux + aux
                 aux_losses[f"layer_{i}_moe"] = aux.detach()
@@ -784,6 +825,9 @@ class DeepSeekTransformer(nn.Module):
     def get_moe_layers(self) -> List[MoEFFNLayer]:
         return [l.ffn for l in self.layers if l.is_moe]
 
-    def make_kv_caches(self, max_len: int = 0) -> List[KVCache]:
-        """max_len > 0 preallocates static buffers (serving hot path)."""
-        return [KVCache(max_len) for _ in self.layers]
+    def make_kv_caches(self, max_len: int = 0,
+                       graph_mode: bool = False) -> List[KVCache]:
+        """max_len > 0 preallocates static buffers (serving hot path);
+        graph_mode additionally keeps the write cursor on device so a
+        single-token decode step is hipGraph-capturable."""
+        return [KVCache(max_len, graph_mode) for _ in self.layers]

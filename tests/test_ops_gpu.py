@@ -366,3 +366,64 @@ def test_fp8_expert_bmm_matches_bf16():
     y.sum().backward()
     assert torch.isfinite(x.grad.float()).all()
     assert torch.isfinite(w.grad.float()).all()
+
+
+# ---------------------------------------------------------------- hipGraph decode
+def test_graph_decode_matches_eager():
+    """The hipGraph-captured decode step must reproduce the eager KV-cached
+    logits (full-buffer cursor-masked attention vs sliced causal)."""
+    from luminaai_amd.models.transformer import DeepSeekConfig, DeepSeekTransformer
+    from luminaai_amd.inference.graph_decode import GraphedDecoder
+    torch.manual_seed(0)
+    cfg = DeepSeekConfig(vocab_size=512, hidden_size=128, num_layers=2,
+                         num_heads=4, num_kv_heads=2, intermediate_size=256,
+                         seq_length=64, use_moe=True, num_experts=4,
+                         moe_top_k=2, use_mod=False, routing_noise_std=0.0)
+    m = DeepSeekTransformer(cfg).to(_dev(), torch.bfloat16).eval()
+    ids = torch.randint(1, 512, (1, 8), device=_dev())
+    toks = torch.randint(1, 512, (6,), device=_dev())
+
+    # eager reference
+    caches = m.make_kv_caches(max_len=32)
+    with torch.no_grad():
+        logits, _, _ = m(ids, kv_caches=caches)
+        eager = [logits[0, -1].float()]
+        for t in range(6):
+            logits, _, _ = m(toks[t].view(1, 1), kv_caches=caches)
+            eager.append(logits[0, -1].float())
+
+    dec = GraphedDecoder(m, max_context=32)
+    got = [dec.prefill(ids)[0].float()]
+    for t in range(6):
+        got.append(dec.step(toks[t].view(1, 1))[0].clone())
+    for i, (a, b) in enumerate(zip(got, eager)):
+        torch.testing.assert_close(a, b, rtol=5e-2, atol=5e-2)
+
+    # reset + reuse across prompts (captured graph survives)
+    dec.reset()
+    again = dec.prefill(ids)[0].float()
+    torch.testing.assert_close(again, eager[0], rtol=5e-2, atol=5e-2)
+    s0 = dec.step(toks[0].view(1, 1))[0].clone()
+    torch.testing.assert_close(s0, eager[1], rtol=5e-2, atol=5e-2)
+
+
+def test_engine_generate_graph_vs_eager():
+    from luminaai_amd.models.transformer import DeepSeekConfig, DeepSeekTransformer
+    from luminaai_amd.inference import GenerationConfig, GenerationEngine
+    from luminaai_amd.data.tokenizer import ConversationTokenizer
+    torch.manual_seed(1)
+    cfg = DeepSeekConfig(vocab_size=512, hidden_size=128, num_layers=2,
+                         num_heads=4, num_kv_heads=2, intermediate_size=256,
+                         seq_length=64, use_moe=False, use_mod=False)
+    m = DeepSeekTransformer(cfg).to(_dev(), torch.bfloat16).eval()
+    eng = GenerationEngine(m, ConversationTokenizer(), _dev())
+    gcfg = GenerationConfig(max_new_tokens=10, temperature=0.0,
+                            max_context=64, stop_token_ids=[-1],
+                            repetition_penalty=1.0)
+    prompt = list(range(5, 15))
+    a = eng.generate(prompt, gcfg, use_graph=False)
+    b = eng.generate(prompt, gcfg, use_graph=True)
+    assert len(b) == 10
+    # greedy over bf16 near-ties may rarely flip; demand strong agreement
+    agree = sum(x == y for x, y in zip(a, b))
+    assert agree >= 8, (a, b)
