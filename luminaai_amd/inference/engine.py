@@ -70,15 +70,22 @@ class GenerationEngine:
 
         dec = None
         if use_graph:
-            from .graph_decode import GraphedDecoder
-            if getattr(self, "_graph_dec", None) is None or \
-                    self._graph_dec.max_context != cfg.max_context:
-                self._graph_dec = GraphedDecoder(self.model, cfg.max_context)
-            dec = self._graph_dec
-            dec.reset()
-            last_logits = dec.prefill(x)[0].float()
-            seq_len = lambda: dec.seq_len  # noqa: E731
-        else:
+            try:
+                from .graph_decode import GraphedDecoder
+                if getattr(self, "_graph_dec", None) is None or \
+                        self._graph_dec.max_context != cfg.max_context:
+                    self._graph_dec = GraphedDecoder(self.model,
+                                                     cfg.max_context)
+                dec = self._graph_dec
+                dec.reset()
+                last_logits = dec.prefill(x)[0].float()
+                dec.ensure_captured()      # capture now; fall back on failure
+                seq_len = lambda: dec.seq_len  # noqa: E731
+            except Exception:  # noqa: BLE001 — e.g. un-capturable MoE ops
+                self._graph_dec = None
+                dec = None
+                use_graph = False
+        if not use_graph:
             caches = self.model.make_kv_caches(max_len=cfg.max_context)
             logits, _, _ = self.model(x, kv_caches=caches)
             last_logits = logits[0, -1].float()

@@ -81,6 +81,10 @@ class GraphedDecoder:
             c.pos_dev.sub_(n)
             c._len -= n
 
+    def ensure_captured(self):
+        if self._graph is None:
+            self._capture()
+
     @torch.no_grad()
     def step(self, token_id: torch.Tensor) -> torch.Tensor:
         """token_id: [B] or [B,1] long on device -> logits [B, V] fp32."""
