@@ -104,6 +104,16 @@ def run_all():
                iters=20),
         flops=flops)
 
+    # decode GEMV: b1 qkv shape vs hipBLASLt batch-1 linear
+    Ng, Kg = 2544, 1908
+    wg = torch.randn(Ng, Kg, device=dev, dtype=dt)
+    xg = torch.randn(Kg, device=dev, dtype=dt)
+    from luminaai_amd.ops import get_ext
+    add(f"gemv [{Ng},{Kg}]",
+        timeit(lambda: get_ext().gemv(xg, wg), iters=200),
+        timeit(lambda: torch.nn.functional.linear(xg, wg), iters=200),
+        bytes_=Ng * Kg * 2)
+
     # AdamW fused step (flat 1e8 params)
     n = 100_000_000
     master = torch.randn(n, device=dev, dtype=torch.float32)

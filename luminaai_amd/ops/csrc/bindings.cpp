@@ -50,6 +50,7 @@ void launch_grouped_gemm_nt(const void*, const void*, void*, int, int, int, int,
 void launch_grouped_gemm_nt_v2(const void*, const void*, void*, int, int, int, int, int64_t, int64_t, int64_t, int, hipStream_t);
 void launch_grouped_gemm_nt_v3(const void*, const void*, void*, int, int, int, int, int64_t, int64_t, int64_t, hipStream_t);
 void launch_grouped_gemm_nt_v4(const void*, const void*, void*, int, int, int, int, int64_t, int64_t, int64_t, hipStream_t);
+hipError_t lumina_gemv(const void*, const void*, void*, int, int64_t, int, hipStream_t);
 hipError_t lumina_moe_gather_rows(const void*, const int64_t*, const bool*, void*, int64_t, int, int, hipStream_t);
 hipError_t lumina_moe_dispatch_bwd(const void*, const int64_t*, void*, int64_t, int, int, int64_t, int, hipStream_t);
 hipError_t lumina_moe_combine_fwd(const void*, const float*, const int64_t*, void*, int64_t, int, int, int64_t, int, hipStream_t);
@@ -404,7 +405,21 @@ at::Tensor grouped_gemm_nt_v4(const at::Tensor& A, const at::Tensor& B) {
   return O;
 }
 
+at::Tensor gemv(const at::Tensor& x, const at::Tensor& w) {
+  // y[N] = w[N,K] @ x[K]
+  TORCH_CHECK(x.is_contiguous() && w.is_contiguous());
+  TORCH_CHECK(x.dim() == 1 && w.dim() == 2 && w.size(1) == x.numel());
+  TORCH_CHECK(x.scalar_type() == w.scalar_type());
+  auto y = at::empty({w.size(0)}, x.options());
+  check_hip(lumina_gemv(w.data_ptr(), x.data_ptr(), y.data_ptr(),
+                        (int)w.size(0), x.numel(), is_bf16(x) ? 1 : 0,
+                        cur_stream()),
+            "gemv");
+  return y;
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
+  mod.def("gemv", &gemv, "batch-1 decode GEMV y = W @ x (gfx950)");
   mod.def("grouped_gemm_nt_v4", &grouped_gemm_nt_v4,
           "2-buffer raw-barrier counted-vmcnt variant (K%64==0)");
   mod.def("grouped_gemm_nt_v3", &grouped_gemm_nt_v3,

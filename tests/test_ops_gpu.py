@@ -448,3 +448,16 @@ def test_engine_moe_graph_fallback():
                             max_context=64, stop_token_ids=[-1])
     out = eng.generate(list(range(5, 12)), gcfg, use_graph=True)
     assert len(out) == 6
+
+
+@pytest.mark.parametrize("shape", [(2544, 1908), (1908, 2544), (512, 100),
+                                   (50304, 1908)])
+def test_gemv_matches_linear(shape):
+    from luminaai_amd.ops import get_ext
+    torch.manual_seed(0)
+    N, K = shape
+    w = torch.randn(N, K, device=_dev(), dtype=torch.bfloat16)
+    x = torch.randn(K, device=_dev(), dtype=torch.bfloat16)
+    y = get_ext().gemv(x, w)
+    ref = torch.nn.functional.linear(x.float(), w.float())
+    torch.testing.assert_close(y.float(), ref, rtol=2e-2, atol=2e-2)
