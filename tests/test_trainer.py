@@ -266,3 +266,45 @@ def test_profile_training_loop_overhead(tiny_config, tokenizer, small_model):
         {"input_ids": ids[:, :-1], "labels": ids[:, 1:]}, iters=2)
     assert set(res) == {"h2d_ms", "forward_ms", "backward_ms", "optimizer_ms"}
     assert all(v >= 0 for v in res.values())
+
+
+def test_scheduler_kinds():
+    import torch
+    from luminaai_amd.training.optimizer import FlatAdamW
+    from luminaai_amd.training.schedulers import WarmupScheduler
+    m = torch.nn.Linear(8, 8)
+    for kind in ("cosine", "linear", "constant"):
+        opt = FlatAdamW(m, lr=1e-3)
+        s = WarmupScheduler(opt, total_steps=100, warmup_steps=10, kind=kind,
+                            min_lr=1e-6)
+        lrs = []
+        for _ in range(100):
+            s.step()
+            lrs.append(opt.groups[0].lr)
+        assert lrs[8] < lrs[9] <= 1e-3            # warmup ramps
+        if kind == "constant":
+            assert lrs[-1] == pytest.approx(1e-3)
+        else:
+            assert lrs[-1] < lrs[20]              # decays
+            assert lrs[-1] >= 1e-6
+        # adaptive override rebasing
+        s.set_base_lr(5e-4)
+        s.step()
+        assert opt.groups[0].lr <= 5e-4 + 1e-9
+
+
+def test_scheduler_state_roundtrip():
+    import torch
+    from luminaai_amd.training.optimizer import FlatAdamW
+    from luminaai_amd.training.schedulers import WarmupScheduler
+    m = torch.nn.Linear(8, 8)
+    opt = FlatAdamW(m, lr=1e-3)
+    s = WarmupScheduler(opt, 100, 10)
+    for _ in range(37):
+        s.step()
+    sd = s.state_dict()
+    opt2 = FlatAdamW(torch.nn.Linear(8, 8), lr=1e-3)
+    s2 = WarmupScheduler(opt2, 100, 10)
+    s2.load_state_dict(sd)
+    assert s2.step_num == 37
+    assert opt2.groups[0].lr == pytest.approx(opt.groups[0].lr)
