@@ -13,6 +13,11 @@
 
 #include "common.h"
 
+// 8-byte vector helpers for the column loops (h % 4 == 0 fast path; rows of
+// 2-byte elems at even h are 8-byte aligned). Scalar tail for other h.
+typedef __attribute__((ext_vector_type(4))) uint16_t ushortx4;
+
+
 // grid-stride over rows helper
 #define ROW_LOOP(row, nrows) \
   for (int64_t row = blockIdx.x; row < (nrows); row += gridDim.x)
@@ -29,6 +34,17 @@ __global__ void moe_gather_rows_kernel(const typename E::storage* __restrict__ x
     const bool ok = fill[s];
     typename E::storage* out = buf + s * h;
     const typename E::storage* in = x + src * h;
+    if constexpr (sizeof(typename E::storage) == 2) {
+      if ((h & 3) == 0) {
+        const int hv = h >> 2;
+        const ushortx4* iv = reinterpret_cast<const ushortx4*>(in);
+        ushortx4* ov = reinterpret_cast<ushortx4*>(out);
+        const ushortx4 z = {0, 0, 0, 0};
+        for (int c = threadIdx.x; c < hv; c += blockDim.x)
+          ov[c] = ok ? iv[c] : z;
+        continue;
+      }
+    }
     for (int c = threadIdx.x; c < h; c += blockDim.x)
       E::store(out + c, ok ? E::load(in + c) : 0.0f);
   }
@@ -43,6 +59,28 @@ __global__ void moe_dispatch_bwd_kernel(const typename E::storage* __restrict__ 
                                         int64_t n_slots) {
   ROW_LOOP(t, n_tok) {
     typename E::storage* out = gx + t * h;
+    if constexpr (sizeof(typename E::storage) == 2) {
+      if ((h & 3) == 0) {
+        const int hv = h >> 2;
+        ushortx4* ov = reinterpret_cast<ushortx4*>(out);
+        for (int c = threadIdx.x; c < hv; c += blockDim.x) {
+          float a0 = 0, a1 = 0, a2 = 0, a3 = 0;
+          for (int j = 0; j < k; ++j) {
+            const int64_t s = slot_tm[t * k + j];
+            if (s < n_slots) {
+              ushortx4 v = reinterpret_cast<const ushortx4*>(
+                  gbuf + s * h)[c];
+              a0 += bf16_to_f32(v[0]); a1 += bf16_to_f32(v[1]);
+              a2 += bf16_to_f32(v[2]); a3 += bf16_to_f32(v[3]);
+            }
+          }
+          ushortx4 o4 = {f32_to_bf16(a0), f32_to_bf16(a1),
+                         f32_to_bf16(a2), f32_to_bf16(a3)};
+          ov[c] = o4;
+        }
+        continue;
+      }
+    }
     for (int c = threadIdx.x; c < h; c += blockDim.x) {
       float acc = 0.0f;
       for (int j = 0; j < k; ++j) {
@@ -64,6 +102,30 @@ __global__ void moe_combine_fwd_kernel(const typename E::storage* __restrict__ y
                                        int64_t n_slots) {
   ROW_LOOP(t, n_tok) {
     typename E::storage* o = out + t * h;
+    if constexpr (sizeof(typename E::storage) == 2) {
+      if ((h & 3) == 0) {
+        const int hv = h >> 2;
+        ushortx4* ov = reinterpret_cast<ushortx4*>(o);
+        for (int c = threadIdx.x; c < hv; c += blockDim.x) {
+          float a0 = 0, a1 = 0, a2 = 0, a3 = 0;
+          for (int j = 0; j < k; ++j) {
+            const int64_t s = slot_tm[t * k + j];
+            if (s < n_slots) {
+              const float w = w_tm[t * k + j];
+              ushortx4 v = reinterpret_cast<const ushortx4*>(y + s * h)[c];
+              a0 = fmaf(w, bf16_to_f32(v[0]), a0);
+              a1 = fmaf(w, bf16_to_f32(v[1]), a1);
+              a2 = fmaf(w, bf16_to_f32(v[2]), a2);
+              a3 = fmaf(w, bf16_to_f32(v[3]), a3);
+            }
+          }
+          ushortx4 o4 = {f32_to_bf16(a0), f32_to_bf16(a1),
+                         f32_to_bf16(a2), f32_to_bf16(a3)};
+          ov[c] = o4;
+        }
+        continue;
+      }
+    }
     for (int c = threadIdx.x; c < h; c += blockDim.x) {
       float acc = 0.0f;
       for (int j = 0; j < k; ++j) {
@@ -91,6 +153,22 @@ __global__ void moe_combine_bwd_y_kernel(const typename E::storage* __restrict__
     const float w = ok ? w_tm[f] : 0.0f;
     const typename E::storage* g = gout + src_tok[s] * h;
     typename E::storage* o = gy + s * h;
+    if constexpr (sizeof(typename E::storage) == 2) {
+      if ((h & 3) == 0) {
+        const int hv = h >> 2;
+        const ushortx4* gv = reinterpret_cast<const ushortx4*>(g);
+        ushortx4* ov = reinterpret_cast<ushortx4*>(o);
+        for (int c = threadIdx.x; c < hv; c += blockDim.x) {
+          ushortx4 in4 = gv[c];
+          ushortx4 o4;
+          #pragma unroll
+          for (int i = 0; i < 4; ++i)
+            o4[i] = f32_to_bf16(w * bf16_to_f32(in4[i]));
+          ov[c] = o4;
+        }
+        continue;
+      }
+    }
     for (int c = threadIdx.x; c < h; c += blockDim.x)
       E::store(o + c, w * E::load(g + c));
   }
