@@ -164,15 +164,30 @@ void grouped_gemm_nt_kernel(const uint16_t* __restrict__ Aall,
       s_koff[i] = (c & 7) * 8;               // in elements
     }
     ushortx8 ra[CHUNKS], rb[CHUNKS];
+    // block-uniform fast path: interior tiles with a complete K-step need no
+    // per-element guards (the guarded path measured ~180 TF vs ~650)
+    const bool intA = tileM + BM <= M;
+    const bool intB = tileN + BN <= N;
 
-    #pragma unroll
-    for (int i = 0; i < CHUNKS; ++i) {
-      int gm = tileM + s_row[i];
-      int gn = tileN + s_row[i];
-      int gk = s_koff[i];
-      ra[i] = load8_guard(A + (int64_t)gm * K + gk, gm < M, gk, K);
-      rb[i] = load8_guard(B + (int64_t)gn * K + gk, gn < N, gk, K);
-    }
+    auto stage_regs = [&](int k0) {
+      const bool kfull = k0 + BK <= K;
+      #pragma unroll
+      for (int i = 0; i < CHUNKS; ++i) {
+        const int gm = tileM + s_row[i];
+        const int gn = tileN + s_row[i];
+        const int gk = k0 + s_koff[i];
+        if (intA && kfull)
+          ra[i] = *reinterpret_cast<const ushortx8*>(A + (int64_t)gm * K + gk);
+        else
+          ra[i] = load8_guard(A + (int64_t)gm * K + gk, gm < M, gk, K);
+        if (intB && kfull)
+          rb[i] = *reinterpret_cast<const ushortx8*>(B + (int64_t)gn * K + gk);
+        else
+          rb[i] = load8_guard(B + (int64_t)gn * K + gk, gn < N, gk, K);
+      }
+    };
+
+    stage_regs(0);
     #pragma unroll
     for (int i = 0; i < CHUNKS; ++i) {
       *reinterpret_cast<ushortx8*>(
@@ -183,17 +198,8 @@ void grouped_gemm_nt_kernel(const uint16_t* __restrict__ Aall,
     __syncthreads();
 
     for (int kt = 0; kt < KT; ++kt) {
-      if (kt + 1 < KT) {
-        const int k0 = (kt + 1) * BK;
-        #pragma unroll
-        for (int i = 0; i < CHUNKS; ++i) {
-          int gm = tileM + s_row[i];
-          int gn = tileN + s_row[i];
-          int gk = k0 + s_koff[i];
-          ra[i] = load8_guard(A + (int64_t)gm * K + gk, gm < M, gk, K);
-          rb[i] = load8_guard(B + (int64_t)gn * K + gk, gn < N, gk, K);
-        }
-      }
+      if (kt + 1 < KT)
+        stage_regs((kt + 1) * BK);
       #pragma unroll
       for (int kk = 0; kk < BK; kk += 32) {
         bf16x8 af[4], bf[4];
