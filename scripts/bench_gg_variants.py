@@ -41,3 +41,12 @@ for _ in range(5):
     d = (ext.grouped_gemm_nt_v4(a, b) - o1).abs().max()
     assert float(d) == 0.0, f"nondeterminism {float(d)}"
 print("v4 race-screen ok")
+# unaligned-K (fallback path) shape: gx2 of the down projection
+Eu, Cu, Nu, Ku = 8, 2560, 5120, 1908
+au = torch.randn(Eu, Cu, Ku, device="cuda", dtype=torch.bfloat16)
+bu = torch.randn(Eu, Nu, Ku, device="cuda", dtype=torch.bfloat16)
+ru = torch.matmul(au.float(), bu.float().transpose(1, 2))
+ou = ext.grouped_gemm_nt(au, bu)
+print("unaligned relerr", float((ou.float()-ru).abs().max()/ru.abs().max()))
+E, M, N, K = Eu, Cu, Nu, Ku
+bench(lambda: ext.grouped_gemm_nt(au, bu), "nt-unalignedK", iters=20)
