@@ -65,7 +65,10 @@ def main():
     ap.add_argument("--steps", type=int, default=10)
     ap.add_argument("--warmup", type=int, default=3)
     ap.add_argument("--micro-batch", type=int, default=16)
-    ap.add_argument("--accum", type=int, default=1)
+    ap.add_argument("--accum", type=int, default=2,
+                    help="grad-accumulation micro-steps (default 2: global "
+                         "batch 16x2=32 matches the reference b1 preset's "
+                         "8x4; also amortises the fused optimizer step)")
     ap.add_argument("--seq-len", type=int, default=None)
     ap.add_argument("--preset", type=str, default=None)
     ap.add_argument("--zero", type=int, default=None)
