@@ -150,6 +150,11 @@ def main(argv: Optional[list] = None) -> dict:
                          "pass --train-data or --synthetic-steps")
             raise SystemExit(2)
 
+    if cfg.compile and rank == 0:
+        logger.warning("config.compile requested but torch.compile is "
+                       "disabled by design on this framework (its ROCm "
+                       "backend is Triton); the native path is HIP kernels "
+                       "+ hipGraphs")
     model = DeepSeekTransformer(config_to_deepseek_config(cfg))
     if rank == 0:
         fp = model.get_memory_footprint()
