@@ -131,6 +131,15 @@ class ZeroEngine:
             return
         if self.stage in (0, 1):
             if self._bucket_plan is not None and self.sync_enabled:
+                # buckets whose params never produced a grad this step (e.g.
+                # a MoD router disabled at capacity 1.0) never fired their
+                # hook — reduce them now or the ranks silently diverge
+                for bi, rem in self._pending.items():
+                    if rem:
+                        g, lo, hi, _ = self._bucket_plan[bi]
+                        self._works.append(
+                            dist.all_reduce(g.flat_g[lo:hi], async_op=True,
+                                            group=self.pg))
                 for w in self._works:
                     w.wait()
                 self._works.clear()

@@ -442,3 +442,38 @@ def dp_sampler_worker(rank, world):
 def test_dp_data_sharding():
     res = _spawn("dp_sampler_worker")
     assert res[0]["sum"] != res[1]["sum"], "ranks received identical data"
+
+
+# ---- gradless-param bucket flush -------------------------------------------
+def gradless_param_worker(rank, world):
+    """A parameter that receives no grad (MoD router at capacity 1.0 never
+    runs) must not leave its bucket unreduced under overlapped ZeRO-1."""
+    from luminaai_amd.config import Config
+    from luminaai_amd.data.tokenizer import ConversationTokenizer
+    from luminaai_amd.models import DeepSeekTransformer, config_to_deepseek_config
+    from luminaai_amd.training import Trainer
+    cfg = Config(vocab_size=512, hidden_size=64, num_layers=2, num_heads=4,
+                 num_kv_heads=2, seq_length=32, intermediate_size=128,
+                 micro_batch_size=2, gradient_accumulation_steps=1,
+                 num_workers=0, use_moe=False, use_mod=True,
+                 mod_capacity_factor=1.0,     # run_mod False -> router unused
+                 zero_stage=1, precision="fp32",
+                 experiment_name=f"gl_{rank}",
+                 eval_every_n_batches=0, save_every_n_batches=0)
+    torch.manual_seed(1234)
+    model = DeepSeekTransformer(config_to_deepseek_config(cfg))
+    t = Trainer(model, ConversationTokenizer(), cfg)
+    t._setup_scheduler(10)
+    torch.manual_seed(800 + rank)            # different data per rank
+    for _ in range(3):
+        ids = torch.randint(1, cfg.vocab_size, (2, cfg.seq_length + 1))
+        t.engine.set_sync(True)
+        t.train_step({"input_ids": ids[:, :-1], "labels": ids[:, 1:]})
+        t.optimizer_step()
+    w = torch.cat([p.detach().reshape(-1) for p in t.model.parameters()])
+    return {"checksum": float(w.sum())}
+
+
+def test_gradless_param_bucket_flushed():
+    res = _spawn("gradless_param_worker")
+    assert res[0]["checksum"] == pytest.approx(res[1]["checksum"], abs=1e-4)
