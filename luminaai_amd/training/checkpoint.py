@@ -140,6 +140,16 @@ class CheckpointManager:
         self.history = [h for h in self.history if h not in removed]
         self._write_history()
 
+    def reload_history(self):
+        """Re-read the on-disk history (ranks that never wrote the main
+        checkpoint share the directory but have an empty in-memory list)."""
+        if self.history_path.exists():
+            try:
+                self.history = json.loads(self.history_path.read_text())
+            except json.JSONDecodeError:
+                pass
+        return self.history
+
     def _write_history(self):
         self.history_path.write_text(json.dumps(self.history, indent=2))
 

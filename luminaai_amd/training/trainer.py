@@ -323,6 +323,17 @@ class Trainer:
         return path
 
     def load_checkpoint(self, which: str = "latest", load_optimizer: bool = True):
+        if comm.is_distributed():
+            self.checkpoints.reload_history()   # non-writer ranks share the dir
+        if self.mesh is not None and self.mesh.ep_size > 1:
+            # every EP rank resumes from ITS expert-shard file
+            import re as _re
+            base = self.checkpoints.resolve(which)
+            if base is not None:
+                fixed = _re.sub(r"_ep_rank_\d+",
+                                f"_ep_rank_{self.mesh.ep_rank}", str(base))
+                if os.path.exists(fixed):
+                    which = fixed
         payload = self.checkpoints.load_checkpoint(which, map_location=self.device)
         if self.engine.stage >= 3:
             with self.engine.gathered_weights():
@@ -581,7 +592,10 @@ class Trainer:
 
     def rollback_steps(self, n_steps: int = 100) -> bool:
         """Reload the most recent checkpoint at least n_steps back
-        (reference trainer.py:1727-1791)."""
+        (reference trainer.py:1727-1791). Under DP all ranks reach the same
+        decision from the same on-disk history (shared experiment dir)."""
+        if comm.is_distributed():
+            self.checkpoints.reload_history()
         target = self.global_step - n_steps
         best = None
         for h in reversed(self.checkpoints.history):
