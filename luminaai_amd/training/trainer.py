@@ -264,6 +264,12 @@ class Trainer:
             tot_acc += float(out["accuracy"])
             n += 1
         self.model.train()
+        # eval shards are disjoint per DP rank: combine before deciding
+        # best/early-stop, or ranks diverge on should_stop and hang
+        if comm.is_distributed():
+            tot_loss = comm.all_reduce_scalar(tot_loss)
+            tot_acc = comm.all_reduce_scalar(tot_acc)
+            n = int(comm.all_reduce_scalar(float(n)))
         if n == 0:
             return {}
         loss = tot_loss / n
