@@ -281,6 +281,7 @@ class GroupedQueryAttention(nn.Module):
         mesh = get_mesh()
         self.sp_size = mesh.sp_size if mesh is not None else 1
         self.sp_group = mesh.sp_group if mesh is not None else None
+        self.tp_group = None    # set by convert_to_tensor_parallel
         if self.sp_size > 1:
             assert self.num_heads % self.sp_size == 0 and \
                 self.num_kv_heads % self.sp_size == 0, \
@@ -290,6 +291,9 @@ class GroupedQueryAttention(nn.Module):
                 pos_offset: int = 0, kv_cache: Optional[KVCache] = None,
                 attn_mask: Optional[torch.Tensor] = None):
         B, S, _ = x.shape
+        if self.tp_group is not None:
+            from ..parallel.tensor_parallel import tp_copy
+            x = tp_copy(x, self.tp_group)
         qkv = _linear(self.qkv_proj, x)
         q, k, v = qkv.split([self.q_size, self.kv_size, self.kv_size], dim=-1)
         q = q.view(B, S, self.num_heads, self.head_dim)
@@ -338,7 +342,11 @@ class GroupedQueryAttention(nn.Module):
             out = scatter_seq_gather_heads(out.contiguous(), self.sp_size,
                                            self.sp_group)
         out = out.reshape(B, S, self.q_size)
-        return _linear(self.o_proj, out)
+        out = _linear(self.o_proj, out)
+        if self.tp_group is not None:
+            from ..parallel.tensor_parallel import tp_reduce
+            out = tp_reduce(out, self.tp_group)
+        return out
 
 
 class SwiGLUExpert(nn.Module):
@@ -350,15 +358,23 @@ class SwiGLUExpert(nn.Module):
         self.intermediate_size = intermediate_size
         self.gate_up_proj = nn.Linear(hidden_size, 2 * intermediate_size, bias=False)
         self.down_proj = nn.Linear(intermediate_size, hidden_size, bias=False)
+        self.tp_group = None    # set by convert_to_tensor_parallel
 
     def forward(self, x):
+        if self.tp_group is not None:
+            from ..parallel.tensor_parallel import tp_copy
+            x = tp_copy(x, self.tp_group)
         gu = _linear(self.gate_up_proj, x)
         shp = gu.shape[:-1]
         gu2 = gu.view(-1, 2 * self.intermediate_size)
         gate = gu2.narrow(1, 0, self.intermediate_size)
         up = gu2.narrow(1, self.intermediate_size, self.intermediate_size)
         act = ops.swiglu(gate, up).view(*shp, self.intermediate_size)
-        return _linear(self.down_proj, act)
+        y = _linear(self.down_proj, act)
+        if self.tp_group is not None:
+            from ..parallel.tensor_parallel import tp_reduce
+            y = tp_reduce(y, self.tp_group)
+        return y
 
 
 DenseSwiGLU = SwiGLUExpert  # reference alias (model.py:1406)

@@ -28,53 +28,93 @@ _MESH: Optional["ParallelMesh"] = None
 
 
 class ParallelMesh:
-    def __init__(self, ep_size: int = 1, sp_size: int = 1):
+    def __init__(self, ep_size: int = 1, sp_size: int = 1, tp_size: int = 1):
         world = comm.get_world_size()
         rank = comm.get_rank()
-        assert world % max(ep_size, 1) == 0, \
-            f"world {world} not divisible by ep_size {ep_size}"
-        assert world % max(sp_size, 1) == 0, \
-            f"world {world} not divisible by sp_size {sp_size}"
-        assert ep_size <= 1 or sp_size <= 1, \
-            "EP and Ulysses-SP composition is not supported yet"
+        for name, sz in (("ep", ep_size), ("sp", sp_size), ("tp", tp_size)):
+            assert world % max(sz, 1) == 0, \
+                f"world {world} not divisible by {name}_size {sz}"
+        assert sum(s > 1 for s in (ep_size, sp_size, tp_size)) <= 1, \
+            "EP / Ulysses-SP / TP composition is not supported yet"
         self.world = world
         self.rank = rank
         self.ep_size = ep_size
         self.sp_size = sp_size
-        self.dp_size = world // (ep_size * max(sp_size, 1))
+        self.tp_size = tp_size
+        self.dp_size = world // (ep_size * max(sp_size, 1) * max(tp_size, 1))
         self.ep_rank = rank % ep_size if ep_size > 1 else 0
-        self.dp_rank = rank // ep_size
+        self.dp_rank = rank // max(ep_size, tp_size, 1)
         self.sp_rank = rank % sp_size if sp_size > 1 else 0
+        self.tp_rank = rank % tp_size if tp_size > 1 else 0
         self.ep_group = None
         self.expert_dp_group = None
         self.sp_group = None
+        self.tp_group = None
+        self.tp_shard_dp_group = None
+
+        def contiguous_groups(size):
+            mine = None
+            for d in range(world // size):
+                ranks = list(range(d * size, (d + 1) * size))
+                g = dist.new_group(ranks)
+                if rank in ranks:
+                    mine = g
+            return mine
+
+        def strided_groups(size):
+            mine = None
+            for e in range(size):
+                ranks = list(range(e, world, size))
+                g = dist.new_group(ranks)
+                if rank in ranks:
+                    mine = g
+            return mine
+
         if world > 1 and ep_size > 1:
-            # build ALL groups on every rank (dist.new_group is collective)
-            for d in range(world // ep_size):
-                ranks = list(range(d * ep_size, (d + 1) * ep_size))
-                g = dist.new_group(ranks)
-                if rank in ranks:
-                    self.ep_group = g
-            for e in range(ep_size):
-                ranks = list(range(e, world, ep_size))
-                g = dist.new_group(ranks)
-                if rank in ranks:
-                    self.expert_dp_group = g
+            self.ep_group = contiguous_groups(ep_size)
+            self.expert_dp_group = strided_groups(ep_size)
         if world > 1 and sp_size > 1:
-            for d in range(world // sp_size):
-                ranks = list(range(d * sp_size, (d + 1) * sp_size))
-                g = dist.new_group(ranks)
-                if rank in ranks:
-                    self.sp_group = g
+            self.sp_group = contiguous_groups(sp_size)
+        if world > 1 and tp_size > 1:
+            self.tp_group = contiguous_groups(tp_size)
+            self.tp_shard_dp_group = strided_groups(tp_size)
 
     @property
     def expert_dp_size(self) -> int:
         return self.world // self.ep_size if self.ep_size > 1 else self.world
 
+    # ---- generic "sharded param" accessors (EP expert shards and TP weight
+    # shards use the same engine machinery: reduce grads across the replica
+    # group; sum norms across one exchange group) -------------------------
+    @property
+    def shard_replica_group(self):
+        if self.tp_size > 1:
+            return self.tp_shard_dp_group
+        return self.expert_dp_group
 
-def init_mesh(ep_size: int = 1, sp_size: int = 1) -> ParallelMesh:
+    @property
+    def shard_replica_size(self) -> int:
+        if self.tp_size > 1:
+            return self.world // self.tp_size
+        if self.ep_size > 1:
+            return self.world // self.ep_size
+        return 1
+
+    @property
+    def shard_exchange_group(self):
+        if self.tp_size > 1:
+            return self.tp_group
+        return self.ep_group
+
+    @property
+    def shard_exchange_size(self) -> int:
+        return self.tp_size if self.tp_size > 1 else self.ep_size
+
+
+def init_mesh(ep_size: int = 1, sp_size: int = 1,
+              tp_size: int = 1) -> ParallelMesh:
     global _MESH
-    _MESH = ParallelMesh(ep_size, sp_size)
+    _MESH = ParallelMesh(ep_size, sp_size, tp_size)
     return _MESH
 
 

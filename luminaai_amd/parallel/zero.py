@@ -157,11 +157,19 @@ class ZeroEngine:
                     tmp = torch.empty_like(shard)
                     dist.reduce_scatter_tensor(tmp, g.flat_g, group=self.pg)
                     shard.copy_(tmp)
-        # expert grads: replicas of the same expert shard live across the
-        # expert-dp group (no-op when every replica set has one member)
-        if self.mesh is not None and self.mesh.expert_dp_size > 1:
+        # sharded-param grads (EP expert shards / TP weight shards):
+        # replicas of the same shard live across the replica group (no-op
+        # when every replica set has one member)
+        if self.mesh is not None and self.mesh.shard_replica_size > 1:
             for g in self._expert_groups():
-                dist.all_reduce(g.flat_g, group=self.mesh.expert_dp_group)
+                dist.all_reduce(g.flat_g, group=self.mesh.shard_replica_group)
+        if self.mesh is not None and self.mesh.tp_size > 1:
+            # TP ranks SHARE their batch: dense grads sum tp duplicates and
+            # the uniform 1/world grad_scale averages them out, but a weight
+            # shard's grad is computed once — pre-scale by tp so the same
+            # grad_scale (and the same clip norm) applies to every group
+            for g in self._expert_groups():
+                g.flat_g.mul_(float(self.mesh.tp_size))
 
     def global_grad_norm_sq(self) -> Optional[torch.Tensor]:
         if self.opt.max_grad_norm <= 0:
@@ -183,10 +191,11 @@ class ZeroEngine:
             for g in exp:
                 n = K.l2norm_sq(g.flat_g)
                 ns_e = n if ns_e is None else ns_e + n
-            if self.mesh is not None and self.mesh.ep_size > 1:
-                # each rank holds E/ep experts (post expert-dp reduce) ->
-                # summing over ONE ep group covers every expert exactly once
-                dist.all_reduce(ns_e, group=self.mesh.ep_group)
+            if self.mesh is not None and self.mesh.shard_exchange_size > 1:
+                # each rank holds 1/N of the sharded params (post replica
+                # reduce) -> summing over ONE exchange group covers every
+                # shard exactly once
+                dist.all_reduce(ns_e, group=self.mesh.shard_exchange_group)
             ns = ns_e if ns is None else ns + ns_e
         return ns
 
@@ -229,8 +238,8 @@ class ZeroEngine:
             if not g._master_is_params:
                 g.master.copy_(
                     g.weight_view()[g.shard_lo:g.shard_hi].float())
-        if self.mesh is not None and self.mesh.expert_dp_size > 1:
-            pg = self.mesh.expert_dp_group
+        if self.mesh is not None and self.mesh.shard_replica_size > 1:
+            pg = self.mesh.shard_replica_group
             src = dist.get_process_group_ranks(pg)[0]
             for g in self._expert_groups():
                 dist.broadcast(g.weight_view(), src=src, group=pg)

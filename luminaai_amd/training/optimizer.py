@@ -46,7 +46,8 @@ def split_decay_groups(model: nn.Module, ep_active: bool = False) -> List[Dict]:
     for name, p in model.named_parameters():
         if not p.requires_grad:
             continue
-        if ep_active and is_expert_param(name):
+        if (ep_active and is_expert_param(name)) \
+                or getattr(p, "_shard_parallel", False):
             expert.append(p)
         elif any(k in name for k in NO_DECAY_KEYWORDS) or p.dim() <= 1:
             no_decay.append(p)
