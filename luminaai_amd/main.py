@@ -60,6 +60,10 @@ def build_arg_parser() -> argparse.ArgumentParser:
     ap.add_argument("--zero", type=int, default=None, choices=(0, 1, 2))
     ap.add_argument("--ep", type=int, default=None,
                     help="expert-parallel degree")
+    ap.add_argument("--tp", type=int, default=1,
+                    help="tensor-parallel degree (dense blocks)")
+    ap.add_argument("--sp", type=int, default=1,
+                    help="Ulysses sequence-parallel degree")
     ap.add_argument("--precision", default=None,
                     choices=("auto", "fp32", "bf16", "fp16", "fp8"))
     ap.add_argument("--no-adaptive", action="store_true",
@@ -114,9 +118,13 @@ def main(argv: Optional[list] = None) -> dict:
     rank = comm.get_rank()
     world = comm.get_world_size()
     ep = cfg.expert_parallel_size or 1
-    if world > 1 and cfg.use_moe and ep <= 1 and cfg.num_experts % world == 0:
+    if world > 1 and cfg.use_moe and ep <= 1 and args.tp <= 1 \
+            and args.sp <= 1 and cfg.num_experts % world == 0:
         ep = world
-    init_mesh(ep if world > 1 else 1)
+    if world > 1:
+        mesh = init_mesh(ep, sp_size=args.sp, tp_size=args.tp)
+    else:
+        mesh = init_mesh(1)
 
     exp_dir = os.path.join("experiments", cfg.experiment_name)
     logger = ProductionLogger("luminaai", log_dir=exp_dir if rank == 0 else None)
@@ -156,6 +164,10 @@ def main(argv: Optional[list] = None) -> dict:
                        "backend is Triton); the native path is HIP kernels "
                        "+ hipGraphs")
     model = DeepSeekTransformer(config_to_deepseek_config(cfg))
+    if mesh.tp_size > 1:
+        from .parallel.tensor_parallel import convert_to_tensor_parallel
+        n_tp = convert_to_tensor_parallel(model, mesh)
+        logger.info(f"tensor parallel: {n_tp} layers sharded {mesh.tp_size}-way")
     if rank == 0:
         fp = model.get_memory_footprint()
         logger.info(f"model: {fp['total_params'] / 1e6:.1f}M total / "
