@@ -142,6 +142,18 @@ class Trainer:
         weights = batch.get("loss_weights")
         if weights is not None:
             weights = weights.to(self.device, non_blocking=True)
+        if self.mesh is not None and self.mesh.sp_size > 1 \
+                and input_ids.shape[1] % self.mesh.sp_size == 0:
+            # Ulysses SP: replicas share the batch (DistributedSampler groups
+            # them) and each rank trains on its contiguous sequence slice
+            from ..parallel.sequence_parallel import shard_sequence
+            input_ids = shard_sequence(input_ids, self.mesh.sp_rank,
+                                       self.mesh.sp_size)
+            labels = shard_sequence(labels, self.mesh.sp_rank,
+                                    self.mesh.sp_size)
+            if weights is not None:
+                weights = shard_sequence(weights, self.mesh.sp_rank,
+                                         self.mesh.sp_size)
 
         logits, aux, _ = self.model(input_ids)
         out = self.compute_loss(logits, labels, weights, aux)
