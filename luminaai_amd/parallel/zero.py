@@ -13,8 +13,10 @@ the FlatAdamW flat buffers:
   stage 2:          reduce-scatter of the flat grad into this rank's shard
                     (half the wire bytes of all-reduce on xGMI), sharded
                     update, bf16 weight all-gather.
-  stage 3:          parameter sharding with prefetch — NOT implemented yet
-                    (roadmap; a ~70B hybrid model fits stage-2 on 288 GB/GPU).
+  stage 3:          parameter sharding lives in its own engine
+                    (parallel/zero3.py: block-unit resizable flat buffers,
+                    gather/free hooks, shard AdamW); the Trainer selects it
+                    for config.zero_stage >= 3.
 
 Bucket size defaults to ~50 MB of grad per collective: xGMI is 7 independent
 point-to-point links, so several in-flight medium buckets keep all links busy
