@@ -46,7 +46,7 @@ class ZeroEngine:
                  bucket_bytes: int = 50_000_000,
                  overlap_comm: bool = True,
                  process_group=None, mesh=None):
-        assert stage in (0, 1, 2), "ZeRO-3 not implemented yet"
+        assert stage in (0, 1, 2), "use parallel/zero3.Zero3Engine for stage 3"
         self.opt = optimizer
         self.stage = stage
         self.bucket_bytes = bucket_bytes
