@@ -57,7 +57,7 @@ def build_arg_parser() -> argparse.ArgumentParser:
     ap.add_argument("--micro-batch", type=int, default=None)
     ap.add_argument("--accum", type=int, default=None)
     ap.add_argument("--seq-len", type=int, default=None)
-    ap.add_argument("--zero", type=int, default=None, choices=(0, 1, 2))
+    ap.add_argument("--zero", type=int, default=None, choices=(0, 1, 2, 3))
     ap.add_argument("--ep", type=int, default=None,
                     help="expert-parallel degree")
     ap.add_argument("--tp", type=int, default=1,

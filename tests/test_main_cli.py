@@ -56,3 +56,22 @@ def test_unknown_config_field_rejected():
     with pytest.raises(SystemExit):
         main(["--preset", "debug", "--set", "not_a_field=1",
               "--synthetic-steps", "1"])
+
+
+def test_main_zero3_synthetic(tmp_path, monkeypatch):
+    """ZeRO-3 engine reachable from the CLI (single process)."""
+    monkeypatch.chdir(tmp_path)
+    from luminaai_amd.main import main
+    result = main([
+        "--preset", "debug", "--synthetic-steps", "2",
+        "--experiment-name", "cli_z3", "--precision", "fp32",
+        "--micro-batch", "2", "--accum", "1", "--seq-len", "32",
+        "--zero", "3",
+        "--set", "hidden_size=64", "--set", "num_layers=2",
+        "--set", "num_heads=4", "--set", "num_kv_heads=2",
+        "--set", "vocab_size=512", "--set", "intermediate_size=128",
+        "--set", "num_workers=0", "--set", "use_moe=false",
+        "--set", "use_mod=false", "--set", "gradient_checkpointing=false",
+        "--set", "eval_every_n_batches=0", "--set", "save_every_n_batches=0",
+    ])
+    assert result["global_step"] >= 2
