@@ -236,3 +236,42 @@ def test_production_logger(tmp_path):
     pl.info("hello")
     pl.log_metrics({"loss": 1.0}, step=5)
     assert pl.jsonl_path and os.path.exists(pl.jsonl_path)
+
+
+def test_rollback_restores_earlier_checkpoint(tiny_config, tokenizer,
+                                              small_model, tmp_path,
+                                              monkeypatch):
+    """The rollback intervention must actually restore earlier weights."""
+    monkeypatch.chdir(tmp_path)
+    import torch
+    from luminaai_amd.training import Trainer
+    t = Trainer(small_model, tokenizer, tiny_config)
+    t._setup_scheduler(50)
+    torch.manual_seed(0)
+
+    def step():
+        ids = torch.randint(1, tiny_config.vocab_size,
+                            (2, tiny_config.seq_length + 1))
+        t.engine.set_sync(True)
+        t.train_step({"input_ids": ids[:, :-1], "labels": ids[:, 1:]})
+        t.optimizer_step()
+
+    step()
+    t.save_checkpoint()
+    w_saved = t.model.embed_tokens.weight.detach().clone()
+    for _ in range(3):
+        step()
+    assert not torch.allclose(t.model.embed_tokens.weight, w_saved)
+    assert t.rollback_steps(2)
+    torch.testing.assert_close(t.model.embed_tokens.weight.detach(), w_saved)
+    assert t.global_step == 1
+
+    # orchestrator-dispatched rollback path
+    from luminaai_amd.training import AdaptiveDecision, AdaptiveTrainingOrchestrator
+    for _ in range(3):
+        step()
+    orch = AdaptiveTrainingOrchestrator(tiny_config, trainer=t)
+    orch.initialize_training()
+    assert orch._execute_decision(AdaptiveDecision("rollback", 2, "test"))
+    torch.testing.assert_close(t.model.embed_tokens.weight.detach(), w_saved)
+    orch.cleanup()
