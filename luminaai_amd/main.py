@@ -70,6 +70,9 @@ def build_arg_parser() -> argparse.ArgumentParser:
                     help="disable the adaptive orchestrator interventions")
     ap.add_argument("--chinchilla", action="store_true",
                     help="auto-set epochs from Chinchilla scaling")
+    ap.add_argument("--eval-only", action="store_true",
+                    help="evaluate --eval-data (or train data) with the "
+                         "resumed checkpoint and exit")
     ap.add_argument("--set", action="append", default=[], metavar="KEY=VALUE",
                     help="override any Config field")
     return ap
@@ -221,6 +224,17 @@ def main(argv: Optional[list] = None) -> dict:
     trainer.set_metrics_hook(_hook)
 
     t0 = time.time()
+    if args.eval_only:
+        from .data.dataset import create_dataloader
+        ds = eval_ds if eval_ds is not None else train_ds
+        dl = create_dataloader(ds, cfg, shuffle=False)
+        stats = trainer.evaluate(dl)
+        if rank == 0:
+            print(json.dumps({"eval": stats}, default=str))
+        orch.cleanup()
+        wb.finish()
+        comm.cleanup()
+        return {"eval": stats, "global_step": trainer.global_step}
     try:
         result = run_with_oom_protection(orch, trainer, cfg, logger,
                                          train_ds, eval_ds)

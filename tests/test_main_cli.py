@@ -75,3 +75,23 @@ def test_main_zero3_synthetic(tmp_path, monkeypatch):
         "--set", "eval_every_n_batches=0", "--set", "save_every_n_batches=0",
     ])
     assert result["global_step"] >= 2
+
+
+def test_main_eval_only(tmp_path, monkeypatch):
+    monkeypatch.chdir(tmp_path)
+    from luminaai_amd.utils import generate_sample_data
+    data = str(tmp_path / "eval.jsonl")
+    generate_sample_data(data, n=6)
+    from luminaai_amd.main import main
+    result = main([
+        "--preset", "debug", "--train-data", data, "--eval-only",
+        "--experiment-name", "cli_eval", "--precision", "fp32",
+        "--micro-batch", "2", "--seq-len", "32",
+        "--set", "hidden_size=64", "--set", "num_layers=2",
+        "--set", "num_heads=4", "--set", "num_kv_heads=2",
+        "--set", "vocab_size=512", "--set", "intermediate_size=128",
+        "--set", "num_workers=0", "--set", "use_moe=false",
+        "--set", "use_mod=false", "--set", "eval_data_path=",
+    ])
+    assert "loss" in result["eval"]
+    assert result["eval"]["batches"] > 0
