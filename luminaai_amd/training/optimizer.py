@@ -86,12 +86,15 @@ class _FlatGroup:
         device = params[0].device
         self.dtype = params[0].dtype
         self.device = device
+        # bf16 AND fp16 params keep low-precision flat weights + an fp32
+        # master copy; fp32 params alias the master directly
+        self.low_prec = self.dtype in (torch.bfloat16, torch.float16)
         self.bf16 = self.dtype == torch.bfloat16
         sharded = shard_world > 1
 
-        if self.bf16:
+        if self.low_prec:
             self.flat_w = torch.zeros(self.padded, device=device,
-                                      dtype=torch.bfloat16)
+                                      dtype=self.dtype)
         else:
             assert not sharded or self.dtype == torch.float32
             self.flat_w = (torch.zeros(self.padded, device=device,

@@ -114,6 +114,11 @@ class Trainer:
         self._lr_override = None        # (lr, until_step, emergency)
         self._tokens_seen = 0
         self._step_t0 = None
+        # dynamic loss scaling (fp16 only; bf16/fp32 train unscaled —
+        # reference PrecisionManager semantics, trainer.py:157-356)
+        self.loss_scale = (float(getattr(config, "fp16_loss_scale", 65536.0))
+                           if self.precision.needs_loss_scale else 1.0)
+        self._scale_good_steps = 0
         exp = config.experiment_name or "default"
         self.checkpoints = CheckpointManager(
             os.path.join("checkpoints", exp), config.save_total_limit)
@@ -159,7 +164,10 @@ class Trainer:
         out = self.compute_loss(logits, labels, weights, aux)
         # grads are SUMMED over micro-batches; normalisation happens once in
         # the fused optimizer kernel via grad_scale = 1/(accum*world).
-        out["loss"].backward()
+        if self.loss_scale != 1.0:
+            (out["loss"] * self.loss_scale).backward()
+        else:
+            out["loss"].backward()
 
         self._tokens_seen += input_ids.numel() * comm.get_world_size()
         self._micro_in_cycle += 1
@@ -175,7 +183,10 @@ class Trainer:
         world = comm.get_world_size()
         grad_scale = 1.0 / (self._micro_in_cycle * world) \
             if self._micro_in_cycle else 1.0
+        grad_scale /= self.loss_scale
         self.engine.step(grad_scale=grad_scale)
+        if self.loss_scale != 1.0:
+            self._adjust_loss_scale()
         self.engine.zero_grad()
         if self.precision.spec.name == "fp8":
             from ..ops import fp8 as _fp8
@@ -186,6 +197,27 @@ class Trainer:
         if self.scheduler is not None and self._lr_override is None:
             self.scheduler.step()
         return {"lr": self.get_lr(), "step": self.global_step}
+
+    def _adjust_loss_scale(self):
+        """Dynamic fp16 scaling: the fused AdamW kernel already SKIPS a
+        non-finite step on-device; here the scale backs off on overflow and
+        grows after a streak of good steps (one device read per step —
+        fp16 is the non-default path)."""
+        ns = self.optimizer._last_norm_sq
+        finite = True
+        if ns is not None:
+            try:
+                finite = bool(torch.isfinite(ns.sum()))
+            except RuntimeError:
+                finite = True
+        if not finite:
+            self.loss_scale = max(1.0, self.loss_scale * 0.5)
+            self._scale_good_steps = 0
+        else:
+            self._scale_good_steps += 1
+            if self._scale_good_steps >= 2000:
+                self.loss_scale = min(self.loss_scale * 2.0, 2.0 ** 24)
+                self._scale_good_steps = 0
 
     # ================================================== epoch / train loops
     def train_epoch(self, dataloader, epoch: int) -> Dict:

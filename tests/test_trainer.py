@@ -308,3 +308,33 @@ def test_scheduler_state_roundtrip():
     s2.load_state_dict(sd)
     assert s2.step_num == 37
     assert opt2.groups[0].lr == pytest.approx(opt.groups[0].lr)
+
+
+def test_fp16_loss_scaling(tokenizer):
+    """fp16 trains with dynamic loss scaling; overflow halves the scale."""
+    import torch
+    from luminaai_amd.config import Config
+    from luminaai_amd.models import DeepSeekTransformer, config_to_deepseek_config
+    from luminaai_amd.training import Trainer
+    cfg = Config(vocab_size=512, hidden_size=64, num_layers=2, num_heads=4,
+                 num_kv_heads=2, seq_length=32, intermediate_size=128,
+                 micro_batch_size=2, gradient_accumulation_steps=1,
+                 num_workers=0, use_moe=False, use_mod=False, zero_stage=0,
+                 precision="fp16", experiment_name="fp16_test",
+                 eval_every_n_batches=0, save_every_n_batches=0)
+    torch.manual_seed(0)
+    model = DeepSeekTransformer(config_to_deepseek_config(cfg))
+    t = Trainer(model, tokenizer, cfg)
+    t._setup_scheduler(10)
+    assert t.loss_scale == 65536.0
+    w0 = t.model.embed_tokens.weight.detach().float().clone()
+    ids = torch.randint(1, 512, (2, 33))
+    t.engine.set_sync(True)
+    out = t.train_step({"input_ids": ids[:, :-1], "labels": ids[:, 1:]})
+    t.optimizer_step()
+    assert torch.isfinite(out["ce_loss"].detach().float())
+    assert not torch.allclose(t.model.embed_tokens.weight.detach().float(), w0)
+    # simulated overflow -> scale backs off
+    t.optimizer._last_norm_sq = torch.tensor([float("inf")])
+    t._adjust_loss_scale()
+    assert t.loss_scale == 32768.0
