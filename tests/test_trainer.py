@@ -338,3 +338,52 @@ def test_fp16_loss_scaling(tokenizer):
     t.optimizer._last_norm_sq = torch.tensor([float("inf")])
     t._adjust_loss_scale()
     assert t.loss_scale == 32768.0
+
+
+def test_early_stopping_via_eval(tiny_config, tokenizer, small_model):
+    import torch
+    from torch.utils.data import DataLoader, TensorDataset
+    from luminaai_amd.training import Trainer
+    tiny_config.early_stopping_patience = 1
+    t = Trainer(small_model, tokenizer, tiny_config)
+    t._setup_scheduler(10)
+    torch.manual_seed(0)
+    ids = torch.randint(1, tiny_config.vocab_size, (4, tiny_config.seq_length + 1))
+    eval_batches = [{"input_ids": ids[:, :-1], "labels": ids[:, 1:]}]
+    r1 = t.evaluate(eval_batches)
+    assert not t.should_stop
+    # train a bit with high LR so eval on the SAME random data worsens or at
+    # best stays: force the pathway by faking a best
+    t.best_eval_loss = -1.0   # anything is "no improvement" now
+    t._no_improve_evals = 0
+    t.evaluate(eval_batches)
+    assert t.should_stop, "patience=1 with no improvement must stop"
+
+
+def test_training_continues_after_expert_add(tiny_moe_config, tokenizer):
+    import torch
+    from luminaai_amd.models import DeepSeekTransformer, config_to_deepseek_config
+    from luminaai_amd.training import Trainer
+    torch.manual_seed(0)
+    model = DeepSeekTransformer(config_to_deepseek_config(tiny_moe_config))
+    t = Trainer(model, tokenizer, tiny_moe_config)
+    t._setup_scheduler(20)
+
+    def step():
+        ids = torch.randint(1, tiny_moe_config.vocab_size,
+                            (2, tiny_moe_config.seq_length + 1))
+        t.engine.set_sync(True)
+        out = t.train_step({"input_ids": ids[:, :-1], "labels": ids[:, 1:]})
+        t.optimizer_step()
+        return float(out["ce_loss"].detach())
+
+    step()
+    e0 = t.model.get_moe_layers()[0].num_experts
+    assert t.add_expert()
+    assert t.model.get_moe_layers()[0].num_experts == e0 + 1
+    loss = step()                      # optimizer rebuilt; training continues
+    assert loss == loss and loss > 0
+    assert t.prune_expert()
+    assert t.model.get_moe_layers()[0].num_experts == e0
+    loss = step()
+    assert loss == loss and loss > 0
