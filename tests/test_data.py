@@ -143,3 +143,32 @@ def test_acquisition_cli(tmp_path, capsys):
                    '{"role": "assistant", "content": "a fine answer here"}]}\n')
     stats = acq_main(["validate", str(raw)])
     assert stats["valid"] == 1
+
+
+def test_hybrid_dataset_modes(tokenizer, sample_conversations, sample_text,
+                              tiny_config):
+    """base_only / finetuning_only / hybrid mode detection (reference
+    FastHybridDatasetManager, dataset.py:566-761)."""
+    from luminaai_amd.data.dataset import setup_datasets
+    cfg = tiny_config
+    # finetuning only
+    cfg.train_data_path = sample_conversations
+    cfg.eval_data_path = ""
+    train, ev = setup_datasets(cfg, tokenizer)
+    assert len(train) > 0
+    row = train[0]
+    assert "loss_weights" in row
+    # base only
+    cfg.train_data_path = sample_text
+    train2, _ = setup_datasets(cfg, tokenizer)
+    assert len(train2) > 0
+    assert "input_ids" in train2[0]
+
+
+def test_interleaved_ratio(tokenizer, sample_conversations):
+    from luminaai_amd.data.dataset import ConversationDataset, InterleavedDataset
+    a = ConversationDataset(sample_conversations, tokenizer, 32)
+    b = ConversationDataset(sample_conversations, tokenizer, 32)
+    mix = InterleavedDataset(a, b, ratio_a=0.75)
+    assert len(mix) > 0
+    _ = [mix[i] for i in range(min(6, len(mix)))]

@@ -275,3 +275,31 @@ def test_rollback_restores_earlier_checkpoint(tiny_config, tokenizer,
     assert orch._execute_decision(AdaptiveDecision("rollback", 2, "test"))
     torch.testing.assert_close(t.model.embed_tokens.weight.detach(), w_saved)
     orch.cleanup()
+
+
+def test_meta_learning_suggestions(tmp_path, tiny_config):
+    from luminaai_amd.training.orchestrator import MetaLearningEngine
+    m = MetaLearningEngine(str(tmp_path / "meta.json"))
+    assert m.suggest_hyperparameters(tiny_config) is None
+    cfg_a = tiny_config
+    cfg_a.learning_rate = 3e-4
+    m.record_run(cfg_a, final_loss=2.0, steps=100, interventions=1)
+    cfg_a.learning_rate = 1e-3
+    m.record_run(cfg_a, final_loss=5.0, steps=100, interventions=0)
+    sug = m.suggest_hyperparameters(tiny_config)
+    assert sug is not None
+    assert sug["learning_rate"] == pytest.approx(3e-4)  # lower-loss run wins
+    # persisted across instances
+    m2 = MetaLearningEngine(str(tmp_path / "meta.json"))
+    assert len(m2.runs) == 2
+
+
+def test_production_monitoring_scorers():
+    from luminaai_amd.training.orchestrator import ProductionMonitoring
+    pm = ProductionMonitoring()
+    rec = pm.score_sample("hello")
+    assert rec["semantic_drift"] is None       # honest stub, not random
+    pm.register_scorers(drift=lambda t: 0.25, toxicity=lambda t: 0.0)
+    rec = pm.score_sample("hello")
+    assert rec["semantic_drift"] == 0.25
+    assert len(pm.history) == 2
