@@ -123,3 +123,57 @@ def test_pp_1f1b_matches_single_process():
                                                   rel=1e-4)
     assert results[0]["rel_grad_err"] < 1e-3
     assert results[1]["rel_grad_err"] < 1e-3
+
+
+def pp_worker_small(rank, world):
+    """Edge schedules: fewer micro-batches than warmup depth."""
+    from luminaai_amd.models import DeepSeekTransformer
+    from luminaai_amd.parallel.pipeline import PipelineParallelEngine
+    mcfg = _model_cfg()
+    torch.manual_seed(1234)
+    model = DeepSeekTransformer(mcfg)
+    engine = PipelineParallelEngine(model, None)
+    torch.manual_seed(901)
+    for n in (1, 2):
+        micro = []
+        for _ in range(n):
+            ids = torch.randint(1, mcfg.vocab_size, (2, 17))
+            micro.append({"input_ids": ids[:, :-1], "labels": ids[:, 1:]})
+        out = engine.train_batch(micro)
+        assert out["n_micro"] == n
+        for p in engine.parameters():
+            if p.grad is not None:
+                p.grad.zero_()
+    return {"ok": True}
+
+
+def test_pp_small_microbatch_counts():
+    ctx = mp.get_context("spawn")
+    q = ctx.SimpleQueue()
+    port = _free_port()
+
+    def _run2(rank, world, port, q):
+        os.environ.update({
+            "MASTER_ADDR": "127.0.0.1", "MASTER_PORT": str(port),
+            "RANK": str(rank), "LOCAL_RANK": str(rank),
+            "WORLD_SIZE": str(world),
+        })
+        dist.init_process_group("gloo", init_method="env://", rank=rank,
+                                world_size=world)
+        try:
+            q.put((rank, "ok", pp_worker_small(rank, world)))
+        except Exception:  # noqa: BLE001
+            import traceback
+            q.put((rank, "err", traceback.format_exc()))
+        finally:
+            dist.destroy_process_group()
+
+    procs = [ctx.Process(target=_run2, args=(r, WORLD, port, q))
+             for r in range(WORLD)]
+    for p in procs:
+        p.start()
+    for _ in range(WORLD):
+        rank, status, payload = q.get()
+        assert status == "ok", f"rank {rank} failed:\n{payload}"
+    for p in procs:
+        p.join(timeout=60)
