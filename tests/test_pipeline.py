@@ -147,28 +147,28 @@ def pp_worker_small(rank, world):
     return {"ok": True}
 
 
+def _run_small(rank, world, port, q):
+    os.environ.update({
+        "MASTER_ADDR": "127.0.0.1", "MASTER_PORT": str(port),
+        "RANK": str(rank), "LOCAL_RANK": str(rank),
+        "WORLD_SIZE": str(world),
+    })
+    dist.init_process_group("gloo", init_method="env://", rank=rank,
+                            world_size=world)
+    try:
+        q.put((rank, "ok", pp_worker_small(rank, world)))
+    except Exception:  # noqa: BLE001
+        import traceback
+        q.put((rank, "err", traceback.format_exc()))
+    finally:
+        dist.destroy_process_group()
+
+
 def test_pp_small_microbatch_counts():
     ctx = mp.get_context("spawn")
     q = ctx.SimpleQueue()
     port = _free_port()
-
-    def _run2(rank, world, port, q):
-        os.environ.update({
-            "MASTER_ADDR": "127.0.0.1", "MASTER_PORT": str(port),
-            "RANK": str(rank), "LOCAL_RANK": str(rank),
-            "WORLD_SIZE": str(world),
-        })
-        dist.init_process_group("gloo", init_method="env://", rank=rank,
-                                world_size=world)
-        try:
-            q.put((rank, "ok", pp_worker_small(rank, world)))
-        except Exception:  # noqa: BLE001
-            import traceback
-            q.put((rank, "err", traceback.format_exc()))
-        finally:
-            dist.destroy_process_group()
-
-    procs = [ctx.Process(target=_run2, args=(r, WORLD, port, q))
+    procs = [ctx.Process(target=_run_small, args=(r, WORLD, port, q))
              for r in range(WORLD)]
     for p in procs:
         p.start()
