@@ -181,3 +181,64 @@ def test_static_kv_cache_matches_dynamic(small_model):
             b, _, _ = m(ids[:, t:t + 1], kv_caches=sta)
             torch.testing.assert_close(a, b, rtol=1e-5, atol=1e-5)
         assert sta[0].k.shape[1] == 16  # preallocated, not grown
+
+
+def test_prompt_truncated_to_max_context(small_model, tokenizer):
+    """A prompt longer than max_context is left-truncated before prefill,
+    and generation halts once the KV cache is full."""
+    eng = GenerationEngine(small_model.eval(), tokenizer)
+    prompt = list(range(2, 42))  # 40 tokens, max_context 16
+    cfg = GenerationConfig(max_new_tokens=8, temperature=0.0, max_context=16)
+    out = eng.generate(prompt, cfg)
+    assert eng.get_stats()["prefill_tokens"] == 16
+    # cache is already full after prefill: at most one sampled token fits
+    assert len(out) <= 1
+
+
+def test_stop_token_ends_generation(small_model, tokenizer):
+    eng = GenerationEngine(small_model.eval(), tokenizer)
+    cfg = GenerationConfig(max_new_tokens=8, temperature=0.0)
+    first = eng.generate(tokenizer.encode("abc"), cfg)
+    assert first
+    cfg2 = GenerationConfig(max_new_tokens=8, temperature=0.0,
+                            stop_token_ids=[first[0]])
+    out = eng.generate(tokenizer.encode("abc"), cfg2)
+    assert out == []  # greedy path re-derives first[0], which now stops
+
+
+def test_load_zero_shards_merges_optimizer(tmp_path, small_model):
+    """Per-rank ZeRO shards merge into one payload: model from rank0,
+    flat-optimizer partitions concatenated per group."""
+    from luminaai_amd.inference.loader import load_zero_shards
+    sd = small_model.state_dict()
+    for r in range(2):
+        opt = {"groups": [{
+            "exp_avg": torch.full((4,), float(r)),
+            "exp_avg_sq": torch.full((4,), float(10 + r)),
+            "master": torch.full((4,), float(20 + r)),
+            "lr": 1e-3,
+        }], "step": 5}
+        torch.save({"model_state_dict": sd, "optimizer_state_dict": opt,
+                    "global_step": 5},
+                   tmp_path / f"optim_shard_ck_rank{r}.pt")
+    merged = load_zero_shards(str(tmp_path))
+    g = merged["optimizer_state_dict"]["groups"][0]
+    assert g["exp_avg"].shape == (8,)
+    assert float(g["exp_avg"][0]) == 0.0 and float(g["exp_avg"][4]) == 1.0
+    assert float(g["master"][4]) == 21.0
+    assert "embed_tokens.weight" in merged["model_state_dict"]
+
+
+def test_find_latest_checkpoint(tmp_path, monkeypatch):
+    from luminaai_amd.inference.loader import find_latest_checkpoint
+    monkeypatch.chdir(tmp_path)
+    assert find_latest_checkpoint() is None
+    import os as _os
+    ckdir = tmp_path / "checkpoints" / "exp"
+    ckdir.mkdir(parents=True)
+    torch.save({"a": 1}, ckdir / "old.pt")
+    _os.utime(ckdir / "old.pt", (1000, 1000))
+    torch.save({"a": 2}, ckdir / "new.pt")
+    (ckdir / "skip.pt.tmp").write_text("x")
+    latest = find_latest_checkpoint()
+    assert latest.endswith("new.pt")
