@@ -69,7 +69,7 @@ class _FlatGroup:
 
     def __init__(self, params: List[torch.Tensor], lr: float, weight_decay: float,
                  shard_rank: int = 0, shard_world: int = 1,
-                 comm: str = "dp"):
+                 comm: str = "dp", offload: bool = False):
         self.params = params
         self.lr = lr
         self.weight_decay = weight_decay
@@ -120,13 +120,30 @@ class _FlatGroup:
             self.offsets.append((off, n))
             off += n
 
-        # optimizer state: shard-size (== padded when world==1)
+        # optimizer state: shard-size (== padded when world==1).
+        # offload=True keeps master/m/v in host (pinned) memory and steps on
+        # the CPU — the MI355X-native stand-in for the reference's DeepSpeed
+        # optimizer offload (backend_deepspeed.py:129-165) and ColossalAI
+        # cpu_adam (extensions/csrc/cuda/cpu_adam.cpp). Unavailable when
+        # params alias the master (fp32, unsharded): the model itself would
+        # leave the GPU.
+        self.offload = bool(offload) and not w_master_full
+        pin = device.type == "cuda"
+        state_dev = torch.device("cpu") if self.offload else device
         if not w_master_full:
-            self.master = self.flat_w[self.shard_lo:self.shard_hi].float()
+            shard = self.flat_w[self.shard_lo:self.shard_hi].float()
+            self.master = (shard.cpu().pin_memory() if self.offload and pin
+                           else shard.cpu() if self.offload else shard)
         state_size = self.padded if w_master_full else self.shard_size
-        self.m = torch.zeros(state_size, device=device, dtype=torch.float32)
+        self.m = torch.zeros(state_size, device=state_dev, dtype=torch.float32)
         self.v = torch.zeros_like(self.m)
         self._master_is_params = w_master_full
+        if self.offload and pin:
+            # pinned staging for async D2H grad / H2D weight transfers
+            self._g_stage = torch.zeros(self.shard_size, dtype=self.dtype,
+                                        pin_memory=True)
+            self._w_stage = torch.zeros(self.shard_size, dtype=self.dtype,
+                                        pin_memory=True)
 
     # ---- views used by step() -------------------------------------------
     def update_grad(self) -> torch.Tensor:
@@ -154,7 +171,7 @@ class FlatAdamW:
                  betas: Tuple[float, float] = (0.9, 0.95), eps: float = 1e-8,
                  weight_decay: float = 0.01, max_grad_norm: float = 1.0,
                  shard_rank: int = 0, shard_world: int = 1,
-                 ep_active: bool = False):
+                 ep_active: bool = False, offload: bool = False):
         if isinstance(model_or_groups, nn.Module):
             groups = split_decay_groups(model_or_groups, ep_active=ep_active)
         else:
@@ -166,6 +183,7 @@ class FlatAdamW:
         self.shard_rank = shard_rank
         self.shard_world = shard_world
         self.ep_active = ep_active
+        self.offload = offload
         self.step_count = 0
         # expert groups are EP-sharded by construction -> no ZeRO shard on top
         self.groups: List[_FlatGroup] = [
@@ -174,7 +192,7 @@ class FlatAdamW:
                                      else g["weight_decay"]),
                        shard_rank=0 if g.get("comm") == "expert" else shard_rank,
                        shard_world=1 if g.get("comm") == "expert" else shard_world,
-                       comm=g.get("comm", "dp"))
+                       comm=g.get("comm", "dp"), offload=offload)
             for g in groups
         ]
         self._last_norm_sq: Optional[torch.Tensor] = None
@@ -222,11 +240,40 @@ class FlatAdamW:
             self._norm_pinned.copy_(norm_sq.sum().reshape(1).float(),
                                     non_blocking=True)
             self._norm_event.record()
+        offloaded = [g for g in self.groups
+                     if g.offload and g.device.type == "cuda"]
         for g in self.groups:
+            if g in offloaded:
+                continue
             K.adamw_step(
                 g.master, g.update_grad(), g.m, g.v, g.update_weight_out(),
                 g.lr, self.betas[0], self.betas[1], self.eps, g.weight_decay,
                 self.step_count, norm_sq, self.max_grad_norm, grad_scale)
+        if offloaded:
+            self._step_offloaded(offloaded, norm_sq, grad_scale)
+
+    @torch.no_grad()
+    def _step_offloaded(self, groups: List["_FlatGroup"],
+                        norm_sq: Optional[torch.Tensor], grad_scale: float):
+        """CPU AdamW on host-resident state: async D2H grad staging, one
+        sync, vectorized host math, async H2D of the refreshed low-precision
+        weight shard. GPU-resident groups' kernels were launched first so
+        they overlap with the transfers."""
+        for g in groups:
+            g._g_stage.copy_(g.update_grad(), non_blocking=True)
+        torch.cuda.synchronize()
+        norm_cpu = (norm_sq.sum().cpu() if norm_sq is not None else None)
+        if norm_cpu is not None:
+            gn = float(norm_cpu) ** 0.5 * grad_scale
+            if not math.isfinite(gn):
+                return  # NaN/Inf grads: skip, and keep flat_w untouched
+        for g in groups:
+            K.adamw_step(
+                g.master, g._g_stage, g.m, g.v, g._w_stage,
+                g.lr, self.betas[0], self.betas[1], self.eps, g.weight_decay,
+                self.step_count, norm_cpu, self.max_grad_norm, grad_scale)
+            g.flat_w[g.shard_lo:g.shard_hi].copy_(g._w_stage,
+                                                  non_blocking=True)
 
     def last_grad_norm(self) -> float:
         """Host-visible grad norm of the most recent COMPLETED step. On GPU
@@ -300,7 +347,7 @@ class FlatAdamW:
                                           if wd is None else wd),
                             shard_rank=0 if expert else self.shard_rank,
                             shard_world=1 if expert else self.shard_world,
-                            comm=gd.get("comm", "dp"))
+                            comm=gd.get("comm", "dp"), offload=self.offload)
             if fg.shard_world == 1:
                 for p, (off, n) in zip(fg.params, fg.offsets):
                     st = old_state.get(id(p))

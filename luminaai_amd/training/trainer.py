@@ -93,7 +93,9 @@ class Trainer:
                 weight_decay=config.weight_decay, max_grad_norm=1.0,
                 shard_rank=comm.get_rank() if config.zero_stage in (1, 2) else 0,
                 shard_world=comm.get_world_size() if config.zero_stage in (1, 2) else 1,
-                ep_active=ep_active)
+                ep_active=ep_active,
+                offload=bool(getattr(config, "cpu_offload_optimizer", False)
+                             or getattr(config, "aggressive_cpu_offload", False)))
             self.engine = ZeroEngine(
                 self.optimizer, stage=min(config.zero_stage, 2),
                 bucket_bytes=config.reduce_bucket_size,
