@@ -27,7 +27,8 @@ class ChatInterface:
                  tokenizer: Optional[ConversationTokenizer] = None,
                  device: Optional[str] = None,
                  history_window: int = 8,
-                 system_prompt: str = "You are a helpful assistant."):
+                 system_prompt: str = "You are a helpful assistant.",
+                 quantize: Optional[str] = None):
         self.device = torch.device(device) if device else (
             torch.device("cuda") if torch.cuda.is_available()
             else torch.device("cpu"))
@@ -44,6 +45,11 @@ class ChatInterface:
             print(f"loaded {path}: {cfg.num_layers}L/{cfg.hidden_size}h, "
                   f"moe={cfg.use_moe}")
         self.model = model.to(self.device).eval()
+        if quantize:
+            from ..ops.quant import quantize_model, quantized_model_bytes
+            n = quantize_model(self.model, mode=quantize)
+            print(f"quantized {n} Linear layers to {quantize} "
+                  f"({quantized_model_bytes(self.model) / 1e6:.1f} MB)")
         self.tokenizer = tokenizer or ConversationTokenizer()
         self.engine = GenerationEngine(self.model, self.tokenizer, self.device)
         self.gen_config = GenerationConfig.from_mode("standard")
@@ -147,8 +153,11 @@ def main():
     ap = argparse.ArgumentParser(description="LuminaAI-AMD chat REPL")
     ap.add_argument("--checkpoint", default=None)
     ap.add_argument("--device", default=None)
+    ap.add_argument("--quantize", default=None, choices=["int8", "int4", "fp8"],
+                    help="weight-only quantization for inference")
     args = ap.parse_args()
-    ChatInterface(checkpoint=args.checkpoint, device=args.device).run()
+    ChatInterface(checkpoint=args.checkpoint, device=args.device,
+                  quantize=args.quantize).run()
 
 
 if __name__ == "__main__":
