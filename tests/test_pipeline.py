@@ -177,3 +177,113 @@ def test_pp_small_microbatch_counts():
         assert status == "ok", f"rank {rank} failed:\n{payload}"
     for p in procs:
         p.join(timeout=60)
+
+
+def test_interleaved_single_process_matches_full():
+    """pp=1, v=2: the chunked local pipeline reproduces the plain model's
+    loss and gradients exactly."""
+    from luminaai_amd.models import DeepSeekTransformer
+    from luminaai_amd.parallel.pipeline import InterleavedPipelineEngine
+    from luminaai_amd.ops import fused_cross_entropy
+    mcfg = _model_cfg()
+    torch.manual_seed(77)
+    model = DeepSeekTransformer(mcfg)
+    eng = InterleavedPipelineEngine(model, None, virtual_stages=2)
+    torch.manual_seed(5)
+    micro = []
+    for _ in range(3):
+        ids = torch.randint(1, mcfg.vocab_size, (2, 17))
+        micro.append({"input_ids": ids[:, :-1], "labels": ids[:, 1:]})
+    out = eng.train_batch(micro)
+
+    torch.manual_seed(77)
+    ref = DeepSeekTransformer(mcfg)
+    ref_losses = []
+    for mb in micro:
+        logits, aux, _ = ref(mb["input_ids"])
+        ce, _, _ = fused_cross_entropy(logits, mb["labels"])
+        (ce + aux).backward()
+        ref_losses.append(float(ce))
+    assert float(out["loss"]) == pytest.approx(
+        sum(ref_losses) / len(ref_losses), rel=1e-5)
+    g_pp = eng.chunks[0].embed_tokens.weight.grad
+    g_ref = ref.embed_tokens.weight.grad
+    torch.testing.assert_close(g_pp, g_ref, rtol=1e-4, atol=1e-6)
+    g_pp = eng.chunks[-1].lm_head.weight.grad
+    g_ref = ref.lm_head.weight.grad
+    torch.testing.assert_close(g_pp, g_ref, rtol=1e-4, atol=1e-6)
+
+
+def interleaved_worker(rank, world):
+    from luminaai_amd.models import DeepSeekTransformer
+    from luminaai_amd.parallel.pipeline import InterleavedPipelineEngine
+    from luminaai_amd.ops import fused_cross_entropy
+    mcfg = _model_cfg()
+    torch.manual_seed(1234)
+    model = DeepSeekTransformer(mcfg)
+    engine = InterleavedPipelineEngine(model, None, virtual_stages=2)
+    assert engine.stage_ids == [rank, 2 + rank]
+    torch.manual_seed(900)
+    micro = []
+    for _ in range(4):
+        ids = torch.randint(1, mcfg.vocab_size, (2, 33))
+        micro.append({"input_ids": ids[:, :-1], "labels": ids[:, 1:]})
+    out = engine.train_batch(micro)
+
+    torch.manual_seed(1234)
+    ref = DeepSeekTransformer(mcfg)
+    ref_losses = []
+    for mb in micro:
+        logits, aux, _ = ref(mb["input_ids"])
+        ce, _, _ = fused_cross_entropy(logits, mb["labels"])
+        (ce + aux).backward()
+        ref_losses.append(float(ce))
+    res = {"pp_loss": float(out["loss"]),
+           "ref_loss": sum(ref_losses) / len(ref_losses)}
+    if rank == 0:
+        g_pp = engine.chunks[0].embed_tokens.weight.grad
+        g_ref = ref.embed_tokens.weight.grad
+    else:
+        g_pp = engine.chunks[-1].lm_head.weight.grad
+        g_ref = ref.lm_head.weight.grad
+    res["rel_grad_err"] = float((g_pp - g_ref).abs().max()
+                                / g_ref.abs().max().clamp_min(1e-12))
+    return res
+
+
+def _run_interleaved(rank, world, port, q):
+    os.environ.update({
+        "MASTER_ADDR": "127.0.0.1", "MASTER_PORT": str(port),
+        "RANK": str(rank), "LOCAL_RANK": str(rank),
+        "WORLD_SIZE": str(world),
+    })
+    dist.init_process_group("gloo", init_method="env://", rank=rank,
+                            world_size=world)
+    try:
+        q.put((rank, "ok", interleaved_worker(rank, world)))
+    except Exception:  # noqa: BLE001
+        import traceback
+        q.put((rank, "err", traceback.format_exc()))
+    finally:
+        dist.destroy_process_group()
+
+
+def test_interleaved_two_ranks_matches_single_process():
+    ctx = mp.get_context("spawn")
+    q = ctx.SimpleQueue()
+    port = _free_port()
+    procs = [ctx.Process(target=_run_interleaved, args=(r, WORLD, port, q))
+             for r in range(WORLD)]
+    for p in procs:
+        p.start()
+    results = {}
+    for _ in range(WORLD):
+        rank, status, payload = q.get()
+        assert status == "ok", f"rank {rank} failed:\n{payload}"
+        results[rank] = payload
+    for p in procs:
+        p.join(timeout=120)
+    assert results[1]["pp_loss"] == pytest.approx(results[1]["ref_loss"],
+                                                  rel=1e-4)
+    assert results[0]["rel_grad_err"] < 1e-3
+    assert results[1]["rel_grad_err"] < 1e-3
