@@ -1,0 +1,188 @@
+"""HTTP inference server: OpenAI-compatible completions over the
+KV-cached GenerationEngine.
+
+The reference ships only a terminal REPL (reference Chat.py:472-937); this
+module adds the serving path a production deployment needs: an ASGI app
+with /v1/completions, /v1/chat/completions (ChatML via the tokenizer's
+conversation encoding), SSE streaming, /health and /v1/models. Decode is
+serialised with a lock (one hipGraph replay chain owns the GPU at a time);
+continuous batching is a round-2 item (ROADMAP.md).
+
+Run: python serve.py --checkpoint PATH [--port 8000] [--quantize int8]
+"""
+
+from __future__ import annotations
+
+import json
+import queue
+import threading
+import time
+from typing import Dict, List, Optional
+
+import torch
+
+from .engine import GenerationConfig, GenerationEngine
+
+
+def _gen_config(body: Dict) -> GenerationConfig:
+    cfg = GenerationConfig.from_mode(body.get("mode", "standard"))
+    if "max_tokens" in body:
+        cfg.max_new_tokens = int(body["max_tokens"])
+    if "temperature" in body:
+        cfg.temperature = float(body["temperature"])
+    if "top_p" in body:
+        cfg.top_p = float(body["top_p"])
+    if "top_k" in body:
+        cfg.top_k = int(body["top_k"])
+    if "repetition_penalty" in body:
+        cfg.repetition_penalty = float(body["repetition_penalty"])
+    if "stop_token_ids" in body:
+        cfg.stop_token_ids = list(body["stop_token_ids"])
+    return cfg
+
+
+def create_app(model, tokenizer, model_name: str = "luminaai-amd"):
+    """Build the ASGI app around an already-loaded model."""
+    from fastapi import FastAPI
+    from fastapi.responses import JSONResponse, StreamingResponse
+
+    app = FastAPI(title="LuminaAI-AMD", version="0.1")
+    device = next(model.parameters()).device
+    engine = GenerationEngine(model, tokenizer, device)
+    lock = threading.Lock()
+    started = time.time()
+    counters = {"requests": 0, "tokens_out": 0}
+
+    def _decode(prompt_ids: List[int], cfg: GenerationConfig,
+                cb=None) -> List[int]:
+        with lock:
+            counters["requests"] += 1
+            out = engine.generate(prompt_ids, cfg, stream_callback=cb)
+            counters["tokens_out"] += len(out)
+            return out
+
+    def _sse_stream(prompt_ids: List[int], cfg: GenerationConfig,
+                    wrap) -> StreamingResponse:
+        q: "queue.Queue[Optional[int]]" = queue.Queue()
+
+        def run():
+            try:
+                _decode(prompt_ids, cfg, cb=q.put)
+            finally:
+                q.put(None)
+
+        threading.Thread(target=run, daemon=True).start()
+
+        def gen():
+            while True:
+                tok = q.get()
+                if tok is None:
+                    break
+                piece = tokenizer.decode([tok])
+                yield f"data: {json.dumps(wrap(piece))}\n\n"
+            yield "data: [DONE]\n\n"
+
+        return StreamingResponse(gen(), media_type="text/event-stream")
+
+    @app.get("/health")
+    def health():
+        return {"status": "ok", "device": str(device),
+                "uptime_s": round(time.time() - started, 1), **counters}
+
+    @app.get("/v1/models")
+    def models():
+        return {"object": "list",
+                "data": [{"id": model_name, "object": "model"}]}
+
+    @app.post("/v1/completions")
+    def completions(body: Dict):
+        prompt = body.get("prompt", "")
+        cfg = _gen_config(body)
+        ids = tokenizer.encode(prompt)
+        if body.get("stream"):
+            return _sse_stream(ids, cfg, lambda piece: {
+                "object": "text_completion.chunk",
+                "choices": [{"text": piece, "index": 0}]})
+        toks = _decode(ids, cfg)
+        return JSONResponse({
+            "object": "text_completion",
+            "model": model_name,
+            "choices": [{"text": tokenizer.decode(toks), "index": 0,
+                         "finish_reason": "stop"}],
+            "usage": {"prompt_tokens": len(ids),
+                      "completion_tokens": len(toks),
+                      "total_tokens": len(ids) + len(toks)},
+        })
+
+    @app.post("/v1/chat/completions")
+    def chat_completions(body: Dict):
+        messages = body.get("messages", [])
+        cfg = _gen_config(body)
+        ids: List[int] = []
+        for m in messages:
+            ids.extend(tokenizer.encode_message(m.get("role", "user"),
+                                                m.get("content", "")))
+        # open the assistant turn so decoding continues it
+        ids.extend(tokenizer.encode_message("assistant", "")[:-1])
+        if body.get("stream"):
+            return _sse_stream(ids, cfg, lambda piece: {
+                "object": "chat.completion.chunk",
+                "choices": [{"delta": {"content": piece}, "index": 0}]})
+        toks = _decode(ids, cfg)
+        return JSONResponse({
+            "object": "chat.completion",
+            "model": model_name,
+            "choices": [{"message": {"role": "assistant",
+                                     "content": tokenizer.decode(toks)},
+                         "index": 0, "finish_reason": "stop"}],
+            "usage": {"prompt_tokens": len(ids),
+                      "completion_tokens": len(toks),
+                      "total_tokens": len(ids) + len(toks)},
+        })
+
+    return app
+
+
+def main(argv=None):
+    import argparse
+
+    from ..data.tokenizer import ConversationTokenizer
+    from ..models import DeepSeekTransformer
+    from .loader import (find_latest_checkpoint, infer_config_from_state_dict,
+                         load_checkpoint_smart)
+
+    ap = argparse.ArgumentParser(description="LuminaAI-AMD inference server")
+    ap.add_argument("--checkpoint", default=None)
+    ap.add_argument("--host", default="127.0.0.1")
+    ap.add_argument("--port", type=int, default=8000)
+    ap.add_argument("--device", default=None)
+    ap.add_argument("--quantize", default=None,
+                    choices=["int8", "int4", "fp8"])
+    args = ap.parse_args(argv)
+
+    device = torch.device(args.device) if args.device else (
+        torch.device("cuda") if torch.cuda.is_available()
+        else torch.device("cpu"))
+    path = args.checkpoint or find_latest_checkpoint()
+    if path is None:
+        raise SystemExit("no checkpoint found; pass --checkpoint")
+    payload = load_checkpoint_smart(path)
+    sd = payload["model_state_dict"]
+    cfg = infer_config_from_state_dict(sd)
+    model = DeepSeekTransformer(cfg)
+    model.load_state_dict(sd, strict=False)
+    model = model.to(device).eval()
+    if device.type == "cuda":
+        model = model.to(torch.bfloat16)
+    if args.quantize:
+        from ..ops.quant import quantize_model
+        n = quantize_model(model, mode=args.quantize)
+        print(f"quantized {n} Linear layers to {args.quantize}")
+
+    app = create_app(model, ConversationTokenizer())
+    import uvicorn
+    uvicorn.run(app, host=args.host, port=args.port, log_level="info")
+
+
+if __name__ == "__main__":
+    main()
