@@ -308,7 +308,8 @@ class FlatAdamW:
     def load_state_dict(self, sd: Dict):
         self.step_count = sd.get("step_count", 0)
         if sd.get("shard_world", 1) != self.shard_world:
-            raise ValueError("FlatAdamW: resharding checkpoints not supported yet "
+            raise ValueError("FlatAdamW: world size changed — collect every "
+                             "saved rank's shard and call load_resharded() "
                              f"(saved world={sd.get('shard_world')}, "
                              f"current={self.shard_world})")
         for g, gs in zip(self.groups, sd["groups"]):
@@ -320,6 +321,43 @@ class FlatAdamW:
             g.v.copy_(gs["v"].to(g.v.device))
             g.lr = gs.get("lr", g.lr)
             g.weight_decay = gs.get("weight_decay", g.weight_decay)
+            if not g._master_is_params:
+                g.flat_w[g.shard_lo:g.shard_hi].copy_(g.master.to(g.dtype))
+
+    def load_resharded(self, shard_sds: List[Dict]):
+        """Elastic resume: rebuild this rank's optimizer state from the
+        per-rank shards of a run saved at a DIFFERENT world size. shard_sds
+        is the saved ranks' state dicts in rank order (rank 0's lives in the
+        main checkpoint payload, the rest in optim_shard_*_rank{r}.pt
+        siblings). Sharding is contiguous slicing of the flat buffer, so
+        concatenating the saved shards in rank order reconstructs the full
+        padded-at-world-A buffer; the first `numel` elements are re-padded
+        for world B and re-sliced."""
+        if any(g.comm == "expert" for g in self.groups):
+            raise ValueError("elastic resharding with expert parallelism is "
+                             "not supported (expert placement changes)")
+        saved_world = shard_sds[0].get("shard_world", 1)
+        if len(shard_sds) != saved_world:
+            raise ValueError(f"need all {saved_world} shards, got "
+                             f"{len(shard_sds)}")
+        self.step_count = shard_sds[0].get("step_count", 0)
+        for gi, g in enumerate(self.groups):
+            saved_numel = shard_sds[0]["groups"][gi]["numel"]
+            if saved_numel != g.numel:
+                raise ValueError(f"group {gi} size mismatch ({g.numel} vs "
+                                 f"{saved_numel})")
+            for key, dst in (("master", g.master), ("m", g.m), ("v", g.v)):
+                full = torch.cat([sd["groups"][gi][key].float()
+                                  for sd in shard_sds])
+                buf = torch.zeros(g.padded, dtype=torch.float32)
+                buf[:g.numel] = full[:g.numel]
+                if g._master_is_params:
+                    dst.copy_(buf.to(dst.device))
+                else:
+                    dst.copy_(buf[g.shard_lo:g.shard_hi].to(dst.device))
+            g.lr = shard_sds[0]["groups"][gi].get("lr", g.lr)
+            g.weight_decay = shard_sds[0]["groups"][gi].get(
+                "weight_decay", g.weight_decay)
             if not g._master_is_params:
                 g.flat_w[g.shard_lo:g.shard_hi].copy_(g.master.to(g.dtype))
 

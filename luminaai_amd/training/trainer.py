@@ -413,12 +413,48 @@ class Trainer:
                                            weights_only=False).get(
                         "optimizer_state_dict")
         if load_optimizer and opt_state:
-            self.optimizer.load_state_dict(opt_state)
+            cur_world = getattr(self.optimizer, "shard_world", 1)
+            if opt_state.get("shard_world", 1) != cur_world \
+                    and hasattr(self.optimizer, "load_resharded"):
+                # elastic resume: world size changed since the save — merge
+                # all saved per-rank shards and re-slice for this topology
+                shards = self._gather_saved_optim_shards(which, payload)
+                if shards is None:
+                    raise FileNotFoundError(
+                        "elastic resume: not all optimizer shard files of "
+                        f"the saved world={opt_state.get('shard_world')} run "
+                        "are present")
+                self.optimizer.load_resharded(shards)
+            else:
+                self.optimizer.load_state_dict(opt_state)
         if load_optimizer and payload.get("scheduler_state_dict") and self.scheduler:
             self.scheduler.load_state_dict(payload["scheduler_state_dict"])
         self.global_step = payload.get("global_step", 0)
         self.epoch = payload.get("epoch", 0)
         return payload
+
+    def _gather_saved_optim_shards(self, which, payload):
+        """All per-rank optimizer shards of a saved run, in rank order
+        (rank 0's state travels in the main payload, ranks >= 1 in
+        optim_shard_*_rank{r}.pt siblings)."""
+        opt0 = payload.get("optimizer_state_dict")
+        if not opt0:
+            return None
+        world = opt0.get("shard_world", 1)
+        shards = [opt0]
+        base = self.checkpoints.resolve(which)
+        if base is None:
+            return None if world > 1 else shards
+        for r in range(1, world):
+            p = os.path.join(
+                os.path.dirname(str(base)),
+                f"optim_shard_{os.path.basename(str(base))[:-3]}_rank{r}.pt")
+            if not os.path.exists(p):
+                return None
+            shards.append(torch.load(p, map_location="cpu",
+                                     weights_only=False)
+                          ["optimizer_state_dict"])
+        return shards
 
     # ================================================== metrics plumbing
     def set_metrics_hook(self, fn):
