@@ -139,6 +139,7 @@ class Config:
     backup_every_n_hours: int = 6
     max_retries: int = 3
     enable_wandb: bool = False
+    prometheus_port: Optional[int] = None   # rank-0 scrape endpoint
     wandb_project: Optional[str] = None
     wandb_entity: Optional[str] = None
 

@@ -52,22 +52,29 @@ def create_app(model, tokenizer, model_name: str = "luminaai-amd"):
     lock = threading.Lock()
     started = time.time()
     counters = {"requests": 0, "tokens_out": 0}
+    from ..monitoring.prometheus import make_server_metrics
+    prom = make_server_metrics()
 
     def _decode(prompt_ids: List[int], cfg: GenerationConfig,
-                cb=None) -> List[int]:
+                cb=None, endpoint: str = "completions") -> List[int]:
         with lock:
             counters["requests"] += 1
+            t0 = time.perf_counter()
             out = engine.generate(prompt_ids, cfg, stream_callback=cb)
             counters["tokens_out"] += len(out)
+            if prom is not None:
+                prom["requests"].labels(endpoint=endpoint).inc()
+                prom["tokens"].inc(len(out))
+                prom["latency"].observe(time.perf_counter() - t0)
             return out
 
     def _sse_stream(prompt_ids: List[int], cfg: GenerationConfig,
-                    wrap) -> StreamingResponse:
+                    wrap, endpoint: str = "completions") -> StreamingResponse:
         q: "queue.Queue[Optional[int]]" = queue.Queue()
 
         def run():
             try:
-                _decode(prompt_ids, cfg, cb=q.put)
+                _decode(prompt_ids, cfg, cb=q.put, endpoint=endpoint)
             finally:
                 q.put(None)
 
@@ -88,6 +95,16 @@ def create_app(model, tokenizer, model_name: str = "luminaai-amd"):
     def health():
         return {"status": "ok", "device": str(device),
                 "uptime_s": round(time.time() - started, 1), **counters}
+
+    @app.get("/metrics")
+    def metrics():
+        if prom is None:
+            return JSONResponse({"error": "prometheus_client not installed"},
+                                status_code=501)
+        from fastapi.responses import Response
+        from prometheus_client import generate_latest
+        return Response(generate_latest(prom["registry"]),
+                        media_type="text/plain; version=0.0.4")
 
     @app.get("/v1/models")
     def models():
@@ -127,8 +144,9 @@ def create_app(model, tokenizer, model_name: str = "luminaai-amd"):
         if body.get("stream"):
             return _sse_stream(ids, cfg, lambda piece: {
                 "object": "chat.completion.chunk",
-                "choices": [{"delta": {"content": piece}, "index": 0}]})
-        toks = _decode(ids, cfg)
+                "choices": [{"delta": {"content": piece}, "index": 0}]},
+                endpoint="chat")
+        toks = _decode(ids, cfg, endpoint="chat")
         return JSONResponse({
             "object": "chat.completion",
             "model": model_name,

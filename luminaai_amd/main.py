@@ -213,9 +213,19 @@ def main(argv: Optional[list] = None) -> dict:
 
     health = TrainingHealthMonitor(check_every=cfg.health_check_interval)
     wb = WandbLogger(cfg, enabled=cfg.enable_wandb and rank == 0)
+    prom = None
+    if cfg.prometheus_port and rank == 0:
+        try:
+            from .monitoring.prometheus import PrometheusExporter
+            prom = PrometheusExporter(port=cfg.prometheus_port)
+            logger.info(f"prometheus scrape endpoint on :{cfg.prometheus_port}")
+        except Exception as e:  # noqa: BLE001
+            logger.warning(f"prometheus exporter unavailable: {e}")
 
     def _hook(m):
         orch._enqueue_metrics(m)
+        if prom is not None:
+            prom(m)
         floats = {k: v for k, v in m.as_dict().items()
                   if isinstance(v, (int, float))}
         health.log_step(floats, m.step)
