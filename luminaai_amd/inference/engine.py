@@ -118,6 +118,75 @@ class GenerationEngine:
         self.stats["time_in_generate"] += time.perf_counter() - t0
         return generated
 
+    @torch.no_grad()
+    def generate_batch(self, prompts: List[List[int]],
+                       configs=None) -> List[List[int]]:
+        """Batched decoding over left-padded prompts with a shared KV cache.
+        configs: one GenerationConfig for all rows, or a list (per-request
+        sampling params — the serving batcher mixes requests freely). RoPE
+        attention scores depend only on relative positions, so the uniform
+        per-row pad shift is exact; pad slots are masked via the cache-wide
+        attention mask (transformer.forward, width pos_offset+S)."""
+        B = len(prompts)
+        if configs is None:
+            configs = GenerationConfig()
+        cfgs = configs if isinstance(configs, list) else [configs] * B
+        assert len(cfgs) == B
+        t0 = time.perf_counter()
+        self.model.eval()
+        max_ctx = max(c.max_context for c in cfgs)
+        prompts = [p[-max_ctx:] for p in prompts]
+        Smax = max(len(p) for p in prompts)
+        max_len = min(max_ctx, Smax + max(c.max_new_tokens for c in cfgs))
+        pad = self.tokenizer.pad_token_id
+        ids = torch.full((B, Smax), pad, dtype=torch.long)
+        mask = torch.zeros(B, Smax, dtype=torch.long)
+        for i, p in enumerate(prompts):
+            ids[i, Smax - len(p):] = torch.tensor(p, dtype=torch.long)
+            mask[i, Smax - len(p):] = 1
+        ids = ids.to(self.device)
+        mask = mask.to(self.device)
+        caches = self.model.make_kv_caches(max_len=max_len)
+        logits, _, _ = self.model(ids, attention_mask=mask, kv_caches=caches)
+        last = logits[:, -1].float()
+
+        stops = [set(c.stop_token_ids) | {self.tokenizer.eos_token_id}
+                 for c in cfgs]
+        outs: List[List[int]] = [[] for _ in range(B)]
+        recent: List[List[int]] = [list(p) for p in prompts]
+        finished = [False] * B
+        for _ in range(max(c.max_new_tokens for c in cfgs)):
+            next_ids = []
+            for b in range(B):
+                if finished[b]:
+                    next_ids.append(pad)
+                    continue
+                nid = self._sample(last[b], cfgs[b],
+                                   recent[b][-cfgs[b].rep_window:])
+                if nid in stops[b] or len(outs[b]) >= cfgs[b].max_new_tokens:
+                    finished[b] = True
+                    next_ids.append(pad)
+                    continue
+                outs[b].append(nid)
+                recent[b].append(nid)
+                next_ids.append(nid)
+                if len(outs[b]) >= cfgs[b].max_new_tokens:
+                    finished[b] = True
+            if all(finished) or caches[0].seq_len >= max_len:
+                break
+            step = torch.tensor(next_ids, dtype=torch.long,
+                                device=self.device).view(B, 1)
+            col = torch.tensor([0 if finished[b] else 1 for b in range(B)],
+                               dtype=torch.long, device=self.device).view(B, 1)
+            mask = torch.cat([mask, col], dim=1)
+            logits, _, _ = self.model(step, attention_mask=mask,
+                                      kv_caches=caches)
+            last = logits[:, -1].float()
+        self.stats["prefill_tokens"] += sum(len(p) for p in prompts)
+        self.stats["tokens_generated"] += sum(len(o) for o in outs)
+        self.stats["time_in_generate"] += time.perf_counter() - t0
+        return outs
+
     def generate_text(self, prompt: str, config: Optional[GenerationConfig] = None,
                       stream: bool = False) -> str:
         ids = self.tokenizer.encode(prompt)

@@ -771,11 +771,18 @@ class DeepSeekTransformer(nn.Module):
         attn_mask = None
         if attention_mask is not None and attention_mask.dim() == 2 \
                 and not bool(attention_mask.all()):
-            # padding mask [B, S] -> additive SDPA mask [B, 1, S, S] + causal
-            causal = torch.ones(S, S, dtype=torch.bool,
-                                device=x.device).tril()
-            keep = attention_mask.bool().view(B, 1, 1, S) & causal
-            attn_mask = torch.zeros(B, 1, S, S, device=x.device,
+            # padding mask -> additive SDPA mask + causal. Width S masks the
+            # current block ([B,1,S,S]); width pos_offset+S also covers the
+            # KV-cache slots ([B,1,S,L] — left-padded batched decode).
+            L = attention_mask.shape[1]
+            if L not in (S, pos_offset + S):
+                raise ValueError(f"attention_mask width {L} matches neither "
+                                 f"S={S} nor cache+S={pos_offset + S}")
+            off = pos_offset if L == pos_offset + S else 0
+            kpos = torch.arange(L, device=x.device).view(1, 1, 1, L)
+            qpos = (off + torch.arange(S, device=x.device)).view(1, 1, S, 1)
+            keep = attention_mask.bool().view(B, 1, 1, L) & (kpos <= qpos)
+            attn_mask = torch.zeros(B, 1, S, L, device=x.device,
                                     dtype=x.dtype).masked_fill(~keep, -1e4)
 
         total_aux = x.new_zeros(())
