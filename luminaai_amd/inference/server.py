@@ -41,7 +41,65 @@ def _gen_config(body: Dict) -> GenerationConfig:
     return cfg
 
 
-def create_app(model, tokenizer, model_name: str = "luminaai-amd"):
+class _DynamicBatcher:
+    """Dynamic request batching: non-streaming requests queue up and are
+    decoded together through GenerationEngine.generate_batch (left-padded
+    shared KV cache, per-row sampling configs). Requests that arrive while
+    a batch is decoding form the next batch — throughput scales with
+    concurrency without a latency penalty for the lone request."""
+
+    def __init__(self, engine: GenerationEngine, max_batch: int = 8,
+                 window_ms: float = 2.0, lock: Optional[threading.Lock] = None):
+        self.engine = engine
+        self.lock = lock or threading.Lock()
+        self.max_batch = max_batch
+        self.window_s = window_ms / 1000.0
+        self.q: "queue.Queue" = queue.Queue()
+        self.batches_run = 0
+        self.rows_seen = 0
+        self._thread = threading.Thread(target=self._loop, daemon=True)
+        self._thread.start()
+
+    def submit(self, prompt_ids: List[int],
+               cfg: GenerationConfig) -> List[int]:
+        done = threading.Event()
+        slot: Dict = {"ids": prompt_ids, "cfg": cfg, "done": done}
+        self.q.put(slot)
+        done.wait()
+        if "error" in slot:
+            raise slot["error"]
+        return slot["out"]
+
+    def _loop(self):
+        while True:
+            batch = [self.q.get()]
+            deadline = time.monotonic() + self.window_s
+            while len(batch) < self.max_batch:
+                remaining = deadline - time.monotonic()
+                if remaining <= 0:
+                    break
+                try:
+                    batch.append(self.q.get(timeout=remaining))
+                except queue.Empty:
+                    break
+            try:
+                with self.lock:
+                    outs = self.engine.generate_batch(
+                        [r["ids"] for r in batch], [r["cfg"] for r in batch])
+                for r, out in zip(batch, outs):
+                    r["out"] = out
+            except Exception as e:  # noqa: BLE001
+                for r in batch:
+                    r["error"] = e
+            finally:
+                self.batches_run += 1
+                self.rows_seen += len(batch)
+                for r in batch:
+                    r["done"].set()
+
+
+def create_app(model, tokenizer, model_name: str = "luminaai-amd",
+               max_batch: int = 8):
     """Build the ASGI app around an already-loaded model."""
     from fastapi import FastAPI
     from fastapi.responses import JSONResponse, StreamingResponse
@@ -50,23 +108,33 @@ def create_app(model, tokenizer, model_name: str = "luminaai-amd"):
     device = next(model.parameters()).device
     engine = GenerationEngine(model, tokenizer, device)
     lock = threading.Lock()
+    batcher = _DynamicBatcher(engine, max_batch=max_batch, lock=lock)
     started = time.time()
     counters = {"requests": 0, "tokens_out": 0}
     from ..monitoring.prometheus import make_server_metrics
     prom = make_server_metrics()
 
+    def _account(endpoint: str, n_out: int, t0: float):
+        counters["requests"] += 1
+        counters["tokens_out"] += n_out
+        if prom is not None:
+            prom["requests"].labels(endpoint=endpoint).inc()
+            prom["tokens"].inc(n_out)
+            prom["latency"].observe(time.perf_counter() - t0)
+
     def _decode(prompt_ids: List[int], cfg: GenerationConfig,
                 cb=None, endpoint: str = "completions") -> List[int]:
-        with lock:
-            counters["requests"] += 1
-            t0 = time.perf_counter()
-            out = engine.generate(prompt_ids, cfg, stream_callback=cb)
-            counters["tokens_out"] += len(out)
-            if prom is not None:
-                prom["requests"].labels(endpoint=endpoint).inc()
-                prom["tokens"].inc(len(out))
-                prom["latency"].observe(time.perf_counter() - t0)
-            return out
+        t0 = time.perf_counter()
+        if cb is None:
+            # non-streaming: ride the dynamic batcher (concurrent requests
+            # share one forward pass)
+            out = batcher.submit(prompt_ids, cfg)
+        else:
+            # streaming: dedicated single-sequence decode (hipGraph path)
+            with lock:
+                out = engine.generate(prompt_ids, cfg, stream_callback=cb)
+        _account(endpoint, len(out), t0)
+        return out
 
     def _sse_stream(prompt_ids: List[int], cfg: GenerationConfig,
                     wrap, endpoint: str = "completions") -> StreamingResponse:
@@ -94,7 +162,9 @@ def create_app(model, tokenizer, model_name: str = "luminaai-amd"):
     @app.get("/health")
     def health():
         return {"status": "ok", "device": str(device),
-                "uptime_s": round(time.time() - started, 1), **counters}
+                "uptime_s": round(time.time() - started, 1),
+                "batches_run": batcher.batches_run,
+                "batched_rows": batcher.rows_seen, **counters}
 
     @app.get("/metrics")
     def metrics():
