@@ -119,6 +119,76 @@ class GenerationEngine:
         return generated
 
     @torch.no_grad()
+    def generate_speculative(self, prompt_ids: List[int], draft_model,
+                             config: Optional[GenerationConfig] = None,
+                             draft_k: int = 4) -> List[int]:
+        """Greedy speculative decoding: a cheap draft model proposes
+        draft_k tokens, the target verifies them in ONE forward (sequence
+        of k+1 tokens — MFMA-shaped work instead of k GEMV chains). Output
+        is EXACTLY the target's greedy decode for any draft; acceptance is
+        argmax equality. Rejections rewind the KV caches via
+        KVCache.truncate. Greedy only (temperature must be 0)."""
+        cfg = config or GenerationConfig(temperature=0.0)
+        if cfg.temperature != 0.0:
+            raise ValueError("speculative decoding is greedy-only here")
+        self.model.eval()
+        draft_model.eval()
+        stop = set(cfg.stop_token_ids) | {self.tokenizer.eos_token_id}
+        ids = prompt_ids[-cfg.max_context:]
+        cap = len(ids) + cfg.max_new_tokens + draft_k + 2
+        ct = self.model.make_kv_caches(max_len=cap)
+        cd = draft_model.make_kv_caches(max_len=cap)
+        x = torch.tensor([ids], dtype=torch.long, device=self.device)
+        lt, _, _ = self.model(x, kv_caches=ct)
+        draft_model(x, kv_caches=cd)
+        nxt = int(lt[0, -1].argmax())
+
+        generated: List[int] = []
+        self.stats["prefill_tokens"] += len(ids)
+        t0 = time.perf_counter()
+        while len(generated) < cfg.max_new_tokens and nxt not in stop:
+            generated.append(nxt)
+            if len(generated) >= cfg.max_new_tokens:
+                break
+            # draft first consumes whatever confirmed history it is missing
+            # (>= 1 token: at least the newest `nxt`), then proposes k
+            confirmed = ids + generated
+            missing = confirmed[cd[0].seq_len:]
+            ld, _, _ = draft_model(
+                torch.tensor([missing], dtype=torch.long,
+                             device=self.device), kv_caches=cd)
+            cur = int(ld[0, -1].argmax())
+            proposal: List[int] = [cur]
+            for _ in range(draft_k - 1):
+                ld, _, _ = draft_model(
+                    torch.tensor([[cur]], dtype=torch.long,
+                                 device=self.device), kv_caches=cd)
+                cur = int(ld[0, -1].argmax())
+                proposal.append(cur)
+            # target verifies the whole block in one forward
+            block = torch.tensor([[nxt] + proposal], dtype=torch.long,
+                                 device=self.device)
+            lt, _, _ = self.model(block, kv_caches=ct)
+            preds = lt[0].argmax(-1).tolist()     # len k+1
+            j = 0
+            while j < len(proposal) and preds[j] == proposal[j] \
+                    and proposal[j] not in stop \
+                    and len(generated) + j + 1 < cfg.max_new_tokens:
+                j += 1
+            generated.extend(proposal[:j])
+            nxt = preds[j]
+            # rewind to the accepted prefix (truncate is a no-op when the
+            # cache holds fewer tokens, e.g. the draft after a full accept)
+            conf = len(ids) + len(generated)
+            for c in ct:
+                c.truncate(conf)
+            for c in cd:
+                c.truncate(conf)
+        self.stats["tokens_generated"] += len(generated)
+        self.stats["time_in_generate"] += time.perf_counter() - t0
+        return generated
+
+    @torch.no_grad()
     def generate_batch(self, prompts: List[List[int]],
                        configs=None) -> List[List[int]]:
         """Batched decoding over left-padded prompts with a shared KV cache.

@@ -258,6 +258,22 @@ class KVCache:
         self._len = self.k.shape[1]
         return self.k, self.v
 
+    def truncate(self, n: int):
+        """Rewind the cache to n tokens (speculative-decode rejection).
+        Preallocated buffers just move the write cursor; growing caches
+        slice."""
+        if n >= self._len:
+            return
+        if self.max_len > 0:
+            self._len = n
+            if self.graph_mode and self.pos_dev is not None:
+                self.pos_dev.fill_(n)
+        else:
+            if self.k is not None:
+                self.k = self.k[:, :n].contiguous()
+                self.v = self.v[:, :n].contiguous()
+            self._len = n
+
 
 class GroupedQueryAttention(nn.Module):
     """GQA with fused QKV projection, HIP RoPE, SDPA core

@@ -354,3 +354,41 @@ def test_generate_batch_moe_model(tiny_moe_config, tokenizer):
                                                temperature=0.0))
     assert len(outs) == 2
     assert all(isinstance(t, int) for o in outs for t in o)
+
+
+@pytest.mark.parametrize("draft_k", [1, 3, 5])
+def test_speculative_matches_greedy(small_model, tokenizer, draft_k):
+    """Speculative output must equal plain target greedy decode for ANY
+    draft — here a different random model of the same vocab."""
+    from luminaai_amd.models.transformer import DeepSeekConfig
+    torch.manual_seed(9)
+    draft = DeepSeekTransformer(DeepSeekConfig(
+        vocab_size=512, hidden_size=32, num_layers=1, num_heads=2,
+        num_kv_heads=1, intermediate_size=64, seq_length=64, use_moe=False,
+        use_mod=False, tie_word_embeddings=False)).eval()
+    eng = GenerationEngine(small_model.eval(), tokenizer)
+    cfg = GenerationConfig(max_new_tokens=8, temperature=0.0,
+                           stop_token_ids=[-1])
+    for prompt in ("hello there", "x"):
+        p = tokenizer.encode(prompt)
+        ref = eng.generate(p, cfg)
+        spec = eng.generate_speculative(p, draft, cfg, draft_k=draft_k)
+        assert spec == ref, (draft_k, prompt)
+
+
+def test_speculative_self_draft_accepts_everything(small_model, tokenizer):
+    """Target drafting for itself accepts every proposal (sanity on the
+    acceptance bookkeeping)."""
+    eng = GenerationEngine(small_model.eval(), tokenizer)
+    cfg = GenerationConfig(max_new_tokens=6, temperature=0.0,
+                           stop_token_ids=[-1])
+    p = tokenizer.encode("self draft")
+    assert eng.generate_speculative(p, small_model, cfg, draft_k=3) == \
+        eng.generate(p, cfg)
+
+
+def test_speculative_rejects_sampling(small_model, tokenizer):
+    eng = GenerationEngine(small_model.eval(), tokenizer)
+    with pytest.raises(ValueError):
+        eng.generate_speculative([1, 2], small_model,
+                                 GenerationConfig(temperature=0.8))
