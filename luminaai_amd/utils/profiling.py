@@ -21,20 +21,37 @@ _lock = threading.Lock()
 _stats: Dict[str, Dict] = defaultdict(
     lambda: {"calls": 0, "total_s": 0.0, "max_s": 0.0})
 _ENABLED = False
-_pending = []          # (name, start_event, end_event)
+_pending = []          # (name, start_event, end_event, tid)
+_TRACE = False
+_trace_events = []     # chrome-trace slices: (name, ts_us, dur_us, tid)
+_TRACE_CAP = 200_000
+_epoch_cpu = 0.0
+_epoch_ev = None       # HIP event anchoring device timestamps
 
 
-def enable_profiling(on: bool = True):
-    global _ENABLED
+def enable_profiling(on: bool = True, trace: bool = False):
+    """trace=True additionally collects per-call slices for
+    export_chrome_trace (bounded to 200k events)."""
+    global _ENABLED, _TRACE, _epoch_cpu, _epoch_ev
     _ENABLED = on
+    _TRACE = on and trace
+    if _TRACE:
+        _epoch_cpu = time.perf_counter()
+        if torch.cuda.is_available():
+            _epoch_ev = torch.cuda.Event(enable_timing=True)
+            _epoch_ev.record()
 
 
-def _record(name: str, dt: float):
+def _record(name: str, dt: float, ts_s: float = None):
     with _lock:
         s = _stats[name]
         s["calls"] += 1
         s["total_s"] += dt
         s["max_s"] = max(s["max_s"], dt)
+        if _TRACE and ts_s is not None and len(_trace_events) < _TRACE_CAP:
+            _trace_events.append(
+                (name, (ts_s - _epoch_cpu) * 1e6, dt * 1e6,
+                 threading.get_ident()))
 
 
 def profile_function(name: str = None):
@@ -57,7 +74,7 @@ def profile_function(name: str = None):
                 return out
             t0 = time.perf_counter()
             out = fn(*a, **kw)
-            _record(label, time.perf_counter() - t0)
+            _record(label, time.perf_counter() - t0, ts_s=t0)
             return out
         return wrapper
     return deco
@@ -79,7 +96,7 @@ def profiling_context(name: str):
     else:
         t0 = time.perf_counter()
         yield
-        _record(name, time.perf_counter() - t0)
+        _record(name, time.perf_counter() - t0, ts_s=t0)
 
 
 def _drain_pending():
@@ -89,7 +106,14 @@ def _drain_pending():
     if pending:
         torch.cuda.synchronize()
         for name, ev0, ev1 in pending:
-            _record(name, ev0.elapsed_time(ev1) / 1000.0)
+            dt = ev0.elapsed_time(ev1) / 1000.0
+            _record(name, dt)
+            if _TRACE and _epoch_ev is not None \
+                    and len(_trace_events) < _TRACE_CAP:
+                with _lock:
+                    _trace_events.append(
+                        (name, _epoch_ev.elapsed_time(ev0) * 1e3, dt * 1e6,
+                         0))
 
 
 def get_profiling_stats() -> Dict[str, Dict]:
@@ -104,3 +128,21 @@ def reset_profiling_stats():
     with _lock:
         _stats.clear()
         _pending.clear()
+        _trace_events.clear()
+
+
+def export_chrome_trace(path: str) -> int:
+    """Write collected slices as a chrome://tracing / Perfetto JSON file
+    (ROADMAP: trace export for the decorator API). Requires
+    enable_profiling(True, trace=True). Returns the event count."""
+    import json
+    if torch.cuda.is_available():
+        _drain_pending()
+    with _lock:
+        events = [{"name": n, "ph": "X", "ts": ts, "dur": dur,
+                   "pid": 0, "tid": tid, "cat": "lumina"}
+                  for n, ts, dur, tid in _trace_events]
+    with open(path, "w") as f:
+        json.dump({"traceEvents": events,
+                   "displayTimeUnit": "ms"}, f)
+    return len(events)

@@ -158,3 +158,34 @@ def test_secure_chat_wrapper(small_model, tokenizer):
     assert sec.respond("<script>x</script>").startswith("[input rejected")
     sec.respond("ok")
     assert sec.respond("third").startswith("[rate limited")
+
+
+def test_chrome_trace_export(tmp_path):
+    import json
+    from luminaai_amd.utils.profiling import (enable_profiling,
+                                              export_chrome_trace,
+                                              profile_function,
+                                              profiling_context,
+                                              reset_profiling_stats)
+    reset_profiling_stats()
+    enable_profiling(True, trace=True)
+    try:
+        @profile_function("traced_fn")
+        def f():
+            return sum(range(1000))
+
+        for _ in range(3):
+            f()
+        with profiling_context("traced_ctx"):
+            f()
+        p = str(tmp_path / "trace.json")
+        n = export_chrome_trace(p)
+        assert n >= 5
+        data = json.loads(open(p).read())
+        names = {e["name"] for e in data["traceEvents"]}
+        assert {"traced_fn", "traced_ctx"} <= names
+        for e in data["traceEvents"]:
+            assert e["ph"] == "X" and e["dur"] >= 0 and e["ts"] >= 0
+    finally:
+        enable_profiling(False)
+        reset_profiling_stats()
