@@ -60,6 +60,12 @@ def build_arg_parser() -> argparse.ArgumentParser:
     ap.add_argument("--zero", type=int, default=None, choices=(0, 1, 2, 3))
     ap.add_argument("--ep", type=int, default=None,
                     help="expert-parallel degree")
+    ap.add_argument("--pp", type=int, default=1,
+                    help="pipeline-parallel stages (= world size; dedicated "
+                         "1F1B loop, see training/pipeline_loop.py)")
+    ap.add_argument("--pp-virtual", type=int, default=1,
+                    help="virtual stage chunks per PP rank (interleaved "
+                         "schedule)")
     ap.add_argument("--tp", type=int, default=1,
                     help="tensor-parallel degree (dense blocks)")
     ap.add_argument("--sp", type=int, default=1,
@@ -167,6 +173,20 @@ def main(argv: Optional[list] = None) -> dict:
                        "backend is Triton); the native path is HIP kernels "
                        "+ hipGraphs")
     model = DeepSeekTransformer(config_to_deepseek_config(cfg))
+    if args.pp > 1:
+        if args.pp != world:
+            raise SystemExit(f"--pp {args.pp} requires world size {args.pp} "
+                             f"(got {world}); PP x DP composition is not "
+                             "supported yet")
+        from .training.pipeline_loop import run_pipeline_training
+        result = run_pipeline_training(
+            model, cfg, train_ds, logger,
+            virtual_stages=max(1, args.pp_virtual),
+            steps=args.synthetic_steps)
+        if rank == world - 1:
+            logger.info(f"pipeline training done: step "
+                        f"{result['global_step']} loss {result['loss']:.4f}")
+        return result
     if mesh.tp_size > 1:
         from .parallel.tensor_parallel import convert_to_tensor_parallel
         n_tp = convert_to_tensor_parallel(model, mesh)
