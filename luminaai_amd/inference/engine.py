@@ -126,11 +126,13 @@ class GenerationEngine:
         draft_k tokens, the target verifies them in ONE forward (sequence
         of k+1 tokens — MFMA-shaped work instead of k GEMV chains). Output
         is EXACTLY the target's greedy decode for any draft; acceptance is
-        argmax equality. Rejections rewind the KV caches via
-        KVCache.truncate. Greedy only (temperature must be 0)."""
+        argmax equality. With temperature > 0 the classic acceptance-
+        rejection scheme runs instead (accept x ~ q with prob
+        min(1, p(x)/q(x)); on reject resample from norm(max(p-q, 0))),
+        which preserves the target's sampling distribution exactly.
+        Rejections rewind the KV caches via KVCache.truncate."""
         cfg = config or GenerationConfig(temperature=0.0)
-        if cfg.temperature != 0.0:
-            raise ValueError("speculative decoding is greedy-only here")
+        greedy = cfg.temperature == 0.0
         self.model.eval()
         draft_model.eval()
         stop = set(cfg.stop_token_ids) | {self.tokenizer.eos_token_id}
@@ -141,7 +143,15 @@ class GenerationEngine:
         x = torch.tensor([ids], dtype=torch.long, device=self.device)
         lt, _, _ = self.model(x, kv_caches=ct)
         draft_model(x, kv_caches=cd)
-        nxt = int(lt[0, -1].argmax())
+        if greedy:
+            nxt = int(lt[0, -1].argmax())
+        else:
+            nxt = int(self._filtered_probs(
+                lt[0, -1].float(), cfg,
+                ids[-cfg.rep_window:]).multinomial(1))
+        if not greedy:
+            return self._speculative_sampling_loop(
+                draft_model, cfg, draft_k, ids, ct, cd, nxt, stop)
 
         generated: List[int] = []
         self.stats["prefill_tokens"] += len(ids)
@@ -179,6 +189,74 @@ class GenerationEngine:
             nxt = preds[j]
             # rewind to the accepted prefix (truncate is a no-op when the
             # cache holds fewer tokens, e.g. the draft after a full accept)
+            conf = len(ids) + len(generated)
+            for c in ct:
+                c.truncate(conf)
+            for c in cd:
+                c.truncate(conf)
+        self.stats["tokens_generated"] += len(generated)
+        self.stats["time_in_generate"] += time.perf_counter() - t0
+        return generated
+
+    def _speculative_sampling_loop(self, draft_model, cfg, draft_k, ids,
+                                   ct, cd, nxt, stop) -> List[int]:
+        """Acceptance-rejection speculative decoding (temperature > 0)."""
+        generated: List[int] = []
+        t0 = time.perf_counter()
+        self.stats["prefill_tokens"] += len(ids)
+        W = cfg.rep_window
+        while len(generated) < cfg.max_new_tokens and nxt not in stop:
+            generated.append(nxt)
+            if len(generated) >= cfg.max_new_tokens:
+                break
+            confirmed = ids + generated
+            missing = confirmed[cd[0].seq_len:]
+            ld, _, _ = draft_model(
+                torch.tensor([missing], dtype=torch.long,
+                             device=self.device), kv_caches=cd)
+            proposal: List[int] = []
+            qs: List[torch.Tensor] = []
+            for step in range(draft_k):
+                q = self._filtered_probs(ld[0, -1].float(), cfg,
+                                         (confirmed + proposal)[-W:])
+                tok = int(q.multinomial(1))
+                proposal.append(tok)
+                qs.append(q)
+                if step < draft_k - 1:
+                    ld, _, _ = draft_model(
+                        torch.tensor([[tok]], dtype=torch.long,
+                                     device=self.device), kv_caches=cd)
+            block = torch.tensor([[nxt] + proposal], dtype=torch.long,
+                                 device=self.device)
+            lt, _, _ = self.model(block, kv_caches=ct)
+            j = 0
+            nxt = None
+            while j < len(proposal):
+                if len(generated) + j + 1 >= cfg.max_new_tokens:
+                    break
+                p = self._filtered_probs(lt[0, j].float(), cfg,
+                                         (confirmed + proposal[:j])[-W:])
+                xj = proposal[j]
+                ratio = float(p[xj]) / max(float(qs[j][xj]), 1e-20)
+                if float(torch.rand(())) < ratio:
+                    if xj in stop:          # accepted stop token ends it
+                        nxt = xj
+                        break
+                    j += 1
+                    continue
+                resid = (p - qs[j]).clamp_min(0)
+                tot = float(resid.sum())
+                nxt = int((resid / tot).multinomial(1)) if tot > 0 \
+                    else int(p.multinomial(1))
+                break
+            generated.extend(proposal[:j])
+            if nxt is None:
+                # no rejection token chosen: sample the next token from the
+                # target's distribution at the accepted position (j ==
+                # len(proposal) is the all-accepted bonus token)
+                p = self._filtered_probs(lt[0, j].float(), cfg,
+                                         (confirmed + proposal[:j])[-W:])
+                nxt = int(p.multinomial(1))
             conf = len(ids) + len(generated)
             for c in ct:
                 c.truncate(conf)
@@ -296,6 +374,31 @@ class GenerationEngine:
                 0, sorted_idx, sorted_logits)
         probs = torch.softmax(logits, dim=-1)
         return int(torch.multinomial(probs, 1))
+
+    def _filtered_probs(self, logits: torch.Tensor, cfg: GenerationConfig,
+                        recent: List[int]) -> torch.Tensor:
+        """The sampling distribution _sample draws from (rep penalty +
+        temperature + top-k + top-p), as an explicit probability vector —
+        needed by speculative acceptance sampling. temperature > 0 only."""
+        logits = logits.clone()
+        if cfg.repetition_penalty != 1.0 and recent:
+            idx = torch.tensor(sorted(set(recent)), device=logits.device)
+            sel = logits[idx]
+            logits[idx] = torch.where(sel > 0, sel / cfg.repetition_penalty,
+                                      sel * cfg.repetition_penalty)
+        logits = logits / cfg.temperature
+        if cfg.top_k and cfg.top_k > 0:
+            kth = torch.topk(logits, min(cfg.top_k, logits.numel())).values[-1]
+            logits = logits.masked_fill(logits < kth, float("-inf"))
+        if cfg.top_p < 1.0:
+            sorted_logits, sorted_idx = torch.sort(logits, descending=True)
+            probs = torch.softmax(sorted_logits, dim=-1)
+            cum = torch.cumsum(probs, dim=-1)
+            cut = cum - probs > cfg.top_p
+            sorted_logits = sorted_logits.masked_fill(cut, float("-inf"))
+            logits = torch.full_like(logits, float("-inf")).scatter(
+                0, sorted_idx, sorted_logits)
+        return torch.softmax(logits, dim=-1)
 
     def get_stats(self) -> Dict:
         t = self.stats["time_in_generate"]

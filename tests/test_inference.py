@@ -387,8 +387,82 @@ def test_speculative_self_draft_accepts_everything(small_model, tokenizer):
         eng.generate(p, cfg)
 
 
-def test_speculative_rejects_sampling(small_model, tokenizer):
+def test_speculative_sampling_runs_on_real_model(small_model, tokenizer):
+    """Sampling-mode speculative decode on a real transformer pair."""
+    from luminaai_amd.models.transformer import DeepSeekConfig
+    torch.manual_seed(4)
+    draft = DeepSeekTransformer(DeepSeekConfig(
+        vocab_size=512, hidden_size=32, num_layers=1, num_heads=2,
+        num_kv_heads=1, intermediate_size=64, seq_length=64, use_moe=False,
+        use_mod=False, tie_word_embeddings=False)).eval()
     eng = GenerationEngine(small_model.eval(), tokenizer)
-    with pytest.raises(ValueError):
-        eng.generate_speculative([1, 2], small_model,
-                                 GenerationConfig(temperature=0.8))
+    cfg = GenerationConfig(max_new_tokens=6, temperature=0.9, top_k=50,
+                           top_p=0.95, stop_token_ids=[-1])
+    out = eng.generate_speculative(tokenizer.encode("abc"), draft, cfg,
+                                   draft_k=3)
+    assert 0 < len(out) <= 6
+    assert all(0 <= t < 512 for t in out)
+
+
+def test_speculative_sampling_preserves_distribution(tokenizer):
+    """Acceptance-rejection speculative sampling must reproduce the
+    target's sampling distribution for any draft. Fixed-logit stub models
+    make the target distribution analytic; compare empirically."""
+    import torch.nn as nn
+
+    V = 8
+
+    class _StubCache:
+        def __init__(self):
+            self._len = 0
+
+        @property
+        def seq_len(self):
+            return self._len
+
+        def truncate(self, n):
+            self._len = min(self._len, n)
+
+    class Stub(nn.Module):
+        """Position-independent logits; KV cache only tracks length."""
+
+        def __init__(self, logits):
+            super().__init__()
+            self.logits = logits
+            self.dummy = nn.Parameter(torch.zeros(1))
+
+        def make_kv_caches(self, max_len=0):
+            self._caches = [_StubCache()]
+            return self._caches
+
+        def forward(self, ids, kv_caches=None, **kw):
+            B, S = ids.shape
+            if kv_caches is not None:
+                kv_caches[0]._len += S
+            out = self.logits.view(1, 1, V).expand(B, S, V)
+            return out, None, None
+
+    torch.manual_seed(0)
+    t_logits = torch.tensor([2.0, 1.5, 1.0, 0.5, 0.0, -0.5, -1.0, -8.0])
+    d_logits = torch.tensor([0.0, 0.0, 2.0, 2.0, 0.0, 1.0, -1.0, -8.0])
+    target, draft = Stub(t_logits), Stub(d_logits)
+
+    from luminaai_amd.inference.engine import (GenerationConfig,
+                                               GenerationEngine)
+    cfg = GenerationConfig(max_new_tokens=3, temperature=1.0, top_k=0,
+                           top_p=1.0, repetition_penalty=1.0,
+                           stop_token_ids=[-1])
+    eng = GenerationEngine(target, tokenizer, torch.device("cpu"))
+    # expected = plain softmax of target logits (no filters active);
+    # token V-1 has ~0 mass and the tokenizer eos (id 256+) is out of range
+    expected = torch.softmax(t_logits, -1)
+    counts = torch.zeros(V)
+    n_runs = 400
+    torch.manual_seed(1234)
+    for _ in range(n_runs):
+        toks = eng.generate_speculative([1, 2], draft, cfg, draft_k=2)
+        for t in toks:
+            counts[t] += 1
+    emp = counts / counts.sum()
+    tv = 0.5 * float((emp - expected).abs().sum())
+    assert tv < 0.06, (tv, emp.tolist(), expected.tolist())
