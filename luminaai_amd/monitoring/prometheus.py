@@ -60,8 +60,12 @@ class PrometheusExporter:
         try:
             if getattr(m, "loss", None) is not None:
                 self.loss.set(float(m.loss))
-            if getattr(m, "learning_rate", None) is not None:
-                self.lr.set(float(m.learning_rate))
+            # TrainingMetrics uses `lr` / `memory_gb`; accept both spellings
+            lr = getattr(m, "lr", None)
+            if lr is None:
+                lr = getattr(m, "learning_rate", None)
+            if lr is not None:
+                self.lr.set(float(lr))
             if getattr(m, "grad_norm", None) is not None:
                 self.grad_norm.set(float(m.grad_norm))
             if getattr(m, "tokens_per_sec", None):
@@ -72,9 +76,15 @@ class PrometheusExporter:
             if getattr(m, "epoch", None) is not None:
                 self.epoch.set(int(m.epoch))
             imb = getattr(m, "expert_imbalance", None)
+            if imb is None:
+                es = getattr(m, "expert_stats", None)
+                if isinstance(es, dict):
+                    imb = es.get("imbalance")
             if imb is not None:
                 self.expert_imbalance.set(float(imb))
-            mem = getattr(m, "memory_allocated_gb", None)
+            mem = getattr(m, "memory_gb", None)
+            if mem is None:
+                mem = getattr(m, "memory_allocated_gb", None)
             if mem is not None:
                 self.memory_gb.set(float(mem))
         finally:

@@ -133,3 +133,19 @@ def test_batched_result_matches_direct(client, small_model, tokenizer):
                                              "max_tokens": 4,
                                              "temperature": 0.0}).json()
     assert r["choices"][0]["text"] == direct
+
+
+def test_prometheus_exporter_with_real_metrics(small_model, tokenizer):
+    """Exporter must map the actual TrainingMetrics field names."""
+    from luminaai_amd.monitoring.prometheus import PrometheusExporter
+    from luminaai_amd.training.trainer import TrainingMetrics
+    m = TrainingMetrics(step=7, epoch=2, loss=3.1, aux_loss=0.01,
+                        grad_norm=0.5, lr=2e-4, tokens_per_sec=123.0,
+                        accuracy=0.4, perplexity=22.0, memory_gb=1.5,
+                        expert_stats={"imbalance": 1.3}, timestamp=0.0)
+    exp = PrometheusExporter(port=None)
+    exp(m)
+    out = exp.scrape().decode()
+    assert "lumina_learning_rate 0.0002" in out
+    assert "lumina_gpu_memory_gb 1.5" in out
+    assert "lumina_expert_imbalance 1.3" in out
