@@ -34,6 +34,7 @@ import torch
 import torch.distributed as dist
 
 from . import comm
+from ..utils.profiling import profiling_context
 
 if TYPE_CHECKING:  # avoid circular import at runtime
     from ..training.optimizer import FlatAdamW
@@ -128,6 +129,10 @@ class ZeroEngine:
 
     # ------------------------------------------------------------- step
     def reduce_gradients(self):
+        with profiling_context("comm.reduce_gradients"):
+            return self._reduce_gradients_impl()
+
+    def _reduce_gradients_impl(self):
         """Complete (or launch) the gradient reduction for this step."""
         if self.world <= 1:
             return
@@ -202,6 +207,10 @@ class ZeroEngine:
         return ns
 
     def step(self, grad_scale: float = 1.0):
+        with profiling_context("comm.zero_step"):
+            return self._step_impl(grad_scale)
+
+    def _step_impl(self, grad_scale: float = 1.0):
         """reduce -> (norm) -> sharded fused update -> weight all-gather."""
         self.reduce_gradients()
         norm_sq = self.global_grad_norm_sq()

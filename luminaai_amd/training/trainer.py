@@ -118,6 +118,12 @@ class Trainer:
         self._step_t0 = None
         # dynamic loss scaling (fp16 only; bf16/fp32 train unscaled —
         # reference PrecisionManager semantics, trainer.py:157-356)
+        if getattr(config, "profile_communication", False) \
+                or getattr(config, "profile_memory", False):
+            # comm sync points (ZeroEngine) carry profiling_context markers;
+            # stats via utils.profiling.get_profiling_stats / chrome trace
+            from ..utils.profiling import enable_profiling
+            enable_profiling(True)
         self.loss_scale = (float(getattr(config, "fp16_loss_scale", 65536.0))
                            if self.precision.needs_loss_scale else 1.0)
         self._scale_good_steps = 0
@@ -494,6 +500,12 @@ class Trainer:
     # Adaptive-intervention API (reference trainer.py:1144-1835)
     # ================================================================== ==
     def get_current_metrics(self) -> TrainingMetrics:
+        if getattr(self.config, "profile_memory", False) \
+                and torch.cuda.is_available():
+            # peak since the previous emission (config flag profile_memory)
+            self._last_metrics["memory_peak_gb"] = \
+                torch.cuda.max_memory_allocated() / 1e9
+            torch.cuda.reset_peak_memory_stats()
         d = self._metric_floats()
         mem = (torch.cuda.memory_allocated() / 1e9
                if torch.cuda.is_available() else 0.0)

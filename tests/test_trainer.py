@@ -452,3 +452,29 @@ def test_elastic_reshard_world2_to_world1():
         o.step()
     for p1, p2 in zip(m1.parameters(), m_new.parameters()):
         torch.testing.assert_close(p1, p2)
+
+
+def test_profile_communication_flag(tiny_config, tokenizer):
+    """profile_communication populates the decorator-profiling stats with
+    the ZeroEngine sync points."""
+    from luminaai_amd.models import (DeepSeekTransformer,
+                                     config_to_deepseek_config)
+    from luminaai_amd.training import Trainer
+    from luminaai_amd.utils.profiling import (enable_profiling,
+                                              get_profiling_stats,
+                                              reset_profiling_stats)
+    reset_profiling_stats()
+    tiny_config.profile_communication = True
+    torch.manual_seed(0)
+    model = DeepSeekTransformer(config_to_deepseek_config(tiny_config))
+    t = Trainer(model, tokenizer, tiny_config)
+    try:
+        ids = torch.randint(1, tiny_config.vocab_size,
+                            (2, tiny_config.seq_length + 1))
+        t.train_step({"input_ids": ids[:, :-1], "labels": ids[:, 1:]})
+        t.optimizer_step()
+        stats = get_profiling_stats()
+        assert any(k.startswith("comm.") for k in stats), stats.keys()
+    finally:
+        enable_profiling(False)
+        reset_profiling_stats()
