@@ -156,8 +156,12 @@ def main(argv: Optional[list] = None) -> dict:
     if args.synthetic_steps:
         n = (cfg.micro_batch_size or 1) * cfg.gradient_accumulation_steps \
             * args.synthetic_steps
+        # PP: every stage of one pipeline column MUST see the same stream
+        # (rank 0 consumes input_ids, the last stage consumes the matching
+        # labels) — seed per DP replica, not per rank
+        seed = cfg.seed + (rank // args.pp if args.pp > 1 else rank)
         train_ds = SyntheticDataset(cfg.vocab_size, cfg.seq_length, n,
-                                    seed=cfg.seed + rank)
+                                    seed=seed)
         eval_ds = None
         cfg.num_epochs = 1
     else:
@@ -174,13 +178,13 @@ def main(argv: Optional[list] = None) -> dict:
                        "+ hipGraphs")
     model = DeepSeekTransformer(config_to_deepseek_config(cfg))
     if args.pp > 1:
-        if args.pp != world:
-            raise SystemExit(f"--pp {args.pp} requires world size {args.pp} "
-                             f"(got {world}); PP x DP composition is not "
-                             "supported yet")
+        if world % args.pp != 0:
+            raise SystemExit(f"--pp {args.pp} requires world size divisible "
+                             f"by {args.pp} (got {world}); ranks lay out as "
+                             "dp_replica x pp_stage")
         from .training.pipeline_loop import run_pipeline_training
         result = run_pipeline_training(
-            model, cfg, train_ds, logger,
+            model, cfg, train_ds, logger, pp=args.pp,
             virtual_stages=max(1, args.pp_virtual),
             steps=args.synthetic_steps)
         if rank == world - 1:

@@ -1,15 +1,19 @@
-"""Dedicated training loop for pipeline parallelism.
+"""Dedicated training loop for pipeline parallelism (PP and PP x DP).
 
 PP composes differently from the data-parallel Trainer (micro-batches flow
 THROUGH ranks instead of being sharded ACROSS them), so `--pp` runs this
-compact loop: partition the model into stages (1F1B) or pp x v chunks
-(interleaved), step a flat fused AdamW over the local stage's parameters,
-and checkpoint per stage. The adaptive orchestrator does not attach here
-(its interventions assume a whole-model rank); full PP x DP mesh
-composition is a round-2 item (ROADMAP.md).
+compact loop. With world > pp the remaining factor is data parallelism:
+ranks lay out as dp_replica x pp_stage (rank = dp_idx * pp + stage), each
+pipeline column consumes its own data shard, and stage gradients
+all-reduce across the per-stage DP group (flat FlatAdamW buffers -> one
+collective per group) before the AdamW step.
+
+The adaptive orchestrator does not attach here (its interventions assume a
+whole-model rank); PP x TP/EP mesh composition is a round-2 item
+(ROADMAP.md).
 
 Reference capability: vendored ColossalAI pipeline (p2p.py, one_f_one_b.py,
-interleaved_pp.py) reached through HybridParallelPlugin.
+interleaved_pp.py) reached through HybridParallelPlugin (pp_size x dp).
 """
 
 from __future__ import annotations
@@ -18,6 +22,7 @@ import os
 from typing import Dict, List, Optional
 
 import torch
+import torch.distributed as dist
 
 from ..parallel import comm
 from .optimizer import FlatAdamW
@@ -35,29 +40,64 @@ def _micro_batches(batch: Dict, n: int) -> List[Dict]:
     return out
 
 
-def run_pipeline_training(model, cfg, train_ds, logger,
+def build_pp_dp_groups(world: int, pp: int):
+    """(pp_group, dp_group, dp_idx, stage) for this rank; every rank must
+    call this with identical arguments (dist.new_group is collective)."""
+    rank = comm.get_rank()
+    dp = world // pp
+    pp_groups = [dist.new_group(list(range(d * pp, (d + 1) * pp)))
+                 for d in range(dp)]
+    dp_groups = [dist.new_group(list(range(s, world, pp)))
+                 for s in range(pp)]
+    dp_idx, stage = rank // pp, rank % pp
+    return pp_groups[dp_idx], dp_groups[stage], dp_idx, stage
+
+
+def run_pipeline_training(model, cfg, train_ds, logger, pp: int = 0,
                           virtual_stages: int = 1,
                           steps: Optional[int] = None) -> Dict:
-    """Train `model` over the whole PP world. Returns summary stats."""
-    from torch.utils.data import DataLoader
+    """Train `model` over a pp (x dp) world. Returns summary stats."""
+    from torch.utils.data import DataLoader, Subset
+
+    world = comm.get_world_size()
+    rank = comm.get_rank()
+    pp = pp or world
+    dp = world // pp
+    pp_group = dp_group = None
+    dp_idx, stage = 0, rank
+    if dp > 1:
+        pp_group, dp_group, dp_idx, stage = build_pp_dp_groups(world, pp)
+        # map-style datasets shard across DP replicas (all stages of one
+        # column already share the stream: per-column seed in main.py)
+        if hasattr(train_ds, "__len__"):
+            idx = list(range(dp_idx, len(train_ds), dp))
+            train_ds = Subset(train_ds, idx)
 
     if virtual_stages > 1:
         from ..parallel.pipeline import InterleavedPipelineEngine
         engine = InterleavedPipelineEngine(model, cfg,
-                                           virtual_stages=virtual_stages)
+                                           virtual_stages=virtual_stages,
+                                           pp_group=pp_group)
         stage_mod = engine.chunks
     else:
         from ..parallel.pipeline import PipelineParallelEngine
-        engine = PipelineParallelEngine(model, cfg)
+        engine = PipelineParallelEngine(model, cfg, pp_group=pp_group)
         stage_mod = engine.stage
-    rank = comm.get_rank()
     opt = FlatAdamW(stage_mod, lr=cfg.learning_rate,
                     weight_decay=cfg.weight_decay, max_grad_norm=1.0)
+    if dp > 1:
+        # replicas start identical (same init seed, but broadcast makes it
+        # robust for ad-hoc callers)
+        src = dist.get_process_group_ranks(dp_group)[0]
+        for g in opt.groups:
+            dist.broadcast(g.weight_view(), src, group=dp_group)
+            if not g._master_is_params:   # refresh the fp32 master copy
+                g.master.copy_(g.flat_w[g.shard_lo:g.shard_hi].float())
     accum = max(1, cfg.gradient_accumulation_steps)
     dl = DataLoader(train_ds, batch_size=(cfg.micro_batch_size or 1) * accum,
                     shuffle=False, drop_last=True,
                     num_workers=0)  # every stage consumes the SAME stream
-    total = len(dl) * max(1, cfg.num_epochs)
+    total = max(1, len(dl)) * max(1, cfg.num_epochs)
     warmup = max(1, int(total * getattr(cfg, "warmup_ratio", 0.1)))
     sched = WarmupScheduler(opt, total_steps=total, warmup_steps=warmup,
                             kind=getattr(cfg, "lr_scheduler", "cosine"))
@@ -68,28 +108,39 @@ def run_pipeline_training(model, cfg, train_ds, logger,
         for batch in dl:
             micro = _micro_batches(batch, accum)
             out = engine.train_batch(micro)
-            # mean over micro-batches; scale grads accordingly
-            opt.step(grad_scale=1.0 / len(micro))
+            if dp > 1:
+                for g in opt.groups:
+                    dist.all_reduce(g.flat_g, group=dp_group)
+            # mean over micro-batches and DP replicas
+            opt.step(grad_scale=1.0 / (len(micro) * dp))
             opt.zero_grad()
             sched.step()
             global_step += 1
-            last_loss = float(out["loss"])
+            loss = out["loss"]
+            if dp > 1:
+                loss = loss.clone()
+                dist.all_reduce(loss, group=dp_group)
+                loss = loss / dp
+            last_loss = float(loss)
             if steps is not None and global_step >= steps:
                 break
-            if global_step % 10 == 0 and engine.pp - 1 == rank:
+            if global_step % 10 == 0 and stage == pp - 1:
                 logger.info(f"pp step {global_step}: loss {last_loss:.4f} "
                             f"lr {opt.groups[0].lr:.2e}")
         if steps is not None and global_step >= steps:
             break
 
-    # per-stage checkpoint: stage_rank{r}.pt holds this rank's slice
+    # per-stage checkpoint: DP replica 0 writes each stage's slice
     exp_dir = os.path.join("experiments", cfg.experiment_name, "checkpoints")
     os.makedirs(exp_dir, exist_ok=True)
-    path = os.path.join(exp_dir, f"pp_stage_rank{rank}.pt")
-    torch.save({"stage_state_dict": stage_mod.state_dict(),
-                "optimizer_state_dict": opt.state_dict(),
-                "global_step": global_step,
-                "pp_rank": rank, "pp_world": engine.pp,
-                "virtual_stages": virtual_stages}, path)
+    path = os.path.join(exp_dir, f"pp_stage_rank{stage}.pt")
+    if dp_idx == 0:
+        torch.save({"stage_state_dict": stage_mod.state_dict(),
+                    "optimizer_state_dict": opt.state_dict(),
+                    "global_step": global_step,
+                    "pp_rank": stage, "pp_world": pp, "dp_world": dp,
+                    "virtual_stages": virtual_stages}, path)
+    if dp > 1:
+        dist.barrier()
     return {"global_step": global_step, "loss": last_loss,
-            "checkpoint": path, "pp_rank": rank}
+            "checkpoint": path, "pp_rank": stage, "dp_idx": dp_idx}
