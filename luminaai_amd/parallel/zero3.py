@@ -70,12 +70,15 @@ class _Zero3Segment:
     """One (params, weight_decay) flat segment of a sharding unit."""
 
     def __init__(self, params: List[torch.Tensor], lr: float,
-                 weight_decay: float, rank: int, world: int):
+                 weight_decay: float, rank: int, world: int, pg=None,
+                 comm_kind: str = "dp"):
         self.params = params
         self.lr = lr
         self.weight_decay = weight_decay
         self.rank = rank
         self.world = world
+        self.pg = pg
+        self.comm = comm_kind
         self.numel = sum(p.numel() for p in params)
         q = ALIGN * world
         self.padded = (self.numel + q - 1) // q * q
@@ -120,7 +123,8 @@ class _Zero3Segment:
             st.resize_(self.padded * self.flat_g.element_size())
         self.flat_g.zero_()
 
-    def gather(self, pg=None):
+    def gather(self, pg="unused"):
+        pg = self.pg
         st = self.flat_w.untyped_storage()
         if st.size() > 0:
             return
@@ -133,8 +137,9 @@ class _Zero3Segment:
         else:
             dist.all_gather_into_tensor(self.flat_w, self.w_shard, group=pg)
 
-    def reduce_into_shard(self, pg=None):
+    def reduce_into_shard(self, pg="unused"):
         """flat_g -> += g_shard (sum over ranks), then free flat_g."""
+        pg = self.pg
         lo = self.rank * self.shard_size
         hi = lo + self.shard_size
         if self.world == 1:
@@ -154,8 +159,9 @@ class _Zero3Segment:
                      self.lr, 0.9, 0.95, 1e-8, self.weight_decay, step_count,
                      norm_sq, max_norm, grad_scale)
 
-    def refresh_from_shard(self, pg=None):
+    def refresh_from_shard(self, pg="unused"):
         """w_shard -> flat_w (persistent units after an optimizer step)."""
+        pg = self.pg
         if self.flat_w.untyped_storage().size() == 0:
             return
         if self.world == 1:
@@ -179,23 +185,44 @@ class _Zero3Unit:
 
     def __init__(self, name: str, module: nn.Module, lr: float,
                  weight_decay: float, rank: int, world: int,
-                 persistent: bool = False):
+                 persistent: bool = False, pg=None, mesh=None):
         self.name = name
         self.module = module
         self.persistent = persistent
-        decay, no_decay = _decay_split(module.named_parameters())
+        named = list(module.named_parameters())
+        expert = []
+        if mesh is not None and mesh.ep_size > 1:
+            from .expert_parallel import is_expert_param
+            expert = [(n, p) for n, p in named
+                      if is_expert_param(n)
+                      or getattr(p, "_shard_parallel", False)]
+            eids = {id(p) for _, p in expert}
+            named = [(n, p) for n, p in named if id(p) not in eids]
+        decay, no_decay = _decay_split(named)
         self.segments: List[_Zero3Segment] = []
         if decay:
             self.segments.append(_Zero3Segment(decay, lr, weight_decay,
-                                               rank, world))
+                                               rank, world, pg=pg))
         if no_decay:
-            self.segments.append(_Zero3Segment(no_decay, lr, 0.0, rank, world))
+            self.segments.append(_Zero3Segment(no_decay, lr, 0.0, rank, world,
+                                               pg=pg))
+        if expert:
+            # EP-sharded expert weights: shard only across the ranks that
+            # hold the SAME experts (the expert replica group); with full EP
+            # (ep == world) that group is this rank alone -> local state
+            r_world = mesh.shard_replica_size
+            r_pg = mesh.shard_replica_group if r_world > 1 else None
+            r_rank = (dist.get_process_group_ranks(r_pg).index(
+                dist.get_rank()) if r_pg is not None else 0)
+            self.segments.append(_Zero3Segment(
+                [p for _, p in expert], lr, weight_decay,
+                r_rank, r_world, pg=r_pg, comm_kind="expert"))
         self.n_params = sum(len(s.params) for s in self.segments)
         self._grads_pending = self.n_params
 
-    def gather(self, pg=None):
+    def gather(self, pg="unused"):
         for s in self.segments:
-            s.gather(pg)
+            s.gather()
 
     def free_weights(self):
         if self.persistent:
@@ -299,9 +326,6 @@ class Zero3Engine:
 
     def __init__(self, model: nn.Module, config, process_group=None,
                  mesh=None):
-        if mesh is not None and mesh.ep_size > 1:
-            raise RuntimeError("ZeRO-3 + expert parallelism not supported "
-                               "yet; use ZeRO-2 with EP")
         self.model = model
         self.pg = process_group
         self.mesh = mesh
@@ -317,10 +341,12 @@ class Zero3Engine:
         self.units: List[_Zero3Unit] = []
         root_mod = _RootShell(model, blocks)
         self.units.append(_Zero3Unit("root", root_mod, lr, wd,
-                                     self.rank, self.world, persistent=True))
+                                     self.rank, self.world, persistent=True,
+                                     pg=self.pg, mesh=mesh))
         for i, b in enumerate(blocks):
             self.units.append(_Zero3Unit(f"block{i}", b, lr, wd,
-                                         self.rank, self.world))
+                                         self.rank, self.world,
+                                         pg=self.pg, mesh=mesh))
         self.optimizer = Zero3Optimizer(self)
         self._unit_of_param: Dict[int, _Zero3Unit] = {}
         self._hooks = []
@@ -340,12 +366,19 @@ class Zero3Engine:
         self._finalized = True
 
     def broadcast_parameters(self):
-        """Rank-0 initial weights to all, then shard + free."""
+        """Initial weights to all replicas, then shard + free. Dense
+        segments broadcast from global rank 0; expert segments only within
+        their replica group (each EP rank keeps its own experts)."""
         if self.world > 1:
             for u in self.units:
                 for s in u.segments:
-                    s.gather(self.pg)  # no-op pre-finalize (still allocated)
-                    dist.broadcast(s.flat_w, src=0, group=self.pg)
+                    s.gather()  # no-op pre-finalize (still allocated)
+                    if s.comm == "expert":
+                        if s.pg is not None:
+                            src = dist.get_process_group_ranks(s.pg)[0]
+                            dist.broadcast(s.flat_w, src=src, group=s.pg)
+                    else:
+                        dist.broadcast(s.flat_w, src=0, group=self.pg)
                     lo = s.rank * s.shard_size
                     s.w_shard.copy_(s.flat_w[lo:lo + s.shard_size])
                     s.master.copy_(s.w_shard.float())
@@ -378,14 +411,14 @@ class Zero3Engine:
             unit = self.units[ui]
             if not self._finalized:
                 return
-            unit.gather(self.pg)
+            unit.gather()
             if _in_backward():
                 # checkpoint recompute: backward follows immediately
                 unit.alloc_grads()
             elif self._prefetch_stream is not None and ui + 1 < len(self.units):
                 nxt = self.units[ui + 1]
                 with torch.cuda.stream(self._prefetch_stream):
-                    nxt.gather(self.pg)
+                    nxt.gather()
         return hook
 
     def _make_fwd_post(self, ui):
@@ -401,7 +434,7 @@ class Zero3Engine:
     def _make_bwd_pre(self, ui):
         def hook(module, grad_output):
             unit = self.units[ui]
-            unit.gather(self.pg)
+            unit.gather()
             unit.alloc_grads()
         return hook
 
@@ -411,7 +444,7 @@ class Zero3Engine:
             if unit._grads_pending == 0:
                 unit._grads_pending = unit.n_params
                 for s in unit.segments:
-                    s.reduce_into_shard(self.pg)
+                    s.reduce_into_shard()
                     if unit.persistent:
                         # persistent units have no backward-pre hook to
                         # re-allocate grads: keep the buffer live (zeroed)
@@ -443,7 +476,7 @@ class Zero3Engine:
             for s in u.segments:
                 if partial or (not u.persistent
                                and s.flat_g.untyped_storage().size() > 0):
-                    s.reduce_into_shard(self.pg)
+                    s.reduce_into_shard()
                     if u.persistent:
                         s.alloc_grads()
             u._grads_pending = u.n_params
@@ -453,7 +486,7 @@ class Zero3Engine:
         for u in self.units:
             if u.persistent:
                 for s in u.segments:
-                    s.refresh_from_shard(self.pg)
+                    s.refresh_from_shard()
 
     def sync_shards_from_full(self):
         """After load_state_dict materialised full weights into flat buffers."""
@@ -476,7 +509,7 @@ class Zero3Engine:
 
         def __enter__(self):
             for u in self.engine.units:
-                u.gather(self.engine.pg)
+                u.gather()
             return self
 
         def __exit__(self, *exc):
