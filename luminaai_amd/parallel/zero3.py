@@ -313,6 +313,37 @@ class Zero3Optimizer:
             g.weight_decay = gs.get("weight_decay", g.weight_decay)
             g.w_shard.copy_(g.master.to(g.w_shard.dtype))
 
+    def load_resharded(self, shard_sds: List[Dict]):
+        """Elastic resume at a different world size: concatenate every
+        saved rank's shard (contiguous slices of the padded flat buffer)
+        and re-slice for this topology (mirrors FlatAdamW.load_resharded;
+        the Trainer routes here when shard_world mismatches)."""
+        if any(g.comm == "expert" for g in self.groups):
+            raise ValueError("elastic resharding with expert parallelism is "
+                             "not supported (expert placement changes)")
+        saved_world = shard_sds[0].get("shard_world", 1)
+        if len(shard_sds) != saved_world:
+            raise ValueError(f"need all {saved_world} shards, got "
+                             f"{len(shard_sds)}")
+        self.step_count = shard_sds[0].get("step_count", 0)
+        for gi, g in enumerate(self.groups):
+            saved_numel = shard_sds[0]["groups"][gi]["numel"]
+            if saved_numel != g.numel:
+                raise ValueError(f"group {gi} size mismatch ({g.numel} vs "
+                                 f"{saved_numel})")
+            lo = g.rank * g.shard_size
+            hi = lo + g.shard_size
+            for key, dst in (("master", g.master), ("m", g.m), ("v", g.v)):
+                full = torch.cat([sd["groups"][gi][key].float()
+                                  for sd in shard_sds])
+                buf = torch.zeros(g.padded, dtype=torch.float32)
+                buf[:g.numel] = full[:g.numel]
+                dst.copy_(buf[lo:hi].to(dst.device))
+            g.lr = shard_sds[0]["groups"][gi].get("lr", g.lr)
+            g.weight_decay = shard_sds[0]["groups"][gi].get(
+                "weight_decay", g.weight_decay)
+            g.w_shard.copy_(g.master.to(g.w_shard.dtype))
+
     def rebuild(self, model):
         raise RuntimeError("dynamic expert add/prune is not supported under "
                            "ZeRO-3 (param shards are fixed); use ZeRO-0/1/2")

@@ -291,3 +291,66 @@ def test_zero3_with_expert_parallelism():
         assert res[r]["gn3"] == pytest.approx(res[r]["gn0"], rel=1e-4), res
     # different EP ranks hold different experts
     assert res[0]["exp_sum"] != res[1]["exp_sum"]
+
+
+def z3_elastic_save_worker(rank, world):
+    """Train ZeRO-3 at world=2 with SAME data on both ranks and save."""
+    from luminaai_amd.training import CheckpointManager
+    tmp = os.environ["Z3_ELASTIC_TMP"]
+    os.chdir(tmp)
+    cfg = _cfg(3, rank=rank)
+    from luminaai_amd.data.tokenizer import ConversationTokenizer
+    from luminaai_amd.models import (DeepSeekTransformer,
+                                     config_to_deepseek_config)
+    from luminaai_amd.training import Trainer
+    torch.manual_seed(1234)
+    model = DeepSeekTransformer(config_to_deepseek_config(cfg))
+    t = Trainer(model, ConversationTokenizer(), cfg)
+    t._setup_scheduler(10)
+    t.checkpoints = CheckpointManager(os.path.join(tmp, "shared_ckpts"))
+    torch.manual_seed(750)
+    for _ in range(2):
+        ids = torch.randint(1, cfg.vocab_size, (2, cfg.seq_length + 1))
+        t.engine.set_sync(True)
+        t.train_step({"input_ids": ids[:, :-1], "labels": ids[:, 1:]})
+        t.optimizer_step()
+    t.save_checkpoint()
+    dist.barrier()
+    m_sum = sum(float(g.m.sum()) for g in t.optimizer.groups)
+    t.engine.remove_hooks()
+    return {"m_sum": m_sum, "steps": t.optimizer.step_count}
+
+
+def test_zero3_elastic_resume_world1(tmp_path):
+    os.environ["Z3_ELASTIC_TMP"] = str(tmp_path)
+    res = _spawn("z3_elastic_save_worker")
+    cwd = os.getcwd()
+    os.chdir(tmp_path)
+    try:
+        from luminaai_amd.training import CheckpointManager
+        # reference arm: identical run entirely in this process at world=1
+        t_ref = _train(_cfg(3, rank=9), steps=2, seed=750)
+        # elastic arm: fresh world-1 trainer resuming the world-2 save
+        from luminaai_amd.data.tokenizer import ConversationTokenizer
+        from luminaai_amd.models import (DeepSeekTransformer,
+                                         config_to_deepseek_config)
+        from luminaai_amd.training import Trainer
+        cfg = _cfg(3, rank=8)
+        torch.manual_seed(1234)
+        model = DeepSeekTransformer(config_to_deepseek_config(cfg))
+        t = Trainer(model, ConversationTokenizer(), cfg)
+        t._setup_scheduler(10)
+        t.checkpoints = CheckpointManager(str(tmp_path / "shared_ckpts"))
+        t.load_checkpoint("latest")
+        assert t.optimizer.step_count == res[0]["steps"]
+        m_here = sum(float(g.m.sum()) for g in t.optimizer.groups)
+        assert m_here == pytest.approx(res[0]["m_sum"] + res[1]["m_sum"],
+                                       rel=1e-5)
+        # resumed state matches the uninterrupted world-1 reference
+        w_ref = _full_weights(t_ref)
+        w_new = _full_weights(t)
+        for k in w_ref:
+            torch.testing.assert_close(w_new[k], w_ref[k], rtol=1e-4,
+                                       atol=1e-5, msg=k)
+    finally:
+        os.chdir(cwd)
