@@ -42,6 +42,35 @@ class GenerationConfig:
         return cls(**kw)
 
 
+def sample_token(logits: torch.Tensor, cfg: GenerationConfig,
+                 recent: List[int]) -> int:
+    """One token from processed logits: repetition penalty over `recent`
+    (reference Chat.py:392-398), temperature, top-k, top-p nucleus, greedy
+    at temperature 0. Mutates `logits` (callers pass a clone or a buffer
+    they are done with)."""
+    if cfg.repetition_penalty != 1.0 and recent:
+        idx = torch.tensor(sorted(set(recent)), device=logits.device)
+        sel = logits[idx]
+        logits[idx] = torch.where(sel > 0, sel / cfg.repetition_penalty,
+                                  sel * cfg.repetition_penalty)
+    if cfg.temperature <= 0.0:
+        return int(logits.argmax())
+    logits = logits / cfg.temperature
+    if cfg.top_k and cfg.top_k > 0:
+        kth = torch.topk(logits, min(cfg.top_k, logits.numel())).values[-1]
+        logits = logits.masked_fill(logits < kth, float("-inf"))
+    if cfg.top_p < 1.0:
+        sorted_logits, sorted_idx = torch.sort(logits, descending=True)
+        probs = torch.softmax(sorted_logits, dim=-1)
+        cum = torch.cumsum(probs, dim=-1)
+        cut = cum - probs > cfg.top_p       # keep first token above p
+        sorted_logits = sorted_logits.masked_fill(cut, float("-inf"))
+        logits = torch.full_like(logits, float("-inf")).scatter(
+            0, sorted_idx, sorted_logits)
+    probs = torch.softmax(logits, dim=-1)
+    return int(torch.multinomial(probs, 1))
+
+
 class GenerationEngine:
     def __init__(self, model, tokenizer, device: Optional[torch.device] = None):
         self.model = model
@@ -352,28 +381,7 @@ class GenerationEngine:
     # ------------------------------------------------------------------
     def _sample(self, logits: torch.Tensor, cfg: GenerationConfig,
                 recent: List[int]) -> int:
-        # repetition penalty (reference Chat.py:392-398)
-        if cfg.repetition_penalty != 1.0 and recent:
-            idx = torch.tensor(sorted(set(recent)), device=logits.device)
-            sel = logits[idx]
-            logits[idx] = torch.where(sel > 0, sel / cfg.repetition_penalty,
-                                      sel * cfg.repetition_penalty)
-        if cfg.temperature <= 0.0:
-            return int(logits.argmax())
-        logits = logits / cfg.temperature
-        if cfg.top_k and cfg.top_k > 0:
-            kth = torch.topk(logits, min(cfg.top_k, logits.numel())).values[-1]
-            logits = logits.masked_fill(logits < kth, float("-inf"))
-        if cfg.top_p < 1.0:
-            sorted_logits, sorted_idx = torch.sort(logits, descending=True)
-            probs = torch.softmax(sorted_logits, dim=-1)
-            cum = torch.cumsum(probs, dim=-1)
-            cut = cum - probs > cfg.top_p       # keep first token above p
-            sorted_logits = sorted_logits.masked_fill(cut, float("-inf"))
-            logits = torch.full_like(logits, float("-inf")).scatter(
-                0, sorted_idx, sorted_logits)
-        probs = torch.softmax(logits, dim=-1)
-        return int(torch.multinomial(probs, 1))
+        return sample_token(logits, cfg, recent)
 
     def _filtered_probs(self, logits: torch.Tensor, cfg: GenerationConfig,
                         recent: List[int]) -> torch.Tensor:

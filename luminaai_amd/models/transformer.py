@@ -258,6 +258,20 @@ class KVCache:
         self._len = self.k.shape[1]
         return self.k, self.v
 
+    def ensure_batch(self, B: int, like: torch.Tensor):
+        """Preallocate the [B, max_len, H, D] buffers before the first
+        append (continuous batching: a row view must write into the full
+        batch buffer before any batch-wide append happened)."""
+        if self.k is None:
+            _, _, H, D = like.shape
+            self.k = like.new_zeros(B, max(self.max_len, 1), H, D)
+            self.v = torch.zeros_like(self.k)
+
+    def row_view(self, row: int, start: int) -> "KVCacheRowView":
+        """A single-row cache facade writing at its own column cursor —
+        lets one sequence prefill into a live batched cache."""
+        return KVCacheRowView(self, row, start)
+
     def truncate(self, n: int):
         """Rewind the cache to n tokens (speculative-decode rejection).
         Preallocated buffers just move the write cursor; growing caches
@@ -273,6 +287,44 @@ class KVCache:
                 self.k = self.k[:, :n].contiguous()
                 self.v = self.v[:, :n].contiguous()
             self._len = n
+
+
+class KVCacheRowView:
+    """One row of a batched KVCache with an independent column cursor.
+
+    Continuous batching admits a new sequence into a RUNNING batch: the
+    prompt forward goes through this view (batch 1), whose appends land in
+    `parent.k[row]` starting at column `start` (the row is left-padded so
+    the prompt ENDS at the batch's shared write position — RoPE attention
+    scores depend only on relative positions, so the uniform per-row shift
+    is exact, and the pad columns are masked by the serving loop's
+    attention mask). The parent's cursor is untouched."""
+
+    graph_mode = False
+    pos_dev = None
+
+    def __init__(self, parent: KVCache, row: int, start: int):
+        self.parent = parent
+        self.row = row
+        self._len = start
+
+    @property
+    def seq_len(self) -> int:
+        return self._len
+
+    def append(self, k: torch.Tensor, v: torch.Tensor):
+        B, S, H, D = k.shape
+        assert B == 1, "row views take batch-1 appends"
+        p = self.parent
+        if p.k is None:
+            raise RuntimeError("call KVCache.ensure_batch before row "
+                               "prefill")
+        end = self._len + S
+        p.k[self.row, self._len:end] = k[0]
+        p.v[self.row, self._len:end] = v[0]
+        self._len = end
+        return (p.k[self.row:self.row + 1, :end],
+                p.v[self.row:self.row + 1, :end])
 
 
 class GroupedQueryAttention(nn.Module):
