@@ -552,11 +552,20 @@ class MoEFFNLayer(nn.Module):
             from ..ops.fp8 import expert_bmm_fp8 as _ebmm
         else:
             _ebmm = ops.interface.expert_bmm
-        gu = _ebmm(bufv, self.w_gate_up.to(x.dtype))
+        if getattr(self, "use_int8_weights", False):
+            # inference-only int8 expert storage (ops/quant.py): dequant at
+            # the GEMM input; the fused dequant kernel is a round-2 item
+            w_gu = (self.w_gate_up_q.float()
+                    * self.w_gate_up_scale).to(x.dtype)
+            w_dn = (self.w_down_q.float() * self.w_down_scale).to(x.dtype)
+        else:
+            w_gu = self.w_gate_up.to(x.dtype)
+            w_dn = self.w_down.to(x.dtype)
+        gu = _ebmm(bufv, w_gu)
         I = self.intermediate_size
         gu2 = gu.reshape(-1, 2 * I)
         act = ops.swiglu(gu2.narrow(1, 0, I), gu2.narrow(1, I, I))
-        y = _ebmm(act.view(EL, -1, I), self.w_down.to(x.dtype))
+        y = _ebmm(act.view(EL, -1, I), w_dn)
 
         if self.ep_size > 1:
             from ..parallel.expert_parallel import all_to_all

@@ -147,6 +147,31 @@ class Int4Linear(nn.Module):
                 f"group={self.group_size}")
 
 
+# ---------------------------------------------------------------- MoE experts
+def quantize_moe_experts(model: nn.Module) -> int:
+    """Inference-only int8 storage for batched expert weights ([E, K, N],
+    per-(expert, out-channel) scales over K). The experts hold most of a
+    MoE model's parameters; this removes the bf16/fp32 copies entirely
+    (the Parameters are deleted — the layer forward reads the int8
+    buffers). Returns the number of MoE layers converted."""
+    n = 0
+    for layer in getattr(model, "get_moe_layers", lambda: [])():
+        if getattr(layer, "use_int8_weights", False):
+            continue
+        for name in ("w_gate_up", "w_down"):
+            w = getattr(layer, name).detach()            # [E, K, N]
+            amax = w.abs().amax(dim=1, keepdim=True).float().clamp_min(1e-12)
+            scale = amax / INT8_MAX                      # [E, 1, N]
+            q = (w.float() / scale).round().clamp(
+                -INT8_MAX, INT8_MAX).to(torch.int8)
+            delattr(layer, name)                         # drop the Parameter
+            layer.register_buffer(name + "_q", q)
+            layer.register_buffer(name + "_scale", scale)
+        layer.use_int8_weights = True
+        n += 1
+    return n
+
+
 # ---------------------------------------------------------------- manager
 def quantize_model(model: nn.Module, mode: str = "int8",
                    min_dim: int = 64, group_size: int = 128) -> int:
@@ -159,6 +184,8 @@ def quantize_model(model: nn.Module, mode: str = "int8",
     if mode not in ("int8", "int4"):
         raise ValueError(f"unknown quantization mode {mode!r}")
     n = 0
+    if mode == "int8":
+        n += quantize_moe_experts(model)
     for mod in model.modules():
         for name, child in list(mod.named_children()):
             if type(child) is not nn.Linear or child.bias is not None:

@@ -466,3 +466,30 @@ def test_speculative_sampling_preserves_distribution(tokenizer):
     emp = counts / counts.sum()
     tv = 0.5 * float((emp - expected).abs().sum())
     assert tv < 0.06, (tv, emp.tolist(), expected.tolist())
+
+
+def test_quantize_moe_experts_int8(tiny_moe_config, tokenizer):
+    """Expert weights (the bulk of a MoE model) drop to int8 storage and
+    the model still decodes with close logits."""
+    from luminaai_amd.ops.quant import (quantize_model,
+                                        quantized_model_bytes)
+    torch.manual_seed(0)
+    m = DeepSeekTransformer(config_to_deepseek_config(tiny_moe_config)).eval()
+    before = quantized_model_bytes(m)
+    ids = torch.randint(0, 500, (1, 12))
+    with torch.no_grad():
+        ref, _, _ = m(ids)
+    n = quantize_model(m, mode="int8", min_dim=32)
+    assert n > 0
+    assert not hasattr(m.layers[1].ffn, "w_gate_up")  # Parameter removed
+    assert m.layers[1].ffn.w_gate_up_q.dtype == torch.int8
+    after = quantized_model_bytes(m)
+    assert after < before * 0.62   # experts fp32->int8 dominate the drop
+    with torch.no_grad():
+        out, _, _ = m(ids)
+    agree = (out.argmax(-1) == ref.argmax(-1)).float().mean()
+    assert float(agree) > 0.5
+    eng = GenerationEngine(m, tokenizer)
+    toks = eng.generate(tokenizer.encode("hello"),
+                        GenerationConfig(max_new_tokens=4, temperature=0.0))
+    assert all(isinstance(t, int) for t in toks)
