@@ -33,6 +33,36 @@ class _Slot:
         self.active = False
 
 
+class _RowsView:
+    """A compacted multi-row facade over a batched KVCache: forward work
+    runs only on the ACTIVE rows (batch len(rows), not max_batch). Appends
+    scatter into the parent rows at the shared cursor; reads gather the
+    selected rows' prefixes (the attention has to stream those bytes
+    anyway)."""
+
+    graph_mode = False
+    pos_dev = None
+
+    def __init__(self, parent, rows: List[int]):
+        self.parent = parent
+        self.rows = torch.as_tensor(rows, dtype=torch.long,
+                                    device=parent.k.device
+                                    if parent.k is not None else None)
+
+    @property
+    def seq_len(self) -> int:
+        return self.parent.seq_len
+
+    def append(self, k: torch.Tensor, v: torch.Tensor):
+        p = self.parent
+        cur = p.seq_len
+        S = k.shape[1]
+        p.k[self.rows, cur:cur + S] = k
+        p.v[self.rows, cur:cur + S] = v
+        p._len = cur + S
+        return (p.k[self.rows, :cur + S], p.v[self.rows, :cur + S])
+
+
 class ContinuousBatchingEngine:
     """Iteration-level scheduler over one shared batched KV cache."""
 
@@ -156,8 +186,6 @@ class ContinuousBatchingEngine:
         if self.n_active() == 0:
             return []
         cur = self.cursor
-        step_ids = torch.full((self.max_batch, 1), self._pad,
-                              dtype=torch.long, device=self.device)
         finished: List[int] = []
         tok_of: Dict[int, int] = {}
         for b, s in enumerate(self.slots):
@@ -171,7 +199,6 @@ class ContinuousBatchingEngine:
             s.out.append(nid)
             s.recent.append(nid)
             tok_of[b] = nid
-            step_ids[b, 0] = nid
             if s.cb is not None:
                 try:
                     s.cb(nid)
@@ -186,16 +213,20 @@ class ContinuousBatchingEngine:
             for b in live:
                 self._retire(b)
             return finished + live
-        mask = torch.cat([self.occupancy[:, :cur],
-                          torch.zeros(self.max_batch, 1, dtype=torch.long,
-                                      device=self.device)], dim=1)
-        for b in live:
-            mask[b, cur] = 1
+        # compacted forward: batch = len(live), not max_batch — a lone
+        # request pays B=1 compute
+        step_ids = torch.tensor([[tok_of[b]] for b in live],
+                                dtype=torch.long, device=self.device)
+        mask = torch.cat([self.occupancy[live, :cur],
+                          torch.ones(len(live), 1, dtype=torch.long,
+                                     device=self.device)], dim=1)
+        views = [c if len(live) == self.max_batch else _RowsView(c, live)
+                 for c in self.caches]
         logits, _, _ = self.model(step_ids, attention_mask=mask,
-                                  kv_caches=self.caches)
-        self.occupancy[:, cur] = mask[:, cur]
-        self.last_logits = logits[:, -1].float()
-        # rows that just finished max_new or will stop are handled next step
+                                  kv_caches=views)
+        for i, b in enumerate(live):
+            self.occupancy[b, cur] = 1
+            self.last_logits[b] = logits[i, -1].float()
         self.stats["steps"] += 1
         self.stats["tokens_out"] += len(tok_of)
         return finished

@@ -86,3 +86,16 @@ def test_continuous_empty_batch_resets_cursor(small_model, tokenizer):
     eng.admit(tokenizer.encode("again"), _cfg(2))
     # the cache was reclaimed before the new prefill
     assert eng.cursor == len(tokenizer.encode("again"))
+
+
+def test_continuous_compaction_equivalence(small_model, tokenizer):
+    """Partially-occupied pool (compacted forward, batch=len(live)) must
+    match a fully-occupied pool's output for the same request."""
+    p = tokenizer.encode("compaction check prompt")
+    solo = GenerationEngine(small_model.eval(), tokenizer)
+    ref = solo.generate(p, _cfg(5))
+    # pool of 8 slots, single request -> compacted path (B=1) throughout
+    eng = ContinuousBatchingEngine(small_model.eval(), tokenizer,
+                                   max_batch=8, max_len=64)
+    out = eng.run_to_completion([p], _cfg(5))
+    assert out == [ref]
