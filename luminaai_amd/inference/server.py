@@ -98,9 +98,88 @@ class _DynamicBatcher:
                     r["done"].set()
 
 
+class _ContinuousWorker:
+    """Serving loop over ContinuousBatchingEngine: requests (streaming or
+    not) join the running decode batch at token granularity; a lone
+    request pays batch-1 compute (active-row compaction)."""
+
+    def __init__(self, model, tokenizer, max_batch: int = 8,
+                 max_len: int = 2048):
+        from .continuous import ContinuousBatchingEngine
+        self.eng = ContinuousBatchingEngine(model, tokenizer,
+                                            max_batch=max_batch,
+                                            max_len=max_len)
+        self.q: "queue.Queue" = queue.Queue()
+        self._thread = threading.Thread(target=self._loop, daemon=True)
+        self._thread.start()
+
+    def submit(self, prompt_ids: List[int], cfg: GenerationConfig,
+               cb=None) -> List[int]:
+        evt = threading.Event()
+        item = {"ids": prompt_ids, "cfg": cfg, "cb": cb, "evt": evt}
+        self.q.put(item)
+        evt.wait()
+        if "err" in item:
+            raise item["err"]
+        return item["out"]
+
+    def _loop(self):
+        tracked: Dict[int, Dict] = {}
+        pending: List[Dict] = []
+        while True:
+            # drain arrivals; block only when completely idle
+            block = not tracked and not pending
+            while True:
+                try:
+                    pending.append(self.q.get(block=block, timeout=None))
+                    block = False
+                except queue.Empty:
+                    break
+                if not block:
+                    break
+            while True:     # opportunistic non-blocking drain
+                try:
+                    pending.append(self.q.get_nowait())
+                except queue.Empty:
+                    break
+            for item in list(pending):
+                if not self.eng.can_admit(len(item["ids"])):
+                    if self.eng.n_active() == 0:
+                        item["err"] = RuntimeError(
+                            "prompt exceeds the serving context window")
+                        item["evt"].set()
+                        pending.remove(item)
+                    continue
+                try:
+                    b = self.eng.admit(item["ids"], item["cfg"],
+                                       item.get("cb"))
+                    tracked[b] = item
+                except Exception as e:  # noqa: BLE001
+                    item["err"] = e
+                    item["evt"].set()
+                pending.remove(item)
+            if self.eng.n_active():
+                try:
+                    self.eng.step()
+                except Exception as e:  # noqa: BLE001
+                    for item in tracked.values():
+                        item["err"] = e
+                        item["evt"].set()
+                    tracked.clear()
+            for b, item in list(tracked.items()):
+                s = self.eng.slots[b]
+                if not s.active and s.result is not None:
+                    item["out"] = s.result
+                    item["evt"].set()
+                    del tracked[b]
+
+
 def create_app(model, tokenizer, model_name: str = "luminaai-amd",
-               max_batch: int = 8):
-    """Build the ASGI app around an already-loaded model."""
+               max_batch: int = 8, serving: str = "continuous",
+               max_len: int = 2048):
+    """Build the ASGI app around an already-loaded model. serving:
+    "continuous" (token-level joins, default) or "dynamic" (request-level
+    batching with the hipGraph single-stream path for streaming)."""
     from fastapi import FastAPI
     from fastapi.responses import JSONResponse, StreamingResponse
 
@@ -108,7 +187,13 @@ def create_app(model, tokenizer, model_name: str = "luminaai-amd",
     device = next(model.parameters()).device
     engine = GenerationEngine(model, tokenizer, device)
     lock = threading.Lock()
-    batcher = _DynamicBatcher(engine, max_batch=max_batch, lock=lock)
+    if serving == "continuous":
+        worker = _ContinuousWorker(model, tokenizer, max_batch=max_batch,
+                                   max_len=max_len)
+        batcher = None
+    else:
+        batcher = _DynamicBatcher(engine, max_batch=max_batch, lock=lock)
+        worker = None
     started = time.time()
     counters = {"requests": 0, "tokens_out": 0}
     from ..monitoring.prometheus import make_server_metrics
@@ -125,12 +210,15 @@ def create_app(model, tokenizer, model_name: str = "luminaai-amd",
     def _decode(prompt_ids: List[int], cfg: GenerationConfig,
                 cb=None, endpoint: str = "completions") -> List[int]:
         t0 = time.perf_counter()
-        if cb is None:
-            # non-streaming: ride the dynamic batcher (concurrent requests
-            # share one forward pass)
+        if worker is not None:
+            # continuous batching: streaming and non-streaming alike join
+            # the running decode batch at token granularity
+            out = worker.submit(prompt_ids, cfg, cb=cb)
+        elif cb is None:
+            # dynamic mode, non-streaming: requests share one batched pass
             out = batcher.submit(prompt_ids, cfg)
         else:
-            # streaming: dedicated single-sequence decode (hipGraph path)
+            # dynamic mode, streaming: single-sequence (hipGraph) decode
             with lock:
                 out = engine.generate(prompt_ids, cfg, stream_callback=cb)
         _account(endpoint, len(out), t0)
@@ -161,10 +249,14 @@ def create_app(model, tokenizer, model_name: str = "luminaai-amd",
 
     @app.get("/health")
     def health():
-        return {"status": "ok", "device": str(device),
-                "uptime_s": round(time.time() - started, 1),
-                "batches_run": batcher.batches_run,
-                "batched_rows": batcher.rows_seen, **counters}
+        d = {"status": "ok", "device": str(device), "serving": serving,
+             "uptime_s": round(time.time() - started, 1), **counters}
+        if worker is not None:
+            d.update(worker.eng.stats)
+        else:
+            d.update(batches_run=batcher.batches_run,
+                     batched_rows=batcher.rows_seen)
+        return d
 
     @app.get("/metrics")
     def metrics():

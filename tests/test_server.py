@@ -106,7 +106,7 @@ def test_prometheus_training_exporter(small_model, tokenizer):
 
 
 def test_concurrent_requests_batch_together(client):
-    """Parallel non-streaming requests share one generate_batch pass."""
+    """Parallel requests join the continuous decode batch."""
     import concurrent.futures as cf
     h0 = client.get("/health").json()
     with cf.ThreadPoolExecutor(max_workers=4) as ex:
@@ -116,9 +116,21 @@ def test_concurrent_requests_batch_together(client):
         results = [f.result() for f in futs]
     assert all(r.status_code == 200 for r in results)
     h1 = client.get("/health").json()
-    assert h1["batched_rows"] - h0["batched_rows"] == 4
-    # fewer batches than rows => at least some requests were coalesced
-    assert h1["batches_run"] - h0["batches_run"] <= 4
+    assert h1["serving"] == "continuous"
+    assert h1["admitted"] - h0["admitted"] == 4
+    assert h1["finished"] - h0["finished"] == 4
+
+
+def test_dynamic_mode_still_available(small_model, tokenizer):
+    from luminaai_amd.inference.server import create_app
+    app = create_app(small_model.eval(), tokenizer, serving="dynamic")
+    c = TestClient(app)
+    r = c.post("/v1/completions", json={"prompt": "dyn", "max_tokens": 3,
+                                        "temperature": 0.0})
+    assert r.status_code == 200
+    h = c.get("/health").json()
+    assert h["serving"] == "dynamic"
+    assert h["batched_rows"] >= 1
 
 
 def test_batched_result_matches_direct(client, small_model, tokenizer):
