@@ -398,6 +398,13 @@ class GroupedQueryAttention(nn.Module):
         kt = k.transpose(1, 2)
         vt = v.transpose(1, 2)
         is_causal = attn_mask is None and (kv_cache is None or S > 1)
+        # flash-vs-fallback call counters (reference model.py:635-637,
+        # 841-853): the AOTriton flash path serves mask-free causal SDPA;
+        # additive masks route to the composite kernel
+        if is_causal:
+            self._flash_calls = getattr(self, "_flash_calls", 0) + 1
+        else:
+            self._masked_calls = getattr(self, "_masked_calls", 0) + 1
         out = F.scaled_dot_product_attention(
             qt, kt, vt,
             attn_mask=attn_mask,
@@ -934,6 +941,16 @@ class DeepSeekTransformer(nn.Module):
                 d["mod_skip_frac"] = layer._mod_skip_frac
             out.append(d)
         return out
+
+    def get_attention_stats(self) -> Dict[str, int]:
+        """Aggregate flash-vs-masked SDPA call counters
+        (reference model.py:841-853)."""
+        flash = masked = 0
+        for layer in self.layers:
+            a = layer.attention
+            flash += getattr(a, "_flash_calls", 0)
+            masked += getattr(a, "_masked_calls", 0)
+        return {"flash_calls": flash, "masked_calls": masked}
 
     def get_moe_layers(self) -> List[MoEFFNLayer]:
         return [l.ffn for l in self.layers if l.is_moe]

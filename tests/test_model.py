@@ -202,3 +202,16 @@ def test_config_conversion_carries_mod():
                use_mod=True, use_moe=False, vocab_size=512)
     mc = config_to_deepseek_config(c)
     assert mc.use_mod is True  # reference dropped this flag (Main.py:572)
+
+
+def test_attention_call_counters(small_model):
+    m = small_model.eval()
+    ids = torch.randint(0, 500, (1, 8))
+    with torch.no_grad():
+        m(ids)  # causal -> flash path
+        mask = torch.ones(1, 8, dtype=torch.long)
+        mask[0, :3] = 0
+        m(ids, attention_mask=mask)  # additive mask -> composite path
+    stats = m.get_attention_stats()
+    assert stats["flash_calls"] >= len(m.layers)
+    assert stats["masked_calls"] >= len(m.layers)
