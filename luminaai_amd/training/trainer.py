@@ -179,6 +179,10 @@ class Trainer:
 
         self._tokens_seen += input_ids.numel() * comm.get_world_size()
         self._micro_in_cycle += 1
+        # detach everything leaving the step: consumers read scalars, and a
+        # live autograd reference would pin the whole graph in memory
+        out = {k: (v.detach() if torch.is_tensor(v) else v)
+               for k, v in out.items()}
         self._last_metrics = {
             "loss": out["loss"], "ce_loss": out["ce_loss"],
             "aux_loss": out["aux_loss"], "accuracy": out["accuracy"],
