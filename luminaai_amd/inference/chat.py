@@ -50,9 +50,16 @@ class ChatInterface:
             n = quantize_model(self.model, mode=quantize)
             print(f"quantized {n} Linear layers to {quantize} "
                   f"({quantized_model_bytes(self.model) / 1e6:.1f} MB)")
+
+    def set_draft_model(self, draft) -> None:
+        """Enable speculative decoding: a small draft proposes, the main
+        model verifies (exact greedy / distribution-preserving sampling —
+        inference/engine.generate_speculative)."""
+        self.draft_model = draft.to(self.device).eval()
         self.tokenizer = tokenizer or ConversationTokenizer()
         self.engine = GenerationEngine(self.model, self.tokenizer, self.device)
         self.gen_config = GenerationConfig.from_mode("standard")
+        self.draft_model = None       # set_draft_model() enables speculative
         self.mode = "standard"
         self.system_prompt = system_prompt
         self.history: List[Dict] = []
@@ -82,7 +89,13 @@ class ChatInterface:
             if stream:
                 print(piece, end="", flush=True)
 
-        self.engine.generate(ids, self.gen_config, stream_callback=cb)
+        if self.draft_model is not None:
+            toks = self.engine.generate_speculative(ids, self.draft_model,
+                                                    self.gen_config)
+            for t in toks:     # speculative emits in verified blocks
+                cb(t)
+        else:
+            self.engine.generate(ids, self.gen_config, stream_callback=cb)
         reply = "".join(pieces).strip()
         self.history.append({"role": "user", "content": user_text})
         self.history.append({"role": "assistant", "content": reply})
@@ -155,9 +168,22 @@ def main():
     ap.add_argument("--device", default=None)
     ap.add_argument("--quantize", default=None, choices=["int8", "int4", "fp8"],
                     help="weight-only quantization for inference")
+    ap.add_argument("--draft-checkpoint", default=None,
+                    help="small draft model checkpoint for speculative "
+                         "decoding")
     args = ap.parse_args()
-    ChatInterface(checkpoint=args.checkpoint, device=args.device,
-                  quantize=args.quantize).run()
+    chat = ChatInterface(checkpoint=args.checkpoint, device=args.device,
+                         quantize=args.quantize)
+    if args.draft_checkpoint:
+        from ..models import DeepSeekTransformer
+        from .loader import infer_config_from_state_dict, load_checkpoint_smart
+        payload = load_checkpoint_smart(args.draft_checkpoint)
+        dsd = payload["model_state_dict"]
+        draft = DeepSeekTransformer(infer_config_from_state_dict(dsd))
+        draft.load_state_dict(dsd, strict=False)
+        chat.set_draft_model(draft)
+        print(f"speculative decoding on (draft: {args.draft_checkpoint})")
+    chat.run()
 
 
 if __name__ == "__main__":

@@ -493,3 +493,22 @@ def test_quantize_moe_experts_int8(tiny_moe_config, tokenizer):
     toks = eng.generate(tokenizer.encode("hello"),
                         GenerationConfig(max_new_tokens=4, temperature=0.0))
     assert all(isinstance(t, int) for t in toks)
+
+
+def test_chat_with_draft_model(small_model, tokenizer):
+    """Chat replies through speculative decoding when a draft is set, and
+    greedy replies match the non-speculative path exactly."""
+    from luminaai_amd.models.transformer import DeepSeekConfig
+    chat = ChatInterface(model=small_model, tokenizer=tokenizer)
+    chat.gen_config = GenerationConfig(max_new_tokens=5, temperature=0.0,
+                                       stop_token_ids=[-1])
+    plain = chat.respond("hello there")
+    chat.history.clear()
+    torch.manual_seed(2)
+    draft = DeepSeekTransformer(DeepSeekConfig(
+        vocab_size=512, hidden_size=32, num_layers=1, num_heads=2,
+        num_kv_heads=1, intermediate_size=64, seq_length=64, use_moe=False,
+        use_mod=False, tie_word_embeddings=False)).eval()
+    chat.set_draft_model(draft)
+    spec = chat.respond("hello there")
+    assert spec == plain
