@@ -51,11 +51,6 @@ class ChatInterface:
             print(f"quantized {n} Linear layers to {quantize} "
                   f"({quantized_model_bytes(self.model) / 1e6:.1f} MB)")
 
-    def set_draft_model(self, draft) -> None:
-        """Enable speculative decoding: a small draft proposes, the main
-        model verifies (exact greedy / distribution-preserving sampling —
-        inference/engine.generate_speculative)."""
-        self.draft_model = draft.to(self.device).eval()
         self.tokenizer = tokenizer or ConversationTokenizer()
         self.engine = GenerationEngine(self.model, self.tokenizer, self.device)
         self.gen_config = GenerationConfig.from_mode("standard")
@@ -65,6 +60,12 @@ class ChatInterface:
         self.history: List[Dict] = []
         self.history_window = history_window
         self.session_stats = {"turns": 0, "start": time.time()}
+
+    def set_draft_model(self, draft) -> None:
+        """Enable speculative decoding: a small draft proposes, the main
+        model verifies (exact greedy / distribution-preserving sampling —
+        inference/engine.generate_speculative)."""
+        self.draft_model = draft.to(self.device).eval()
 
     # ------------------------------------------------------------------
     def _prompt_ids(self, user_text: str) -> List[int]:
