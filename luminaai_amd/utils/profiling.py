@@ -131,6 +131,48 @@ def reset_profiling_stats():
         _trace_events.clear()
 
 
+def attach_module_profilers(model, prefix: str = "layer"):
+    """Per-module forward timing (the reference decorated every module
+    forward, reference model.py:142-221; here it is opt-in so the hot loop
+    pays nothing by default). Returns hook handles; remove() them to
+    detach. Works with enable_profiling(True[, trace=True]) — each module
+    forward becomes a stat entry and (with trace) a chrome-trace slice."""
+    handles = []
+
+    def make_pre(name):
+        def pre(module, args):
+            if not _ENABLED:
+                return
+            if torch.cuda.is_available():
+                ev = torch.cuda.Event(enable_timing=True)
+                ev.record()
+                module.__dict__["_prof_start"] = ev
+            else:
+                module.__dict__["_prof_start"] = time.perf_counter()
+        return pre
+
+    def make_post(name):
+        def post(module, args, output):
+            start = module.__dict__.pop("_prof_start", None)
+            if not _ENABLED or start is None:
+                return
+            if torch.cuda.is_available():
+                ev = torch.cuda.Event(enable_timing=True)
+                ev.record()
+                with _lock:
+                    _pending.append((name, start, ev))
+            else:
+                t0 = start
+                _record(name, time.perf_counter() - t0, ts_s=t0)
+        return post
+
+    for i, layer in enumerate(getattr(model, "layers", [])):
+        name = f"{prefix}{i}"
+        handles.append(layer.register_forward_pre_hook(make_pre(name)))
+        handles.append(layer.register_forward_hook(make_post(name)))
+    return handles
+
+
 def export_chrome_trace(path: str) -> int:
     """Write collected slices as a chrome://tracing / Perfetto JSON file
     (ROADMAP: trace export for the decorator API). Requires

@@ -189,3 +189,26 @@ def test_chrome_trace_export(tmp_path):
     finally:
         enable_profiling(False)
         reset_profiling_stats()
+
+
+def test_module_profilers(small_model):
+    import torch
+    from luminaai_amd.utils.profiling import (attach_module_profilers,
+                                              enable_profiling,
+                                              get_profiling_stats,
+                                              reset_profiling_stats)
+    reset_profiling_stats()
+    enable_profiling(True)
+    handles = attach_module_profilers(small_model)
+    try:
+        with torch.no_grad():
+            small_model(torch.randint(0, 500, (1, 8)))
+        stats = get_profiling_stats()
+        layer_keys = [k for k in stats if k.startswith("layer")]
+        assert len(layer_keys) == len(small_model.layers)
+        assert all(stats[k]["calls"] == 1 for k in layer_keys)
+    finally:
+        for h in handles:
+            h.remove()
+        enable_profiling(False)
+        reset_profiling_stats()
