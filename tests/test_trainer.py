@@ -478,3 +478,52 @@ def test_profile_communication_flag(tiny_config, tokenizer):
     finally:
         enable_profiling(False)
         reset_profiling_stats()
+
+
+def test_adjust_batch_size_and_weight_decay(tiny_config, tokenizer):
+    from luminaai_amd.models import (DeepSeekTransformer,
+                                     config_to_deepseek_config)
+    from luminaai_amd.training import Trainer
+    torch.manual_seed(0)
+    t = Trainer(DeepSeekTransformer(config_to_deepseek_config(tiny_config)),
+                tokenizer, tiny_config)
+    t.adjust_batch_size(4)
+    assert t.config.micro_batch_size == 4
+    t.adjust_batch_size(0)       # clamps to 1
+    assert t.config.micro_batch_size == 1
+    t.adjust_weight_decay(0.05)
+    decayed = [g for g in t.optimizer.groups if g.weight_decay > 0]
+    assert decayed and all(g.weight_decay == 0.05 for g in decayed)
+    zero_wd = [g for g in t.optimizer.groups if g.weight_decay == 0.0]
+    assert zero_wd  # norms/biases stay decay-free
+    from luminaai_amd.data.dataset import SyntheticDataset
+    dl = t._recreate_dataloader(SyntheticDataset(tiny_config.vocab_size,
+                                                 tiny_config.seq_length, 8,
+                                                 seed=0))
+    assert dl.batch_size == 1
+
+
+def test_train_with_oom_fallback(tiny_config, tokenizer, monkeypatch):
+    """OOM in train() halves micro-batch / doubles accumulation and
+    retries (reference Main.py:292-501)."""
+    from luminaai_amd.models import (DeepSeekTransformer,
+                                     config_to_deepseek_config)
+    from luminaai_amd.training import Trainer
+    torch.manual_seed(0)
+    tiny_config.micro_batch_size = 4
+    t = Trainer(DeepSeekTransformer(config_to_deepseek_config(tiny_config)),
+                tokenizer, tiny_config)
+    calls = {"n": 0}
+
+    def fake_train(*a, **kw):
+        calls["n"] += 1
+        if calls["n"] < 3:
+            raise torch.cuda.OutOfMemoryError("synthetic OOM")
+        return {"global_step": 1}
+
+    monkeypatch.setattr(t, "train", fake_train)
+    monkeypatch.setattr(torch.cuda, "empty_cache", lambda: None)
+    out = t.train_with_oom_fallback()
+    assert out["global_step"] == 1
+    assert calls["n"] == 3
+    assert t.config.micro_batch_size == 1        # 4 -> 2 -> 1
