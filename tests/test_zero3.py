@@ -354,3 +354,76 @@ def test_zero3_elastic_resume_world1(tmp_path):
                                        atol=1e-5, msg=k)
     finally:
         os.chdir(cwd)
+
+
+def z3_ep_subworld_worker(rank, world):
+    """world=4, ep=2: expert replica groups have TWO members, so expert
+    segments really shard (the ep==world case degenerates to local
+    state). ZeRO-3 must still match ZeRO-0 on the same mesh."""
+    from luminaai_amd.config import Config
+    from luminaai_amd.data.tokenizer import ConversationTokenizer
+    from luminaai_amd.models import (DeepSeekTransformer,
+                                     config_to_deepseek_config)
+    from luminaai_amd.parallel.mesh import init_mesh, reset_mesh
+    from luminaai_amd.training import Trainer
+
+    def run(stage):
+        reset_mesh()
+        cfg = Config(vocab_size=512, hidden_size=64, num_layers=2,
+                     num_heads=4, num_kv_heads=2, seq_length=32,
+                     intermediate_size=128, micro_batch_size=2,
+                     gradient_accumulation_steps=1, num_workers=0,
+                     use_moe=True, num_experts=4, moe_top_k=2,
+                     routing_noise_std=0.0, use_mod=False,
+                     zero_stage=stage, precision="fp32",
+                     experiment_name=f"z3ep2_s{stage}_r{rank}",
+                     eval_every_n_batches=0, save_every_n_batches=0,
+                     gradient_checkpointing=False)
+        init_mesh(ep_size=2)
+        torch.manual_seed(1234)
+        model = DeepSeekTransformer(config_to_deepseek_config(cfg))
+        t = Trainer(model, ConversationTokenizer(), cfg)
+        t._setup_scheduler(10)
+        torch.manual_seed(800 + rank)
+        for _ in range(2):
+            ids = torch.randint(1, cfg.vocab_size, (2, cfg.seq_length + 1))
+            t.engine.set_sync(True)
+            t.train_step({"input_ids": ids[:, :-1], "labels": ids[:, 1:]})
+            t.optimizer_step()
+        if stage >= 3:
+            with t.engine.gathered_weights():
+                dense = t.model.embed_tokens.weight.detach().clone()
+                exp = t.model.layers[1].ffn.w_gate_up.detach().clone()
+        else:
+            dense = t.model.embed_tokens.weight.detach().clone()
+            exp = t.model.layers[1].ffn.w_gate_up.detach().clone()
+        if hasattr(t.engine, "remove_hooks"):
+            t.engine.remove_hooks()
+        return dense, exp
+
+    d0, e0 = run(0)
+    d3, e3 = run(3)
+    return {"dense_err": float((d0 - d3).abs().max()),
+            "exp_err": float((e0 - e3).abs().max())}
+
+
+def test_zero3_ep_subworld():
+    import torch.multiprocessing as mp
+    ctx = mp.get_context("spawn")
+    q = ctx.SimpleQueue()
+    port = _free_port()
+    procs = [ctx.Process(target=_run, args=(r, 4, port,
+                                            "z3_ep_subworld_worker", q))
+             for r in range(4)]
+    for p in procs:
+        p.start()
+    results = {}
+    for _ in range(4):
+        rank, status, payload = q.get()
+        assert status == "ok", f"rank {rank} failed:\n{payload}"
+        results[rank] = payload
+    for p in procs:
+        p.join(timeout=180)
+    for r in range(4):
+        assert results[r]["dense_err"] < 1e-4, results
+        assert results[r]["exp_err"] < 1e-4, results
