@@ -123,7 +123,7 @@ class _Zero3Segment:
             st.resize_(self.padded * self.flat_g.element_size())
         self.flat_g.zero_()
 
-    def gather(self, pg="unused"):
+    def gather(self):
         pg = self.pg
         st = self.flat_w.untyped_storage()
         if st.size() > 0:
@@ -137,7 +137,7 @@ class _Zero3Segment:
         else:
             dist.all_gather_into_tensor(self.flat_w, self.w_shard, group=pg)
 
-    def reduce_into_shard(self, pg="unused"):
+    def reduce_into_shard(self):
         """flat_g -> += g_shard (sum over ranks), then free flat_g."""
         pg = self.pg
         lo = self.rank * self.shard_size
@@ -159,7 +159,7 @@ class _Zero3Segment:
                      self.lr, 0.9, 0.95, 1e-8, self.weight_decay, step_count,
                      norm_sq, max_norm, grad_scale)
 
-    def refresh_from_shard(self, pg="unused"):
+    def refresh_from_shard(self):
         """w_shard -> flat_w (persistent units after an optimizer step)."""
         pg = self.pg
         if self.flat_w.untyped_storage().size() == 0:
@@ -220,7 +220,7 @@ class _Zero3Unit:
         self.n_params = sum(len(s.params) for s in self.segments)
         self._grads_pending = self.n_params
 
-    def gather(self, pg="unused"):
+    def gather(self):
         for s in self.segments:
             s.gather()
 
