@@ -739,3 +739,54 @@ def test_zero1_elastic_resume_world1(tmp_path):
         assert torch.isfinite(t.model.embed_tokens.weight.detach()).all()
     finally:
         _os.chdir(cwd)
+
+
+def zero1_elastic_grow_worker(rank, world):
+    """Resume a single-process (world=1) save at world=2 under ZeRO-1."""
+    from luminaai_amd.training import CheckpointManager
+    tmp = os.environ["Z1_GROW_TMP"]
+    os.chdir(tmp)
+    t, cfg = _make_trainer(rank, world, zero_stage=1)
+    t.checkpoints = CheckpointManager(os.path.join(tmp, "shared_ckpts"))
+    t.load_checkpoint("latest")
+    assert t.global_step == 2
+    m_sum = float(t.optimizer.groups[0].m.sum())
+    master_sum = float(t.optimizer.groups[0].master.sum())
+    # training continues after the reshard
+    torch.manual_seed(820)
+    ids = torch.randint(1, cfg.vocab_size, (2, cfg.seq_length + 1))
+    t.engine.set_sync(True)
+    t.train_step({"input_ids": ids[:, :-1], "labels": ids[:, 1:]})
+    t.optimizer_step()
+    return {"m_sum": m_sum, "master_sum": master_sum,
+            "step": t.optimizer.step_count}
+
+
+def test_zero1_elastic_resume_grow(tmp_path):
+    import os as _os
+    _os.environ["Z1_GROW_TMP"] = str(tmp_path)
+    cwd = _os.getcwd()
+    _os.chdir(tmp_path)
+    try:
+        from luminaai_amd.training import CheckpointManager
+        t, cfg = _make_trainer(0, 1, zero_stage=1)
+        t.checkpoints = CheckpointManager(str(tmp_path / "shared_ckpts"))
+        torch.manual_seed(810)
+        for _ in range(2):
+            ids = torch.randint(1, cfg.vocab_size, (2, cfg.seq_length + 1))
+            t.engine.set_sync(True)
+            t.train_step({"input_ids": ids[:, :-1], "labels": ids[:, 1:]})
+            t.optimizer_step()
+        t.save_checkpoint()
+        g = t.optimizer.groups[0]
+        full_m = float(g.m.sum())
+        full_master = float(g.master.sum())
+    finally:
+        _os.chdir(cwd)
+    res = _spawn("zero1_elastic_grow_worker")
+    # the two world-2 shards partition the world-1 state exactly
+    assert res[0]["m_sum"] + res[1]["m_sum"] == pytest.approx(full_m,
+                                                              rel=1e-5)
+    assert res[0]["master_sum"] + res[1]["master_sum"] == \
+        pytest.approx(full_master, rel=1e-5)
+    assert res[0]["step"] == 3
