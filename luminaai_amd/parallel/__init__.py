@@ -1,4 +1,5 @@
 from . import comm
+from .context_parallel import ring_attention
 from .expert_parallel import all_to_all, is_expert_param, sync_expert_grads
 from .mesh import ParallelMesh, get_mesh, init_mesh, reset_mesh
 from .pipeline import (InterleavedPipelineEngine, PipelineParallelEngine,
@@ -13,6 +14,6 @@ __all__ = ["InterleavedPipelineEngine", "ParallelMesh",
            "PipelineParallelEngine", "PipelineStage",
            "Zero3Engine", "ZeroEngine", "all_to_all", "comm",
            "convert_to_tensor_parallel", "create_engine", "get_mesh",
-           "init_mesh", "is_expert_param", "partition_layers", "reset_mesh",
+           "init_mesh", "is_expert_param", "partition_layers", "reset_mesh", "ring_attention",
            "scatter_heads_gather_seq", "scatter_seq_gather_heads",
            "shard_sequence", "sync_expert_grads", "tp_copy", "tp_reduce"]
