@@ -68,6 +68,10 @@ def build_arg_parser() -> argparse.ArgumentParser:
                          "schedule)")
     ap.add_argument("--tp", type=int, default=1,
                     help="tensor-parallel degree (dense blocks)")
+    ap.add_argument("--sp-mode", default="ulysses",
+                    choices=["ulysses", "ring"],
+                    help="sequence-parallel exchange: head<->seq all-to-all "
+                         "or ring attention (no head-count limit)")
     ap.add_argument("--sp", type=int, default=1,
                     help="Ulysses sequence-parallel degree")
     ap.add_argument("--precision", default=None,
@@ -131,7 +135,8 @@ def main(argv: Optional[list] = None) -> dict:
             and args.sp <= 1 and cfg.num_experts % world == 0:
         ep = world
     if world > 1:
-        mesh = init_mesh(ep, sp_size=args.sp, tp_size=args.tp)
+        mesh = init_mesh(ep, sp_size=args.sp, tp_size=args.tp,
+                         sp_mode=args.sp_mode)
     else:
         mesh = init_mesh(1)
 

@@ -349,8 +349,11 @@ class GroupedQueryAttention(nn.Module):
         mesh = get_mesh()
         self.sp_size = mesh.sp_size if mesh is not None else 1
         self.sp_group = mesh.sp_group if mesh is not None else None
+        self.sp_mode = getattr(mesh, "sp_mode", "ulysses") \
+            if mesh is not None else "ulysses"
         self.tp_group = None    # set by convert_to_tensor_parallel
-        if self.sp_size > 1:
+        if self.sp_size > 1 and self.sp_mode == "ulysses":
+            # ring mode has no head-count constraint (that is its point)
             assert self.num_heads % self.sp_size == 0 and \
                 self.num_kv_heads % self.sp_size == 0, \
                 "num_heads and num_kv_heads must divide sp_size"
@@ -384,8 +387,27 @@ class GroupedQueryAttention(nn.Module):
                 attn_mask = torch.zeros(1, 1, S, L, device=x.device,
                                         dtype=x.dtype).masked_fill(~live, -1e4)
 
-        run_sp = self.sp_size > 1 and kv_cache is None and S > 1 \
-            and attn_mask is None
+        if self.sp_size > 1 and self.sp_mode == "ring" \
+                and kv_cache is None and S > 1 and attn_mask is None:
+            # ring context parallelism: K/V blocks travel the xGMI ring and
+            # softmax accumulates blockwise (parallel/context_parallel.py).
+            # Scales past the Ulysses num_kv_heads ceiling; GQA KV heads
+            # are expanded for the exchange.
+            from ..parallel.context_parallel import ring_attention
+            rep = self.num_heads // self.num_kv_heads
+            kt = k.repeat_interleave(rep, dim=2).transpose(1, 2)
+            vt = v.repeat_interleave(rep, dim=2).transpose(1, 2)
+            out = ring_attention(q.transpose(1, 2), kt, vt,
+                                 group=self.sp_group, causal=True)
+            out = out.transpose(1, 2).reshape(B, S, self.q_size)
+            out = _linear(self.o_proj, out)
+            if self.tp_group is not None:
+                from ..parallel.tensor_parallel import tp_reduce
+                out = tp_reduce(out, self.tp_group)
+            return out
+
+        run_sp = self.sp_size > 1 and self.sp_mode == "ulysses" \
+            and kv_cache is None and S > 1 and attn_mask is None
         if run_sp:
             from ..parallel.sequence_parallel import (
                 scatter_heads_gather_seq, scatter_seq_gather_heads)

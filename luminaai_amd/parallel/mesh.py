@@ -28,7 +28,8 @@ _MESH: Optional["ParallelMesh"] = None
 
 
 class ParallelMesh:
-    def __init__(self, ep_size: int = 1, sp_size: int = 1, tp_size: int = 1):
+    def __init__(self, ep_size: int = 1, sp_size: int = 1, tp_size: int = 1,
+                 sp_mode: str = "ulysses"):
         world = comm.get_world_size()
         rank = comm.get_rank()
         for name, sz in (("ep", ep_size), ("sp", sp_size), ("tp", tp_size)):
@@ -40,6 +41,12 @@ class ParallelMesh:
         self.rank = rank
         self.ep_size = ep_size
         self.sp_size = sp_size
+        # sequence-parallel exchange pattern (reference shardformer
+        # SUPPORT_SP_MODE, shard_config.py:13): "ulysses" = head<->sequence
+        # all-to-all; "ring" = blockwise ring attention
+        # (parallel/context_parallel.py) for sp_size beyond num_kv_heads
+        assert sp_mode in ("ulysses", "ring")
+        self.sp_mode = sp_mode
         self.tp_size = tp_size
         self.dp_size = world // (ep_size * max(sp_size, 1) * max(tp_size, 1))
         self.ep_rank = rank % ep_size if ep_size > 1 else 0
@@ -112,9 +119,9 @@ class ParallelMesh:
 
 
 def init_mesh(ep_size: int = 1, sp_size: int = 1,
-              tp_size: int = 1) -> ParallelMesh:
+              tp_size: int = 1, sp_mode: str = "ulysses") -> ParallelMesh:
     global _MESH
-    _MESH = ParallelMesh(ep_size, sp_size, tp_size)
+    _MESH = ParallelMesh(ep_size, sp_size, tp_size, sp_mode=sp_mode)
     return _MESH
 
 

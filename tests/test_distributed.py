@@ -790,3 +790,60 @@ def test_zero1_elastic_resume_grow(tmp_path):
     assert res[0]["master_sum"] + res[1]["master_sum"] == \
         pytest.approx(full_master, rel=1e-5)
     assert res[0]["step"] == 3
+
+
+# ---- ring sequence parallelism ---------------------------------------------
+def sp_ring_worker(rank, world):
+    """sp_mode="ring": forward AND replicated-param grads must match the
+    full-sequence single-rank model — including head counts that do NOT
+    divide sp (Ulysses' hard limit)."""
+    from luminaai_amd.models import DeepSeekTransformer
+    from luminaai_amd.models.transformer import DeepSeekConfig
+    from luminaai_amd.parallel.mesh import init_mesh, reset_mesh
+    from luminaai_amd.ops import fused_cross_entropy
+
+    # 3 heads / 1 kv head: indivisible by sp=2 -> only ring can shard this
+    mcfg = DeepSeekConfig(vocab_size=512, hidden_size=48, num_layers=2,
+                          num_heads=3, num_kv_heads=1, intermediate_size=96,
+                          seq_length=64, use_moe=False, use_mod=False,
+                          tie_word_embeddings=False)
+    reset_mesh()
+    torch.manual_seed(1234)
+    full = DeepSeekTransformer(mcfg)
+
+    init_mesh(sp_size=world, sp_mode="ring")
+    torch.manual_seed(1234)
+    sp_model = DeepSeekTransformer(mcfg)
+    with torch.no_grad():
+        for p_sp, p_f in zip(sp_model.parameters(), full.parameters()):
+            p_sp.copy_(p_f)
+
+    torch.manual_seed(777)
+    ids = torch.randint(1, mcfg.vocab_size, (2, 65))
+    inp, lab = ids[:, :-1], ids[:, 1:]
+    S_loc = 64 // world
+    lo, hi = rank * S_loc, (rank + 1) * S_loc
+
+    logits_f, _, _ = full(inp)
+    ce_f, _, _ = fused_cross_entropy(logits_f, lab)
+    ce_f.backward()
+
+    logits_sp, _, _ = sp_model(inp[:, lo:hi])
+    torch.testing.assert_close(logits_sp, logits_f[:, lo:hi].detach(),
+                               rtol=1e-4, atol=1e-4)
+    ce_sp, _, _ = fused_cross_entropy(logits_sp, lab[:, lo:hi])
+    # full-model CE averages over ALL tokens; local CE over the shard —
+    # rescale so the grad contribution matches, then sum across ranks
+    (ce_sp / world).backward()
+    g_local = sp_model.embed_tokens.weight.grad.clone()
+    dist.all_reduce(g_local)
+    err = float((g_local - full.embed_tokens.weight.grad).abs().max()
+                / full.embed_tokens.weight.grad.abs().max().clamp_min(1e-12))
+    reset_mesh()
+    return {"grad_err": err}
+
+
+def test_sp_ring_matches_full_model():
+    res = _spawn("sp_ring_worker")
+    for r in range(2):
+        assert res[r]["grad_err"] < 1e-3, res
