@@ -99,3 +99,19 @@ def test_continuous_compaction_equivalence(small_model, tokenizer):
                                    max_batch=8, max_len=64)
     out = eng.run_to_completion([p], _cfg(5))
     assert out == [ref]
+
+
+def test_continuous_with_quantized_model(tiny_moe_config, tokenizer):
+    """Serving combos: int8-quantized MoE model through continuous
+    batching."""
+    from luminaai_amd.models import (DeepSeekTransformer,
+                                     config_to_deepseek_config)
+    from luminaai_amd.ops.quant import quantize_model
+    torch.manual_seed(0)
+    m = DeepSeekTransformer(config_to_deepseek_config(tiny_moe_config)).eval()
+    quantize_model(m, mode="int8", min_dim=32)
+    eng = ContinuousBatchingEngine(m, tokenizer, max_batch=2, max_len=48)
+    outs = eng.run_to_completion([tokenizer.encode("quantized"),
+                                  tokenizer.encode("serving")], _cfg(4))
+    assert len(outs) == 2
+    assert all(isinstance(t, int) for o in outs for t in o)

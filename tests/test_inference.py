@@ -512,3 +512,24 @@ def test_chat_with_draft_model(small_model, tokenizer):
     chat.set_draft_model(draft)
     spec = chat.respond("hello there")
     assert spec == plain
+
+
+def test_speculative_with_quantized_target(small_model, tokenizer):
+    """int8 target + fp draft still decodes (quantization changes the
+    target's argmax sequence, so just check validity + determinism)."""
+    import copy
+    from luminaai_amd.models.transformer import DeepSeekConfig
+    from luminaai_amd.ops.quant import quantize_model
+    target = copy.deepcopy(small_model).eval()
+    quantize_model(target, mode="int8", min_dim=32)
+    torch.manual_seed(5)
+    draft = DeepSeekTransformer(DeepSeekConfig(
+        vocab_size=512, hidden_size=32, num_layers=1, num_heads=2,
+        num_kv_heads=1, intermediate_size=64, seq_length=64, use_moe=False,
+        use_mod=False, tie_word_embeddings=False)).eval()
+    eng = GenerationEngine(target, tokenizer)
+    cfg = GenerationConfig(max_new_tokens=5, temperature=0.0,
+                           stop_token_ids=[-1])
+    a = eng.generate_speculative(tokenizer.encode("q"), draft, cfg)
+    b = eng.generate(tokenizer.encode("q"), cfg)
+    assert a == b  # speculative == plain greedy on the SAME (int8) target
