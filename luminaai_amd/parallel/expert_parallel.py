@@ -59,7 +59,9 @@ def is_expert_param(name: str) -> bool:
 
 def sync_expert_grads(model, expert_dp_group):
     """All-reduce expert-weight grads across replicas owning the same shard
-    (no-op when expert_dp_size == 1). Used by the engine when EP is active."""
+    (no-op when expert_dp_size == 1). Standalone utility for custom
+    training loops; the ZeRO engines do this on their flat buffers
+    (zero.py reduce_gradients / zero3.py expert segments) instead."""
     if expert_dp_group is None:
         return
     for name, p in model.named_parameters():

@@ -847,3 +847,32 @@ def test_sp_ring_matches_full_model():
     res = _spawn("sp_ring_worker")
     for r in range(2):
         assert res[r]["grad_err"] < 1e-3, res
+
+
+def sync_expert_grads_worker(rank, world):
+    """Standalone sync_expert_grads utility: expert grads all-reduce over
+    the given replica group, dense grads untouched."""
+    from luminaai_amd.models import DeepSeekTransformer
+    from luminaai_amd.models.transformer import DeepSeekConfig
+    from luminaai_amd.parallel.expert_parallel import sync_expert_grads
+    torch.manual_seed(1234)
+    m = DeepSeekTransformer(DeepSeekConfig(
+        vocab_size=128, hidden_size=32, num_layers=1, num_heads=2,
+        num_kv_heads=1, intermediate_size=64, seq_length=16, use_moe=True,
+        num_experts=2, moe_top_k=1, routing_noise_std=0.0,
+        moe_pattern="all", dense_start_layers=0, use_mod=False,
+        tie_word_embeddings=False))
+    for name, p in m.named_parameters():
+        p.grad = torch.full_like(p, float(rank + 1))
+    sync_expert_grads(m, dist.group.WORLD)
+    gu = m.layers[0].ffn.w_gate_up.grad
+    emb = m.embed_tokens.weight.grad
+    return {"expert_g": float(gu.flatten()[0]),
+            "dense_g": float(emb.flatten()[0])}
+
+
+def test_sync_expert_grads_utility():
+    res = _spawn("sync_expert_grads_worker")
+    for r in range(2):
+        assert res[r]["expert_g"] == pytest.approx(3.0)  # 1 + 2 summed
+        assert res[r]["dense_g"] == pytest.approx(r + 1.0)  # untouched
