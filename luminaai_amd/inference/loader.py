@@ -97,6 +97,58 @@ def merge_ep_checkpoints(paths: List[str], map_location="cpu") -> Dict:
     return out
 
 
+def merge_pp_checkpoints(paths: List[str], map_location="cpu") -> Dict:
+    """Reassemble per-stage pipeline checkpoints (pp_stage_rank{r}.pt from
+    training/pipeline_loop.py) into one full model_state_dict. Handles the
+    1F1B layout (one PipelineStage per rank) and interleaved virtual
+    stages (a ModuleList of chunks per rank, global stage s = c*pp + r);
+    local layer indices are re-based onto the global layer numbering."""
+    def _rank(p):
+        m = re.search(r"pp_stage_rank(\d+)", os.path.basename(p))
+        return int(m.group(1)) if m else 0
+
+    paths = sorted(paths, key=_rank)
+    payloads = [torch.load(p, map_location=map_location, weights_only=False)
+                for p in paths]
+    pp = payloads[0].get("pp_world", len(payloads))
+    v = payloads[0].get("virtual_stages", 1)
+    if len(payloads) != pp:
+        raise ValueError(f"need all {pp} stage files, got {len(payloads)}")
+
+    # chunk (global stage) -> its layer key prefix and state source
+    chunks: List[Dict] = [None] * (pp * v)
+    for r, pl in enumerate(payloads):
+        sd = pl["stage_state_dict"]
+        if v == 1:
+            chunks[r] = sd
+        else:
+            for c in range(v):
+                pref = f"{c}."
+                chunks[c * pp + r] = {k[len(pref):]: t for k, t in sd.items()
+                                      if k.startswith(pref)}
+    merged: Dict[str, torch.Tensor] = {}
+    layer_off = 0
+    for s, sd in enumerate(chunks):
+        n_local = 0
+        for k, t in sd.items():
+            if k.startswith("layers."):
+                rest = k.split(".", 2)
+                li = int(rest[1])
+                n_local = max(n_local, li + 1)
+                merged[f"layers.{layer_off + li}.{rest[2]}"] = t
+            else:   # embed (stage 0), final_norm/lm_head (last stage)
+                merged[k] = t
+        layer_off += n_local
+    return {"model_state_dict": merged,
+            "global_step": payloads[0].get("global_step", 0)}
+
+
+def find_pp_stages(any_stage_path: str) -> List[str]:
+    """All sibling pp_stage_rank* files of one pipeline checkpoint."""
+    base = re.sub(r"pp_stage_rank\d+", "pp_stage_rank*", any_stage_path)
+    return sorted(glob.glob(base))
+
+
 def find_ep_shards(any_shard_path: str) -> List[str]:
     """All sibling _ep_rank_* files of one EP checkpoint shard."""
     base = re.sub(r"_ep_rank_\d+", "_ep_rank_*", any_shard_path)
