@@ -33,6 +33,10 @@ def load_checkpoint_smart(path: str, map_location="cpu") -> Dict:
     """Load a checkpoint file; returns {"model_state_dict": ..., ...} with
     wrapper prefixes stripped. Accepts raw state dicts too."""
     payload = torch.load(path, map_location=map_location, weights_only=False)
+    if isinstance(payload, dict) and "stage_state_dict" in payload:
+        # pipeline-parallel per-stage save: merge all sibling stages
+        return merge_pp_checkpoints(find_pp_stages(path),
+                                    map_location=map_location)
     if isinstance(payload, dict) and "model_state_dict" in payload:
         payload["model_state_dict"] = _strip_prefixes(payload["model_state_dict"])
         return payload

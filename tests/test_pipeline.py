@@ -435,3 +435,30 @@ def test_merge_pp_checkpoints_roundtrip(tmp_path, virtual):
     for k, vref in ref_sd.items():
         torch.testing.assert_close(merged["model_state_dict"][k], vref,
                                    msg=k)
+
+
+def test_load_checkpoint_smart_handles_pp_stage(tmp_path):
+    """Pointing the smart loader at ANY stage file reassembles the full
+    model automatically (chat.py / serve.py just work on PP output)."""
+    from luminaai_amd.inference.loader import load_checkpoint_smart
+    from luminaai_amd.models import DeepSeekTransformer
+    from luminaai_amd.parallel.pipeline import (PipelineStage,
+                                                partition_layers)
+    mcfg = _model_cfg()
+    torch.manual_seed(11)
+    model = DeepSeekTransformer(mcfg)
+    bounds = partition_layers(mcfg.num_layers, 2)
+    for r in range(2):
+        lo, hi = bounds[r]
+        stage = PipelineStage(model, lo, hi, is_first=r == 0,
+                              is_last=r == 1)
+        torch.save({"stage_state_dict": stage.state_dict(), "pp_rank": r,
+                    "pp_world": 2, "virtual_stages": 1, "global_step": 4},
+                   tmp_path / f"pp_stage_rank{r}.pt")
+    payload = load_checkpoint_smart(str(tmp_path / "pp_stage_rank1.pt"))
+    m2 = DeepSeekTransformer(mcfg)
+    missing, unexpected = m2.load_state_dict(payload["model_state_dict"],
+                                             strict=False)
+    assert not unexpected
+    for k, v in model.state_dict().items():
+        torch.testing.assert_close(payload["model_state_dict"][k], v, msg=k)
