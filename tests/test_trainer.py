@@ -527,3 +527,61 @@ def test_train_with_oom_fallback(tiny_config, tokenizer, monkeypatch):
     assert out["global_step"] == 1
     assert calls["n"] == 3
     assert t.config.micro_batch_size == 1        # 4 -> 2 -> 1
+
+
+def test_loss_curve_regression(tiny_config, tokenizer):
+    """Fixed-seed 30-step run must actually LEARN (loss drops >30% and
+    lands under an absolute bound) — catches silent training-math
+    regressions (a gap in the reference's own suite)."""
+    from luminaai_amd.models import (DeepSeekTransformer,
+                                     config_to_deepseek_config)
+    from luminaai_amd.training import Trainer
+    tiny_config.learning_rate = 3e-3   # tiny model memorising 4 batches
+    torch.manual_seed(1234)
+    model = DeepSeekTransformer(config_to_deepseek_config(tiny_config))
+    t = Trainer(model, tokenizer, tiny_config)
+    t._setup_scheduler(80)
+    torch.manual_seed(4321)
+    # a SMALL fixed corpus, repeated: the model must memorise it
+    data = [torch.randint(1, tiny_config.vocab_size,
+                          (2, tiny_config.seq_length + 1)) for _ in range(4)]
+    losses = []
+    for step in range(60):
+        ids = data[step % len(data)]
+        t.engine.set_sync(True)
+        out = t.train_step({"input_ids": ids[:, :-1], "labels": ids[:, 1:]})
+        t.optimizer_step()
+        losses.append(float(out["ce_loss"]))
+    first = sum(losses[:4]) / 4
+    last = sum(losses[-4:]) / 4
+    assert last < first * 0.7, (first, last)
+    assert last < 5.0, losses[-4:]
+
+
+def test_training_does_not_leak_objects(tiny_config, tokenizer):
+    """Steady-state steps must not grow live Python objects (reference
+    test_performance.py:108-121 gc leak check)."""
+    import gc
+    from luminaai_amd.models import (DeepSeekTransformer,
+                                     config_to_deepseek_config)
+    from luminaai_amd.training import Trainer
+    torch.manual_seed(0)
+    model = DeepSeekTransformer(config_to_deepseek_config(tiny_config))
+    t = Trainer(model, tokenizer, tiny_config)
+    ids = torch.randint(1, tiny_config.vocab_size,
+                        (2, tiny_config.seq_length + 1))
+    batch = {"input_ids": ids[:, :-1], "labels": ids[:, 1:]}
+
+    def steps(n):
+        for _ in range(n):
+            t.engine.set_sync(True)
+            t.train_step(batch)
+            t.optimizer_step()
+
+    steps(5)                 # warm the caches
+    gc.collect()
+    before = len(gc.get_objects())
+    steps(10)
+    gc.collect()
+    after = len(gc.get_objects())
+    assert after - before < 500, (before, after)
