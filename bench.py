@@ -179,6 +179,13 @@ def main():
                 "final_loss": loss,
             },
         }
+        # MoE routing health over the run (outside the timed region)
+        stats = trainer._extract_moe_routing_stats()
+        if stats:
+            result["config"]["expert_imbalance"] = round(
+                stats["max_imbalance"], 3)
+            result["config"]["expert_utilization"] = round(
+                stats["mean_utilization"], 3)
         print(json.dumps(result))
     comm.cleanup()
 
