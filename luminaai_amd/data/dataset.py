@@ -270,15 +270,36 @@ def _load_texts(path: str) -> List[str]:
             else:
                 out.append(str(obj))
         return out
-    if p.suffix == ".parquet" and _HAS_HF:
-        ds = hf_datasets.load_dataset("parquet", data_files=str(p), split="train")
-        return [r.get("text", "") for r in ds]
+    if p.suffix == ".parquet":
+        try:
+            import pyarrow.parquet as pq
+            table = pq.read_table(str(p))
+            col = "text" if "text" in table.column_names \
+                else table.column_names[0]
+            return [str(x) for x in table.column(col).to_pylist()
+                    if x is not None]
+        except ImportError:
+            if _HAS_HF:
+                ds = hf_datasets.load_dataset("parquet", data_files=str(p),
+                                              split="train")
+                return [r.get("text", "") for r in ds]
+            raise
     raise ValueError(f"unsupported data file {path}")
 
 
 def _iter_texts(path: str) -> Iterator[str]:
     p = Path(path)
-    if p.suffix == ".txt":
+    if p.suffix == ".parquet":
+        # row-group streaming: memory stays bounded for >10 GB files
+        import pyarrow.parquet as pq
+        pf = pq.ParquetFile(str(p))
+        col = "text" if "text" in pf.schema_arrow.names \
+            else pf.schema_arrow.names[0]
+        for batch in pf.iter_batches(columns=[col]):
+            for x in batch.column(0).to_pylist():
+                if x:
+                    yield str(x)
+    elif p.suffix == ".txt":
         with open(p, errors="replace") as f:
             for line in f:
                 if line.strip():

@@ -1,5 +1,6 @@
 """Tokenizer + dataset pipeline tests."""
 
+import pytest
 import torch
 
 from luminaai_amd.data import (
@@ -194,3 +195,24 @@ def test_streaming_dataset_worker_sharding(tokenizer, sample_text):
     # the tails; the essential property is no duplication blow-up
     assert sum(cm.values()) <= sum(cs.values()) + 2
     assert max(cm.values()) <= max(cs.values())
+
+
+def test_parquet_dataset(tmp_path, tokenizer):
+    """Parquet corpora load (pyarrow direct) and stream (row groups)."""
+    pa = pytest.importorskip("pyarrow")
+    import pyarrow.parquet as pq
+    from luminaai_amd.data.dataset import (BaseTrainingDataset,
+                                           StreamingBaseTrainingDataset)
+    texts = [f"document number {i} with several words of content {i}"
+             for i in range(12)]
+    p = str(tmp_path / "corpus.parquet")
+    pq.write_table(pa.table({"text": texts}), p, row_group_size=4)
+    ds = BaseTrainingDataset(p, tokenizer, seq_length=16)
+    assert len(ds) > 0
+    row = ds[0]
+    assert row["input_ids"].shape == (16,)
+    stream = StreamingBaseTrainingDataset(p, tokenizer, seq_length=16)
+    rows = list(stream)
+    assert len(rows) > 0
+    # map-style and streaming chunk the same token stream
+    assert torch.equal(rows[0]["input_ids"], ds[0]["input_ids"])
