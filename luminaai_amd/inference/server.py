@@ -4,9 +4,12 @@ KV-cached GenerationEngine.
 The reference ships only a terminal REPL (reference Chat.py:472-937); this
 module adds the serving path a production deployment needs: an ASGI app
 with /v1/completions, /v1/chat/completions (ChatML via the tokenizer's
-conversation encoding), SSE streaming, /health and /v1/models. Decode is
-serialised with a lock (one hipGraph replay chain owns the GPU at a time);
-continuous batching is a round-2 item (ROADMAP.md).
+conversation encoding), SSE streaming, /health, /v1/models and a
+Prometheus /metrics endpoint. The default serving mode is token-level
+continuous batching (inference/continuous.py — requests join the running
+decode batch between steps; a lone request pays batch-1 compute);
+serving="dynamic" keeps request-level batching with the hipGraph
+single-stream path for streaming.
 
 Run: python serve.py --checkpoint PATH [--port 8000] [--quantize int8]
 """
