@@ -10,8 +10,11 @@ running max/denominator so the result is EXACTLY full attention over the
 global sequence.
 
 This module is the primitive (forward + backward as an autograd.Function,
-bit-checked against full SDPA in tests/test_context_parallel.py);
-attention-layer integration behind a mesh cp_size is round-2 (ROADMAP).
+bit-checked against full attention in tests/test_context_parallel.py);
+GroupedQueryAttention uses it when the mesh is built with
+sp_mode="ring" (`--sp-mode ring`), which lifts Ulysses' head-count
+divisibility limit. Fusing the block step with a CDNA4 flash kernel and
+overlapping the ring hop with compute are round-2 (ROADMAP).
 
 Layout: q, k, v are [B, H, S_local, D]; the global sequence is the rank-
 order concatenation. Causality is block-causal: a K/V block from rank
