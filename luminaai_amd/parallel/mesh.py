@@ -36,7 +36,7 @@ class ParallelMesh:
             assert world % max(sz, 1) == 0, \
                 f"world {world} not divisible by {name}_size {sz}"
         assert sum(s > 1 for s in (ep_size, sp_size, tp_size)) <= 1, \
-            "EP / Ulysses-SP / TP composition is not supported yet"
+            "EP / SP / TP are mutually exclusive in the mesh (PPxDP and ZeRO-3+EP compose elsewhere)"
         self.world = world
         self.rank = rank
         self.ep_size = ep_size
