@@ -341,6 +341,12 @@ def main(argv=None):
     ap.add_argument("--device", default=None)
     ap.add_argument("--quantize", default=None,
                     choices=["int8", "int4", "fp8"])
+    ap.add_argument("--max-batch", type=int, default=8,
+                    help="continuous-batching slot count")
+    ap.add_argument("--max-len", type=int, default=2048,
+                    help="serving context window (KV cache columns)")
+    ap.add_argument("--serving", default="continuous",
+                    choices=["continuous", "dynamic"])
     args = ap.parse_args(argv)
 
     device = torch.device(args.device) if args.device else (
@@ -362,7 +368,9 @@ def main(argv=None):
         n = quantize_model(model, mode=args.quantize)
         print(f"quantized {n} Linear layers to {args.quantize}")
 
-    app = create_app(model, ConversationTokenizer())
+    app = create_app(model, ConversationTokenizer(),
+                     max_batch=args.max_batch, serving=args.serving,
+                     max_len=args.max_len)
     import uvicorn
     uvicorn.run(app, host=args.host, port=args.port, log_level="info")
 
