@@ -876,3 +876,41 @@ def test_sync_expert_grads_utility():
     for r in range(2):
         assert res[r]["expert_g"] == pytest.approx(3.0)  # 1 + 2 summed
         assert res[r]["dense_g"] == pytest.approx(r + 1.0)  # untouched
+
+
+def sp_ring_trainer_worker(rank, world):
+    """Ring SP through the full Trainer (data sharding + optimizer):
+    replicas stay in sync after steps."""
+    from luminaai_amd.config import Config
+    from luminaai_amd.data.tokenizer import ConversationTokenizer
+    from luminaai_amd.models import (DeepSeekTransformer,
+                                     config_to_deepseek_config)
+    from luminaai_amd.parallel.mesh import init_mesh, reset_mesh
+    from luminaai_amd.training import Trainer
+    cfg = Config(vocab_size=512, hidden_size=64, num_layers=2, num_heads=4,
+                 num_kv_heads=2, seq_length=64, intermediate_size=128,
+                 micro_batch_size=2, gradient_accumulation_steps=1,
+                 num_workers=0, use_moe=False, use_mod=False,
+                 zero_stage=0, precision="fp32",
+                 experiment_name=f"spr_t_{rank}",
+                 eval_every_n_batches=0, save_every_n_batches=0)
+    init_mesh(sp_size=world, sp_mode="ring")
+    torch.manual_seed(1234)
+    model = DeepSeekTransformer(config_to_deepseek_config(cfg))
+    t = Trainer(model, ConversationTokenizer(), cfg)
+    t._setup_scheduler(10)
+    torch.manual_seed(911)             # same batch on every SP rank
+    for _ in range(2):
+        ids = torch.randint(1, cfg.vocab_size, (2, cfg.seq_length + 1))
+        t.engine.set_sync(True)
+        out = t.train_step({"input_ids": ids[:, :-1], "labels": ids[:, 1:]})
+        t.optimizer_step()
+    w = float(t.model.embed_tokens.weight.detach().sum())
+    loss = float(out["ce_loss"])
+    reset_mesh()
+    return {"w": w, "loss": loss}
+
+
+def test_sp_ring_through_trainer():
+    res = _spawn("sp_ring_trainer_worker")
+    assert res[0]["w"] == pytest.approx(res[1]["w"], abs=1e-4)
