@@ -914,3 +914,52 @@ def sp_ring_trainer_worker(rank, world):
 def test_sp_ring_through_trainer():
     res = _spawn("sp_ring_trainer_worker")
     assert res[0]["w"] == pytest.approx(res[1]["w"], abs=1e-4)
+
+
+def sp_ring_ckpt_worker(rank, world):
+    """Ring SP + activation checkpointing: the recompute re-runs the ring
+    exchange inside backward — collective order must stay aligned."""
+    from luminaai_amd.models import DeepSeekTransformer
+    from luminaai_amd.models.transformer import DeepSeekConfig
+    from luminaai_amd.parallel.mesh import init_mesh, reset_mesh
+    from luminaai_amd.ops import fused_cross_entropy
+    mcfg = DeepSeekConfig(vocab_size=512, hidden_size=64, num_layers=2,
+                          num_heads=4, num_kv_heads=2, intermediate_size=128,
+                          seq_length=64, use_moe=False, use_mod=False,
+                          gradient_checkpointing=True,
+                          tie_word_embeddings=False)
+    reset_mesh()
+    torch.manual_seed(1234)
+    full = DeepSeekTransformer(mcfg)
+    init_mesh(sp_size=world, sp_mode="ring")
+    torch.manual_seed(1234)
+    m = DeepSeekTransformer(mcfg)
+    with torch.no_grad():
+        for a, b in zip(m.parameters(), full.parameters()):
+            a.copy_(b)
+    m.train()
+    full.train()
+    torch.manual_seed(777)
+    ids = torch.randint(1, mcfg.vocab_size, (2, 65))
+    inp, lab = ids[:, :-1], ids[:, 1:]
+    S_loc = 64 // world
+    lo, hi = rank * S_loc, (rank + 1) * S_loc
+
+    lf, _, _ = full(inp)
+    cf, _, _ = fused_cross_entropy(lf, lab)
+    cf.backward()
+    ls, _, _ = m(inp[:, lo:hi])
+    cs, _, _ = fused_cross_entropy(ls, lab[:, lo:hi])
+    (cs / world).backward()
+    g = m.embed_tokens.weight.grad.clone()
+    dist.all_reduce(g)
+    err = float((g - full.embed_tokens.weight.grad).abs().max()
+                / full.embed_tokens.weight.grad.abs().max().clamp_min(1e-12))
+    reset_mesh()
+    return {"err": err}
+
+
+def test_sp_ring_with_checkpointing():
+    res = _spawn("sp_ring_ckpt_worker")
+    for r in range(2):
+        assert res[r]["err"] < 1e-3, res
