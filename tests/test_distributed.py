@@ -916,9 +916,7 @@ def test_sp_ring_through_trainer():
     assert res[0]["w"] == pytest.approx(res[1]["w"], abs=1e-4)
 
 
-def sp_ring_ckpt_worker(rank, world):
-    """Ring SP + activation checkpointing: the recompute re-runs the ring
-    exchange inside backward — collective order must stay aligned."""
+def _sp_ckpt_body(rank, world, mode):
     from luminaai_amd.models import DeepSeekTransformer
     from luminaai_amd.models.transformer import DeepSeekConfig
     from luminaai_amd.parallel.mesh import init_mesh, reset_mesh
@@ -931,7 +929,7 @@ def sp_ring_ckpt_worker(rank, world):
     reset_mesh()
     torch.manual_seed(1234)
     full = DeepSeekTransformer(mcfg)
-    init_mesh(sp_size=world, sp_mode="ring")
+    init_mesh(sp_size=world, sp_mode=mode)
     torch.manual_seed(1234)
     m = DeepSeekTransformer(mcfg)
     with torch.no_grad():
@@ -959,7 +957,25 @@ def sp_ring_ckpt_worker(rank, world):
     return {"err": err}
 
 
+def sp_ring_ckpt_worker(rank, world):
+    """Ring SP + activation checkpointing: the recompute re-runs the ring
+    exchange inside backward — collective order must stay aligned."""
+    return _sp_ckpt_body(rank, world, "ring")
+
+
+def sp_ulysses_ckpt_worker(rank, world):
+    """Ulysses SP + activation checkpointing: recompute re-fires the
+    head<->sequence all-to-alls in backward."""
+    return _sp_ckpt_body(rank, world, "ulysses")
+
+
 def test_sp_ring_with_checkpointing():
     res = _spawn("sp_ring_ckpt_worker")
+    for r in range(2):
+        assert res[r]["err"] < 1e-3, res
+
+
+def test_sp_ulysses_with_checkpointing():
+    res = _spawn("sp_ulysses_ckpt_worker")
     for r in range(2):
         assert res[r]["err"] < 1e-3, res
