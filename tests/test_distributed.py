@@ -979,3 +979,46 @@ def test_sp_ulysses_with_checkpointing():
     res = _spawn("sp_ulysses_ckpt_worker")
     for r in range(2):
         assert res[r]["err"] < 1e-3, res
+
+
+def ep_ckpt_worker(rank, world):
+    """EP + activation checkpointing: token all-to-alls re-fire during
+    recompute; grads must equal the non-checkpointed EP run."""
+    from luminaai_amd.models import DeepSeekTransformer
+    from luminaai_amd.models.transformer import DeepSeekConfig
+    from luminaai_amd.parallel.mesh import init_mesh, reset_mesh
+    from luminaai_amd.ops import fused_cross_entropy
+
+    def run(ckpt):
+        reset_mesh()
+        init_mesh(world)
+        mcfg = DeepSeekConfig(vocab_size=512, hidden_size=64, num_layers=2,
+                              num_heads=4, num_kv_heads=2,
+                              intermediate_size=128, seq_length=32,
+                              use_moe=True, num_experts=4, moe_top_k=2,
+                              routing_noise_std=0.0, moe_pattern="all",
+                              dense_start_layers=0, use_mod=False,
+                              gradient_checkpointing=ckpt,
+                              tie_word_embeddings=False)
+        torch.manual_seed(1234)
+        m = DeepSeekTransformer(mcfg)
+        m.train()
+        torch.manual_seed(930 + rank)
+        ids = torch.randint(1, mcfg.vocab_size, (2, 33))
+        logits, aux, _ = m(ids[:, :-1])
+        ce, _, _ = fused_cross_entropy(logits, ids[:, 1:])
+        (ce + aux).backward()
+        return (m.embed_tokens.weight.grad.clone(),
+                m.layers[0].ffn.w_gate_up.grad.clone())
+
+    g_plain, e_plain = run(False)
+    g_ckpt, e_ckpt = run(True)
+    return {"g_err": float((g_plain - g_ckpt).abs().max()),
+            "e_err": float((e_plain - e_ckpt).abs().max())}
+
+
+def test_ep_with_checkpointing():
+    res = _spawn("ep_ckpt_worker")
+    for r in range(2):
+        assert res[r]["g_err"] < 1e-5, res
+        assert res[r]["e_err"] < 1e-5, res
