@@ -131,8 +131,10 @@ def main(argv: Optional[list] = None) -> dict:
     rank = comm.get_rank()
     world = comm.get_world_size()
     ep = cfg.expert_parallel_size or 1
-    if world > 1 and cfg.use_moe and ep <= 1 and args.tp <= 1 \
-            and args.sp <= 1 and cfg.num_experts % world == 0:
+    if args.sp > 1 or args.tp > 1:
+        ep = 1        # SP/TP claim the mesh; auto-EP must stand down
+    elif world > 1 and cfg.use_moe and ep <= 1 \
+            and cfg.num_experts % world == 0:
         ep = world
     if world > 1:
         mesh = init_mesh(ep, sp_size=args.sp, tp_size=args.tp,
