@@ -585,3 +585,30 @@ def test_training_does_not_leak_objects(tiny_config, tokenizer):
     gc.collect()
     after = len(gc.get_objects())
     assert after - before < 500, (before, after)
+
+
+def test_precision_manager_registry():
+    from luminaai_amd.training.precision import PrecisionManager
+
+    class C:
+        precision = "bf16"
+
+    pm = PrecisionManager(C())
+    assert pm.param_dtype == torch.bfloat16
+    assert not pm.needs_loss_scale
+    C.precision = "fp16"
+    assert PrecisionManager(C()).needs_loss_scale
+    C.precision = "auto"
+    assert PrecisionManager(C()).spec.name in ("fp32", "bf16")
+    C.precision = "int8"         # reference-compat alias -> bf16 training
+    assert PrecisionManager(C()).param_dtype == torch.bfloat16
+    C.precision = "nope"
+    import pytest as _pytest
+    with _pytest.raises(ValueError):
+        PrecisionManager(C())
+    assert "fp8" in PrecisionManager.available()
+    # cast_model applies the param dtype
+    import torch.nn as nn
+    C.precision = "fp16"
+    m = PrecisionManager(C()).cast_model(nn.Linear(4, 4))
+    assert m.weight.dtype == torch.float16
