@@ -161,3 +161,38 @@ def test_prometheus_exporter_with_real_metrics(small_model, tokenizer):
     assert "lumina_learning_rate 0.0002" in out
     assert "lumina_gpu_memory_gb 1.5" in out
     assert "lumina_expert_imbalance 1.3" in out
+
+
+def test_mixed_stream_nonstream_soak(client):
+    """Concurrency soak: interleaved streaming and non-streaming requests
+    through the continuous worker all complete correctly."""
+    import concurrent.futures as cf
+    import json as _json
+
+    def nonstream(i):
+        r = client.post("/v1/completions",
+                        json={"prompt": f"soak {i}", "max_tokens": 3,
+                              "temperature": 0.0})
+        assert r.status_code == 200
+        return r.json()["usage"]["completion_tokens"]
+
+    def stream(i):
+        toks = 0
+        with client.stream("POST", "/v1/completions",
+                           json={"prompt": f"soak s{i}", "max_tokens": 3,
+                                 "temperature": 0.0, "stream": True}) as r:
+            assert r.status_code == 200
+            for line in r.iter_lines():
+                if line.startswith("data: ") and line != "data: [DONE]":
+                    _json.loads(line[6:])
+                    toks += 1
+        return toks
+
+    with cf.ThreadPoolExecutor(max_workers=8) as ex:
+        futs = []
+        for i in range(8):
+            futs.append(ex.submit(stream if i % 2 else nonstream, i))
+        results = [f.result(timeout=60) for f in futs]
+    assert all(0 <= r <= 3 for r in results)
+    h = client.get("/health").json()
+    assert h["finished"] >= 8
