@@ -121,7 +121,8 @@ class ContinuousBatchingEngine:
         b = self.free_slot()
         if b is None:
             raise RuntimeError("no free slot — poll can_admit first")
-        ids = prompt_ids[-min(cfg.max_context, self.max_len - 1):]
+        ids = prompt_ids[-min(cfg.max_context, self.max_len - 1):] \
+            or [self._pad]
         P = len(ids)
         if self.n_active() == 0 and self.cursor > 0:
             # empty batch: reclaim the whole cache

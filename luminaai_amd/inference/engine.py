@@ -92,7 +92,7 @@ class GenerationEngine:
         self.model.eval()
         stop = set(cfg.stop_token_ids) | {self.tokenizer.eos_token_id}
 
-        ids = prompt_ids[-cfg.max_context:]
+        ids = prompt_ids[-cfg.max_context:] or [self.tokenizer.pad_token_id]
         x = torch.tensor([ids], dtype=torch.long, device=self.device)
         if use_graph is None:
             use_graph = self.device.type == "cuda"

@@ -196,3 +196,10 @@ def test_mixed_stream_nonstream_soak(client):
     assert all(0 <= r <= 3 for r in results)
     h = client.get("/health").json()
     assert h["finished"] >= 8
+
+
+def test_empty_prompt(client):
+    r = client.post("/v1/completions", json={"prompt": "", "max_tokens": 2,
+                                             "temperature": 0.0})
+    assert r.status_code == 200
+    assert isinstance(r.json()["choices"][0]["text"], str)
