@@ -165,7 +165,7 @@ class GenerationEngine:
         self.model.eval()
         draft_model.eval()
         stop = set(cfg.stop_token_ids) | {self.tokenizer.eos_token_id}
-        ids = prompt_ids[-cfg.max_context:]
+        ids = prompt_ids[-cfg.max_context:] or [self.tokenizer.pad_token_id]
         cap = len(ids) + cfg.max_new_tokens + draft_k + 2
         ct = self.model.make_kv_caches(max_len=cap)
         cd = draft_model.make_kv_caches(max_len=cap)
@@ -312,7 +312,8 @@ class GenerationEngine:
         t0 = time.perf_counter()
         self.model.eval()
         max_ctx = max(c.max_context for c in cfgs)
-        prompts = [p[-max_ctx:] for p in prompts]
+        prompts = [p[-max_ctx:] or [self.tokenizer.pad_token_id]
+                   for p in prompts]
         Smax = max(len(p) for p in prompts)
         max_len = min(max_ctx, Smax + max(c.max_new_tokens for c in cfgs))
         pad = self.tokenizer.pad_token_id
