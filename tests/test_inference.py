@@ -533,3 +533,17 @@ def test_speculative_with_quantized_target(small_model, tokenizer):
     a = eng.generate_speculative(tokenizer.encode("q"), draft, cfg)
     b = eng.generate(tokenizer.encode("q"), cfg)
     assert a == b  # speculative == plain greedy on the SAME (int8) target
+
+
+def test_chat_run_loop(small_model, tokenizer, monkeypatch, capsys):
+    """The REPL loop: commands, a chat turn, and /quit."""
+    chat = ChatInterface(model=small_model, tokenizer=tokenizer)
+    chat.gen_config.max_new_tokens = 2
+    lines = iter(["/help", "", "hello there", "/stats", "/quit"])
+    monkeypatch.setattr("builtins.input", lambda *_: next(lines))
+    chat.run()
+    out = capsys.readouterr().out
+    assert "commands:" in out
+    assert "ai> " in out
+    assert "tokens_generated" in out
+    assert len(chat.history) == 2
