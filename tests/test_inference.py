@@ -567,3 +567,30 @@ def test_quantized_model_state_dict_roundtrip(tiny_moe_config, tokenizer):
     with torch.no_grad():
         out, _, _ = m2(ids)
     torch.testing.assert_close(out, ref)
+
+
+def test_int4_group_size_constraint():
+    import torch.nn as nn
+    from luminaai_amd.ops.quant import Int4Linear, quantize_model
+    with pytest.raises(AssertionError):
+        Int4Linear(100, 32, group_size=64)   # 100 % 64 != 0
+    # quantize_model silently skips incompatible layers instead
+    m = nn.Sequential(nn.Linear(100, 64, bias=False))
+    assert quantize_model(m, mode="int4", min_dim=32, group_size=64) == 0
+    assert type(m[0]) is nn.Linear
+
+
+def test_generation_mode_fallback():
+    g = GenerationConfig.from_mode("no_such_mode")
+    assert g.temperature == pytest.approx(0.8)   # standard
+
+
+def test_server_gen_config_parsing():
+    from luminaai_amd.inference.server import _gen_config
+    cfg = _gen_config({"mode": "precise", "max_tokens": 7,
+                       "stop_token_ids": [5, 9],
+                       "repetition_penalty": 1.3})
+    assert cfg.max_new_tokens == 7
+    assert cfg.temperature == pytest.approx(0.3)
+    assert cfg.stop_token_ids == [5, 9]
+    assert cfg.repetition_penalty == pytest.approx(1.3)
