@@ -48,3 +48,33 @@ def test_auto_configure():
     assert c.vocab_size % 64 == 0          # padded up
     assert c.intermediate_size is not None
     assert c.intermediate_size % 256 == 0
+
+
+def test_param_and_memory_estimators():
+    """Estimators track the real built model within ~15%."""
+    from luminaai_amd.config import ConfigPresets
+    from luminaai_amd.models import (DeepSeekTransformer,
+                                     config_to_deepseek_config)
+    cfg = ConfigPresets.debug()
+    cfg.use_moe = True
+    cfg.num_experts = 4
+    cfg.moe_top_k = 2
+    est_total = cfg.estimate_total_params()
+    est_active = cfg.estimate_active_params()
+    m = DeepSeekTransformer(config_to_deepseek_config(cfg))
+    real_total = m.count_parameters()
+    real_active = m.count_active_parameters()
+    assert abs(est_total - real_total) / real_total < 0.15, \
+        (est_total, real_total)
+    assert abs(est_active - real_active) / real_active < 0.20, \
+        (est_active, real_active)
+    assert est_active <= est_total
+    assert cfg.estimate_memory_gb() > 0
+
+
+def test_largest_divisor_helper():
+    from luminaai_amd.config import largest_divisor_leq
+    assert largest_divisor_leq(8, 8) == 8
+    assert largest_divisor_leq(8, 5) == 4
+    assert largest_divisor_leq(12, 5) == 4
+    assert largest_divisor_leq(7, 3) == 1
