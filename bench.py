@@ -31,7 +31,17 @@ from luminaai_amd.models import DeepSeekTransformer, config_to_deepseek_config  
 from luminaai_amd.parallel import comm  # noqa: E402
 from luminaai_amd.training import Trainer  # noqa: E402
 
-BASELINE_TOKENS_PER_SEC = 73000.0  # reference b1 MoE 8E top-2 on A100 40GB
+# Per-config reference baselines (config-matched rows).  The
+# vs_baseline ratio is only meaningful against the SAME architecture row:
+# b1_moe vs the A100 MoE number, the b1 MoD preset vs the published MoD
+# number, etc.  Presets without a published row report vs_baseline=null.
+BASELINES = {
+    "b1_moe": 73000.0,      # b1 MoE 8E top-2, A100 40GB (BENCHMARKS.md:109-120)
+    "b1": 132000.0,         # b1 MoD cap 0.6, RTX 4090 (BENCHMARKS.md:122-132)
+    "b7_moe": 54000.0,      # b7 MoE 8E top-2, 2x A100 80GB (BENCHMARKS.md:165-176)
+    "b7": 110800.0,         # b7 MoD cap 0.5, A100 80GB (BENCHMARKS.md:178-189)
+    "debug_200m": 172000.0, # debug_200m MoD cap 0.5, RTX 3090 (BENCHMARKS.md:67-77)
+}
 
 
 def build_config(args):
@@ -152,6 +162,7 @@ def main():
     tokens_per_sec = tokens_per_step * args.steps / elapsed
     ms_per_step = elapsed / args.steps * 1000.0
 
+    baseline = BASELINES.get(args.preset or "b1_moe")
     if rank == 0:
         loss = trainer._metric_floats().get("ce_loss")
         result = {
@@ -164,7 +175,8 @@ def main():
             "ms_per_step": round(ms_per_step, 2),
             "higher_is_better": True,
             "scaling": "weak",
-            "vs_baseline": round(tokens_per_sec / BASELINE_TOKENS_PER_SEC, 3),
+            "vs_baseline": (round(tokens_per_sec / baseline, 3)
+                            if baseline else None),
             "dtype": trainer.precision.spec.name,
             "data": "synthetic",
             "config": {

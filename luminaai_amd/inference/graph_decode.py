@@ -82,12 +82,22 @@ class GraphedDecoder:
             c._len -= n
 
     def ensure_captured(self):
+        # A prompt that exactly fills max_context leaves the cache cursor at
+        # capacity; the warmup/capture single-token pass would then
+        # index_copy_ at idx == capacity (a sticky device-side assert on
+        # GPU). Skip capture -- the decode loop stops before any step.
+        if self.seq_len >= self.max_context:
+            return
         if self._graph is None:
             self._capture()
 
     @torch.no_grad()
     def step(self, token_id: torch.Tensor) -> torch.Tensor:
         """token_id: [B] or [B,1] long on device -> logits [B, V] fp32."""
+        if self.seq_len >= self.max_context:
+            raise RuntimeError(
+                f"KV cache full (seq_len={self.seq_len} == max_context); "
+                "cannot decode further")
         if self._graph is None:
             self._capture()
         self._in_tok.copy_(token_id.view(self.batch, 1))

@@ -356,7 +356,7 @@ class GroupedQueryAttention(nn.Module):
             # ring mode has no head-count constraint (that is its point)
             assert self.num_heads % self.sp_size == 0 and \
                 self.num_kv_heads % self.sp_size == 0, \
-                "num_heads and num_kv_heads must divide sp_size"
+                "num_heads and num_kv_heads must be divisible by sp_size"
 
     def forward(self, x, rope_cs, pos: Optional[torch.Tensor] = None,
                 pos_offset: int = 0, kv_cache: Optional[KVCache] = None,
@@ -406,6 +406,15 @@ class GroupedQueryAttention(nn.Module):
                 out = tp_reduce(out, self.tp_group)
             return out
 
+        if self.sp_size > 1 and kv_cache is None and S > 1 \
+                and attn_mask is not None and self.training:
+            # Neither the Ulysses nor the ring path supports an explicit
+            # additive mask; silently attending only over the local sequence
+            # chunk would produce wrong results. Fail loudly.
+            raise ValueError(
+                "sequence parallelism (sp_size > 1) does not support an "
+                "explicit attention mask during a sharded training forward; "
+                "use causal masking (attn_mask=None) or disable SP")
         run_sp = self.sp_size > 1 and self.sp_mode == "ulysses" \
             and kv_cache is None and S > 1 and attn_mask is None
         if run_sp:

@@ -221,6 +221,19 @@ class _Zero3Unit:
         self._grads_pending = self.n_params
 
     def gather(self):
+        # If a prefetch stream gathered this unit's weights, the storage is
+        # already allocated and segment gather() early-returns -- but the
+        # all-gather may still be in flight on that stream.  Make the
+        # consuming stream wait on the recorded event and tell the caching
+        # allocator the buffers are now used by this stream too.
+        ev = getattr(self, "_prefetch_event", None)
+        if ev is not None:
+            self._prefetch_event = None
+            cur = torch.cuda.current_stream()
+            cur.wait_event(ev)
+            for s in self.segments:
+                if s.flat_w.is_cuda:
+                    s.flat_w.record_stream(cur)
         for s in self.segments:
             s.gather()
 
@@ -448,8 +461,14 @@ class Zero3Engine:
                 unit.alloc_grads()
             elif self._prefetch_stream is not None and ui + 1 < len(self.units):
                 nxt = self.units[ui + 1]
+                # The prefetch stream must not race ahead of the compute
+                # stream's last use of the buffers it is about to refill.
+                self._prefetch_stream.wait_stream(torch.cuda.current_stream())
                 with torch.cuda.stream(self._prefetch_stream):
                     nxt.gather()
+                ev = torch.cuda.Event()
+                ev.record(self._prefetch_stream)
+                nxt._prefetch_event = ev
         return hook
 
     def _make_fwd_post(self, ui):
@@ -467,6 +486,17 @@ class Zero3Engine:
             unit = self.units[ui]
             unit.gather()
             unit.alloc_grads()
+            # backward visits units in reverse order: prefetch ui-1
+            if self._prefetch_stream is not None and ui > 0:
+                prev = self.units[ui - 1]
+                if not prev.persistent:
+                    self._prefetch_stream.wait_stream(
+                        torch.cuda.current_stream())
+                    with torch.cuda.stream(self._prefetch_stream):
+                        prev.gather()
+                    ev = torch.cuda.Event()
+                    ev.record(self._prefetch_stream)
+                    prev._prefetch_event = ev
         return hook
 
     def _make_grad_hook(self, unit):

@@ -155,8 +155,16 @@ class Trainer:
         weights = batch.get("loss_weights")
         if weights is not None:
             weights = weights.to(self.device, non_blocking=True)
-        if self.mesh is not None and self.mesh.sp_size > 1 \
-                and input_ids.shape[1] % self.mesh.sp_size == 0:
+        if self.mesh is not None and self.mesh.sp_size > 1:
+            if input_ids.shape[1] % self.mesh.sp_size != 0:
+                # A silent skip here would leave every rank training on the
+                # FULL sequence while the model still applies sp_rank-shifted
+                # RoPE offsets and the Ulysses all-to-all -- corrupted
+                # training. Fail loudly instead.
+                raise ValueError(
+                    f"sequence length {input_ids.shape[1]} is not divisible "
+                    f"by sp_size {self.mesh.sp_size}; pad or trim the batch "
+                    f"(seq_length % sp_size == 0 is required for SP)")
             # Ulysses SP: replicas share the batch (DistributedSampler groups
             # them) and each rank trains on its contiguous sequence slice
             from ..parallel.sequence_parallel import shard_sequence
@@ -475,14 +483,54 @@ class Trainer:
     def _emit_metrics(self, out: Dict):
         if self._metrics_hook is None:
             return
-        m = self.get_current_metrics()
-        # Distributed: the orchestrators on every rank must reach IDENTICAL
-        # decisions from their independent monitor threads — average the
-        # loss across ranks here (training thread, so the collective is safe
-        # and aligned across ranks; grad_norm/lr/step are already global).
-        if comm.is_distributed() and m.loss is not None:
-            m.loss = comm.all_reduce_scalar(float(m.loss), op="sum") \
-                / comm.get_world_size()
+        dev = {k: v for k, v in self._last_metrics.items()
+               if torch.is_tensor(v) and v.is_cuda}
+        if dev:
+            # Async emission: stage this step's device scalars into a pinned
+            # host buffer (non-blocking D2H behind an event) and consume the
+            # PREVIOUS emission's staged values, whose copy completed during
+            # the intervening step. The hot path never calls .item() on a
+            # live device tensor -> no stream sync (measured 37% overhead at
+            # debug scale in round 1). Metrics lag one emission interval.
+            keys = sorted(dev)
+            stacked = torch.stack([dev[k].detach().float().reshape(())
+                                   for k in keys])
+            if comm.is_distributed():
+                # Device-side average across ranks (RCCL, stays on-stream):
+                # all ranks' monitor threads then see IDENTICAL metric values
+                # and reach identical adaptive decisions.
+                import torch.distributed as dist
+                dist.all_reduce(stacked)
+                stacked /= comm.get_world_size()
+            bufs = getattr(self, "_emit_bufs", None)
+            if bufs is None or bufs[0].numel() != len(keys):
+                self._emit_bufs = tuple(
+                    torch.empty(len(keys), dtype=torch.float32,
+                                pin_memory=True) for _ in range(2))
+                self._emit_tick = 0
+                self._pending_emit = None
+            self._emit_tick = getattr(self, "_emit_tick", 0) + 1
+            pinned = self._emit_bufs[self._emit_tick & 1]
+            pinned.copy_(stacked, non_blocking=True)
+            ev = torch.cuda.Event()
+            ev.record()
+            prev = getattr(self, "_pending_emit", None)
+            self._pending_emit = (keys, pinned, ev, self.global_step,
+                                  self.epoch)
+            if prev is None:
+                return
+            pkeys, ppinned, pev, pstep, pepoch = prev
+            pev.synchronize()   # a full emission interval old: ~always done
+            floats = {k: float(ppinned[i]) for i, k in enumerate(pkeys)}
+            m = self.get_current_metrics(_floats=floats, _step=pstep,
+                                         _epoch=pepoch)
+        else:
+            m = self.get_current_metrics()
+            # CPU/gloo path: average the loss across ranks synchronously
+            # (cheap on CPU; keeps rank decisions identical).
+            if comm.is_distributed() and m.loss is not None:
+                m.loss = comm.all_reduce_scalar(float(m.loss), op="sum") \
+                    / comm.get_world_size()
         try:
             self._metrics_hook(m)
         except Exception:  # noqa: BLE001 — monitoring must never kill training
@@ -503,14 +551,16 @@ class Trainer:
     # ============================================================== =====
     # Adaptive-intervention API (reference trainer.py:1144-1835)
     # ================================================================== ==
-    def get_current_metrics(self) -> TrainingMetrics:
+    def get_current_metrics(self, _floats: Optional[Dict] = None,
+                            _step: Optional[int] = None,
+                            _epoch: Optional[int] = None) -> TrainingMetrics:
         if getattr(self.config, "profile_memory", False) \
                 and torch.cuda.is_available():
             # peak since the previous emission (config flag profile_memory)
             self._last_metrics["memory_peak_gb"] = \
                 torch.cuda.max_memory_allocated() / 1e9
             torch.cuda.reset_peak_memory_stats()
-        d = self._metric_floats()
+        d = _floats if _floats is not None else self._metric_floats()
         mem = (torch.cuda.memory_allocated() / 1e9
                if torch.cuda.is_available() else 0.0)
         # MoE routing stats walk every layer and sync device scalars —
@@ -519,7 +569,8 @@ class Trainer:
         if self._stats_tick % 10 == 1 or not hasattr(self, "_stats_cache"):
             self._stats_cache = self._extract_moe_routing_stats()
         return TrainingMetrics(
-            step=self.global_step, epoch=self.epoch,
+            step=self.global_step if _step is None else _step,
+            epoch=self.epoch if _epoch is None else _epoch,
             loss=d.get("loss"), aux_loss=d.get("aux_loss"),
             grad_norm=self.optimizer.last_grad_norm(),
             lr=self.get_lr(), tokens_per_sec=self._calculate_throughput(),
