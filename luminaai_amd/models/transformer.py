@@ -424,26 +424,29 @@ class GroupedQueryAttention(nn.Module):
             k = scatter_heads_gather_seq(k, self.sp_size, self.sp_group)
             v = scatter_heads_gather_seq(v, self.sp_size, self.sp_group)
 
-        # SDPA wants [B, H, S, D]
-        qt = q.transpose(1, 2)
-        kt = k.transpose(1, 2)
-        vt = v.transpose(1, 2)
         is_causal = attn_mask is None and (kv_cache is None or S > 1)
         # flash-vs-fallback call counters (reference model.py:635-637,
-        # 841-853): the AOTriton flash path serves mask-free causal SDPA;
-        # additive masks route to the composite kernel
+        # 841-853): the hand-written CDNA4 kernel serves mask-free causal
+        # attention; additive masks / decode route to SDPA
+        drop = self.dropout if self.training else 0.0
+        use_own_flash = (is_causal and kv_cache is None and S > 1
+                         and ops.can_flash_attention(q, drop))
         if is_causal:
             self._flash_calls = getattr(self, "_flash_calls", 0) + 1
         else:
             self._masked_calls = getattr(self, "_masked_calls", 0) + 1
-        out = F.scaled_dot_product_attention(
-            qt, kt, vt,
-            attn_mask=attn_mask,
-            dropout_p=self.dropout if self.training else 0.0,
-            is_causal=is_causal,
-            enable_gqa=self.num_kv_heads != self.num_heads,
-        )
-        out = out.transpose(1, 2)
+        if use_own_flash:
+            # hand-written CDNA4 flash kernel (ops/csrc/attention.hip):
+            # native GQA, no KV-head repeat, fused online softmax
+            out = ops.flash_attention(q, k, v, scale=self.head_dim ** -0.5)
+        else:
+            out = F.scaled_dot_product_attention(
+                q.transpose(1, 2), k.transpose(1, 2), v.transpose(1, 2),
+                attn_mask=attn_mask,
+                dropout_p=drop,
+                is_causal=is_causal,
+                enable_gqa=self.num_kv_heads != self.num_heads,
+            ).transpose(1, 2)
         if run_sp:
             out = scatter_seq_gather_heads(out.contiguous(), self.sp_size,
                                            self.sp_group)

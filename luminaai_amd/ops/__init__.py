@@ -4,6 +4,8 @@ plus pure-PyTorch CPU reference implementations."""
 from . import reference
 from .interface import (
     adamw_step,
+    can_flash_attention,
+    flash_attention,
     fused_cross_entropy,
     get_ext,
     has_ext,
@@ -15,6 +17,7 @@ from .interface import (
 from .reference import rope_cache
 
 __all__ = [
-    "adamw_step", "fused_cross_entropy", "get_ext", "has_ext", "l2norm_sq",
+    "adamw_step", "can_flash_attention", "flash_attention",
+    "fused_cross_entropy", "get_ext", "has_ext", "l2norm_sq",
     "reference", "rmsnorm", "rope", "rope_cache", "swiglu",
 ]

@@ -1,0 +1,769 @@
+// Causal GQA flash attention (forward + backward) for CDNA4 / gfx950.
+//
+// Replaces torch SDPA -> AOTriton on the training hot path (reference
+// capability: ColossalAI/extensions/flash_attention/flash_attention_dao_cuda
+// .py:1 and csrc/cuda/scaled_masked_softmax_cuda.cu:1).  Round-1 profiling
+// showed AOTriton's bwd_kernel_dk_dv + bwd_kernel_dq at ~156 TF = 11.4% of
+// the b1 training step; these kernels are the MI355X-native replacement.
+//
+// Layout contract (host wrapper packs): Q,K,V,O,dO are [B, H, S, DP] bf16
+// contiguous with DP = head_dim padded up to a multiple of 32 (pad channels
+// ZERO).  LSE2/Delta are [B, H, SP] fp32 with SP = S padded to 128.
+// K,V use HKV heads (GQA); kv_head = q_head / (H/HKV).
+//
+// Orientation scheme (one packing helper serves all four kernels):
+//   fwd   : S^T tile = mfma(A=K, B=Q^T)   -> rows(regs)=kv, cols(lanes)=q
+//           softmax state (m, l) is lane-local; P^T packed via
+//           v_cvt_pk_bf16_f32 + v_permlane32_swap into PV A-fragments.
+//   dk/dv : S tile   = mfma(A=Q, B=K^T)   -> rows(regs)=q, cols(lanes)=kv
+//           lse/delta broadcast per-reg from LDS; P / dS packed the same
+//           way into the A-fragments of dV = P^T dO and dK = dS^T Q.
+//   dq    : S^T tile = mfma(A=K, B=Q^T)   -> lse/delta lane-local;
+//           dS^T packed into the A-fragment of dQ = dS K.
+// MFMA: v_mfma_f32_32x32x16_bf16.  A/B frag: lane holds 8 contiguous k at
+// row/col = l&31, k-half = l>>5.  C/D: col = l&31, row = crow(r, l>>5) =
+// (r&3) + 8*(r>>2) + 4*(l>>5).
+//
+// Staging is REGISTER staging (T14 split: global loads issued before the
+// compute phase, LDS writes after the barrier).  No global_load_lds here:
+// the kernels also issue ordinary per-tile loads (lse/delta slabs), and
+// hipcc drains a glds queue with vmcnt(0) at any ordinary load's use
+// (guide Sec.5 "Three .s-level traps", b).  Two LDS images per staged
+// operand where both orientations are consumed: a row-major [32][DP] image
+// (XOR slot swizzle keyed by row>>2 for conflict-free ds_read_b128) and a
+// transposed [DP][32] image (swizzle keyed by d>>2) for the k-contiguous
+// B-fragments that a transposed operand needs.
+
+#include "common.h"
+
+typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
+typedef __attribute__((ext_vector_type(4))) float  f32x4;
+typedef __attribute__((ext_vector_type(16))) float f32x16;
+
+#define ATTN_THREADS 256
+#define QB 128                 // q rows per workgroup (fwd, dq)
+#define KVB 128                // kv rows per workgroup (dkdv)
+#define TS 32                  // tile side (q-tile / kv-tile rows)
+#define LOG2E 1.4426950408889634f
+
+// C/D row map of mfma_f32_32x32x16_bf16
+DEV_INLINE constexpr int crow(int r, int hi) {
+  return (r & 3) + 8 * (r >> 2) + 4 * hi;
+}
+
+// ---- LDS image addressing -------------------------------------------------
+// row-major [TS][DP] bf16 image; 16B slot XOR within 4-slot windows keyed by
+// row>>2 (rows within a b128 lane group are distinct mod 16; colliding rows
+// differ by 4 -> distinct keys -> conflict-free).
+template <int DP>
+DEV_INLINE int rimg(int row, int dbyte) {
+  const int slot = dbyte >> 4;
+  const int sw = (slot & ~3) | ((slot ^ (row >> 2)) & 3);
+  return row * (DP * 2) + sw * 16 + (dbyte & 15);
+}
+
+// transposed [DP][TS] bf16 image (row = d, 64B rows); same XOR keyed by d>>2.
+DEV_INLINE int timg(int d, int idxbyte) {
+  const int slot = idxbyte >> 4;                  // 0..3
+  const int sw = slot ^ ((d >> 2) & 3);
+  return d * (TS * 2) + sw * 16 + (idxbyte & 15);
+}
+
+// ---- P/dS packing: 32x32 f32 C-tile -> bf16 A-fragments -------------------
+// In-tile: lane holds C[colC = l&31][rowC = crow(r, l>>5)] over 16 regs.
+// Out: frag[s] (s = 0,1) = A-fragment where lane l holds
+//      A[row = l&31][k = 16 s + 8 (l>>5) + j], j = 0..7 over the rowC dim.
+DEV_INLINE uint32_t cvt_pk_bf16(float lo, float hi) {
+  uint32_t r;
+  asm("v_cvt_pk_bf16_f32 %0, %1, %2" : "=v"(r) : "v"(lo), "v"(hi));
+  return r;
+}
+
+DEV_INLINE void pack_frags(const float p[16], bf16x8 frag[2]) {
+  #pragma unroll
+  for (int s = 0; s < 2; ++s) {
+    uint32_t a0 = cvt_pk_bf16(p[8 * s + 0], p[8 * s + 1]);
+    uint32_t a1 = cvt_pk_bf16(p[8 * s + 2], p[8 * s + 3]);
+    uint32_t b0 = cvt_pk_bf16(p[8 * s + 4], p[8 * s + 5]);
+    uint32_t b1 = cvt_pk_bf16(p[8 * s + 6], p[8 * s + 7]);
+    // half-exchange: after this, frag dwords run k = 16s .. 16s+7 (lanes<32)
+    // and k = 16s+8 .. 16s+15 (lanes>=32)
+    auto r0 = __builtin_amdgcn_permlane32_swap(a0, b0, false, false);
+    auto r1 = __builtin_amdgcn_permlane32_swap(a1, b1, false, false);
+    uintx4 u;
+    u[0] = r0[0]; u[1] = r1[0]; u[2] = r0[1]; u[3] = r1[1];
+    frag[s] = __builtin_bit_cast(bf16x8, u);
+  }
+}
+
+DEV_INLINE float shfl_xor32(float v) { return __shfl_xor(v, 32, 64); }
+
+// ---- register staging of a [TS rows][DP] global tile ----------------------
+// Each thread owns chunks c = tid + i*256 (c < TS*DP/8) of 8 bf16.
+// Loads are guarded by clamping the global row to S-1 (garbage-safe: the
+// compute path masks pad rows/cols).
+template <int DP, int NCH>
+struct Stage {
+  ushortx8 v[NCH];
+  DEV_INLINE void load(const uint16_t* __restrict__ base, int row0, int S) {
+    const int t = threadIdx.x;
+    #pragma unroll
+    for (int i = 0; i < NCH; ++i) {
+      const int c = t + i * ATTN_THREADS;
+      if (c < TS * DP / 8) {
+        int row = c / (DP / 8);
+        const int col8 = c - row * (DP / 8);
+        int g = row0 + row;
+        g = g >= S ? S - 1 : g;
+        v[i] = *reinterpret_cast<const ushortx8*>(base + (int64_t)g * DP
+                                                  + col8 * 8);
+      }
+    }
+  }
+  // optional pre-scale (bf16 -> f32 -> *s -> bf16), for Q in the QK^T image
+  DEV_INLINE void scale(float s) {
+    #pragma unroll
+    for (int i = 0; i < NCH; ++i)
+      #pragma unroll
+      for (int j = 0; j < 8; ++j)
+        v[i][j] = f32_to_bf16(bf16_to_f32(v[i][j]) * s);
+  }
+  DEV_INLINE void write_row(uint16_t* img) const {
+    const int t = threadIdx.x;
+    #pragma unroll
+    for (int i = 0; i < NCH; ++i) {
+      const int c = t + i * ATTN_THREADS;
+      if (c < TS * DP / 8) {
+        const int row = c / (DP / 8);
+        const int col8 = c - row * (DP / 8);
+        *reinterpret_cast<ushortx8*>(
+            reinterpret_cast<char*>(img) + rimg<DP>(row, col8 * 16)) = v[i];
+      }
+    }
+  }
+  DEV_INLINE void write_tr(uint16_t* img) const {
+    const int t = threadIdx.x;
+    #pragma unroll
+    for (int i = 0; i < NCH; ++i) {
+      const int c = t + i * ATTN_THREADS;
+      if (c < TS * DP / 8) {
+        const int row = c / (DP / 8);
+        const int col8 = c - row * (DP / 8);
+        #pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          const int d = col8 * 8 + j;
+          *reinterpret_cast<uint16_t*>(
+              reinterpret_cast<char*>(img) + timg(d, row * 2)) = v[i][j];
+        }
+      }
+    }
+  }
+};
+
+// read one A/B fragment (8 bf16) from a row-major image: row = l&31,
+// k-offset = 16*ks + 8*hi
+template <int DP>
+DEV_INLINE bf16x8 rfrag(const uint16_t* img, int lane, int ks) {
+  const int row = lane & 31, hi = lane >> 5;
+  return *reinterpret_cast<const bf16x8*>(
+      reinterpret_cast<const char*>(img) + rimg<DP>(row, 32 * ks + 16 * hi));
+}
+
+// read one B fragment from a transposed image: col(d-row of image) = dtile*32
+// + l&31, idx-offset = 16*s + 8*hi  (idx = the contraction index)
+DEV_INLINE bf16x8 tfrag(const uint16_t* img, int lane, int dtile, int s) {
+  const int d = dtile * 32 + (lane & 31), hi = lane >> 5;
+  return *reinterpret_cast<const bf16x8*>(
+      reinterpret_cast<const char*>(img) + timg(d, 32 * s + 16 * hi));
+}
+
+// ===========================================================================
+// Forward: O = softmax(scale * Q K^T + causal) V, LSE2 = m2 + log2(l)
+// 4 waves, wave w owns q rows [q0 + 32w, q0 + 32w + 32); kv tiles of 32.
+// ===========================================================================
+template <int DP>
+__global__ __launch_bounds__(ATTN_THREADS, 2)
+void attn_fwd_kernel(const uint16_t* __restrict__ Q,
+                     const uint16_t* __restrict__ K,
+                     const uint16_t* __restrict__ V,
+                     uint16_t* __restrict__ O,
+                     float* __restrict__ LSE2,
+                     int B, int H, int HKV, int S, int SP, float scale) {
+  constexpr int NCH = (TS * DP / 8 + ATTN_THREADS - 1) / ATTN_THREADS;
+  constexpr int DT = DP / 32;          // d-tiles of the O accumulator
+  constexpr int KS = DP / 16;          // k-steps per 32x32 S^T tile
+  __shared__ uint16_t lsK[2][TS * DP];     // K row image (double buffered)
+  __shared__ uint16_t lsV[2][TS * DP];     // V transposed image
+  __shared__ float    lsA[4][TS];          // per-wave alpha broadcast
+
+  const int qb   = gridDim.x - 1 - blockIdx.x;   // longest blocks first
+  const int h    = blockIdx.y;
+  const int b    = blockIdx.z;
+  const int hkv  = h / (H / HKV);
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int q0   = qb * QB;
+  const int q0w  = q0 + wave * TS;
+
+  const uint16_t* Qp = Q + ((int64_t)b * H + h) * S * DP;
+  const uint16_t* Kp = K + ((int64_t)b * HKV + hkv) * S * DP;
+  const uint16_t* Vp = V + ((int64_t)b * HKV + hkv) * S * DP;
+  uint16_t* Op = O + ((int64_t)b * H + h) * S * DP;
+  float* Lp = LSE2 + ((int64_t)b * H + h) * SP;
+
+  // ---- per-lane Q fragments (pre-scaled by scale*log2e), B-frag layout:
+  // col(q) = l&31, k(d) = 16 ks + 8 hi .. +7
+  const float c2 = scale * LOG2E;
+  bf16x8 qf[KS];
+  {
+    const int qg0 = q0w + (lane & 31);
+    const int qg = qg0 >= S ? S - 1 : qg0;
+    const uint16_t* qrow = Qp + (int64_t)qg * DP;
+    #pragma unroll
+    for (int ks = 0; ks < KS; ++ks) {
+      ushortx8 raw = *reinterpret_cast<const ushortx8*>(
+          qrow + 16 * ks + 8 * (lane >> 5));
+      bf16x8 f;
+      #pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        uint16_t u = f32_to_bf16(bf16_to_f32(raw[j]) * c2);
+        f[j] = __builtin_bit_cast(__bf16, u);
+      }
+      qf[ks] = f;
+    }
+  }
+
+  f32x16 oacc[DT] = {};
+  float m = -1e30f, l = 0.0f;
+
+  const int smax = q0 + QB < S ? q0 + QB : S;    // kv rows needed
+  const int nkv = (smax + TS - 1) / TS;
+
+  Stage<DP, NCH> sk, sv;
+  sk.load(Kp, 0, S);
+  sv.load(Vp, 0, S);
+  sk.write_row(lsK[0]);
+  sv.write_tr(lsV[0]);
+  __syncthreads();
+
+  for (int t = 0; t < nkv; ++t) {
+    const int buf = t & 1;
+    if (t + 1 < nkv) {                      // T14: issue loads early
+      sk.load(Kp, (t + 1) * TS, S);
+      sv.load(Vp, (t + 1) * TS, S);
+    }
+    // ---- S^T tile: rows(regs) = kv, cols(lanes) = q
+    f32x16 st = {};
+    #pragma unroll
+    for (int ks = 0; ks < KS; ++ks)
+      st = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+          rfrag<DP>(lsK[buf], lane, ks), qf[ks], st, 0, 0, 0);
+
+    float p[16];
+    const int qg = q0w + (lane & 31);
+    const int kvbase = t * TS;
+    const bool edge = (kvbase + TS > q0w + 1) || (kvbase + TS > S);
+    #pragma unroll
+    for (int r = 0; r < 16; ++r) p[r] = st[r];
+    if (edge) {
+      #pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const int kvg = kvbase + crow(r, lane >> 5);
+        p[r] = (kvg > qg || kvg >= S) ? -3e38f : p[r];
+      }
+    }
+    // ---- online softmax (state per lane = per q row)
+    float pm = p[0];
+    #pragma unroll
+    for (int r = 1; r < 16; ++r) pm = fmaxf(pm, p[r]);
+    pm = fmaxf(pm, shfl_xor32(pm));
+    const float mn = fmaxf(m, pm);
+    const float alpha = __builtin_exp2f(m - mn);
+    m = mn;
+    float rs = 0.0f;
+    #pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      p[r] = __builtin_exp2f(p[r] - mn);
+      rs += p[r];
+    }
+    l = l * alpha + rs + shfl_xor32(rs);
+
+    // ---- rescale O accumulators: alpha is per-q (lane) but O rows are in
+    // regs -> broadcast through this wave's LDS slab
+    if (lane < 32) lsA[wave][lane] = alpha;
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    #pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      const float a_r = lsA[wave][crow(r, lane >> 5)];
+      #pragma unroll
+      for (int n = 0; n < DT; ++n) oacc[n][r] *= a_r;
+    }
+
+    // ---- P^T -> A-fragments, PV
+    bf16x8 pf[2];
+    pack_frags(p, pf);
+    #pragma unroll
+    for (int n = 0; n < DT; ++n)
+      #pragma unroll
+      for (int s = 0; s < 2; ++s)
+        oacc[n] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+            pf[s], tfrag(lsV[buf], lane, n, s), oacc[n], 0, 0, 0);
+
+    __syncthreads();                        // done reading buf
+    if (t + 1 < nkv) {
+      sk.write_row(lsK[buf ^ 1]);
+      sv.write_tr(lsV[buf ^ 1]);
+    }
+    __syncthreads();
+  }
+
+  // ---- epilogue: O = acc / l, LSE2 = m + log2(l)
+  if (lane < 32) lsA[wave][lane] = 1.0f / fmaxf(l, 1e-30f);
+  asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+  #pragma unroll
+  for (int r = 0; r < 16; ++r) {
+    const int row = q0w + crow(r, lane >> 5);
+    if (row < S) {
+      const float inv = lsA[wave][crow(r, lane >> 5)];
+      #pragma unroll
+      for (int n = 0; n < DT; ++n)
+        Op[(int64_t)row * DP + n * 32 + (lane & 31)] =
+            f32_to_bf16(oacc[n][r] * inv);
+    }
+  }
+  if (lane < 32) {
+    const int row = q0w + lane;
+    if (row < SP)
+      Lp[row] = m + __log2f(fmaxf(l, 1e-30f));
+  }
+}
+
+// ===========================================================================
+// Delta: delta[b,h,q] = sum_d dO * O   (fp32, [B,H,SP])
+// ===========================================================================
+template <int DP>
+__global__ void attn_delta_kernel(const uint16_t* __restrict__ dO,
+                                  const uint16_t* __restrict__ O,
+                                  float* __restrict__ Delta,
+                                  int64_t rows, int S, int SP) {
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  for (int64_t row = (int64_t)blockIdx.x * 4 + wave; row < rows;
+       row += (int64_t)gridDim.x * 4) {
+    const uint16_t* a = dO + row * DP;
+    const uint16_t* b = O + row * DP;
+    float acc = 0.0f;
+    #pragma unroll
+    for (int i = 0; i < (DP + 255) / 256; ++i) {
+      const int e = (lane + i * 64) * 4;
+      if (e < DP) {
+        const uint64_t xa = *reinterpret_cast<const uint64_t*>(a + e);
+        const uint64_t ya = *reinterpret_cast<const uint64_t*>(b + e);
+        #pragma unroll
+        for (int j = 0; j < 4; ++j)
+          acc += bf16_to_f32((uint16_t)(xa >> (16 * j)))
+               * bf16_to_f32((uint16_t)(ya >> (16 * j)));
+      }
+    }
+    acc = wave_reduce_sum(acc);
+    if (lane == 0) {
+      const int64_t bh = row / S;
+      const int64_t q = row - bh * S;
+      Delta[bh * SP + q] = acc;
+    }
+  }
+}
+
+// ===========================================================================
+// dK/dV: 4 waves, wave owns 32 kv rows of a 128-row kv block; iterates
+// (q-head of the GQA group) x (q-tiles >= diagonal).  S-orientation.
+// ===========================================================================
+template <int DP>
+__global__ __launch_bounds__(ATTN_THREADS, 1)
+void attn_bwd_dkdv_kernel(const uint16_t* __restrict__ Q,
+                          const uint16_t* __restrict__ K,
+                          const uint16_t* __restrict__ V,
+                          const uint16_t* __restrict__ dO,
+                          const float* __restrict__ LSE2,
+                          const float* __restrict__ Delta,
+                          uint16_t* __restrict__ dK,
+                          uint16_t* __restrict__ dV,
+                          int B, int H, int HKV, int S, int SP, float scale) {
+  constexpr int NCH = (TS * DP / 8 + ATTN_THREADS - 1) / ATTN_THREADS;
+  constexpr int DT = DP / 32;
+  constexpr int KS = DP / 16;
+  // q-tile images: row-major Q (pre-scaled) + dO, transposed Q + dO,
+  // lse/delta slabs; double buffered.
+  __shared__ uint16_t lsQ[2][TS * DP];
+  __shared__ uint16_t lsO[2][TS * DP];
+  __shared__ uint16_t lsQt[2][TS * DP];
+  __shared__ uint16_t lsOt[2][TS * DP];
+  __shared__ float    lsL[2][TS];
+  __shared__ float    lsD[2][TS];
+
+  const int kb   = blockIdx.x;
+  const int hkv  = blockIdx.y;
+  const int b    = blockIdx.z;
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int G    = H / HKV;                 // q heads per kv head
+  const int kv0w = kb * KVB + wave * TS;    // this wave's kv rows
+
+  const uint16_t* Kp = K + ((int64_t)b * HKV + hkv) * S * DP;
+  const uint16_t* Vp = V + ((int64_t)b * HKV + hkv) * S * DP;
+  uint16_t* dKp = dK + ((int64_t)b * HKV + hkv) * S * DP;
+  uint16_t* dVp = dV + ((int64_t)b * HKV + hkv) * S * DP;
+
+  const float c2 = scale * LOG2E;
+
+  // ---- resident K / V B-fragments: col(kv) = l&31, k(d) contiguous
+  bf16x8 kf[KS], vf[KS];
+  {
+    const int kg0 = kv0w + (lane & 31);
+    const int kg = kg0 >= S ? S - 1 : kg0;
+    #pragma unroll
+    for (int ks = 0; ks < KS; ++ks) {
+      kf[ks] = *reinterpret_cast<const bf16x8*>(
+          Kp + (int64_t)kg * DP + 16 * ks + 8 * (lane >> 5));
+      vf[ks] = *reinterpret_cast<const bf16x8*>(
+          Vp + (int64_t)kg * DP + 16 * ks + 8 * (lane >> 5));
+    }
+  }
+
+  f32x16 dkacc[DT] = {};
+  f32x16 dvacc[DT] = {};
+
+  const int tq0 = kb * (KVB / TS);          // first q-tile on the diagonal
+  const int ntq = (S + TS - 1) / TS - tq0;  // q-tiles per head
+  const int total = ntq * G;                // flattened (head, q-tile) steps
+
+  auto stage_load = [&](Stage<DP, NCH>& sq, Stage<DP, NCH>& so, float lse[1],
+                        float del[1], int step) {
+    const int gi = step / ntq;              // group member
+    const int qt = tq0 + (step - gi * ntq);
+    const int h = hkv * G + gi;
+    const uint16_t* Qp = Q + ((int64_t)b * H + h) * S * DP;
+    const uint16_t* Op = dO + ((int64_t)b * H + h) * S * DP;
+    sq.load(Qp, qt * TS, S);
+    so.load(Op, qt * TS, S);
+    if (threadIdx.x < TS) {
+      const float* Lp = LSE2 + ((int64_t)b * H + h) * SP + qt * TS;
+      lse[0] = Lp[threadIdx.x];
+    } else if (threadIdx.x < 2 * TS) {
+      const float* Dp = Delta + ((int64_t)b * H + h) * SP + qt * TS;
+      del[0] = Dp[threadIdx.x - TS];
+    }
+  };
+  Stage<DP, NCH> sq, so;
+  float lse1[1], del1[1];
+  stage_load(sq, so, lse1, del1, 0);
+  // The dK product wants RAW Q in the transposed image while the QK^T row
+  // image wants Q pre-scaled: write the transposed image BEFORE scaling.
+  sq.write_tr(lsQt[0]);
+  sq.scale(c2);
+  sq.write_row(lsQ[0]);
+  so.write_row(lsO[0]);
+  so.write_tr(lsOt[0]);
+  if (threadIdx.x < TS) lsL[0][threadIdx.x] = lse1[0];
+  else if (threadIdx.x < 2 * TS) lsD[0][threadIdx.x - TS] = del1[0];
+  __syncthreads();
+
+  for (int step = 0; step < total; ++step) {
+    const int buf = step & 1;
+    const int gi = step / ntq;
+    const int qt = tq0 + (step - gi * ntq);
+    if (step + 1 < total)
+      stage_load(sq, so, lse1, del1, step + 1);
+
+    // waves whose kv rows sit entirely above this q-tile skip compute
+    const bool active = qt * TS + TS - 1 >= kv0w;
+    if (active) {
+      // ---- S tile: rows(regs) = q, cols(lanes) = kv  (Q pre-scaled)
+      f32x16 st = {};
+      #pragma unroll
+      for (int ks = 0; ks < KS; ++ks)
+        st = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+            rfrag<DP>(lsQ[buf], lane, ks), kf[ks], st, 0, 0, 0);
+      // ---- dP tile
+      f32x16 dp = {};
+      #pragma unroll
+      for (int ks = 0; ks < KS; ++ks)
+        dp = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+            rfrag<DP>(lsO[buf], lane, ks), vf[ks], dp, 0, 0, 0);
+
+      // ---- P = exp2(s2 - lse2[q]), masked; dS = P*(dP - delta[q])*scale
+      const int kvg = kv0w + (lane & 31);
+      float p[16], ds[16];
+      #pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const int qg = qt * TS + crow(r, lane >> 5);
+        const float lse = lsL[buf][crow(r, lane >> 5)];
+        const float dlt = lsD[buf][crow(r, lane >> 5)];
+        float pr = __builtin_exp2f(st[r] - lse);
+        pr = (kvg <= qg && qg < S) ? pr : 0.0f;
+        p[r] = pr;
+        ds[r] = pr * (dp[r] - dlt) * scale;
+      }
+      bf16x8 pf[2], df[2];
+      pack_frags(p, pf);
+      pack_frags(ds, df);
+
+      // ---- dV += P^T dO ; dK += dS^T Q   (B-frags from transposed images)
+      #pragma unroll
+      for (int n = 0; n < DT; ++n)
+        #pragma unroll
+        for (int s = 0; s < 2; ++s) {
+          dvacc[n] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+              pf[s], tfrag(lsOt[buf], lane, n, s), dvacc[n], 0, 0, 0);
+          dkacc[n] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+              df[s], tfrag(lsQt[buf], lane, n, s), dkacc[n], 0, 0, 0);
+        }
+    }
+
+    __syncthreads();
+    if (step + 1 < total) {
+      const int bi = buf ^ 1;
+      sq.write_tr(lsQt[bi]);
+      sq.scale(c2);
+      sq.write_row(lsQ[bi]);
+      so.write_row(lsO[bi]);
+      so.write_tr(lsOt[bi]);
+      if (threadIdx.x < TS) lsL[bi][threadIdx.x] = lse1[0];
+      else if (threadIdx.x < 2 * TS) lsD[bi][threadIdx.x - TS] = del1[0];
+    }
+    __syncthreads();
+  }
+
+  // ---- epilogue: C rows = kv (crow regs), cols = d (lanes)
+  #pragma unroll
+  for (int r = 0; r < 16; ++r) {
+    const int row = kv0w + crow(r, lane >> 5);
+    if (row < S) {
+      #pragma unroll
+      for (int n = 0; n < DT; ++n) {
+        dKp[(int64_t)row * DP + n * 32 + (lane & 31)] =
+            f32_to_bf16(dkacc[n][r]);
+        dVp[(int64_t)row * DP + n * 32 + (lane & 31)] =
+            f32_to_bf16(dvacc[n][r]);
+      }
+    }
+  }
+}
+
+// ===========================================================================
+// dQ: 4 waves, wave owns 32 q rows of a 128-row q block; iterates kv tiles.
+// S^T orientation (lse/delta lane-local).
+// ===========================================================================
+template <int DP>
+__global__ __launch_bounds__(ATTN_THREADS, 1)
+void attn_bwd_dq_kernel(const uint16_t* __restrict__ Q,
+                        const uint16_t* __restrict__ K,
+                        const uint16_t* __restrict__ V,
+                        const uint16_t* __restrict__ dO,
+                        const float* __restrict__ LSE2,
+                        const float* __restrict__ Delta,
+                        uint16_t* __restrict__ dQ,
+                        int B, int H, int HKV, int S, int SP, float scale) {
+  constexpr int NCH = (TS * DP / 8 + ATTN_THREADS - 1) / ATTN_THREADS;
+  constexpr int DT = DP / 32;
+  constexpr int KS = DP / 16;
+  __shared__ uint16_t lsK[2][TS * DP];      // K row image (raw)
+  __shared__ uint16_t lsKt[2][TS * DP];     // K transposed image
+  __shared__ uint16_t lsV[2][TS * DP];      // V row image
+
+  const int qb   = gridDim.x - 1 - blockIdx.x;
+  const int h    = blockIdx.y;
+  const int b    = blockIdx.z;
+  const int hkv  = h / (H / HKV);
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int q0w  = qb * QB + wave * TS;
+
+  const uint16_t* Qp = Q + ((int64_t)b * H + h) * S * DP;
+  const uint16_t* Kp = K + ((int64_t)b * HKV + hkv) * S * DP;
+  const uint16_t* Vp = V + ((int64_t)b * HKV + hkv) * S * DP;
+  const uint16_t* Op = dO + ((int64_t)b * H + h) * S * DP;
+  uint16_t* dQp = dQ + ((int64_t)b * H + h) * S * DP;
+
+  const float c2 = scale * LOG2E;
+
+  // resident B-fragments: Q^T (pre-scaled by c2) and dO^T; lane-local
+  // lse2/delta (q = l&31 of this wave's rows)
+  bf16x8 qtf[KS], otf[KS];
+  float lse, dlt;
+  {
+    const int qg0 = q0w + (lane & 31);
+    const int qg = qg0 >= S ? S - 1 : qg0;
+    #pragma unroll
+    for (int ks = 0; ks < KS; ++ks) {
+      ushortx8 raw = *reinterpret_cast<const ushortx8*>(
+          Qp + (int64_t)qg * DP + 16 * ks + 8 * (lane >> 5));
+      bf16x8 f;
+      #pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        uint16_t u = f32_to_bf16(bf16_to_f32(raw[j]) * c2);
+        f[j] = __builtin_bit_cast(__bf16, u);
+      }
+      qtf[ks] = f;
+      otf[ks] = *reinterpret_cast<const bf16x8*>(
+          Op + (int64_t)qg * DP + 16 * ks + 8 * (lane >> 5));
+    }
+    const int64_t bh = (int64_t)blockIdx.z * H + h;
+    lse = LSE2[bh * SP + (q0w + (lane & 31) < SP ? q0w + (lane & 31) : SP - 1)];
+    dlt = Delta[bh * SP + (q0w + (lane & 31) < SP ? q0w + (lane & 31) : SP - 1)];
+  }
+
+  f32x16 dqacc[DT] = {};
+  // block-uniform kv range (waves skip via `active`, barriers stay aligned)
+  const int hi_q = qb * QB + QB;
+  const int nkv = ((hi_q < S ? hi_q : S) + TS - 1) / TS;
+
+  Stage<DP, NCH> sk, sv;
+  sk.load(Kp, 0, S);
+  sv.load(Vp, 0, S);
+  sk.write_row(lsK[0]);
+  sk.write_tr(lsKt[0]);
+  sv.write_row(lsV[0]);
+  __syncthreads();
+
+  for (int t = 0; t < nkv; ++t) {
+    const int buf = t & 1;
+    if (t + 1 < nkv) {
+      sk.load(Kp, (t + 1) * TS, S);
+      sv.load(Vp, (t + 1) * TS, S);
+    }
+    const bool active = t * TS <= q0w + TS - 1;   // below/at the diagonal
+    if (active) {
+      // ---- S^T tile: rows(regs) = kv, cols(lanes) = q
+      f32x16 st = {};
+      #pragma unroll
+      for (int ks = 0; ks < KS; ++ks)
+        st = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+            rfrag<DP>(lsK[buf], lane, ks), qtf[ks], st, 0, 0, 0);
+      // ---- dP^T tile
+      f32x16 dp = {};
+      #pragma unroll
+      for (int ks = 0; ks < KS; ++ks)
+        dp = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+            rfrag<DP>(lsV[buf], lane, ks), otf[ks], dp, 0, 0, 0);
+
+      const int qg = q0w + (lane & 31);
+      float ds[16];
+      #pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const int kvg = t * TS + crow(r, lane >> 5);
+        float pr = __builtin_exp2f(st[r] - lse);
+        pr = (kvg <= qg) ? pr : 0.0f;
+        ds[r] = pr * (dp[r] - dlt) * scale;
+      }
+      bf16x8 df[2];
+      pack_frags(ds, df);
+      // ---- dQ += dS K  (B-frags from the transposed K image)
+      #pragma unroll
+      for (int n = 0; n < DT; ++n)
+        #pragma unroll
+        for (int s = 0; s < 2; ++s)
+          dqacc[n] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+              df[s], tfrag(lsKt[buf], lane, n, s), dqacc[n], 0, 0, 0);
+    }
+
+    __syncthreads();
+    if (t + 1 < nkv) {
+      const int bi = buf ^ 1;
+      sk.write_row(lsK[bi]);
+      sk.write_tr(lsKt[bi]);
+      sv.write_row(lsV[bi]);
+    }
+    __syncthreads();
+  }
+
+  #pragma unroll
+  for (int r = 0; r < 16; ++r) {
+    const int row = q0w + crow(r, lane >> 5);
+    if (row < S) {
+      #pragma unroll
+      for (int n = 0; n < DT; ++n)
+        dQp[(int64_t)row * DP + n * 32 + (lane & 31)] =
+            f32_to_bf16(dqacc[n][r]);
+    }
+  }
+}
+
+// ===========================================================================
+// launchers
+// ===========================================================================
+#define INSTANT_DP(DPV)                                                      \
+  if (DP == DPV) {                                                           \
+    hipLaunchKernelGGL(attn_fwd_kernel<DPV>, grid, dim3(ATTN_THREADS), 0,    \
+                       stream, (const uint16_t*)Q, (const uint16_t*)K,       \
+                       (const uint16_t*)V, (uint16_t*)O, (float*)LSE2,       \
+                       B, H, HKV, S, SP, scale);                             \
+    return hipGetLastError();                                                \
+  }
+
+extern "C" hipError_t lumina_attn_fwd(const void* Q, const void* K,
+                                      const void* V, void* O, void* LSE2,
+                                      int B, int H, int HKV, int S, int SP,
+                                      int DP, float scale,
+                                      hipStream_t stream) {
+  dim3 grid((S + QB - 1) / QB, H, B);
+  INSTANT_DP(64)
+  INSTANT_DP(128)
+  INSTANT_DP(160)
+  return hipErrorInvalidValue;
+}
+#undef INSTANT_DP
+
+#define INSTANT_DP(DPV)                                                      \
+  if (DP == DPV) {                                                           \
+    hipLaunchKernelGGL(attn_delta_kernel<DPV>, dim3(grid), dim3(256), 0,     \
+                       stream, (const uint16_t*)dO, (const uint16_t*)O,      \
+                       (float*)Delta, rows, S, SP);                          \
+    return hipGetLastError();                                                \
+  }
+
+extern "C" hipError_t lumina_attn_delta(const void* dO, const void* O,
+                                        void* Delta, int64_t rows, int S,
+                                        int SP, int DP, hipStream_t stream) {
+  int grid = (int)((rows / 4 + 1) < 4096 ? (rows / 4 + 1) : 4096);
+  INSTANT_DP(64)
+  INSTANT_DP(128)
+  INSTANT_DP(160)
+  return hipErrorInvalidValue;
+}
+#undef INSTANT_DP
+
+#define INSTANT_DP(DPV)                                                      \
+  if (DP == DPV) {                                                           \
+    hipLaunchKernelGGL(attn_bwd_dkdv_kernel<DPV>, gridkv,                    \
+                       dim3(ATTN_THREADS), 0, stream,                        \
+                       (const uint16_t*)Q, (const uint16_t*)K,               \
+                       (const uint16_t*)V, (const uint16_t*)dO,              \
+                       (const float*)LSE2, (const float*)Delta,              \
+                       (uint16_t*)dK, (uint16_t*)dV, B, H, HKV, S, SP,       \
+                       scale);                                               \
+    hipError_t e = hipGetLastError();                                        \
+    if (e != hipSuccess) return e;                                           \
+    hipLaunchKernelGGL(attn_bwd_dq_kernel<DPV>, gridq, dim3(ATTN_THREADS),   \
+                       0, stream, (const uint16_t*)Q, (const uint16_t*)K,    \
+                       (const uint16_t*)V, (const uint16_t*)dO,              \
+                       (const float*)LSE2, (const float*)Delta,              \
+                       (uint16_t*)dQ, B, H, HKV, S, SP, scale);              \
+    return hipGetLastError();                                                \
+  }
+
+extern "C" hipError_t lumina_attn_bwd(const void* Q, const void* K,
+                                      const void* V, const void* dO,
+                                      const void* LSE2, const void* Delta,
+                                      void* dQ, void* dK, void* dV,
+                                      int B, int H, int HKV, int S, int SP,
+                                      int DP, float scale,
+                                      hipStream_t stream) {
+  dim3 gridkv((S + KVB - 1) / KVB, HKV, B);
+  dim3 gridq((S + QB - 1) / QB, H, B);
+  INSTANT_DP(64)
+  INSTANT_DP(128)
+  INSTANT_DP(160)
+  return hipErrorInvalidValue;
+}
+#undef INSTANT_DP

@@ -29,6 +29,9 @@ void check_dtype(const at::Tensor& t, const char* name) {
 
 // ---- launchers from the .hip TUs ----------------------------------------
 extern "C" {
+hipError_t lumina_attn_fwd(const void*, const void*, const void*, void*, void*, int, int, int, int, int, int, float, hipStream_t);
+hipError_t lumina_attn_delta(const void*, const void*, void*, int64_t, int, int, int, hipStream_t);
+hipError_t lumina_attn_bwd(const void*, const void*, const void*, const void*, const void*, const void*, void*, void*, void*, int, int, int, int, int, int, float, hipStream_t);
 hipError_t lumina_rmsnorm_fwd_bf16(const void*, const void*, void*, float*, int64_t, int, float, hipStream_t);
 hipError_t lumina_rmsnorm_fwd_f32(const void*, const void*, void*, float*, int64_t, int, float, hipStream_t);
 hipError_t lumina_rmsnorm_bwd_bf16(const void*, const void*, const void*, const float*, void*, float*, int64_t, int, hipStream_t);
@@ -405,6 +408,58 @@ at::Tensor grouped_gemm_nt_v4(const at::Tensor& A, const at::Tensor& B) {
   return O;
 }
 
+// ---- causal GQA flash attention ------------------------------------------
+// q,k,v: [B,H|HKV,S,DP] bf16 contiguous, DP % 32 == 0 (pad channels zero).
+std::tuple<at::Tensor, at::Tensor> attn_fwd(const at::Tensor& q,
+                                            const at::Tensor& k,
+                                            const at::Tensor& v,
+                                            double scale) {
+  TORCH_CHECK(q.is_contiguous() && k.is_contiguous() && v.is_contiguous());
+  TORCH_CHECK(is_bf16(q) && is_bf16(k) && is_bf16(v));
+  TORCH_CHECK(q.dim() == 4 && k.dim() == 4 && v.dim() == 4);
+  const int B = (int)q.size(0), H = (int)q.size(1);
+  const int S = (int)q.size(2), DP = (int)q.size(3);
+  const int HKV = (int)k.size(1);
+  TORCH_CHECK(DP % 32 == 0 && (DP == 64 || DP == 128 || DP == 160),
+              "unsupported padded head dim ", DP);
+  TORCH_CHECK(H % HKV == 0 && k.size(2) == S && v.size(1) == HKV);
+  const int SP = (S + 127) / 128 * 128;
+  auto o = at::empty_like(q);
+  auto lse2 = at::empty({B, H, SP}, q.options().dtype(at::kFloat));
+  check_hip(lumina_attn_fwd(q.data_ptr(), k.data_ptr(), v.data_ptr(),
+                            o.data_ptr(), lse2.data_ptr(), B, H, HKV, S, SP,
+                            DP, (float)scale, cur_stream()),
+            "attn_fwd");
+  return {o, lse2};
+}
+
+std::tuple<at::Tensor, at::Tensor, at::Tensor> attn_bwd(
+    const at::Tensor& q, const at::Tensor& k, const at::Tensor& v,
+    const at::Tensor& o, const at::Tensor& do_, const at::Tensor& lse2,
+    double scale) {
+  TORCH_CHECK(q.is_contiguous() && k.is_contiguous() && v.is_contiguous() &&
+              o.is_contiguous() && do_.is_contiguous() && lse2.is_contiguous());
+  TORCH_CHECK(is_bf16(do_) && lse2.scalar_type() == at::kFloat);
+  const int B = (int)q.size(0), H = (int)q.size(1);
+  const int S = (int)q.size(2), DP = (int)q.size(3);
+  const int HKV = (int)k.size(1);
+  const int SP = (S + 127) / 128 * 128;
+  TORCH_CHECK(lse2.size(2) == SP);
+  auto delta = at::empty({B, H, SP}, q.options().dtype(at::kFloat));
+  check_hip(lumina_attn_delta(do_.data_ptr(), o.data_ptr(), delta.data_ptr(),
+                              (int64_t)B * H * S, S, SP, DP, cur_stream()),
+            "attn_delta");
+  auto dq = at::empty_like(q);
+  auto dk = at::empty_like(k);
+  auto dv = at::empty_like(v);
+  check_hip(lumina_attn_bwd(q.data_ptr(), k.data_ptr(), v.data_ptr(),
+                            do_.data_ptr(), lse2.data_ptr(), delta.data_ptr(),
+                            dq.data_ptr(), dk.data_ptr(), dv.data_ptr(),
+                            B, H, HKV, S, SP, DP, (float)scale, cur_stream()),
+            "attn_bwd");
+  return {dq, dk, dv};
+}
+
 at::Tensor gemv(const at::Tensor& x, const at::Tensor& w) {
   // y[N] = w[N,K] @ x[K]
   TORCH_CHECK(x.is_contiguous() && w.is_contiguous());
@@ -419,6 +474,10 @@ at::Tensor gemv(const at::Tensor& x, const at::Tensor& w) {
 }
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
+  mod.def("attn_fwd", &attn_fwd,
+          "causal GQA flash attention forward -> (O, LSE2) (gfx950)");
+  mod.def("attn_bwd", &attn_bwd,
+          "causal GQA flash attention backward -> (dQ, dK, dV) (gfx950)");
   mod.def("gemv", &gemv, "batch-1 decode GEMV y = W @ x (gfx950)");
   mod.def("grouped_gemm_nt_v4", &grouped_gemm_nt_v4,
           "2-buffer raw-barrier counted-vmcnt variant (K%64==0)");

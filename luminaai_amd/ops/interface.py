@@ -266,6 +266,70 @@ def expert_bmm(x: torch.Tensor, w: torch.Tensor) -> torch.Tensor:
     return BatchedLinearFn.apply(x, w)
 
 
+# ------------------------------------------------------- flash attention
+class _FlashAttnFn(torch.autograd.Function):
+    """Causal GQA flash attention on the hand-written CDNA4 kernels
+    (csrc/attention.hip). Inputs are PACKED [B, H, S, DP] bf16 contiguous
+    with DP = padded head dim (zero pad channels); GQA handled natively
+    (no KV repeat). Forward saves LSE2 (log2-sum-exp); backward runs the
+    delta / dkdv / dq kernel pipeline.
+
+    Replaces torch SDPA -> AOTriton on the training path (reference:
+    flash_attention_dao_cuda.py:1, scaled_masked_softmax_cuda.cu:1)."""
+
+    @staticmethod
+    def forward(ctx, q, k, v, scale):
+        o, lse2 = get_ext().attn_fwd(q, k, v, scale)
+        ctx.save_for_backward(q, k, v, o, lse2)
+        ctx.scale = scale
+        return o
+
+    @staticmethod
+    def backward(ctx, do):
+        q, k, v, o, lse2 = ctx.saved_tensors
+        dq, dk, dv = get_ext().attn_bwd(q, k, v, o, do.contiguous(), lse2,
+                                        ctx.scale)
+        return dq, dk, dv, None
+
+
+def _attn_pad_dim(d: int) -> Optional[int]:
+    for dp in (64, 128, 160):
+        if d <= dp:
+            return dp
+    return None
+
+
+def can_flash_attention(q: torch.Tensor, dropout: float) -> bool:
+    """True when the hand-written causal-GQA kernel path applies."""
+    return (q.is_cuda and q.dtype == torch.bfloat16 and dropout == 0.0
+            and _attn_pad_dim(q.shape[-1]) is not None and has_ext())
+
+
+def flash_attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
+                    scale: Optional[float] = None) -> torch.Tensor:
+    """Causal GQA attention. q [B,S,H,D], k/v [B,S,HKV,D] -> [B,S,H,D].
+
+    Pads D up to the kernel's tile dim and repacks to [B,H,S,DP]; the pad
+    and transpose run through autograd, so gradients flow back to the
+    original layout automatically."""
+    B, S, H, D = q.shape
+    dp = _attn_pad_dim(D)
+    if scale is None:
+        scale = D ** -0.5
+    pad = dp - D
+
+    def pack(t):
+        t = t.transpose(1, 2)
+        if pad:
+            t = torch.nn.functional.pad(t, (0, pad))
+        return t.contiguous()
+
+    o = _FlashAttnFn.apply(pack(q), pack(k), pack(v), scale)
+    if pad:
+        o = o[..., :D]
+    return o.transpose(1, 2)
+
+
 def grouped_gemm_nt(a: torch.Tensor, b: torch.Tensor) -> torch.Tensor:
     """out[e] = a[e] @ b[e]^T with a [E,M,K], b [E,N,K] both row-major
     (contraction over the trailing dim). HIP MFMA kernel on GPU bf16;
