@@ -35,6 +35,7 @@
 // B-fragments that a transposed operand needs.
 
 #include "common.h"
+#include <stdlib.h>
 
 typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
 typedef __attribute__((ext_vector_type(4))) float  f32x4;
@@ -62,11 +63,26 @@ DEV_INLINE int rimg(int row, int dbyte) {
   return row * (DP * 2) + sw * 16 + (dbyte & 15);
 }
 
-// transposed [DP][TS] bf16 image (row = d, 64B rows); same XOR keyed by d>>2.
-DEV_INLINE int timg(int d, int idxbyte) {
-  const int slot = idxbyte >> 4;                  // 0..3
-  const int sw = slot ^ ((d >> 2) & 3);
-  return d * (TS * 2) + sw * 16 + (idxbyte & 15);
+// Blocked "transpose-read" image for ds_read_b64_tr_b16 consumers: element
+// (idx, d) at ELEMENT offset
+//   (idx>>2)*(DP*4) + (d>>4)*64 + (idx&3)*16 + (d&15)
+// i.e. [4 idx][16 d] sub-blocks of 64 elements (the guide's V-subtile
+// layout).  A 16-byte chunk (idx, d0..d0+7) stays contiguous, so staging
+// writes are plain ds_write_b128 of the same chunks the row image uses
+// (no scalar scatter -> no bank-conflict disaster).  The tr read gives
+// lane l four consecutive idx at fixed d (stride 32B = 16 elements).
+template <int DP>
+DEV_INLINE int blk_off(int idx, int d) {       // in elements
+  return (idx >> 2) * (DP * 4) + ((d >> 4) << 6) + ((idx & 3) << 4)
+         + (d & 15);
+}
+
+typedef __attribute__((ext_vector_type(4))) __bf16 bf16x4;
+typedef __attribute__((address_space(3))) bf16x4 lds_bf16x4;
+
+DEV_INLINE bf16x4 tr16_read(const uint16_t* p) {
+  return __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+      (lds_bf16x4*)(p));
 }
 
 // ---- P/dS packing: 32x32 f32 C-tile -> bf16 A-fragments -------------------
@@ -141,20 +157,15 @@ struct Stage {
       }
     }
   }
-  DEV_INLINE void write_tr(uint16_t* img) const {
+  DEV_INLINE void write_blk(uint16_t* img) const {
     const int t = threadIdx.x;
     #pragma unroll
     for (int i = 0; i < NCH; ++i) {
       const int c = t + i * ATTN_THREADS;
       if (c < TS * DP / 8) {
         const int row = c / (DP / 8);
-        const int col8 = c - row * (DP / 8);
-        #pragma unroll
-        for (int j = 0; j < 8; ++j) {
-          const int d = col8 * 8 + j;
-          *reinterpret_cast<uint16_t*>(
-              reinterpret_cast<char*>(img) + timg(d, row * 2)) = v[i][j];
-        }
+        const int d0 = (c - row * (DP / 8)) * 8;
+        *reinterpret_cast<ushortx8*>(img + blk_off<DP>(row, d0)) = v[i];
       }
     }
   }
@@ -169,12 +180,31 @@ DEV_INLINE bf16x8 rfrag(const uint16_t* img, int lane, int ks) {
       reinterpret_cast<const char*>(img) + rimg<DP>(row, 32 * ks + 16 * hi));
 }
 
-// read one B fragment from a transposed image: col(d-row of image) = dtile*32
-// + l&31, idx-offset = 16*s + 8*hi  (idx = the contraction index)
+// read one B fragment from a blocked image with two hardware transpose
+// reads: lane l needs 8 consecutive idx at d = dtile*32 + (l&31),
+// idx0 = 16*s + 8*hi.
+//
+// Probed tr16 semantics (scripts/probe_tr.hip + probe_blk.hip): each lane
+// loads 4 bf16 at its OWN 8-byte-aligned address; the hardware then
+// redistributes across the 16-lane group as  out[l][j] = in[4j + ((l>>2)&3)]
+// [l&3].  With the [4 idx][16 d] block layout, lane l must therefore point
+// at element ((d>>2)&3)*16 + (d&3)*4 of the block -- the lane's quad index
+// selects the idx row it SUPPLIES, not the row it receives.
+template <int DP>
+DEV_INLINE int blk_raddr(int idx, int d) {     // tr16 read address (elements)
+  return (idx >> 2) * (DP * 4) + ((d >> 4) << 6) + (((d >> 2) & 3) << 4)
+         + ((d & 3) << 2);
+}
+
+template <int DP>
 DEV_INLINE bf16x8 tfrag(const uint16_t* img, int lane, int dtile, int s) {
   const int d = dtile * 32 + (lane & 31), hi = lane >> 5;
-  return *reinterpret_cast<const bf16x8*>(
-      reinterpret_cast<const char*>(img) + timg(d, 32 * s + 16 * hi));
+  const int idx0 = 16 * s + 8 * hi;
+  bf16x4 lo = tr16_read(img + blk_raddr<DP>(idx0, d));
+  bf16x4 hi4 = tr16_read(img + blk_raddr<DP>(idx0 + 4, d));
+  union { struct { bf16x4 a, b; } p; bf16x8 v; } u;
+  u.p.a = lo; u.p.b = hi4;
+  return u.v;
 }
 
 // ===========================================================================
@@ -188,7 +218,8 @@ void attn_fwd_kernel(const uint16_t* __restrict__ Q,
                      const uint16_t* __restrict__ V,
                      uint16_t* __restrict__ O,
                      float* __restrict__ LSE2,
-                     int B, int H, int HKV, int S, int SP, float scale) {
+                     int B, int H, int HKV, int S, int SP, float scale,
+                     float thr) {
   constexpr int NCH = (TS * DP / 8 + ATTN_THREADS - 1) / ATTN_THREADS;
   constexpr int DT = DP / 32;          // d-tiles of the O accumulator
   constexpr int KS = DP / 16;          // k-steps per 32x32 S^T tile
@@ -243,7 +274,7 @@ void attn_fwd_kernel(const uint16_t* __restrict__ Q,
   sk.load(Kp, 0, S);
   sv.load(Vp, 0, S);
   sk.write_row(lsK[0]);
-  sv.write_tr(lsV[0]);
+  sv.write_blk(lsV[0]);
   __syncthreads();
 
   for (int t = 0; t < nkv; ++t) {
@@ -254,10 +285,12 @@ void attn_fwd_kernel(const uint16_t* __restrict__ Q,
     }
     // ---- S^T tile: rows(regs) = kv, cols(lanes) = q
     f32x16 st = {};
+    __builtin_amdgcn_s_setprio(1);
     #pragma unroll
     for (int ks = 0; ks < KS; ++ks)
       st = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
           rfrag<DP>(lsK[buf], lane, ks), qf[ks], st, 0, 0, 0);
+    __builtin_amdgcn_s_setprio(0);
 
     float p[16];
     const int qg = q0w + (lane & 31);
@@ -272,47 +305,54 @@ void attn_fwd_kernel(const uint16_t* __restrict__ Q,
         p[r] = (kvg > qg || kvg >= S) ? -3e38f : p[r];
       }
     }
-    // ---- online softmax (state per lane = per q row)
+    // ---- online softmax (state per lane = per q row); defer-max (T13):
+    // skip the O rescale while the running max has not grown past THR --
+    // P is then bounded by exp2(THR), which fp32 l and bf16 P absorb.
     float pm = p[0];
     #pragma unroll
     for (int r = 1; r < 16; ++r) pm = fmaxf(pm, p[r]);
     pm = fmaxf(pm, shfl_xor32(pm));
-    const float mn = fmaxf(m, pm);
-    const float alpha = __builtin_exp2f(m - mn);
-    m = mn;
+    float mn = m;
+    if (!__all(pm - m <= thr)) {
+      mn = fmaxf(m, pm);
+      const float alpha = __builtin_exp2f(m - mn);
+      m = mn;
+      l *= alpha;
+      // alpha is per-q (lane) but O rows are in regs -> broadcast through
+      // this wave's LDS slab
+      if (lane < 32) lsA[wave][lane] = alpha;
+      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+      #pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const float a_r = lsA[wave][crow(r, lane >> 5)];
+        #pragma unroll
+        for (int n = 0; n < DT; ++n) oacc[n][r] *= a_r;
+      }
+    }
     float rs = 0.0f;
     #pragma unroll
     for (int r = 0; r < 16; ++r) {
       p[r] = __builtin_exp2f(p[r] - mn);
       rs += p[r];
     }
-    l = l * alpha + rs + shfl_xor32(rs);
-
-    // ---- rescale O accumulators: alpha is per-q (lane) but O rows are in
-    // regs -> broadcast through this wave's LDS slab
-    if (lane < 32) lsA[wave][lane] = alpha;
-    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-    #pragma unroll
-    for (int r = 0; r < 16; ++r) {
-      const float a_r = lsA[wave][crow(r, lane >> 5)];
-      #pragma unroll
-      for (int n = 0; n < DT; ++n) oacc[n][r] *= a_r;
-    }
+    l += rs + shfl_xor32(rs);
 
     // ---- P^T -> A-fragments, PV
     bf16x8 pf[2];
     pack_frags(p, pf);
+    __builtin_amdgcn_s_setprio(1);
     #pragma unroll
     for (int n = 0; n < DT; ++n)
       #pragma unroll
       for (int s = 0; s < 2; ++s)
         oacc[n] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
-            pf[s], tfrag(lsV[buf], lane, n, s), oacc[n], 0, 0, 0);
+            pf[s], tfrag<DP>(lsV[buf], lane, n, s), oacc[n], 0, 0, 0);
+    __builtin_amdgcn_s_setprio(0);
 
     __syncthreads();                        // done reading buf
     if (t + 1 < nkv) {
       sk.write_row(lsK[buf ^ 1]);
-      sv.write_tr(lsV[buf ^ 1]);
+      sv.write_blk(lsV[buf ^ 1]);
     }
     __syncthreads();
   }
@@ -459,11 +499,11 @@ void attn_bwd_dkdv_kernel(const uint16_t* __restrict__ Q,
   stage_load(sq, so, lse1, del1, 0);
   // The dK product wants RAW Q in the transposed image while the QK^T row
   // image wants Q pre-scaled: write the transposed image BEFORE scaling.
-  sq.write_tr(lsQt[0]);
+  sq.write_blk(lsQt[0]);
   sq.scale(c2);
   sq.write_row(lsQ[0]);
   so.write_row(lsO[0]);
-  so.write_tr(lsOt[0]);
+  so.write_blk(lsOt[0]);
   if (threadIdx.x < TS) lsL[0][threadIdx.x] = lse1[0];
   else if (threadIdx.x < 2 * TS) lsD[0][threadIdx.x - TS] = del1[0];
   __syncthreads();
@@ -480,16 +520,18 @@ void attn_bwd_dkdv_kernel(const uint16_t* __restrict__ Q,
     if (active) {
       // ---- S tile: rows(regs) = q, cols(lanes) = kv  (Q pre-scaled)
       f32x16 st = {};
+      f32x16 dp = {};
+      __builtin_amdgcn_s_setprio(1);
       #pragma unroll
       for (int ks = 0; ks < KS; ++ks)
         st = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
             rfrag<DP>(lsQ[buf], lane, ks), kf[ks], st, 0, 0, 0);
       // ---- dP tile
-      f32x16 dp = {};
       #pragma unroll
       for (int ks = 0; ks < KS; ++ks)
         dp = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
             rfrag<DP>(lsO[buf], lane, ks), vf[ks], dp, 0, 0, 0);
+      __builtin_amdgcn_s_setprio(0);
 
       // ---- P = exp2(s2 - lse2[q]), masked; dS = P*(dP - delta[q])*scale
       const int kvg = kv0w + (lane & 31);
@@ -509,25 +551,27 @@ void attn_bwd_dkdv_kernel(const uint16_t* __restrict__ Q,
       pack_frags(ds, df);
 
       // ---- dV += P^T dO ; dK += dS^T Q   (B-frags from transposed images)
+      __builtin_amdgcn_s_setprio(1);
       #pragma unroll
       for (int n = 0; n < DT; ++n)
         #pragma unroll
         for (int s = 0; s < 2; ++s) {
           dvacc[n] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
-              pf[s], tfrag(lsOt[buf], lane, n, s), dvacc[n], 0, 0, 0);
+              pf[s], tfrag<DP>(lsOt[buf], lane, n, s), dvacc[n], 0, 0, 0);
           dkacc[n] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
-              df[s], tfrag(lsQt[buf], lane, n, s), dkacc[n], 0, 0, 0);
+              df[s], tfrag<DP>(lsQt[buf], lane, n, s), dkacc[n], 0, 0, 0);
         }
+      __builtin_amdgcn_s_setprio(0);
     }
 
     __syncthreads();
     if (step + 1 < total) {
       const int bi = buf ^ 1;
-      sq.write_tr(lsQt[bi]);
+      sq.write_blk(lsQt[bi]);
       sq.scale(c2);
       sq.write_row(lsQ[bi]);
       so.write_row(lsO[bi]);
-      so.write_tr(lsOt[bi]);
+      so.write_blk(lsOt[bi]);
       if (threadIdx.x < TS) lsL[bi][threadIdx.x] = lse1[0];
       else if (threadIdx.x < 2 * TS) lsD[bi][threadIdx.x - TS] = del1[0];
     }
@@ -622,7 +666,7 @@ void attn_bwd_dq_kernel(const uint16_t* __restrict__ Q,
   sk.load(Kp, 0, S);
   sv.load(Vp, 0, S);
   sk.write_row(lsK[0]);
-  sk.write_tr(lsKt[0]);
+  sk.write_blk(lsKt[0]);
   sv.write_row(lsV[0]);
   __syncthreads();
 
@@ -636,16 +680,18 @@ void attn_bwd_dq_kernel(const uint16_t* __restrict__ Q,
     if (active) {
       // ---- S^T tile: rows(regs) = kv, cols(lanes) = q
       f32x16 st = {};
+      f32x16 dp = {};
+      __builtin_amdgcn_s_setprio(1);
       #pragma unroll
       for (int ks = 0; ks < KS; ++ks)
         st = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
             rfrag<DP>(lsK[buf], lane, ks), qtf[ks], st, 0, 0, 0);
       // ---- dP^T tile
-      f32x16 dp = {};
       #pragma unroll
       for (int ks = 0; ks < KS; ++ks)
         dp = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
             rfrag<DP>(lsV[buf], lane, ks), otf[ks], dp, 0, 0, 0);
+      __builtin_amdgcn_s_setprio(0);
 
       const int qg = q0w + (lane & 31);
       float ds[16];
@@ -659,19 +705,21 @@ void attn_bwd_dq_kernel(const uint16_t* __restrict__ Q,
       bf16x8 df[2];
       pack_frags(ds, df);
       // ---- dQ += dS K  (B-frags from the transposed K image)
+      __builtin_amdgcn_s_setprio(1);
       #pragma unroll
       for (int n = 0; n < DT; ++n)
         #pragma unroll
         for (int s = 0; s < 2; ++s)
           dqacc[n] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
-              df[s], tfrag(lsKt[buf], lane, n, s), dqacc[n], 0, 0, 0);
+              df[s], tfrag<DP>(lsKt[buf], lane, n, s), dqacc[n], 0, 0, 0);
+      __builtin_amdgcn_s_setprio(0);
     }
 
     __syncthreads();
     if (t + 1 < nkv) {
       const int bi = buf ^ 1;
       sk.write_row(lsK[bi]);
-      sk.write_tr(lsKt[bi]);
+      sk.write_blk(lsKt[bi]);
       sv.write_row(lsV[bi]);
     }
     __syncthreads();
@@ -697,7 +745,7 @@ void attn_bwd_dq_kernel(const uint16_t* __restrict__ Q,
     hipLaunchKernelGGL(attn_fwd_kernel<DPV>, grid, dim3(ATTN_THREADS), 0,    \
                        stream, (const uint16_t*)Q, (const uint16_t*)K,       \
                        (const uint16_t*)V, (uint16_t*)O, (float*)LSE2,       \
-                       B, H, HKV, S, SP, scale);                             \
+                       B, H, HKV, S, SP, scale, defer_thr);                  \
     return hipGetLastError();                                                \
   }
 
@@ -707,6 +755,10 @@ extern "C" hipError_t lumina_attn_fwd(const void* Q, const void* K,
                                       int DP, float scale,
                                       hipStream_t stream) {
   dim3 grid((S + QB - 1) / QB, H, B);
+  // defer-max threshold (T13); LUMINA_ATTN_NODEFER=1 forces the always-
+  // rescale path (A/B + numerics bisection knob)
+  static const float defer_thr =
+      getenv("LUMINA_ATTN_NODEFER") ? -3e38f : 8.0f;
   INSTANT_DP(64)
   INSTANT_DP(128)
   INSTANT_DP(160)
