@@ -121,52 +121,53 @@ DEV_INLINE float shfl_xor32(float v) { return __shfl_xor(v, 32, 64); }
 template <int DP, int NCH>
 struct Stage {
   ushortx8 v[NCH];
-  DEV_INLINE void load(const uint16_t* __restrict__ base, int row0, int S) {
+  // per-thread chunk coordinates and image offsets are threadIdx-only:
+  // computed once, reused every step (the div-by-DP/8 and image swizzles
+  // were a measurable VALU cost when recomputed per tile)
+  int rowoff[NCH];     // row * DP  (global element offset of the row)
+  int roff[NCH];       // row-image byte offset
+  int boff[NCH];       // blocked-image element offset
+  int row_[NCH];
+  DEV_INLINE void init() {
     const int t = threadIdx.x;
     #pragma unroll
     for (int i = 0; i < NCH; ++i) {
       const int c = t + i * ATTN_THREADS;
+      const int row = c < TS * DP / 8 ? c / (DP / 8) : 0;
+      const int col8 = c - row * (DP / 8);
+      row_[i] = row;
+      rowoff[i] = row * DP + col8 * 8;
+      roff[i] = rimg<DP>(row, col8 * 16);
+      boff[i] = blk_off<DP>(row, col8 * 8);
+    }
+  }
+  DEV_INLINE void load(const uint16_t* __restrict__ base, int row0, int S) {
+    #pragma unroll
+    for (int i = 0; i < NCH; ++i) {
+      const int c = threadIdx.x + i * ATTN_THREADS;
       if (c < TS * DP / 8) {
-        int row = c / (DP / 8);
-        const int col8 = c - row * (DP / 8);
-        int g = row0 + row;
-        g = g >= S ? S - 1 : g;
-        v[i] = *reinterpret_cast<const ushortx8*>(base + (int64_t)g * DP
-                                                  + col8 * 8);
+        const int over = row0 + row_[i] - (S - 1);   // clamp row to S-1
+        const int off = rowoff[i] - (over > 0 ? over * DP : 0);
+        v[i] = *reinterpret_cast<const ushortx8*>(base + (int64_t)row0 * DP
+                                                  + off);
       }
     }
   }
-  // optional pre-scale (bf16 -> f32 -> *s -> bf16), for Q in the QK^T image
-  DEV_INLINE void scale(float s) {
-    #pragma unroll
-    for (int i = 0; i < NCH; ++i)
-      #pragma unroll
-      for (int j = 0; j < 8; ++j)
-        v[i][j] = f32_to_bf16(bf16_to_f32(v[i][j]) * s);
-  }
   DEV_INLINE void write_row(uint16_t* img) const {
-    const int t = threadIdx.x;
     #pragma unroll
     for (int i = 0; i < NCH; ++i) {
-      const int c = t + i * ATTN_THREADS;
-      if (c < TS * DP / 8) {
-        const int row = c / (DP / 8);
-        const int col8 = c - row * (DP / 8);
+      const int c = threadIdx.x + i * ATTN_THREADS;
+      if (c < TS * DP / 8)
         *reinterpret_cast<ushortx8*>(
-            reinterpret_cast<char*>(img) + rimg<DP>(row, col8 * 16)) = v[i];
-      }
+            reinterpret_cast<char*>(img) + roff[i]) = v[i];
     }
   }
   DEV_INLINE void write_blk(uint16_t* img) const {
-    const int t = threadIdx.x;
     #pragma unroll
     for (int i = 0; i < NCH; ++i) {
-      const int c = t + i * ATTN_THREADS;
-      if (c < TS * DP / 8) {
-        const int row = c / (DP / 8);
-        const int d0 = (c - row * (DP / 8)) * 8;
-        *reinterpret_cast<ushortx8*>(img + blk_off<DP>(row, d0)) = v[i];
-      }
+      const int c = threadIdx.x + i * ATTN_THREADS;
+      if (c < TS * DP / 8)
+        *reinterpret_cast<ushortx8*>(img + boff[i]) = v[i];
     }
   }
 };
@@ -271,6 +272,8 @@ void attn_fwd_kernel(const uint16_t* __restrict__ Q,
   const int nkv = (smax + TS - 1) / TS;
 
   Stage<DP, NCH> sk, sv;
+  sk.init();
+  sv.init();
   sk.load(Kp, 0, S);
   sv.load(Vp, 0, S);
   sk.write_row(lsK[0]);
@@ -349,7 +352,8 @@ void attn_fwd_kernel(const uint16_t* __restrict__ Q,
             pf[s], tfrag<DP>(lsV[buf], lane, n, s), oacc[n], 0, 0, 0);
     __builtin_amdgcn_s_setprio(0);
 
-    __syncthreads();                        // done reading buf
+    // write the next tile into the OTHER buffer: safe while slower waves
+    // are still reading buf (disjoint); one barrier publishes it
     if (t + 1 < nkv) {
       sk.write_row(lsK[buf ^ 1]);
       sv.write_blk(lsV[buf ^ 1]);
@@ -441,12 +445,19 @@ void attn_bwd_dkdv_kernel(const uint16_t* __restrict__ Q,
   __shared__ float    lsL[2][TS];
   __shared__ float    lsD[2][TS];
 
-  const int kb   = blockIdx.x;
   const int hkv  = blockIdx.y;
   const int b    = blockIdx.z;
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
   const int G    = H / HKV;                 // q heads per kv head
+  const int nkb  = (S + KVB - 1) / KVB;
+  // causal balance: pair kv-block bx (long: many q-tiles) with its mirror
+  // nkb-1-bx (short) in one workgroup -> near-constant work per block
+  #pragma unroll 1
+  for (int pass = 0; pass < 2; ++pass) {
+  const int kb = pass == 0 ? blockIdx.x : nkb - 1 - blockIdx.x;
+  if (pass == 1 && kb <= (int)blockIdx.x) break;
+  if (pass == 1) __syncthreads();           // epilogue/LDS reuse fence
   const int kv0w = kb * KVB + wave * TS;    // this wave's kv rows
 
   const uint16_t* Kp = K + ((int64_t)b * HKV + hkv) * S * DP;
@@ -495,12 +506,11 @@ void attn_bwd_dkdv_kernel(const uint16_t* __restrict__ Q,
     }
   };
   Stage<DP, NCH> sq, so;
+  sq.init();
+  so.init();
   float lse1[1], del1[1];
   stage_load(sq, so, lse1, del1, 0);
-  // The dK product wants RAW Q in the transposed image while the QK^T row
-  // image wants Q pre-scaled: write the transposed image BEFORE scaling.
   sq.write_blk(lsQt[0]);
-  sq.scale(c2);
   sq.write_row(lsQ[0]);
   so.write_row(lsO[0]);
   so.write_blk(lsOt[0]);
@@ -533,16 +543,21 @@ void attn_bwd_dkdv_kernel(const uint16_t* __restrict__ Q,
             rfrag<DP>(lsO[buf], lane, ks), vf[ks], dp, 0, 0, 0);
       __builtin_amdgcn_s_setprio(0);
 
-      // ---- P = exp2(s2 - lse2[q]), masked; dS = P*(dP - delta[q])*scale
+      // ---- P = exp2(s2*c2 - lse2[q]), masked; dS = P*(dP-delta[q])*scale.
+      // Per-element masking only on edge tiles (diagonal overlap or q>=S);
+      // interior tiles skip the compares entirely.
       const int kvg = kv0w + (lane & 31);
+      const bool interior = (qt * TS >= kv0w + TS) && (qt * TS + TS <= S);
       float p[16], ds[16];
       #pragma unroll
       for (int r = 0; r < 16; ++r) {
-        const int qg = qt * TS + crow(r, lane >> 5);
         const float lse = lsL[buf][crow(r, lane >> 5)];
         const float dlt = lsD[buf][crow(r, lane >> 5)];
-        float pr = __builtin_exp2f(st[r] - lse);
-        pr = (kvg <= qg && qg < S) ? pr : 0.0f;
+        float pr = __builtin_exp2f(fmaf(st[r], c2, -lse));
+        if (!interior) {
+          const int qg = qt * TS + crow(r, lane >> 5);
+          pr = (kvg <= qg && qg < S) ? pr : 0.0f;
+        }
         p[r] = pr;
         ds[r] = pr * (dp[r] - dlt) * scale;
       }
@@ -564,11 +579,9 @@ void attn_bwd_dkdv_kernel(const uint16_t* __restrict__ Q,
       __builtin_amdgcn_s_setprio(0);
     }
 
-    __syncthreads();
     if (step + 1 < total) {
       const int bi = buf ^ 1;
       sq.write_blk(lsQt[bi]);
-      sq.scale(c2);
       sq.write_row(lsQ[bi]);
       so.write_row(lsO[bi]);
       so.write_blk(lsOt[bi]);
@@ -592,6 +605,7 @@ void attn_bwd_dkdv_kernel(const uint16_t* __restrict__ Q,
       }
     }
   }
+  }  // pass loop
 }
 
 // ===========================================================================
@@ -663,6 +677,8 @@ void attn_bwd_dq_kernel(const uint16_t* __restrict__ Q,
   const int nkv = ((hi_q < S ? hi_q : S) + TS - 1) / TS;
 
   Stage<DP, NCH> sk, sv;
+  sk.init();
+  sv.init();
   sk.load(Kp, 0, S);
   sv.load(Vp, 0, S);
   sk.write_row(lsK[0]);
@@ -715,7 +731,6 @@ void attn_bwd_dq_kernel(const uint16_t* __restrict__ Q,
       __builtin_amdgcn_s_setprio(0);
     }
 
-    __syncthreads();
     if (t + 1 < nkv) {
       const int bi = buf ^ 1;
       sk.write_row(lsK[bi]);
@@ -811,7 +826,8 @@ extern "C" hipError_t lumina_attn_bwd(const void* Q, const void* K,
                                       int B, int H, int HKV, int S, int SP,
                                       int DP, float scale,
                                       hipStream_t stream) {
-  dim3 gridkv((S + KVB - 1) / KVB, HKV, B);
+  const int nkb = (S + KVB - 1) / KVB;
+  dim3 gridkv((nkb + 1) / 2, HKV, B);
   dim3 gridq((S + QB - 1) / QB, H, B);
   INSTANT_DP(64)
   INSTANT_DP(128)
