@@ -73,9 +73,10 @@ DEV_INLINE int rimg(int row, int dbyte) {
 // lane l four consecutive idx at fixed d (stride 32B = 16 elements).
 template <int DP>
 DEV_INLINE int blk_off(int idx, int d) {       // in elements
-  return (idx >> 2) * (DP * 4) + ((d >> 4) << 6) + ((idx & 3) << 4)
+  return (idx >> 2) * (DP * 4 + 8) + ((d >> 4) << 6) + ((idx & 3) << 4)
          + (d & 15);
 }
+#define BLK_ELEMS(DP) (TS / 4 * (DP * 4 + 8))
 
 typedef __attribute__((ext_vector_type(4))) __bf16 bf16x4;
 typedef __attribute__((address_space(3))) bf16x4 lds_bf16x4;
@@ -118,7 +119,7 @@ DEV_INLINE float shfl_xor32(float v) { return __shfl_xor(v, 32, 64); }
 // Each thread owns chunks c = tid + i*256 (c < TS*DP/8) of 8 bf16.
 // Loads are guarded by clamping the global row to S-1 (garbage-safe: the
 // compute path masks pad rows/cols).
-template <int DP, int NCH>
+template <int DP, int NCH, int NT = ATTN_THREADS>
 struct Stage {
   ushortx8 v[NCH];
   // per-thread chunk coordinates and image offsets are threadIdx-only:
@@ -132,7 +133,7 @@ struct Stage {
     const int t = threadIdx.x;
     #pragma unroll
     for (int i = 0; i < NCH; ++i) {
-      const int c = t + i * ATTN_THREADS;
+      const int c = t + i * NT;
       const int row = c < TS * DP / 8 ? c / (DP / 8) : 0;
       const int col8 = c - row * (DP / 8);
       row_[i] = row;
@@ -144,7 +145,7 @@ struct Stage {
   DEV_INLINE void load(const uint16_t* __restrict__ base, int row0, int S) {
     #pragma unroll
     for (int i = 0; i < NCH; ++i) {
-      const int c = threadIdx.x + i * ATTN_THREADS;
+      const int c = threadIdx.x + i * NT;
       if (c < TS * DP / 8) {
         const int over = row0 + row_[i] - (S - 1);   // clamp row to S-1
         const int off = rowoff[i] - (over > 0 ? over * DP : 0);
@@ -156,7 +157,7 @@ struct Stage {
   DEV_INLINE void write_row(uint16_t* img) const {
     #pragma unroll
     for (int i = 0; i < NCH; ++i) {
-      const int c = threadIdx.x + i * ATTN_THREADS;
+      const int c = threadIdx.x + i * NT;
       if (c < TS * DP / 8)
         *reinterpret_cast<ushortx8*>(
             reinterpret_cast<char*>(img) + roff[i]) = v[i];
@@ -165,7 +166,7 @@ struct Stage {
   DEV_INLINE void write_blk(uint16_t* img) const {
     #pragma unroll
     for (int i = 0; i < NCH; ++i) {
-      const int c = threadIdx.x + i * ATTN_THREADS;
+      const int c = threadIdx.x + i * NT;
       if (c < TS * DP / 8)
         *reinterpret_cast<ushortx8*>(img + boff[i]) = v[i];
     }
@@ -193,8 +194,16 @@ DEV_INLINE bf16x8 rfrag(const uint16_t* img, int lane, int ks) {
 // selects the idx row it SUPPLIES, not the row it receives.
 template <int DP>
 DEV_INLINE int blk_raddr(int idx, int d) {     // tr16 read address (elements)
-  return (idx >> 2) * (DP * 4) + ((d >> 4) << 6) + (((d >> 2) & 3) << 4)
+  return (idx >> 2) * (DP * 4 + 8) + ((d >> 4) << 6) + (((d >> 2) & 3) << 4)
          + ((d & 3) << 2);
+}
+
+// plain row fragment (8 contiguous d at row = l&31) read from the SAME
+// blocked image -- a written chunk is contiguous, so this is one b128
+template <int DP>
+DEV_INLINE bf16x8 brfrag(const uint16_t* img, int lane, int ks) {
+  return *reinterpret_cast<const bf16x8*>(
+      img + blk_off<DP>(lane & 31, 16 * ks + 8 * (lane >> 5)));
 }
 
 template <int DP>
@@ -212,8 +221,10 @@ DEV_INLINE bf16x8 tfrag(const uint16_t* img, int lane, int dtile, int s) {
 // Forward: O = softmax(scale * Q K^T + causal) V, LSE2 = m2 + log2(l)
 // 4 waves, wave w owns q rows [q0 + 32w, q0 + 32w + 32); kv tiles of 32.
 // ===========================================================================
+#define FWD_THREADS 512
+#define FWD_QB 256
 template <int DP>
-__global__ __launch_bounds__(ATTN_THREADS, 2)
+__global__ __launch_bounds__(FWD_THREADS, 2)
 void attn_fwd_kernel(const uint16_t* __restrict__ Q,
                      const uint16_t* __restrict__ K,
                      const uint16_t* __restrict__ V,
@@ -221,12 +232,12 @@ void attn_fwd_kernel(const uint16_t* __restrict__ Q,
                      float* __restrict__ LSE2,
                      int B, int H, int HKV, int S, int SP, float scale,
                      float thr) {
-  constexpr int NCH = (TS * DP / 8 + ATTN_THREADS - 1) / ATTN_THREADS;
+  constexpr int NCH = (TS * DP / 8 + FWD_THREADS - 1) / FWD_THREADS;
   constexpr int DT = DP / 32;          // d-tiles of the O accumulator
   constexpr int KS = DP / 16;          // k-steps per 32x32 S^T tile
   __shared__ uint16_t lsK[2][TS * DP];     // K row image (double buffered)
-  __shared__ uint16_t lsV[2][TS * DP];     // V transposed image
-  __shared__ float    lsA[4][TS];          // per-wave alpha broadcast
+  __shared__ uint16_t lsV[2][BLK_ELEMS(DP)];   // V blocked image
+  __shared__ float    lsA[8][TS];          // per-wave alpha broadcast
 
   const int qb   = gridDim.x - 1 - blockIdx.x;   // longest blocks first
   const int h    = blockIdx.y;
@@ -234,7 +245,7 @@ void attn_fwd_kernel(const uint16_t* __restrict__ Q,
   const int hkv  = h / (H / HKV);
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
-  const int q0   = qb * QB;
+  const int q0   = qb * FWD_QB;
   const int q0w  = q0 + wave * TS;
 
   const uint16_t* Qp = Q + ((int64_t)b * H + h) * S * DP;
@@ -268,10 +279,10 @@ void attn_fwd_kernel(const uint16_t* __restrict__ Q,
   f32x16 oacc[DT] = {};
   float m = -1e30f, l = 0.0f;
 
-  const int smax = q0 + QB < S ? q0 + QB : S;    // kv rows needed
+  const int smax = q0 + FWD_QB < S ? q0 + FWD_QB : S;  // kv rows needed
   const int nkv = (smax + TS - 1) / TS;
 
-  Stage<DP, NCH> sk, sv;
+  Stage<DP, NCH, FWD_THREADS> sk, sv;
   sk.init();
   sv.init();
   sk.load(Kp, 0, S);
@@ -438,10 +449,10 @@ void attn_bwd_dkdv_kernel(const uint16_t* __restrict__ Q,
   constexpr int KS = DP / 16;
   // q-tile images: row-major Q (pre-scaled) + dO, transposed Q + dO,
   // lse/delta slabs; double buffered.
-  __shared__ uint16_t lsQ[2][TS * DP];
-  __shared__ uint16_t lsO[2][TS * DP];
-  __shared__ uint16_t lsQt[2][TS * DP];
-  __shared__ uint16_t lsOt[2][TS * DP];
+  // single blocked image per operand serves BOTH the row fragments (plain
+  // b128 of the written chunks) and the transposed fragments (tr16 reads)
+  __shared__ uint16_t lsQt[2][BLK_ELEMS(DP)];
+  __shared__ uint16_t lsOt[2][BLK_ELEMS(DP)];
   __shared__ float    lsL[2][TS];
   __shared__ float    lsD[2][TS];
 
@@ -511,8 +522,6 @@ void attn_bwd_dkdv_kernel(const uint16_t* __restrict__ Q,
   float lse1[1], del1[1];
   stage_load(sq, so, lse1, del1, 0);
   sq.write_blk(lsQt[0]);
-  sq.write_row(lsQ[0]);
-  so.write_row(lsO[0]);
   so.write_blk(lsOt[0]);
   if (threadIdx.x < TS) lsL[0][threadIdx.x] = lse1[0];
   else if (threadIdx.x < 2 * TS) lsD[0][threadIdx.x - TS] = del1[0];
@@ -535,12 +544,12 @@ void attn_bwd_dkdv_kernel(const uint16_t* __restrict__ Q,
       #pragma unroll
       for (int ks = 0; ks < KS; ++ks)
         st = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
-            rfrag<DP>(lsQ[buf], lane, ks), kf[ks], st, 0, 0, 0);
+            brfrag<DP>(lsQt[buf], lane, ks), kf[ks], st, 0, 0, 0);
       // ---- dP tile
       #pragma unroll
       for (int ks = 0; ks < KS; ++ks)
         dp = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
-            rfrag<DP>(lsO[buf], lane, ks), vf[ks], dp, 0, 0, 0);
+            brfrag<DP>(lsOt[buf], lane, ks), vf[ks], dp, 0, 0, 0);
       __builtin_amdgcn_s_setprio(0);
 
       // ---- P = exp2(s2*c2 - lse2[q]), masked; dS = P*(dP-delta[q])*scale.
@@ -582,8 +591,6 @@ void attn_bwd_dkdv_kernel(const uint16_t* __restrict__ Q,
     if (step + 1 < total) {
       const int bi = buf ^ 1;
       sq.write_blk(lsQt[bi]);
-      sq.write_row(lsQ[bi]);
-      so.write_row(lsO[bi]);
       so.write_blk(lsOt[bi]);
       if (threadIdx.x < TS) lsL[bi][threadIdx.x] = lse1[0];
       else if (threadIdx.x < 2 * TS) lsD[bi][threadIdx.x - TS] = del1[0];
@@ -625,9 +632,8 @@ void attn_bwd_dq_kernel(const uint16_t* __restrict__ Q,
   constexpr int NCH = (TS * DP / 8 + ATTN_THREADS - 1) / ATTN_THREADS;
   constexpr int DT = DP / 32;
   constexpr int KS = DP / 16;
-  __shared__ uint16_t lsK[2][TS * DP];      // K row image (raw)
-  __shared__ uint16_t lsKt[2][TS * DP];     // K transposed image
-  __shared__ uint16_t lsV[2][TS * DP];      // V row image
+  __shared__ uint16_t lsKt[2][BLK_ELEMS(DP)];  // K blocked image (raw)
+  __shared__ uint16_t lsV[2][TS * DP];         // V row image
 
   const int qb   = gridDim.x - 1 - blockIdx.x;
   const int h    = blockIdx.y;
@@ -681,7 +687,6 @@ void attn_bwd_dq_kernel(const uint16_t* __restrict__ Q,
   sv.init();
   sk.load(Kp, 0, S);
   sv.load(Vp, 0, S);
-  sk.write_row(lsK[0]);
   sk.write_blk(lsKt[0]);
   sv.write_row(lsV[0]);
   __syncthreads();
@@ -701,7 +706,7 @@ void attn_bwd_dq_kernel(const uint16_t* __restrict__ Q,
       #pragma unroll
       for (int ks = 0; ks < KS; ++ks)
         st = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
-            rfrag<DP>(lsK[buf], lane, ks), qtf[ks], st, 0, 0, 0);
+            brfrag<DP>(lsKt[buf], lane, ks), qtf[ks], st, 0, 0, 0);
       // ---- dP^T tile
       #pragma unroll
       for (int ks = 0; ks < KS; ++ks)
@@ -733,7 +738,6 @@ void attn_bwd_dq_kernel(const uint16_t* __restrict__ Q,
 
     if (t + 1 < nkv) {
       const int bi = buf ^ 1;
-      sk.write_row(lsK[bi]);
       sk.write_blk(lsKt[bi]);
       sv.write_row(lsV[bi]);
     }
@@ -757,7 +761,7 @@ void attn_bwd_dq_kernel(const uint16_t* __restrict__ Q,
 // ===========================================================================
 #define INSTANT_DP(DPV)                                                      \
   if (DP == DPV) {                                                           \
-    hipLaunchKernelGGL(attn_fwd_kernel<DPV>, grid, dim3(ATTN_THREADS), 0,    \
+    hipLaunchKernelGGL(attn_fwd_kernel<DPV>, grid, dim3(FWD_THREADS), 0,     \
                        stream, (const uint16_t*)Q, (const uint16_t*)K,       \
                        (const uint16_t*)V, (uint16_t*)O, (float*)LSE2,       \
                        B, H, HKV, S, SP, scale, defer_thr);                  \
@@ -769,7 +773,7 @@ extern "C" hipError_t lumina_attn_fwd(const void* Q, const void* K,
                                       int B, int H, int HKV, int S, int SP,
                                       int DP, float scale,
                                       hipStream_t stream) {
-  dim3 grid((S + QB - 1) / QB, H, B);
+  dim3 grid((S + FWD_QB - 1) / FWD_QB, H, B);
   // defer-max threshold (T13); LUMINA_ATTN_NODEFER=1 forces the always-
   // rescale path (A/B + numerics bisection knob)
   static const float defer_thr =

@@ -13,6 +13,8 @@ recipe — wgrad/dgrad keep full bf16 fidelity).
 
 from __future__ import annotations
 
+import weakref
+
 import torch
 import torch.nn as nn
 
@@ -32,11 +34,14 @@ def quantize_e4m3(t: torch.Tensor):
 
 
 def _quantized_weight_2d(w: torch.Tensor):
+    # id() keys can be recycled after a weight is freed -- verify the cache
+    # entry still refers to THIS tensor object via a weakref before trusting
+    # it (a stale hit returned a wrong-shape quantized weight).
     ent = _WCACHE.get(id(w))
-    if ent is not None and ent[2] == _EPOCH:
-        return ent[0], ent[1]
+    if ent is not None and ent[3] == _EPOCH and ent[0]() is w:
+        return ent[1], ent[2]
     q, s = quantize_e4m3(w)
-    _WCACHE[id(w)] = (q, s, _EPOCH)
+    _WCACHE[id(w)] = (weakref.ref(w), q, s, _EPOCH)
     return q, s
 
 
@@ -96,14 +101,14 @@ def invalidate_weight_cache():
 
 def _quantized_weight(w: torch.Tensor):
     ent = _WCACHE.get(id(w))
-    if ent is not None and ent[2] == _EPOCH:
-        return ent[0], ent[1]
+    if ent is not None and ent[3] == _EPOCH and ent[0]() is w:
+        return ent[1], ent[2]
     qs, ss = [], []
     for e in range(w.shape[0]):
         q, s = quantize_e4m3(w[e].t().contiguous())     # [N, K] row-major
         qs.append(q)
         ss.append(s)
-    _WCACHE[id(w)] = (qs, ss, _EPOCH)
+    _WCACHE[id(w)] = (weakref.ref(w), qs, ss, _EPOCH)
     return qs, ss
 
 
