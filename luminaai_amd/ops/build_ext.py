@@ -20,7 +20,8 @@ ROCM = os.environ.get("ROCM_PATH", "/opt/rocm")
 ARCH = os.environ.get("PYTORCH_ROCM_ARCH", "gfx950")
 
 KERNELS = ["rmsnorm.hip", "rope.hip", "swiglu.hip", "ce_loss.hip", "optim.hip",
-           "grouped_gemm.hip", "moe.hip", "gemv.hip", "attention.hip"]
+           "grouped_gemm.hip", "moe.hip", "gemv.hip", "attention.hip",
+           "gemm8p.hip"]
 
 
 def _newer(a: Path, b: Path) -> bool:

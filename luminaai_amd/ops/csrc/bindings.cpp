@@ -50,6 +50,7 @@ hipError_t lumina_l2norm_sq_bf16(const void*, int64_t, float*, hipStream_t);
 hipError_t lumina_l2norm_sq_f32(const void*, int64_t, float*, hipStream_t);
 hipError_t lumina_adamw_step(float*, const void*, int, float*, float*, void*, int, int64_t, float, float, float, float, float, float, float, const float*, float, float, hipStream_t);
 void launch_grouped_gemm_nt(const void*, const void*, void*, int, int, int, int, int64_t, int64_t, int64_t, hipStream_t);
+void launch_gg8p(const void*, const void*, void*, int, int, int, int, int, int64_t, int64_t, int64_t, int, hipStream_t);
 void launch_grouped_gemm_nt_v2(const void*, const void*, void*, int, int, int, int, int64_t, int64_t, int64_t, int, hipStream_t);
 void launch_grouped_gemm_nt_v3(const void*, const void*, void*, int, int, int, int, int64_t, int64_t, int64_t, hipStream_t);
 void launch_grouped_gemm_nt_v4(const void*, const void*, void*, int, int, int, int, int64_t, int64_t, int64_t, hipStream_t);
@@ -460,6 +461,42 @@ std::tuple<at::Tensor, at::Tensor, at::Tensor> attn_bwd(
   return {dq, dk, dv};
 }
 
+// deep-pipelined 256^2 grouped GEMM (gemm8p.hip)
+// nt: out[e] = A[e] @ B[e]^T, A [E,M,K], B [E,N,K]; K % 32 == 0
+// nn: out[e] = A[e] @ B[e],   A [E,M,K], B [E,Kb,N]; K % 32, Kb <= K
+//     (A zero-padded along K up to a multiple of 32; B rows clamped)
+at::Tensor gg8p_nt(const at::Tensor& A, const at::Tensor& B) {
+  TORCH_CHECK(A.is_contiguous() && B.is_contiguous());
+  TORCH_CHECK(is_bf16(A) && is_bf16(B));
+  TORCH_CHECK(A.dim() == 3 && B.dim() == 3 && A.size(2) == B.size(2));
+  TORCH_CHECK(A.size(2) % 32 == 0, "gg8p requires K % 32 == 0");
+  const int E = (int)A.size(0), M = (int)A.size(1);
+  const int K = (int)A.size(2), N = (int)B.size(1);
+  auto O = at::empty({E, M, N}, A.options());
+  launch_gg8p(A.data_ptr(), B.data_ptr(), O.data_ptr(), E, M, N, K, K,
+              (int64_t)M * K, (int64_t)N * K, (int64_t)M * N, 0,
+              cur_stream());
+  check_hip(hipGetLastError(), "gg8p_nt");
+  return O;
+}
+
+at::Tensor gg8p_nn(const at::Tensor& A, const at::Tensor& B) {
+  TORCH_CHECK(A.is_contiguous() && B.is_contiguous());
+  TORCH_CHECK(is_bf16(A) && is_bf16(B));
+  TORCH_CHECK(A.dim() == 3 && B.dim() == 3);
+  const int E = (int)A.size(0), M = (int)A.size(1);
+  const int K = (int)A.size(2), Kb = (int)B.size(1), N = (int)B.size(2);
+  TORCH_CHECK(K % 32 == 0, "gg8p requires (padded) K % 32 == 0");
+  TORCH_CHECK(Kb <= K && K - Kb < 32, "K must be Kb padded up to 32");
+  TORCH_CHECK(N % 8 == 0, "gg8p_nn requires N % 8 == 0");
+  auto O = at::empty({E, M, N}, A.options());
+  launch_gg8p(A.data_ptr(), B.data_ptr(), O.data_ptr(), E, M, N, K, Kb,
+              (int64_t)M * K, (int64_t)Kb * N, (int64_t)M * N, 1,
+              cur_stream());
+  check_hip(hipGetLastError(), "gg8p_nn");
+  return O;
+}
+
 at::Tensor gemv(const at::Tensor& x, const at::Tensor& w) {
   // y[N] = w[N,K] @ x[K]
   TORCH_CHECK(x.is_contiguous() && w.is_contiguous());
@@ -479,6 +516,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("attn_bwd", &attn_bwd,
           "causal GQA flash attention backward -> (dQ, dK, dV) (gfx950)");
   mod.def("gemv", &gemv, "batch-1 decode GEMV y = W @ x (gfx950)");
+  mod.def("gg8p_nt", &gg8p_nt,
+          "256^2 pipelined grouped GEMM, A.B^T (gfx950)");
+  mod.def("gg8p_nn", &gg8p_nn,
+          "256^2 pipelined grouped GEMM, A.B with K-major B (gfx950)");
   mod.def("grouped_gemm_nt_v4", &grouped_gemm_nt_v4,
           "2-buffer raw-barrier counted-vmcnt variant (K%64==0)");
   mod.def("grouped_gemm_nt_v3", &grouped_gemm_nt_v3,
