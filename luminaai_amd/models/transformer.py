@@ -558,7 +558,8 @@ class MoEFFNLayer(nn.Module):
         logits = self.gate(xf).float()
         topw, topi, probs = ref_ops.topk_gating(
             logits, k, self.routing_temperature,
-            self.routing_noise_std, self.training)
+            self.routing_noise_std, self.training,
+            generator=getattr(self, "_routing_gen", None))
         if self.training and self.expert_dropout > 0:
             keep = (torch.rand_like(topw) > self.expert_dropout).float()
             topw = topw * keep
@@ -577,18 +578,9 @@ class MoEFFNLayer(nn.Module):
             else plan.counts[placement]        # back to per-expert order
         bufv = ops.interface.moe_dispatch(xf, plan).view(E, C, h)
 
-        # --- EP token exchange: [E, C, h] -> peers owning each expert shard.
-        # Shapes are static (capacity-bucketed), so the all-to-all needs no
-        # length metadata and maps 1:1 onto the 7 xGMI links.
+        # --- expert MLP on the local shard (grouped hipBLASLt GEMMs;
+        # optional fp8 e4m3 MFMA forward via the precision manager)
         EL = self.num_local_experts
-        if self.ep_size > 1:
-            from ..parallel.expert_parallel import all_to_all
-            bufv = all_to_all(bufv, self.ep_group)      # [ep*EL, C, h]
-            bufv = bufv.view(self.ep_size, EL, C, h).transpose(0, 1) \
-                .reshape(EL, self.ep_size * C, h)       # tokens per local expert
-
-        # --- grouped expert GEMMs (hipBLASLt strided-batched; optional fp8
-        # e4m3 MFMA forward when the precision manager enables it)
         if getattr(self, "use_fp8", False):
             from ..ops.fp8 import expert_bmm_fp8 as _ebmm
         else:
@@ -602,17 +594,34 @@ class MoEFFNLayer(nn.Module):
         else:
             w_gu = self.w_gate_up.to(x.dtype)
             w_dn = self.w_down.to(x.dtype)
-        gu = _ebmm(bufv, w_gu)
         I = self.intermediate_size
-        gu2 = gu.reshape(-1, 2 * I)
-        act = ops.swiglu(gu2.narrow(1, 0, I), gu2.narrow(1, I, I))
-        y = _ebmm(act.view(EL, -1, I), w_dn)
+        tp_group = getattr(self, "tp_group", None)
+
+        def _mlp(z):
+            gu = _ebmm(z, w_gu)
+            gu2 = gu.reshape(-1, 2 * I)
+            act = ops.swiglu(gu2.narrow(1, 0, I), gu2.narrow(1, I, I))
+            out = _ebmm(act.view(EL, -1, I), w_dn)
+            if tp_group is not None:
+                # TP over experts: w_gate_up column-sharded, w_down
+                # row-sharded -> partial sums reduce across the TP group
+                from ..parallel.tensor_parallel import tp_reduce
+                out = tp_reduce(out, tp_group)
+            return out
 
         if self.ep_size > 1:
-            from ..parallel.expert_parallel import all_to_all
-            y = y.view(EL, self.ep_size, C, h).transpose(0, 1) \
-                .reshape(self.ep_size * EL, C, h)       # back to [E, C, h] order
-            y = all_to_all(y, self.ep_group)
+            # EP token exchange pipelined against the expert GEMMs: the
+            # capacity dim is chunked so chunk i+1's all-to-all rides the
+            # RCCL stream under chunk i's GEMMs (reference DeepSpeed
+            # `overlap_alltoall: True`, trainer.py:842-843). Chunking is
+            # numerics-neutral, so CPU/gloo tests run the same path.
+            from ..parallel.expert_parallel import expert_pipeline
+            nch = 2 if C >= 2 and getattr(self, "overlap_alltoall", True) \
+                else 1
+            y = expert_pipeline(bufv, _mlp, self.ep_group, self.ep_size,
+                                n_chunks=nch)
+        else:
+            y = _mlp(bufv)
 
         # --- weighted combine back to token order (gather + k-reduce)
         yf = y.reshape(E * C, h)
@@ -647,55 +656,115 @@ class MoEFFNLayer(nn.Module):
         self._usage_counts.zero_()
 
     # ---- adaptive interventions -----------------------------------------
+    # ---- EP-aware resharding helpers ------------------------------------
+    def _gather_full_experts(self):
+        """All-gather the EP-sharded expert weights in global expert order.
+        Returns (w_gate_up [E,h,2I], w_down [E,I,h]) full tensors."""
+        import torch.distributed as dist
+        if self.ep_size <= 1:
+            return self.w_gate_up.data, self.w_down.data
+        gus = [torch.empty_like(self.w_gate_up.data)
+               for _ in range(self.ep_size)]
+        dns = [torch.empty_like(self.w_down.data)
+               for _ in range(self.ep_size)]
+        dist.all_gather(gus, self.w_gate_up.data.contiguous(),
+                        group=self.ep_group)
+        dist.all_gather(dns, self.w_down.data.contiguous(),
+                        group=self.ep_group)
+        return torch.cat(gus), torch.cat(dns)
+
+    def _reshard_experts(self, full_gu, full_dn, new_E):
+        """Install this rank's shard of the (re)built full expert stack."""
+        import torch.distributed as dist
+        self.num_experts = new_E
+        self.num_local_experts = new_E // self.ep_size
+        if self.ep_size > 1:
+            r = dist.get_process_group_ranks(self.ep_group).index(
+                dist.get_rank())
+            lo = r * self.num_local_experts
+            self.w_gate_up = nn.Parameter(
+                full_gu[lo:lo + self.num_local_experts].contiguous())
+            self.w_down = nn.Parameter(
+                full_dn[lo:lo + self.num_local_experts].contiguous())
+        else:
+            self.w_gate_up = nn.Parameter(full_gu.contiguous())
+            self.w_down = nn.Parameter(full_dn.contiguous())
+        self.placement = None   # load-balancer placement is stale
+        self._usage_counts = torch.zeros(
+            new_E, device=self._usage_counts.device)
+
+    def _shared_noise(self, shape, std, device, dtype):
+        """Noise identical on every EP rank (seed broadcast from the group's
+        first rank) so resharded weights agree bit-for-bit."""
+        import torch.distributed as dist
+        if self.ep_size > 1:
+            seed = torch.randint(0, 2 ** 31 - 1, (1,), device=device)
+            src = dist.get_process_group_ranks(self.ep_group)[0]
+            dist.broadcast(seed, src=src, group=self.ep_group)
+            g = torch.Generator(device=device)
+            g.manual_seed(int(seed.item()))
+            return torch.randn(shape, generator=g, device=device,
+                               dtype=torch.float32).to(dtype) * std
+        return torch.randn(shape, device=device, dtype=dtype) * std
+
     @torch.no_grad()
     def add_expert(self, noise_std: float = 0.01):
-        """Append one expert initialised to the mean of existing experts + noise
-        (reference trainer.py:1337-1376) and grow the gate."""
-        if self.ep_size > 1:
-            raise RuntimeError("dynamic expert add/prune is not supported "
-                               "under expert parallelism (EP shards are "
-                               "fixed-size); run with ep_size=1 to evolve "
-                               "the architecture")
-        E, h, I2 = self.w_gate_up.shape
-        new_gu = self.w_gate_up.mean(0, keepdim=True) + \
-            torch.randn(1, h, I2, device=self.w_gate_up.device,
-                        dtype=self.w_gate_up.dtype) * noise_std
-        new_dn = self.w_down.mean(0, keepdim=True) + \
-            torch.randn_like(self.w_down[:1]) * noise_std
-        self.w_gate_up = nn.Parameter(torch.cat([self.w_gate_up.data, new_gu]))
-        self.w_down = nn.Parameter(torch.cat([self.w_down.data, new_dn]))
+        """Grow the expert pool: +1 expert at ep_size == 1 (reference
+        trainer.py:1337-1376), +ep_size experts (one per shard) under EP so
+        the shards stay even.  Under EP the full stack is all-gathered,
+        extended with rank-identical new experts, and resharded (SURVEY
+        build plan 7.6: "expert add/prune must also rebalance EP shards")."""
+        grow = max(1, self.ep_size)
+        full_gu, full_dn = self._gather_full_experts()
+        E, h, I2 = full_gu.shape
+        mean_gu = full_gu.mean(0, keepdim=True)
+        mean_dn = full_dn.mean(0, keepdim=True)
+        news_gu = mean_gu + self._shared_noise(
+            (grow, h, I2), noise_std, full_gu.device, full_gu.dtype)
+        news_dn = mean_dn + self._shared_noise(
+            (grow,) + tuple(full_dn.shape[1:]), noise_std,
+            full_dn.device, full_dn.dtype)
+        full_gu = torch.cat([full_gu, news_gu])
+        full_dn = torch.cat([full_dn, news_dn])
         old_gate = self.gate
-        self.gate = nn.Linear(h, E + 1, bias=False,
+        self.gate = nn.Linear(h, E + grow, bias=False,
                               device=old_gate.weight.device,
                               dtype=old_gate.weight.dtype)
         self.gate.weight.data[:E] = old_gate.weight.data
-        self.gate.weight.data[E] = old_gate.weight.data.mean(0)
-        self.num_experts = E + 1
-        self.num_local_experts = E + 1
-        self._usage_counts = torch.zeros(E + 1, device=self._usage_counts.device)
+        self.gate.weight.data[E:] = old_gate.weight.data.mean(0)
+        self._reshard_experts(full_gu, full_dn, E + grow)
 
     @torch.no_grad()
-    def prune_expert(self, idx: int):
-        """Remove expert `idx` and its gate row (reference trainer.py:1378-1448)."""
-        if self.ep_size > 1:
-            raise RuntimeError("dynamic expert add/prune is not supported "
-                               "under expert parallelism")
+    def prune_expert(self, idx: Optional[int] = None):
+        """Shrink the expert pool: remove expert `idx` (or the least-used)
+        at ep_size == 1 (reference trainer.py:1378-1448); under EP remove
+        the ep_size least-used experts so shards stay even, then reshard."""
         E = self.num_experts
-        assert E > 1 and 0 <= idx < E
-        keep = [i for i in range(E) if i != idx]
-        kt = torch.tensor(keep, device=self.w_gate_up.device)
-        self.w_gate_up = nn.Parameter(self.w_gate_up.data[kt])
-        self.w_down = nn.Parameter(self.w_down.data[kt])
+        drop = max(1, self.ep_size)
+        assert E - drop >= max(self.top_k, 1), "too few experts to prune"
+        if idx is not None and self.ep_size <= 1:
+            drops = {idx}
+        else:
+            usage = self._usage_counts.clone()
+            if self.ep_size > 1:
+                # every EP rank must pick the SAME experts: use global usage
+                import torch.distributed as dist
+                dist.all_reduce(usage, group=self.ep_group)
+            order = usage.argsort()
+            drops = set(order[:drop].tolist())
+            if idx is not None:
+                drops = set(list(drops - {idx})[:drop - 1]) | {idx}
+        keep = [i for i in range(E) if i not in drops]
+        full_gu, full_dn = self._gather_full_experts()
+        kt = torch.tensor(keep, device=full_gu.device)
         old_gate = self.gate
         h = self.hidden_size
-        self.gate = nn.Linear(h, E - 1, bias=False,
+        self.gate = nn.Linear(h, len(keep), bias=False,
                               device=old_gate.weight.device,
                               dtype=old_gate.weight.dtype)
         self.gate.weight.data.copy_(old_gate.weight.data[kt])
-        self.num_experts = E - 1
-        self.num_local_experts = E - 1
-        self.top_k = min(self.top_k, self.num_experts)
-        self._usage_counts = torch.zeros(E - 1, device=self._usage_counts.device)
+        self.top_k = min(self.top_k, len(keep))
+        self._reshard_experts(full_gu[kt], full_dn[kt], len(keep))
 
 
 # ======================================================================

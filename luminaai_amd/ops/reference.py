@@ -164,15 +164,20 @@ def clip_grads_(grads: List[torch.Tensor], max_norm: float) -> torch.Tensor:
 def topk_gating(router_logits: torch.Tensor, top_k: int,
                 temperature: float = 1.0,
                 noise_std: float = 0.0,
-                training: bool = False):
+                training: bool = False,
+                generator=None):
     """Top-k softmax gating (reference model.py:1200-1217).
 
     router_logits: [N, E]. Returns (weights [N,k] fp32 normalized over the k,
     indices [N,k] long, full softmax probs [N,E] fp32 for the aux loss).
+    `generator` makes the routing noise reproducible -- tensor-parallel
+    ranks share a seeded generator so routing agrees bit-for-bit.
     """
     logits = router_logits.float()
     if training and noise_std > 0:
-        logits = logits + torch.randn_like(logits) * noise_std
+        noise = torch.randn(logits.shape, generator=generator,
+                            device=logits.device, dtype=logits.dtype)
+        logits = logits + noise * noise_std
     logits = logits / max(temperature, 1e-6)
     probs = logits.softmax(-1)
     topw, topi = probs.topk(top_k, dim=-1)

@@ -52,6 +52,102 @@ def all_to_all(x: torch.Tensor, group=None) -> torch.Tensor:
     return _AllToAllFn.apply(x, group)
 
 
+# ---------------------------------------------------------------------------
+# Overlapped (chunked) all-to-all: the exchange of chunk i+1 runs on the
+# RCCL communication stream while the expert GEMMs of chunk i occupy the
+# compute stream (reference capability: DeepSpeed `overlap_alltoall: True`,
+# trainer.py:842-843).  Implemented as a start/wait autograd pair (the
+# Tutel pattern): `a2a_start` launches the collective with async_op=True and
+# parks the work handle in a token registry; `a2a_wait` makes the compute
+# stream wait on it.  The backward mirrors this, so grad exchanges also
+# overlap grad GEMMs in reverse order.  Under gloo (CPU tests) wait() is a
+# host join -- numerics identical, overlap absent.
+_A2A_WORKS: dict = {}
+_A2A_NEXT = [0]
+
+
+def _launch_a2a(x: torch.Tensor, group):
+    out = torch.empty(x.shape, dtype=x.dtype, device=x.device)
+    work = dist.all_to_all_single(out, x.contiguous(), group=group,
+                                  async_op=True)
+    return out, work
+
+
+class _A2AStartFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, group, token):
+        ctx.group = group
+        ctx.token = token
+        out, work = _launch_a2a(x, group)
+        _A2A_WORKS[token] = work
+        return out
+
+    @staticmethod
+    def backward(ctx, gy):
+        # matching _A2AWaitFn.backward launched the grad exchange; join it
+        w = _A2A_WORKS.pop(("g", ctx.token))
+        w[1].wait()
+        return w[0], None, None
+
+
+class _A2AWaitFn(torch.autograd.Function):
+    """Joins the exchange started by the matching _A2AStartFn.
+
+    Backward INVERTS the pair: it STARTS the grad exchange and returns the
+    (possibly still in-flight) output buffer -- autograd hands that buffer
+    only to _A2AStartFn.backward (the matching producer), which JOINS it
+    before anything reads the data."""
+
+    @staticmethod
+    def forward(ctx, x, group, token):
+        ctx.group = group
+        ctx.token = token
+        w = _A2A_WORKS.pop(token)
+        w.wait()
+        return x.view_as(x)
+
+    @staticmethod
+    def backward(ctx, gy):
+        out, work = _launch_a2a(gy, ctx.group)
+        _A2A_WORKS[("g", ctx.token)] = (out, work)
+        return out, None, None
+
+
+def expert_pipeline(bufv: torch.Tensor, mlp_fn, group, ep: int,
+                    n_chunks: int = 2):
+    """Dispatch-exchange -> expert MLP -> return-exchange with the capacity
+    dim split into `n_chunks` so exchange and compute overlap.
+
+    bufv: [E, C, h] capacity-bucketed tokens (global expert order).
+    mlp_fn: [E_local, ep*Cc, h] -> [E_local, ep*Cc, h].
+    Returns [E, C, h] in global expert order.
+    """
+    E, C, h = bufv.shape
+    EL = E // ep
+    n = min(n_chunks, C) if n_chunks > 1 else 1
+    chunks = list(bufv.chunk(n, dim=1))
+    tokens = []
+    for c in chunks:
+        t = _A2A_NEXT[0]
+        _A2A_NEXT[0] += 1
+        tokens.append(t)
+    started = [_A2AStartFn.apply(c.contiguous(), group, t)
+               for c, t in zip(chunks, tokens)]
+    rets, rtokens = [], []
+    for s, t, c in zip(started, tokens, chunks):
+        cc = c.shape[1]
+        z = _A2AWaitFn.apply(s, group, t)
+        z = z.view(ep, EL, cc, h).transpose(0, 1).reshape(EL, ep * cc, h)
+        y = mlp_fn(z)
+        y = y.view(EL, ep, cc, h).transpose(0, 1).reshape(ep * EL, cc, h)
+        rt = _A2A_NEXT[0]
+        _A2A_NEXT[0] += 1
+        rtokens.append(rt)
+        rets.append(_A2AStartFn.apply(y.contiguous(), group, rt))
+    outs = [_A2AWaitFn.apply(r, group, t) for r, t in zip(rets, rtokens)]
+    return torch.cat(outs, dim=1)
+
+
 def is_expert_param(name: str) -> bool:
     """Parameters sharded along EP (not DP-replicated)."""
     return ".w_gate_up" in name or ".w_down" in name

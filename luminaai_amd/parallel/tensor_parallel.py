@@ -119,7 +119,7 @@ def convert_to_tensor_parallel(model, mesh) -> int:
         attn.tp_group = group
 
         ffn = layer.ffn
-        if hasattr(ffn, "gate_up_proj"):       # dense SwiGLU only
+        if hasattr(ffn, "gate_up_proj"):       # dense SwiGLU
             I = ffn.intermediate_size
             assert I % tp == 0
             li = I // tp
@@ -131,5 +131,38 @@ def convert_to_tensor_parallel(model, mesh) -> int:
             _replace_linear(ffn, "down_proj", dw[:, r * li:(r + 1) * li])
             ffn.intermediate_size = li
             ffn.tp_group = group
+        elif hasattr(ffn, "w_gate_up"):        # MoE expert stack
+            # Megatron-style TP over the batched expert weights (reference
+            # capability: ColossalAI hybrid_parallel_plugin.py:880 /
+            # moe_hybrid_parallel_plugin.py:92): gate/up column-parallel on
+            # the intermediate dim, down row-parallel; the partial expert
+            # outputs all-reduce across the TP group inside the layer
+            # forward (transformer.py MoEFFNLayer._mlp).
+            I = ffn.intermediate_size
+            assert I % tp == 0
+            li = I // tp
+            gu = ffn.w_gate_up.data            # [EL, h, 2I]
+            gslice = gu[:, :, r * li:(r + 1) * li]
+            uslice = gu[:, :, I + r * li:I + (r + 1) * li]
+            wg = torch.cat([gslice, uslice], dim=2).contiguous()
+            dn = ffn.w_down.data[:, r * li:(r + 1) * li, :].contiguous()
+            ffn.w_gate_up = nn.Parameter(wg)
+            ffn.w_down = nn.Parameter(dn)
+            ffn.w_gate_up._shard_parallel = True
+            ffn.w_down._shard_parallel = True
+            ffn.intermediate_size = li
+            ffn.tp_group = group
+            # routing must be IDENTICAL on every TP rank: give the layer a
+            # shared-seed generator for its gating noise
+            seed = torch.tensor([torch.initial_seed() % (2 ** 31)],
+                                dtype=torch.int64)
+            if dist.is_initialized():
+                seed = seed.to("cuda" if dist.get_backend(group) == "nccl"
+                               else "cpu")
+                dist.broadcast(seed, src=dist.get_process_group_ranks(group)[0],
+                               group=group)
+            g = torch.Generator(device=ffn.w_gate_up.device)
+            g.manual_seed(int(seed.item()))
+            ffn._routing_gen = g
         n += 1
     return n

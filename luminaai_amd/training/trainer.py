@@ -633,9 +633,10 @@ class Trainer:
 
     def add_expert(self) -> bool:
         layers = self._moe_layers()
-        if not layers or self.engine.stage >= 3 or \
-                (self.mesh is not None and self.mesh.ep_size > 1):
-            return False  # EP shards are fixed-size; use load balancing
+        if not layers or self.engine.stage >= 3:
+            return False  # ZeRO-3 flat shards cannot grow in place
+        # under EP each layer grows by ep_size experts (one per shard) and
+        # reshards; the fresh optimizer below re-flattens the new shapes
         for l in layers:
             l.add_expert()
         self.optimizer.rebuild(self.model)
@@ -646,9 +647,9 @@ class Trainer:
 
     def prune_expert(self, expert_idx: Optional[int] = None) -> bool:
         layers = self._moe_layers()
-        if not layers or layers[0].num_experts <= 2 or \
-                self.engine.stage >= 3 or \
-                (self.mesh is not None and self.mesh.ep_size > 1):
+        ep = self.mesh.ep_size if self.mesh is not None else 1
+        if not layers or layers[0].num_experts - max(1, ep) < \
+                max(layers[0].top_k, 2) or self.engine.stage >= 3:
             return False
         for l in layers:
             idx = expert_idx
