@@ -51,6 +51,9 @@ hipError_t lumina_l2norm_sq_f32(const void*, int64_t, float*, hipStream_t);
 hipError_t lumina_adamw_step(float*, const void*, int, float*, float*, void*, int, int64_t, float, float, float, float, float, float, float, const float*, float, float, hipStream_t);
 void launch_grouped_gemm_nt(const void*, const void*, void*, int, int, int, int, int64_t, int64_t, int64_t, hipStream_t);
 void launch_gg8p(const void*, const void*, void*, int, int, int, int, int, int64_t, int64_t, int64_t, int, hipStream_t);
+hipError_t lumina_mx_quant_rows(const void*, void*, void*, int64_t, int, int, hipStream_t);
+hipError_t lumina_mx_quant_cols(const void*, void*, void*, void*, int, int, int, hipStream_t);
+hipError_t lumina_gg_mx_nt(const void*, const void*, const void*, const void*, void*, int, int, int, int, int64_t, int64_t, int64_t, hipStream_t);
 void launch_grouped_gemm_nt_v2(const void*, const void*, void*, int, int, int, int, int64_t, int64_t, int64_t, int, hipStream_t);
 void launch_grouped_gemm_nt_v3(const void*, const void*, void*, int, int, int, int, int64_t, int64_t, int64_t, hipStream_t);
 void launch_grouped_gemm_nt_v4(const void*, const void*, void*, int, int, int, int, int64_t, int64_t, int64_t, hipStream_t);
@@ -497,6 +500,60 @@ at::Tensor gg8p_nn(const at::Tensor& A, const at::Tensor& B) {
   return O;
 }
 
+// ---- MX-fp8 rowwise path (mxfp8.hip) -------------------------------------
+std::tuple<at::Tensor, at::Tensor> mx_quant_rows(const at::Tensor& x,
+                                                 int64_t kp) {
+  TORCH_CHECK(x.is_contiguous() && is_bf16(x));
+  const int K = (int)x.size(-1);
+  const int Kp = (int)(kp > 0 ? kp : (K + 127) / 128 * 128);
+  TORCH_CHECK(Kp % 128 == 0 && Kp >= K);
+  const int64_t R = x.numel() / K;
+  auto sizes = x.sizes().vec();
+  sizes.back() = Kp;
+  auto q = at::empty(sizes, x.options().dtype(at::kByte));
+  auto ssz = x.sizes().vec();
+  ssz.pop_back();
+  auto s = at::empty(ssz, x.options().dtype(at::kByte));
+  check_hip(lumina_mx_quant_rows(x.data_ptr(), q.data_ptr(), s.data_ptr(),
+                                 R, K, Kp, cur_stream()),
+            "mx_quant_rows");
+  return {q, s};
+}
+
+std::tuple<at::Tensor, at::Tensor> mx_quant_cols(const at::Tensor& w,
+                                                 int64_t kp) {
+  TORCH_CHECK(w.is_contiguous() && is_bf16(w) && w.dim() == 2);
+  const int K = (int)w.size(0), N = (int)w.size(1);
+  const int Kp = (int)(kp > 0 ? kp : (K + 127) / 128 * 128);
+  TORCH_CHECK(Kp % 128 == 0 && Kp >= K);
+  auto q = at::empty({N, Kp}, w.options().dtype(at::kByte));
+  auto s = at::empty({N}, w.options().dtype(at::kByte));
+  auto ws = at::empty({N}, w.options().dtype(at::kFloat));
+  check_hip(lumina_mx_quant_cols(w.data_ptr(), q.data_ptr(), s.data_ptr(),
+                                 ws.data_ptr(), K, N, Kp, cur_stream()),
+            "mx_quant_cols");
+  return {q, s};
+}
+
+at::Tensor gg_mx_nt(const at::Tensor& Aq, const at::Tensor& As,
+                    const at::Tensor& Bq, const at::Tensor& Bs) {
+  TORCH_CHECK(Aq.is_contiguous() && Bq.is_contiguous());
+  TORCH_CHECK(Aq.scalar_type() == at::kByte && Bq.scalar_type() == at::kByte);
+  TORCH_CHECK(Aq.dim() == 3 && Bq.dim() == 3 && Aq.size(2) == Bq.size(2));
+  TORCH_CHECK(Aq.size(2) % 128 == 0, "MX GEMM needs padded K % 128 == 0");
+  const int E = (int)Aq.size(0), M = (int)Aq.size(1);
+  const int K = (int)Aq.size(2), N = (int)Bq.size(1);
+  TORCH_CHECK(As.numel() == (int64_t)E * M && Bs.numel() == (int64_t)E * N);
+  auto O = at::empty({E, M, N},
+                     Aq.options().dtype(at::kBFloat16));
+  check_hip(lumina_gg_mx_nt(Aq.data_ptr(), Bq.data_ptr(), As.data_ptr(),
+                            Bs.data_ptr(), O.data_ptr(), E, M, N, K,
+                            (int64_t)M * K, (int64_t)N * K, (int64_t)M * N,
+                            cur_stream()),
+            "gg_mx_nt");
+  return O;
+}
+
 at::Tensor gemv(const at::Tensor& x, const at::Tensor& w) {
   // y[N] = w[N,K] @ x[K]
   TORCH_CHECK(x.is_contiguous() && w.is_contiguous());
@@ -516,6 +573,12 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("attn_bwd", &attn_bwd,
           "causal GQA flash attention backward -> (dQ, dK, dV) (gfx950)");
   mod.def("gemv", &gemv, "batch-1 decode GEMV y = W @ x (gfx950)");
+  mod.def("mx_quant_rows", &mx_quant_rows,
+          "rowwise e8m0 fp8 quantization, K zero-padded (gfx950)");
+  mod.def("mx_quant_cols", &mx_quant_cols,
+          "columnwise e8m0 fp8 quantization + transpose (gfx950)");
+  mod.def("gg_mx_nt", &gg_mx_nt,
+          "grouped MX-fp8 GEMM A.B^T at the ~5PF fp8 MFMA rate (gfx950)");
   mod.def("gg8p_nt", &gg8p_nt,
           "256^2 pipelined grouped GEMM, A.B^T (gfx950)");
   mod.def("gg8p_nn", &gg8p_nn,

@@ -40,6 +40,46 @@ def _micro_batches(batch: Dict, n: int) -> List[Dict]:
     return out
 
 
+def build_pp_tp_groups(world: int, pp: int, tp: int):
+    """Groups for a dp x pp x tp mesh, rank = dp*(pp*tp) + stage*tp + tp_rank
+    (tp innermost: a stage's TP shard group is contiguous -> one xGMI hop).
+    Returns (pp_group, tp_group, dp_group, dp_idx, stage, tp_rank); every
+    rank must call with identical args (dist.new_group is collective).
+    Reference capability: ColossalAI HybridParallelPlugin's TPxPPxDP mesh
+    (booster/plugin/hybrid_parallel_plugin.py:880)."""
+    rank = comm.get_rank()
+    assert world % (pp * tp) == 0
+    dp = world // (pp * tp)
+    pp_groups, tp_groups, dp_groups = {}, {}, {}
+    for d in range(dp):
+        for t in range(tp):
+            pp_groups[(d, t)] = dist.new_group(
+                [d * pp * tp + s * tp + t for s in range(pp)])
+        for s in range(pp):
+            tp_groups[(d, s)] = dist.new_group(
+                [d * pp * tp + s * tp + t for t in range(tp)])
+    for s in range(pp):
+        for t in range(tp):
+            dp_groups[(s, t)] = dist.new_group(
+                [d * pp * tp + s * tp + t for d in range(dp)])
+    dp_idx = rank // (pp * tp)
+    stage = (rank // tp) % pp
+    tp_rank = rank % tp
+    return (pp_groups[(dp_idx, tp_rank)], tp_groups[(dp_idx, stage)],
+            dp_groups[(stage, tp_rank)], dp_idx, stage, tp_rank)
+
+
+class _TPShim:
+    """Minimal mesh view for convert_to_tensor_parallel inside a pp x tp
+    composition (the global mesh stays unset; groups come from
+    build_pp_tp_groups)."""
+
+    def __init__(self, tp_size, tp_rank, tp_group):
+        self.tp_size = tp_size
+        self.tp_rank = tp_rank
+        self.tp_group = tp_group
+
+
 def build_pp_dp_groups(world: int, pp: int):
     """(pp_group, dp_group, dp_idx, stage) for this rank; every rank must
     call this with identical arguments (dist.new_group is collective)."""
@@ -54,18 +94,28 @@ def build_pp_dp_groups(world: int, pp: int):
 
 
 def run_pipeline_training(model, cfg, train_ds, logger, pp: int = 0,
-                          virtual_stages: int = 1,
+                          virtual_stages: int = 1, tp: int = 1,
                           steps: Optional[int] = None) -> Dict:
-    """Train `model` over a pp (x dp) world. Returns summary stats."""
+    """Train `model` over a pp (x tp) (x dp) world. Returns summary stats."""
     from torch.utils.data import DataLoader, Subset
 
     world = comm.get_world_size()
     rank = comm.get_rank()
-    pp = pp or world
-    dp = world // pp
+    pp = pp or (world // max(tp, 1))
+    dp = world // (pp * max(tp, 1))
     pp_group = dp_group = None
     dp_idx, stage = 0, rank
-    if dp > 1:
+    if tp > 1:
+        from ..parallel.tensor_parallel import convert_to_tensor_parallel
+        pp_group, tp_group, dp_group, dp_idx, stage, tp_rank = \
+            build_pp_tp_groups(world, pp, tp)
+        convert_to_tensor_parallel(model, _TPShim(tp, tp_rank, tp_group))
+        if dp > 1 and hasattr(train_ds, "__len__"):
+            idx = list(range(dp_idx, len(train_ds), dp))
+            train_ds = Subset(train_ds, idx)
+        if dp == 1:
+            dp_group = None
+    elif dp > 1:
         pp_group, dp_group, dp_idx, stage = build_pp_dp_groups(world, pp)
         # map-style datasets shard across DP replicas (all stages of one
         # column already share the stream: per-column seed in main.py)

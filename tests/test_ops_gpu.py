@@ -600,3 +600,70 @@ def test_attn_in_model_training_path():
     for p in model.parameters():
         if p.grad is not None:
             assert torch.isfinite(p.grad.float()).all()
+
+
+# ------------------------------------------------------------- MX-fp8 path
+def test_mx_quant_roundtrip():
+    from luminaai_amd.ops import get_ext
+    ext = get_ext()
+    torch.manual_seed(0)
+    x = torch.randn(64, 200, device=_dev(), dtype=torch.bfloat16) * 3
+    q, s = ext.mx_quant_rows(x, 0)
+    assert q.shape == (64, 256) and s.shape == (64,)
+    # decode: value = e4m3(q) * 2^(s-127)
+    scale = (s.float() - 127).exp2().unsqueeze(1)
+    deq = q[:, :200].view(torch.float8_e4m3fn).float() * scale
+    err = (deq - x.float()).abs().max(dim=1).values
+    amax = x.float().abs().max(dim=1).values
+    assert (err <= amax * 0.075 + 1e-3).all(), err.max()
+    assert (q[:, 200:].view(torch.float8_e4m3fn).float() == 0).all()
+
+
+def test_mx_quant_cols_transpose():
+    from luminaai_amd.ops import get_ext
+    ext = get_ext()
+    torch.manual_seed(1)
+    w = torch.randn(200, 96, device=_dev(), dtype=torch.bfloat16)
+    q, s = ext.mx_quant_cols(w, 0)
+    assert q.shape == (96, 256)
+    scale = (s.float() - 127).exp2().unsqueeze(1)
+    deq = q[:, :200].view(torch.float8_e4m3fn).float() * scale
+    ref = w.float().t()
+    err = (deq - ref).abs().max()
+    assert err <= ref.abs().max() * 0.075 + 1e-3, err
+
+
+def test_mx_gemm_matches_bf16():
+    from luminaai_amd.ops import get_ext
+    ext = get_ext()
+    torch.manual_seed(2)
+    E, M, N, K = 3, 300, 256, 200
+    a = torch.randn(E, M, K, device=_dev(), dtype=torch.bfloat16)
+    b = torch.randn(E, N, K, device=_dev(), dtype=torch.bfloat16)
+    qa, sa = ext.mx_quant_rows(a, 0)
+    qb, sb = ext.mx_quant_rows(b.reshape(E * N, K), 0)
+    out = ext.gg_mx_nt(qa, sa, qb.view(E, N, -1), sb.view(E, N))
+    ref = torch.matmul(a.float(), b.float().transpose(1, 2))
+    rel = (out.float() - ref).abs().max() / ref.abs().max()
+    assert rel < 0.05, float(rel)
+
+
+def test_mx_expert_bmm_autograd():
+    from luminaai_amd.ops.fp8 import expert_bmm_fp8, mx_available, \
+        invalidate_weight_cache
+    if not mx_available():
+        pytest.skip("mx kernels not built")
+    invalidate_weight_cache()
+    torch.manual_seed(3)
+    E, C, K, N = 2, 64, 96, 128
+    x = torch.randn(E, C, K, device=_dev(), dtype=torch.bfloat16,
+                    requires_grad=True)
+    w = torch.randn(E, K, N, device=_dev(), dtype=torch.bfloat16,
+                    requires_grad=True) * 0.1
+    y = expert_bmm_fp8(x, w)
+    ref = torch.matmul(x.detach().float(), w.detach().float())
+    rel = (y.float() - ref).abs().max() / ref.abs().max().clamp_min(1e-3)
+    assert rel < 0.06, float(rel)
+    y.sum().backward()
+    assert torch.isfinite(x.grad.float()).all()
+    assert torch.isfinite(w.grad.float()).all()

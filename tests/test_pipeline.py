@@ -462,3 +462,119 @@ def test_load_checkpoint_smart_handles_pp_stage(tmp_path):
     assert not unexpected
     for k, v in model.state_dict().items():
         torch.testing.assert_close(payload["model_state_dict"][k], v, msg=k)
+
+
+# ---- PP x TP composition (round-2: VERDICT item 10) ------------------------
+def pptp_worker(rank, world):
+    """2 PP stages x 2 TP shards: after one optimizer step the stage/shard
+    weights must match a single-process full-model run on the same data."""
+    from luminaai_amd.training.pipeline_loop import (_micro_batches,
+                                                     build_pp_tp_groups,
+                                                     _TPShim)
+    from luminaai_amd.models import DeepSeekTransformer
+    from luminaai_amd.parallel.pipeline import PipelineParallelEngine
+    from luminaai_amd.parallel.tensor_parallel import convert_to_tensor_parallel
+    from luminaai_amd.training.optimizer import FlatAdamW
+    from luminaai_amd.ops import fused_cross_entropy
+
+    pp, tp = 2, 2
+    # dense untied config: TP reorders float sums, and MoE top-k routing can
+    # flip on near-ties under that noise -- TP-over-MoE equivalence is
+    # covered by test_distributed.test_tp_moe_forward_backward (world 2);
+    # here the PP x TP COMPOSITION is what must be exact.
+    from luminaai_amd.models.transformer import DeepSeekConfig
+    mcfg = DeepSeekConfig(vocab_size=512, hidden_size=64, num_layers=4,
+                          num_heads=4, num_kv_heads=2, intermediate_size=128,
+                          seq_length=32, use_moe=False, use_mod=False,
+                          tie_word_embeddings=False)
+    torch.manual_seed(1234)
+    model = DeepSeekTransformer(mcfg)
+    pp_group, tp_group, dp_group, dp_idx, stage, tp_rank = \
+        build_pp_tp_groups(world, pp, tp)
+    convert_to_tensor_parallel(model, _TPShim(tp, tp_rank, tp_group))
+    engine = PipelineParallelEngine(model, None, pp_group=pp_group)
+    opt = FlatAdamW(engine.stage, lr=1e-2, weight_decay=0.0,
+                    max_grad_norm=0.0)
+
+    torch.manual_seed(900)                    # SAME data everywhere
+    ids = torch.randint(1, mcfg.vocab_size, (4, 17))
+    batch = {"input_ids": ids[:, :-1], "labels": ids[:, 1:]}
+    micro = _micro_batches(batch, 2)
+    engine.train_batch(micro)
+    opt.step(grad_scale=1.0 / len(micro))
+
+    torch.manual_seed(1234)
+    ref = DeepSeekTransformer(mcfg)
+    ref_opt = FlatAdamW(ref, lr=1e-2, weight_decay=0.0, max_grad_norm=0.0)
+    torch.manual_seed(900)
+    rids = torch.randint(1, mcfg.vocab_size, (4, 17))
+    for mb in _micro_batches({"input_ids": rids[:, :-1],
+                              "labels": rids[:, 1:]}, 2):
+        logits, aux, _ = ref(mb["input_ids"])
+        ce, _, _ = fused_cross_entropy(logits, mb["labels"])
+        (ce + aux).backward()
+    ref_opt.step(grad_scale=1.0 / 2)
+
+    if stage == 0:
+        a = engine.stage.embed_tokens.weight.detach()
+        b = ref.embed_tokens.weight.detach()
+        # TP-sharded qkv of the first layer: compare this rank's shard
+        aw = engine.stage.layers[0].attention.qkv_proj.weight.detach()
+        hd = mcfg.hidden_size // mcfg.num_heads
+        lh = mcfg.num_heads // tp
+        lkv = mcfg.num_kv_heads // tp
+        W = ref.layers[0].attention.qkv_proj.weight.detach()
+        qs = mcfg.num_heads * hd
+        kvs = mcfg.num_kv_heads * hd
+        qw = W[:qs].view(mcfg.num_heads, hd, -1)[tp_rank*lh:(tp_rank+1)*lh]
+        kw = W[qs:qs+kvs].view(mcfg.num_kv_heads, hd, -1)[
+            tp_rank*lkv:(tp_rank+1)*lkv]
+        vw = W[qs+kvs:].view(mcfg.num_kv_heads, hd, -1)[
+            tp_rank*lkv:(tp_rank+1)*lkv]
+        bw = torch.cat([qw.reshape(-1, W.shape[1]),
+                        kw.reshape(-1, W.shape[1]),
+                        vw.reshape(-1, W.shape[1])])
+        shard_err = float((aw - bw).abs().max())
+    else:
+        a = engine.stage.lm_head.weight.detach()
+        b = ref.lm_head.weight.detach()
+        shard_err = 0.0
+    return {"err": float((a - b).abs().max()), "shard_err": shard_err,
+            "stage": stage, "tp_rank": tp_rank}
+
+
+def _run_pptp(rank, world, port, q):
+    os.environ.update({
+        "MASTER_ADDR": "127.0.0.1", "MASTER_PORT": str(port),
+        "RANK": str(rank), "LOCAL_RANK": str(rank),
+        "WORLD_SIZE": str(world),
+    })
+    dist.init_process_group("gloo", init_method="env://", rank=rank,
+                            world_size=world)
+    try:
+        q.put((rank, "ok", pptp_worker(rank, world)))
+    except Exception:  # noqa: BLE001
+        import traceback
+        q.put((rank, "err", traceback.format_exc()))
+    finally:
+        dist.destroy_process_group()
+
+
+def test_pp_tp_composition_matches_single_process():
+    ctx = mp.get_context("spawn")
+    q = ctx.SimpleQueue()
+    port = _free_port()
+    procs = [ctx.Process(target=_run_pptp, args=(r, 4, port, q))
+             for r in range(4)]
+    for p in procs:
+        p.start()
+    results = {}
+    for _ in range(4):
+        rank, status, payload = q.get()
+        assert status == "ok", f"rank {rank} failed:\n{payload}"
+        results[rank] = payload
+    for p in procs:
+        p.join(timeout=180)
+    for r in range(4):
+        assert results[r]["err"] < 2e-4, results
+        assert results[r]["shard_err"] < 2e-4, results
