@@ -52,7 +52,7 @@ hipError_t lumina_adamw_step(float*, const void*, int, float*, float*, void*, in
 void launch_grouped_gemm_nt(const void*, const void*, void*, int, int, int, int, int64_t, int64_t, int64_t, hipStream_t);
 void launch_gg8p(const void*, const void*, void*, int, int, int, int, int, int64_t, int64_t, int64_t, int, hipStream_t);
 hipError_t lumina_mx_quant_rows(const void*, void*, void*, int64_t, int, int, hipStream_t);
-hipError_t lumina_mx_quant_cols(const void*, void*, void*, void*, int, int, int, hipStream_t);
+hipError_t lumina_mx_quant_cols(const void*, void*, void*, void*, int, int, int, int, hipStream_t);
 hipError_t lumina_gg_mx_nt(const void*, const void*, const void*, const void*, void*, int, int, int, int, int64_t, int64_t, int64_t, hipStream_t);
 void launch_grouped_gemm_nt_v2(const void*, const void*, void*, int, int, int, int, int64_t, int64_t, int64_t, int, hipStream_t);
 void launch_grouped_gemm_nt_v3(const void*, const void*, void*, int, int, int, int, int64_t, int64_t, int64_t, hipStream_t);
@@ -522,15 +522,22 @@ std::tuple<at::Tensor, at::Tensor> mx_quant_rows(const at::Tensor& x,
 
 std::tuple<at::Tensor, at::Tensor> mx_quant_cols(const at::Tensor& w,
                                                  int64_t kp) {
-  TORCH_CHECK(w.is_contiguous() && is_bf16(w) && w.dim() == 2);
-  const int K = (int)w.size(0), N = (int)w.size(1);
+  // w [K, N] or batched [E, K, N] -> transposed quant [.., N, Kp]
+  TORCH_CHECK(w.is_contiguous() && is_bf16(w));
+  TORCH_CHECK(w.dim() == 2 || w.dim() == 3);
+  const int E = w.dim() == 3 ? (int)w.size(0) : 1;
+  const int K = (int)w.size(-2), N = (int)w.size(-1);
   const int Kp = (int)(kp > 0 ? kp : (K + 127) / 128 * 128);
   TORCH_CHECK(Kp % 128 == 0 && Kp >= K);
-  auto q = at::empty({N, Kp}, w.options().dtype(at::kByte));
-  auto s = at::empty({N}, w.options().dtype(at::kByte));
-  auto ws = at::empty({N}, w.options().dtype(at::kFloat));
+  auto q = w.dim() == 3
+      ? at::empty({E, N, Kp}, w.options().dtype(at::kByte))
+      : at::empty({N, Kp}, w.options().dtype(at::kByte));
+  auto s = w.dim() == 3
+      ? at::empty({E, N}, w.options().dtype(at::kByte))
+      : at::empty({N}, w.options().dtype(at::kByte));
+  auto ws = at::empty({(int64_t)E * N}, w.options().dtype(at::kFloat));
   check_hip(lumina_mx_quant_cols(w.data_ptr(), q.data_ptr(), s.data_ptr(),
-                                 ws.data_ptr(), K, N, Kp, cur_stream()),
+                                 ws.data_ptr(), E, K, N, Kp, cur_stream()),
             "mx_quant_cols");
   return {q, s};
 }

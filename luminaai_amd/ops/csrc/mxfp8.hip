@@ -91,22 +91,34 @@ __global__ void mx_quant_rows_kernel(const uint16_t* __restrict__ X,
 // column quant + transpose: W [K, N] bf16 -> Wq [N, Kp] fp8, S [N].
 // Phase kernels: col-amax (coalesced along N), then 32x32 LDS-tiled
 // transpose with quantization.
-__global__ void mx_colmax_kernel(const uint16_t* __restrict__ W,
-                                 float* __restrict__ amax,
+// tiled column-amax: block covers [512 k x 256 n], coalesced along N;
+// per-column partials combine via atomicMax on the uint32 view (valid for
+// non-negative floats).  grid (N/256, K/512, E).
+__global__ void mx_colmax_kernel(const uint16_t* __restrict__ Wall,
+                                 float* __restrict__ amaxall,
                                  int K, int N) {
+  const uint16_t* W = Wall + (int64_t)blockIdx.z * K * N;
+  float* amax = amaxall + (int64_t)blockIdx.z * N;
   const int n = blockIdx.x * blockDim.x + threadIdx.x;
   if (n >= N) return;
+  const int k0 = blockIdx.y * 512;
+  const int k1 = k0 + 512 < K ? k0 + 512 : K;
   float m = 0.0f;
-  for (int k = 0; k < K; ++k)
+  for (int k = k0; k < k1; ++k)
     m = fmaxf(m, fabsf(bf16_to_f32(W[(int64_t)k * N + n])));
-  amax[n] = m;
+  atomicMax(reinterpret_cast<unsigned*>(amax + n),
+            __float_as_uint(m));
 }
 
-__global__ void mx_quant_t_kernel(const uint16_t* __restrict__ W,
-                                  const float* __restrict__ amax,
-                                  uint8_t* __restrict__ Q,
-                                  uint8_t* __restrict__ S,
+__global__ void mx_quant_t_kernel(const uint16_t* __restrict__ Wall,
+                                  const float* __restrict__ amaxall,
+                                  uint8_t* __restrict__ Qall,
+                                  uint8_t* __restrict__ Sall,
                                   int K, int N, int Kp) {
+  const uint16_t* W = Wall + (int64_t)blockIdx.z * K * N;
+  const float* amax = amaxall + (int64_t)blockIdx.z * N;
+  uint8_t* Q = Qall + (int64_t)blockIdx.z * N * Kp;
+  uint8_t* S = Sall + (int64_t)blockIdx.z * N;
   __shared__ float tile[32][33];
   const int kb = blockIdx.x * 32;
   const int nb = blockIdx.y * 32;
@@ -141,11 +153,11 @@ __global__ void mx_quant_t_kernel(const uint16_t* __restrict__ W,
 }
 
 // zero the K-pad tail of the transposed quant output ([N, Kp], K..Kp-1)
-__global__ void mx_zero_tail_kernel(uint8_t* __restrict__ Q, int64_t N,
+__global__ void mx_zero_tail_kernel(uint8_t* __restrict__ Q, int64_t NE,
                                     int K, int Kp) {
   const int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
   const int tail = Kp - K;
-  if (i < N * tail)
+  if (i < NE * tail)
     Q[(i / tail) * Kp + K + (i % tail)] = 0;
 }
 
@@ -168,7 +180,7 @@ DEV_INLINE int mx_img(int row, int kbyte) {
   return row * MX_BK + slot * 16 + (kbyte & 15);
 }
 
-__global__ __launch_bounds__(256, 1)
+__global__ __launch_bounds__(256, 2)
 void gg_mx_nt_kernel(const uint8_t* __restrict__ Aall,
                      const uint8_t* __restrict__ Ball,
                      const uint8_t* __restrict__ SAall,
@@ -176,7 +188,11 @@ void gg_mx_nt_kernel(const uint8_t* __restrict__ Aall,
                      uint16_t* __restrict__ Oall,
                      int M, int N, int K,
                      int64_t sA, int64_t sB, int64_t sO) {
-  __shared__ uint8_t lds[4 * MX_TILE];         // [A0|B0|A1|B1]
+  // 2-deep tile-pair ring (64 KiB -> 2 blocks/CU): the j+1 prefetch issued
+  // at the top of step j targets the buffer step j-1 retired; the counted
+  // drain at the step end exposes ~300 cycles of HBM latency per step,
+  // which the co-resident second block covers.
+  __shared__ uint8_t lds[4 * MX_TILE];
 
   const int e = blockIdx.z;
   const uint8_t* A = Aall + e * sA;
@@ -205,10 +221,10 @@ void gg_mx_nt_kernel(const uint8_t* __restrict__ Aall,
     sb_e[m] = SB[r > maxB ? maxB : r];
   }
 
-  // staging: per thread 2 chunks of 16B per operand tile (128*128/16/256)
-  int c_row[2], c_koff[2];
+  // staging: 16 KiB tile = 1024 chunks of 16B -> 4 chunks per thread
+  int c_row[4], c_koff[4];
   #pragma unroll
-  for (int i = 0; i < 2; ++i) {
+  for (int i = 0; i < 4; ++i) {
     const int c = t + i * 256;
     c_row[i] = c >> 3;                         // 8 chunks per row
     const int s = c & 7;
@@ -219,7 +235,7 @@ void gg_mx_nt_kernel(const uint8_t* __restrict__ Aall,
     uint8_t* base = lds + ((j & 1) * 2 + which) * MX_TILE;
     const int k0 = j * MX_BK;
     #pragma unroll
-    for (int i = 0; i < 2; ++i) {
+    for (int i = 0; i < 4; ++i) {
       int r = c_row[i];
       r = r > maxR - tileR ? (maxR - tileR < 0 ? 0 : maxR - tileR) : r;
       const char* gp = reinterpret_cast<const char*>(
@@ -234,14 +250,13 @@ void gg_mx_nt_kernel(const uint8_t* __restrict__ Aall,
 
   issue(A, maxA, tileM, 0, 0);
   issue(B, maxB, tileN, 0, 1);
-  if (NKT > 1) { issue(A, maxA, tileM, 1, 0); issue(B, maxB, tileN, 1, 1); }
-  asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
   __builtin_amdgcn_s_barrier();
 
   for (int j = 0; j < NKT; ++j) {
-    if (j + 2 < NKT) {
-      issue(A, maxA, tileM, j + 2, 0);
-      issue(B, maxB, tileN, j + 2, 1);
+    if (j + 1 < NKT) {
+      issue(A, maxA, tileM, j + 1, 0);
+      issue(B, maxB, tileN, j + 1, 1);
     }
     const uint8_t* As = lds + (j & 1) * 2 * MX_TILE;
     const uint8_t* Bs = As + MX_TILE;
@@ -279,10 +294,7 @@ void gg_mx_nt_kernel(const uint8_t* __restrict__ Aall,
       __builtin_amdgcn_s_setprio(0);
     }
     if (j + 1 < NKT) {
-      if (j + 2 < NKT)
-        asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
-      else
-        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
       asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
       __builtin_amdgcn_s_barrier();
     }
@@ -315,23 +327,27 @@ extern "C" hipError_t lumina_mx_quant_rows(const void* X, void* Q, void* S,
 }
 
 extern "C" hipError_t lumina_mx_quant_cols(const void* W, void* Q, void* S,
-                                           void* amax_ws, int K, int N,
-                                           int Kp, hipStream_t stream) {
-  hipLaunchKernelGGL(mx_colmax_kernel, dim3((N + 255) / 256), dim3(256), 0,
-                     stream, (const uint16_t*)W, (float*)amax_ws, K, N);
-  hipError_t e = hipGetLastError();
+                                           void* amax_ws, int E, int K,
+                                           int N, int Kp,
+                                           hipStream_t stream) {
+  hipError_t e = hipMemsetAsync(amax_ws, 0, (int64_t)E * N * 4, stream);
   if (e != hipSuccess) return e;
-  dim3 g((K + 31) / 32, (N + 31) / 32);
+  hipLaunchKernelGGL(mx_colmax_kernel,
+                     dim3((N + 255) / 256, (K + 511) / 512, E), dim3(256), 0,
+                     stream, (const uint16_t*)W, (float*)amax_ws, K, N);
+  e = hipGetLastError();
+  if (e != hipSuccess) return e;
+  dim3 g((K + 31) / 32, (N + 31) / 32, E);
   hipLaunchKernelGGL(mx_quant_t_kernel, g, dim3(256), 0, stream,
                      (const uint16_t*)W, (const float*)amax_ws, (uint8_t*)Q,
                      (uint8_t*)S, K, N, Kp);
   e = hipGetLastError();
   if (e != hipSuccess) return e;
   if (Kp > K) {
-    int64_t tot = (int64_t)N * (Kp - K);
+    int64_t tot = (int64_t)E * N * (Kp - K);
     hipLaunchKernelGGL(mx_zero_tail_kernel,
                        dim3((int)((tot + 255) / 256)), dim3(256), 0, stream,
-                       (uint8_t*)Q, (int64_t)N, K, Kp);
+                       (uint8_t*)Q, (int64_t)E * N, K, Kp);
     e = hipGetLastError();
   }
   return e;

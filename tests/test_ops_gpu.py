@@ -658,8 +658,8 @@ def test_mx_expert_bmm_autograd():
     E, C, K, N = 2, 64, 96, 128
     x = torch.randn(E, C, K, device=_dev(), dtype=torch.bfloat16,
                     requires_grad=True)
-    w = torch.randn(E, K, N, device=_dev(), dtype=torch.bfloat16,
-                    requires_grad=True) * 0.1
+    w = (torch.randn(E, K, N, device=_dev(), dtype=torch.bfloat16) * 0.1
+         ).requires_grad_()
     y = expert_bmm_fp8(x, w)
     ref = torch.matmul(x.detach().float(), w.detach().float())
     rel = (y.float() - ref).abs().max() / ref.abs().max().clamp_min(1e-3)
