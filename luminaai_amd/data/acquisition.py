@@ -120,6 +120,11 @@ def download_oasst(out_dir: str, **kw) -> Dict:
 
 
 # ---------------------------------------------------------- text sources
+# Per-source cleaning pipelines (reference multi_source_dataset.py:277-1350:
+# WikipediaProcessor :277, GutenbergProcessor :511, ArXivProcessor :616,
+# StackOverflowProcessor :729, PubMedProcessor :852, OpenWebTextProcessor
+# :1012, PhilPapersProcessor :1125, CommonCrawlNewsProcessor :1229).  The
+# download legs are network-gated; these run on local dumps.
 _WS = re.compile(r"[ \t]+")
 _MULTI_NL = re.compile(r"\n{3,}")
 
@@ -130,28 +135,176 @@ def _clean_generic(text: str) -> str:
     return text.strip()
 
 
+def _strip_html(text: str) -> str:
+    text = re.sub(r"<code>.*?</code>", "[CODE]", text, flags=re.S | re.I)
+    text = re.sub(r"<[^>]+>", "", text)
+    for ent, ch in (("&lt;", "<"), ("&gt;", ">"), ("&amp;", "&"),
+                    ("&quot;", '"'), ("&#39;", "'"), ("&nbsp;", " ")):
+        text = text.replace(ent, ch)
+    return text
+
+
 def _clean_wikipedia(text: str) -> str:
-    text = re.sub(r"==+ *(References|External links|See also)"
-                  r" *==+.*", "", text, flags=re.S)
-    text = re.sub(r"==+ *([^=]+?) *==+", r"\1.", text)
+    """Wikitext -> plain prose (reference clean_wiki_text :316)."""
+    text = re.sub(r"<!--.*?-->", "", text, flags=re.S)
+    text = re.sub(r"<ref[^>]*>.*?</ref>", "", text, flags=re.S | re.I)
+    text = re.sub(r"<ref[^/>]*/>", "", text, flags=re.I)
+    for _ in range(5):                       # nested templates
+        text = re.sub(r"\{\{[^{}]*\}\}", "", text)
+    text = re.sub(r"\{\{.*?\}\}", "", text, flags=re.S)
+    text = re.sub(r"\{\|.*?\|\}", "", text, flags=re.S)    # tables/infoboxes
+    text = re.sub(r"\[\[(?:File|Image|\u0424\u0430\u0439\u043b):.*?\]\]", "", text,
+                  flags=re.I | re.S)
+    text = re.sub(r"<gallery.*?>.*?</gallery>", "", text, flags=re.S | re.I)
+    text = re.sub(r"\[\[Category:.*?\]\]", "", text, flags=re.I)
+    text = re.sub(r"\[\[(?:[^|\]]*\|)?([^\]]+)\]\]", r"\1", text)  # links
+    text = re.sub(r"\[http[^\]]*\]", "", text)
+    text = re.sub(r"http[s]?://\S+", "", text)
+    text = re.sub(r"<[^>]+>", "", text)
+    # drop trailing reference sections, flatten remaining headers
+    text = re.sub(r"==+ *(References|External links|See also|Further "
+                  r"reading|Notes|Bibliography) *==+.*", "", text, flags=re.S)
+    text = re.sub(r"==+ *([^=\n]+?) *==+", r"\1.", text)
+    text = text.replace("]]", "").replace("[[", "")
+    text = re.sub(r"\'{2,}", "", text)       # bold/italic quotes
     return _clean_generic(text)
 
 
-def _clean_code_qa(text: str) -> str:
-    text = re.sub(r"<[^>]+>", "", text)      # html tags
+def _clean_gutenberg(text: str) -> str:
+    """Strip Project Gutenberg boilerplate (reference :552)."""
+    m = re.search(r"\*\*\* ?START OF (THIS|THE) PROJECT GUTENBERG EBOOK"
+                  r".*?\*\*\*", text, re.I | re.S)
+    if m:
+        text = text[m.end():]
+    m = re.search(r"\*\*\* ?END OF (THIS|THE) PROJECT GUTENBERG EBOOK",
+                  text, re.I)
+    if m:
+        text = text[:m.start()]
+    text = re.sub(r"\[Illustration:?[^\]]*\]", "", text)
     return _clean_generic(text)
 
 
-SOURCE_PROCESSORS: Dict[str, Callable[[str], str]] = {
+_LATEX_CMD = re.compile(r"\\[a-zA-Z]+\*?(\[[^\]]*\])?(\{[^{}]*\})?")
+
+
+def _clean_arxiv(row) -> str:
+    """Title + abstract with LaTeX de-noising (reference ArXivProcessor
+    :616 -- abstracts only, 'Title: .. Abstract: ..' records)."""
+    if isinstance(row, dict):
+        title = (row.get("title") or "").strip()
+        abstract = (row.get("abstract") or row.get("summary") or "").strip()
+    else:
+        title, abstract = "", str(row)
+    body = f"Title: {title}\n\nAbstract: {abstract}" if title else abstract
+    body = re.sub(r"\$+[^$]*\$+", " [MATH] ", body)
+    body = _LATEX_CMD.sub(" ", body)
+    body = body.replace("{", "").replace("}", "")
+    return _clean_generic(body)
+
+
+def _clean_stackoverflow(row) -> str:
+    """Q/A records -> 'Question: .. Answer: ..' with HTML stripped and
+    code spans tokenised (reference StackOverflowProcessor :729)."""
+    if isinstance(row, dict):
+        q = row.get("question") or row.get("title") or ""
+        body = row.get("body") or ""
+        a = row.get("answer") or row.get("accepted_answer") or ""
+        parts = []
+        if q:
+            parts.append("Question: " + _strip_html(q))
+        if body:
+            parts.append(_strip_html(body))
+        if a:
+            parts.append("Answer: " + _strip_html(a))
+        return _clean_generic("\n".join(parts))
+    return _clean_generic(_strip_html(str(row)))
+
+
+_PUBMED_LABELS = re.compile(
+    r"\b(BACKGROUND|OBJECTIVES?|METHODS?|RESULTS?|CONCLUSIONS?|PURPOSE|"
+    r"DESIGN|SETTING|PARTICIPANTS|INTERVENTIONS?|MAIN OUTCOME MEASURES?|"
+    r"MEASUREMENTS)\s*:\s*")
+
+
+def _clean_pubmed(row) -> str:
+    """Structured-abstract labels stripped, title+abstract records
+    (reference PubMedProcessor :852)."""
+    if isinstance(row, dict):
+        title = (row.get("title") or "").strip().rstrip(".")
+        abstract = (row.get("abstract") or row.get("text") or "").strip()
+        body = f"{title}. {abstract}" if title else abstract
+    else:
+        body = str(row)
+    body = _PUBMED_LABELS.sub("", body)
+    body = re.sub(r"\[[0-9,\- ]+\]", "", body)      # citation brackets
+    return _clean_generic(body)
+
+
+_NAVLINE = re.compile(
+    r"^(\s*(home|menu|login|sign ?in|sign ?up|subscribe|share|tweet|"
+    r"advertisement|cookie[s]? (policy|notice)|privacy policy|terms of "
+    r"(use|service)|all rights reserved).*|[\W\d\s]*)$", re.I)
+
+
+def _clean_openwebtext(text: str) -> str:
+    """Boilerplate/navigation line filtering + dedupe (reference
+    OpenWebTextProcessor :1012)."""
+    seen = set()
+    keep = []
+    for line in _strip_html(text).splitlines():
+        ls = line.strip()
+        if len(ls) < 3 or _NAVLINE.match(ls):
+            continue
+        # drop exact repeats (nav fragments recur on every page)
+        if len(ls) < 80:
+            if ls in seen:
+                continue
+            seen.add(ls)
+        keep.append(ls)
+    return _clean_generic("\n".join(keep))
+
+
+def _clean_philpapers(row) -> str:
+    """Philosophy abstracts: same record shape as arXiv (reference
+    PhilPapersProcessor :1125)."""
+    return _clean_arxiv(row)
+
+
+_BYLINE = re.compile(r"^(by [A-Z][\w.\- ]+|published:? .*|updated:? .*|"
+                     r"\d{1,2} [A-Z][a-z]+ \d{4}.*)$", re.I | re.M)
+
+
+def _clean_cc_news(row) -> str:
+    """Headline + body with bylines/dates and site boilerplate removed
+    (reference CommonCrawlNewsProcessor :1229)."""
+    if isinstance(row, dict):
+        title = (row.get("title") or "").strip()
+        body = (row.get("text") or row.get("body") or "").strip()
+        text = f"{title}\n\n{body}" if title else body
+    else:
+        text = str(row)
+    text = _BYLINE.sub("", _strip_html(text))
+    return _clean_openwebtext(text)
+
+
+def _clean_code_qa(text: str) -> str:    # retained alias (round-1 API)
+    return _clean_generic(_strip_html(text))
+
+
+SOURCE_PROCESSORS: Dict[str, Callable] = {
     "wikipedia": _clean_wikipedia,
-    "gutenberg": _clean_generic,
-    "arxiv": _clean_generic,
-    "stackoverflow": _clean_code_qa,
-    "pubmed": _clean_generic,
-    "openwebtext": _clean_generic,
-    "philpapers": _clean_generic,
-    "cc_news": _clean_generic,
+    "gutenberg": _clean_gutenberg,
+    "arxiv": _clean_arxiv,
+    "stackoverflow": _clean_stackoverflow,
+    "pubmed": _clean_pubmed,
+    "openwebtext": _clean_openwebtext,
+    "philpapers": _clean_philpapers,
+    "cc_news": _clean_cc_news,
 }
+
+# processors that want the parsed JSON row rather than a text line
+_ROW_PROCESSORS = {"arxiv", "stackoverflow", "pubmed", "philpapers",
+                   "cc_news"}
 
 
 def prepare_text_corpus(input_path: str, out_path: str, source: str = "generic",
@@ -159,20 +312,22 @@ def prepare_text_corpus(input_path: str, out_path: str, source: str = "generic",
     """Clean a local raw-text/JSONL dump into a training .txt corpus
     (reference multi_source_dataset.py per-source processors)."""
     proc = SOURCE_PROCESSORS.get(source, _clean_generic)
+    row_mode = source in _ROW_PROCESSORS
     n_in = n_out = 0
     os.makedirs(os.path.dirname(out_path) or ".", exist_ok=True)
     with open(input_path, encoding="utf-8", errors="replace") as fin, \
             open(out_path, "w") as fout:
         for line in fin:
             n_in += 1
-            text = line
+            payload = line
             if line.lstrip().startswith("{"):
                 try:
                     row = json.loads(line)
-                    text = row.get("text") or row.get("content") or ""
+                    payload = row if row_mode else (
+                        row.get("text") or row.get("content") or "")
                 except json.JSONDecodeError:
                     pass
-            text = proc(text)
+            text = proc(payload)
             if len(text) >= min_chars:
                 fout.write(text + "\n")
                 n_out += 1

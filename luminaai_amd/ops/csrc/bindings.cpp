@@ -58,6 +58,10 @@ void launch_grouped_gemm_nt_v2(const void*, const void*, void*, int, int, int, i
 void launch_grouped_gemm_nt_v3(const void*, const void*, void*, int, int, int, int, int64_t, int64_t, int64_t, hipStream_t);
 void launch_grouped_gemm_nt_v4(const void*, const void*, void*, int, int, int, int, int64_t, int64_t, int64_t, hipStream_t);
 hipError_t lumina_gemv(const void*, const void*, void*, int, int64_t, int, hipStream_t);
+hipError_t lumina_dec_gemv(const void*, const void*, const void*, const void*, void*, int, int, float, int, hipStream_t);
+hipError_t lumina_dec_rope_cache(const void*, void*, void*, void*, const float*, const float*, const int*, int, int, int, hipStream_t);
+hipError_t lumina_dec_attn(const void*, const void*, const void*, void*, const int*, int, int, int, float, hipStream_t);
+hipError_t lumina_dec_advance(int*, hipStream_t);
 hipError_t lumina_moe_gather_rows(const void*, const int64_t*, const bool*, void*, int64_t, int, int, hipStream_t);
 hipError_t lumina_moe_dispatch_bwd(const void*, const int64_t*, void*, int64_t, int, int, int64_t, int, hipStream_t);
 hipError_t lumina_moe_combine_fwd(const void*, const float*, const int64_t*, void*, int64_t, int, int, int64_t, int, hipStream_t);
@@ -561,6 +565,48 @@ at::Tensor gg_mx_nt(const at::Tensor& Aq, const at::Tensor& As,
   return O;
 }
 
+// ---- fused batch-1 decode kernels (decode.hip) ---------------------------
+void dec_gemv(const at::Tensor& W, const at::Tensor& x,
+              const std::optional<at::Tensor>& wn,
+              const std::optional<at::Tensor>& resid, at::Tensor& y,
+              double eps, int64_t flags) {
+  const int K = (int)x.numel();
+  const int N = (int)y.numel();
+  check_hip(lumina_dec_gemv(W.data_ptr(), x.data_ptr(),
+                            wn ? wn->data_ptr() : nullptr,
+                            resid ? resid->data_ptr() : nullptr,
+                            y.data_ptr(), N, K, (float)eps, (int)flags,
+                            cur_stream()),
+            "dec_gemv");
+}
+
+void dec_rope_cache(const at::Tensor& qkv, at::Tensor& q_out, at::Tensor& kc,
+                    at::Tensor& vc, const at::Tensor& cost,
+                    const at::Tensor& sint, const at::Tensor& pos_dev,
+                    int64_t H, int64_t HKV, int64_t D) {
+  check_hip(lumina_dec_rope_cache(qkv.data_ptr(), q_out.data_ptr(),
+                                  kc.data_ptr(), vc.data_ptr(),
+                                  cost.data_ptr<float>(),
+                                  sint.data_ptr<float>(),
+                                  pos_dev.data_ptr<int>(), (int)H, (int)HKV,
+                                  (int)D, cur_stream()),
+            "dec_rope_cache");
+}
+
+void dec_attn(const at::Tensor& q, const at::Tensor& kc, const at::Tensor& vc,
+              at::Tensor& out, const at::Tensor& pos_dev, int64_t H,
+              int64_t HKV, int64_t D, double scale) {
+  check_hip(lumina_dec_attn(q.data_ptr(), kc.data_ptr(), vc.data_ptr(),
+                            out.data_ptr(), pos_dev.data_ptr<int>(), (int)H,
+                            (int)HKV, (int)D, (float)scale, cur_stream()),
+            "dec_attn");
+}
+
+void dec_advance(const at::Tensor& pos_dev) {
+  check_hip(lumina_dec_advance(pos_dev.data_ptr<int>(), cur_stream()),
+            "dec_advance");
+}
+
 at::Tensor gemv(const at::Tensor& x, const at::Tensor& w) {
   // y[N] = w[N,K] @ x[K]
   TORCH_CHECK(x.is_contiguous() && w.is_contiguous());
@@ -580,6 +626,12 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("attn_bwd", &attn_bwd,
           "causal GQA flash attention backward -> (dQ, dK, dV) (gfx950)");
   mod.def("gemv", &gemv, "batch-1 decode GEMV y = W @ x (gfx950)");
+  mod.def("dec_gemv", &dec_gemv,
+          "fused decode GEMV (norm/residual/swiglu epilogues, nt loads)");
+  mod.def("dec_rope_cache", &dec_rope_cache,
+          "decode RoPE + KV-cache append at the device cursor");
+  mod.def("dec_attn", &dec_attn, "single-token GQA attention over the cache");
+  mod.def("dec_advance", &dec_advance, "advance the decode cursor");
   mod.def("mx_quant_rows", &mx_quant_rows,
           "rowwise e8m0 fp8 quantization, K zero-padded (gfx950)");
   mod.def("mx_quant_cols", &mx_quant_cols,

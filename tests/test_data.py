@@ -216,3 +216,39 @@ def test_parquet_dataset(tmp_path, tokenizer):
     assert len(rows) > 0
     # map-style and streaming chunk the same token stream
     assert torch.equal(rows[0]["input_ids"], ds[0]["input_ids"])
+
+
+# ---- per-source processors (round-2: reference multi_source_dataset.py) ----
+def test_source_processors_clean_each_format(tmp_path):
+    import json
+    from luminaai_amd.data import acquisition as a
+
+    assert a._clean_wikipedia(
+        "{{Infobox|x}}Text <ref>c</ref> [[L|shown]] ==References== junk"
+    ) == "Text shown"
+    assert a._clean_gutenberg(
+        "hdr *** START OF THE PROJECT GUTENBERG EBOOK T *** body "
+        "*** END OF THE PROJECT GUTENBERG EBOOK") == "body"
+    ar = a._clean_arxiv({"title": "T \\emph{x}", "abstract": "sum $y$"})
+    assert ar.startswith("Title: T") and "[MATH]" in ar and "\\emph" not in ar
+    so = a._clean_stackoverflow({"question": "<b>Q</b>?",
+                                 "answer": "use <code>f()</code>"})
+    assert "Question: Q?" in so and "[CODE]" in so and "<" not in so
+    pm = a._clean_pubmed({"title": "T", "abstract":
+                          "BACKGROUND: a. RESULTS: b [3]."})
+    assert "BACKGROUND" not in pm and "[3]" not in pm
+    ow = a._clean_openwebtext("Home\nShare\nA real sentence long enough.\n"
+                              "Share\nCookie policy\n")
+    assert ow == "A real sentence long enough."
+    cc = a._clean_cc_news({"title": "Headline", "text":
+                           "By Jane Doe\nThe story body is long enough."})
+    assert cc.startswith("Headline") and "Jane" not in cc
+
+    # end-to-end through prepare_text_corpus with JSONL row dispatch
+    p = tmp_path / "arxiv.jsonl"
+    rows = [{"title": f"Paper {i}", "abstract": "x " * 150} for i in range(3)]
+    p.write_text("\n".join(json.dumps(r) for r in rows))
+    out = tmp_path / "corpus.txt"
+    stats = a.prepare_text_corpus(str(p), str(out), source="arxiv")
+    assert stats["kept"] == 3
+    assert out.read_text().count("Title: Paper") == 3
