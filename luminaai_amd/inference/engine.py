@@ -99,6 +99,32 @@ class GenerationEngine:
 
         dec = None
         if use_graph:
+            # prefer the fused 5-kernel decode path (inference/fused_decode
+            # .py: 2x the graphed eager decoder) for dense/MoD models
+            try:
+                from .fused_decode import FusedDecoder, can_fuse_decode
+                if can_fuse_decode(self.model):
+                    if getattr(self, "_fused_dec", None) is None or \
+                            self._fused_dec.max_context != cfg.max_context:
+                        self._fused_dec = FusedDecoder(self.model,
+                                                       cfg.max_context)
+                    fd = self._fused_dec
+                    fd.reset()
+                    last_logits = fd.prefill(x)[0].float()
+
+                    class _FDAdapter:
+                        seq_len = property(lambda s: fd.seq_len)
+
+                        @staticmethod
+                        def step(t):
+                            return fd.step(t.view(1)).float().view(1, -1)
+
+                    dec = _FDAdapter()
+                    seq_len = lambda: fd.seq_len  # noqa: E731
+            except Exception:  # noqa: BLE001
+                self._fused_dec = None
+                dec = None
+        if use_graph and dec is None:
             try:
                 from .graph_decode import GraphedDecoder
                 if getattr(self, "_graph_dec", None) is None or \

@@ -152,8 +152,10 @@ class FusedDecoder:
 
     @torch.no_grad()
     def _embed_in(self):
-        self.x.copy_((self.embed[self._in_tok[0]] * self.embed_scale)
-                     .to(self.x.dtype))
+        # index_select keeps the token index on-device (int indexing would
+        # sync and is hipGraph-capture-unsafe)
+        row = torch.index_select(self.embed, 0, self._in_tok)
+        self.x.copy_((row[0] * self.embed_scale).to(self.x.dtype))
 
     @torch.no_grad()
     def step(self, token_id: torch.Tensor) -> torch.Tensor:

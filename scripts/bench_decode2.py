@@ -17,26 +17,25 @@ print("can_fuse:", can_fuse_decode(model))
 
 dec = FusedDecoder(model, 2048)
 prompt = torch.randint(1, 1000, (1, 64), device="cuda")
-logits = dec.prefill(prompt)
-tok = logits.argmax(-1).view(1)
 
-def run(n):
-    global tok
-    for _ in range(n):
-        lg = dec.step(tok)
-        tok = lg.float().argmax().view(1)
+def measure(label, use_graph):
+    dec.reset()
+    logits = dec.prefill(prompt)
+    if use_graph:
+        dec.capture()
+    tok = logits.argmax(-1).view(1)
+    def run(n):
+        nonlocal tok
+        for _ in range(n):
+            lg = dec.step(tok)
+            tok = lg.float().argmax().view(1)
+    run(10)
+    torch.cuda.synchronize(); t0 = time.perf_counter()
+    run(100)
+    torch.cuda.synchronize()
+    dt = (time.perf_counter() - t0) / 100
+    print(f"{label}: {dt*1e3:.2f} ms/token = {1/dt:.0f} tok/s")
 
-run(10)
-torch.cuda.synchronize(); t0 = time.perf_counter()
-run(100)
-torch.cuda.synchronize()
-dt = (time.perf_counter() - t0) / 100
-print(f"fused eager: {dt*1e3:.2f} ms/token = {1/dt:.0f} tok/s")
-
-dec.capture()
-run(5)
-torch.cuda.synchronize(); t0 = time.perf_counter()
-run(100)
-torch.cuda.synchronize()
-dt = (time.perf_counter() - t0) / 100
-print(f"fused graph: {dt*1e3:.2f} ms/token = {1/dt:.0f} tok/s")
+measure("fused eager", False)
+measure("fused graph", True)
+measure("fused graph (2nd)", True)
