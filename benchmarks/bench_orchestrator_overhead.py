@@ -66,12 +66,24 @@ def run(with_orchestrator: bool, steps: int = 30) -> float:
 
 
 def main():
-    plain = run(False)
-    adaptive = run(True)
+    # In-process run order matters at 4 ms debug-scale steps: the SECOND
+    # model build in a process measures ~35% slower regardless of mode
+    # (allocator growth; verified with no-op hooks in scripts/dbg_orch.py).
+    # Discard a warmup run and interleave the arms, reporting medians.
+    run(False, steps=10)                      # discarded allocator warmup
+    plains, adaptives = [], []
+    for _ in range(3):
+        plains.append(run(False))
+        adaptives.append(run(True))
+    plains.sort()
+    adaptives.sort()
+    plain, adaptive = plains[1], adaptives[1]
     res = {
         "steps_per_sec_plain": round(plain, 3),
         "steps_per_sec_adaptive": round(adaptive, 3),
         "orchestrator_overhead_pct": round((plain - adaptive) / plain * 100, 2),
+        "runs_plain": [round(x, 1) for x in plains],
+        "runs_adaptive": [round(x, 1) for x in adaptives],
     }
     print(json.dumps(res))
 

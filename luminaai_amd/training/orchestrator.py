@@ -375,16 +375,23 @@ class AdaptiveTrainingOrchestrator:
             self._monitor_thread = None
 
     def _monitor_loop(self):
+        # Fixed 50 ms drain cadence instead of per-item queue wakeups: at
+        # millisecond step times a wakeup per emitted metric costs a GIL
+        # handoff per step (measured 36% at debug scale in round 2's first
+        # pass); decision latency stays far below the adaptive cooldowns
+        # (>= 50 steps, reference orchestrator.py:312).
         while not self._stop_event.is_set():
-            try:
-                m = self.metrics_queue.get(timeout=0.25)
-            except queue.Empty:
-                continue
-            try:
-                self._process_metrics(m)
-            except Exception:  # noqa: BLE001 — monitoring must not kill training
-                if self.logger:
-                    self.logger.exception("monitor error")
+            self._stop_event.wait(0.05)
+            while True:
+                try:
+                    m = self.metrics_queue.get_nowait()
+                except queue.Empty:
+                    break
+                try:
+                    self._process_metrics(m)
+                except Exception:  # noqa: BLE001 — must not kill training
+                    if self.logger:
+                        self.logger.exception("monitor error")
 
     def _process_metrics(self, m: TrainingMetrics):
         self.analytics.observe(m)

@@ -486,44 +486,65 @@ class Trainer:
         dev = {k: v for k, v in self._last_metrics.items()
                if torch.is_tensor(v) and v.is_cuda}
         if dev:
-            # Async emission: stage this step's device scalars into a pinned
-            # host buffer (non-blocking D2H behind an event) and consume the
-            # PREVIOUS emission's staged values, whose copy completed during
-            # the intervening step. The hot path never calls .item() on a
-            # live device tensor -> no stream sync (measured 37% overhead at
-            # debug scale in round 1). Metrics lag one emission interval.
+            # Fully non-blocking emission: stage this step's device scalars
+            # into a rotating pinned buffer behind an event and consume the
+            # OLDEST staged emission only once its event has completed --
+            # never .item(), never event.synchronize() (an every-step
+            # synchronize pins the host to the GPU's progress and destroys
+            # the launch run-ahead: measured 36% at debug scale).  When the
+            # GPU runs deep behind, staging self-throttles by skipping
+            # emissions instead of stalling the training thread.
             keys = sorted(dev)
-            stacked = torch.stack([dev[k].detach().float().reshape(())
-                                   for k in keys])
-            if comm.is_distributed():
-                # Device-side average across ranks (RCCL, stays on-stream):
-                # all ranks' monitor threads then see IDENTICAL metric values
-                # and reach identical adaptive decisions.
-                import torch.distributed as dist
-                dist.all_reduce(stacked)
-                stacked /= comm.get_world_size()
-            bufs = getattr(self, "_emit_bufs", None)
-            if bufs is None or bufs[0].numel() != len(keys):
-                self._emit_bufs = tuple(
-                    torch.empty(len(keys), dtype=torch.float32,
-                                pin_memory=True) for _ in range(2))
-                self._emit_tick = 0
-                self._pending_emit = None
-            self._emit_tick = getattr(self, "_emit_tick", 0) + 1
-            pinned = self._emit_bufs[self._emit_tick & 1]
-            pinned.copy_(stacked, non_blocking=True)
-            ev = torch.cuda.Event()
-            ev.record()
-            prev = getattr(self, "_pending_emit", None)
-            self._pending_emit = (keys, pinned, ev, self.global_step,
-                                  self.epoch)
-            if prev is None:
-                return
-            pkeys, ppinned, pev, pstep, pepoch = prev
-            pev.synchronize()   # a full emission interval old: ~always done
-            floats = {k: float(ppinned[i]) for i, k in enumerate(pkeys)}
-            m = self.get_current_metrics(_floats=floats, _step=pstep,
-                                         _epoch=pepoch)
+            st = getattr(self, "_emit_state", None)
+            if st is None or len(st["bufs"][0]) != len(keys):
+                st = self._emit_state = {
+                    "bufs": [torch.empty(len(keys), dtype=torch.float32,
+                                         pin_memory=True) for _ in range(4)],
+                    "events": [None] * 4,
+                    "meta": [None] * 4,
+                    "tick": 0,
+                    "pending": [],
+                }
+            slot = st["tick"] & 3
+            ev_old = st["events"][slot]
+            if ev_old is None or ev_old.query():
+                if slot in st["pending"]:
+                    # completed but never consumed: superseded -> dropped
+                    st["pending"].remove(slot)
+                stacked = torch.stack([dev[k].detach().float().reshape(())
+                                       for k in keys])
+                if comm.is_distributed():
+                    # device-side average across ranks (stays on-stream) so
+                    # every rank's monitor reaches identical decisions
+                    import torch.distributed as dist
+                    dist.all_reduce(stacked)
+                    stacked /= comm.get_world_size()
+                st["bufs"][slot].copy_(stacked, non_blocking=True)
+                ev = torch.cuda.Event()
+                ev.record()
+                st["events"][slot] = ev
+                st["meta"][slot] = (self.global_step, self.epoch)
+                st["pending"].append(slot)
+                st["tick"] += 1
+            # consume every completed staged emission (usually exactly one)
+            m = None
+            while st["pending"]:
+                s0 = st["pending"][0]
+                if not st["events"][s0].query():
+                    break
+                st["pending"].pop(0)
+                floats = {k: float(st["bufs"][s0][i])
+                          for i, k in enumerate(keys)}
+                pstep, pepoch = st["meta"][s0]
+                m = self.get_current_metrics(_floats=floats, _step=pstep,
+                                             _epoch=pepoch)
+                st["events"][s0] = None
+                if m is not None:
+                    try:
+                        self._metrics_hook(m)
+                    except Exception:  # noqa: BLE001
+                        pass
+            return
         else:
             m = self.get_current_metrics()
             # CPU/gloo path: average the loss across ranks synchronously

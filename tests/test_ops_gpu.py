@@ -721,3 +721,23 @@ def test_fused_decoder_matches_eager():
         lg = dec.step(tok)
         tok = lg.float().argmax().view(1)
     assert got2 == got, (got2, got)
+
+
+def test_expert_bmm_fp32_tight():
+    """fp32-path expert grouped GEMM with TIGHT tolerances (round-2: the
+    bf16 test's loose rtol could hide a sign/magnitude bug in a tail)."""
+    from luminaai_amd.ops.interface import expert_bmm
+    torch.manual_seed(11)
+    E, C, K, N = 3, 96, 64, 80
+    x = torch.randn(E, C, K, device=_dev(), requires_grad=True)
+    w = torch.randn(E, K, N, device=_dev(), requires_grad=True)
+    y = expert_bmm(x, w)
+    go = torch.randn_like(y)
+    y.backward(go)
+    x2 = x.detach().clone().requires_grad_(True)
+    w2 = w.detach().clone().requires_grad_(True)
+    ref = torch.matmul(x2, w2)
+    ref.backward(go)
+    assert torch.allclose(y, ref.detach(), atol=1e-4, rtol=1e-4)
+    assert torch.allclose(x.grad, x2.grad, atol=1e-4, rtol=1e-4)
+    assert torch.allclose(w.grad, w2.grad, atol=1e-4, rtol=1e-4)
