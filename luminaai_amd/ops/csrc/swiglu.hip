@@ -7,6 +7,67 @@
 // torch.nn.Linear; this kernel removes the three elementwise round trips.
 #include "common.h"
 
+// vectorized bf16 path: 8 elements per lane per pointer (scalar bf16 loads
+// measured ~2x slower -- guide common-mistake 2; the round-2 profile showed
+// the scalar version at 3x its memory floor)
+__global__ void swiglu_fwd_vec_kernel(const uint16_t* __restrict__ g,
+                                      const uint16_t* __restrict__ u,
+                                      uint16_t* __restrict__ y,
+                                      int64_t rows, int I8,
+                                      int64_t g_stride, int64_t u_stride) {
+  const int64_t total = rows * (int64_t)I8;
+  for (int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       idx < total; idx += gstride()) {
+    const int64_t r = idx / I8;
+    const int64_t i = (idx - r * I8) * 8;
+    const ushortx8 gv = *reinterpret_cast<const ushortx8*>(
+        g + r * g_stride + i);
+    const ushortx8 uv = *reinterpret_cast<const ushortx8*>(
+        u + r * u_stride + i);
+    ushortx8 yv;
+    #pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      const float gf = bf16_to_f32(gv[j]);
+      const float s = gf / (1.0f + __expf(-gf));
+      yv[j] = f32_to_bf16(s * bf16_to_f32(uv[j]));
+    }
+    *reinterpret_cast<ushortx8*>(y + idx * 8) = yv;
+  }
+}
+
+__global__ void swiglu_bwd_vec_kernel(const uint16_t* __restrict__ dy,
+                                      const uint16_t* __restrict__ g,
+                                      const uint16_t* __restrict__ u,
+                                      uint16_t* __restrict__ dg,
+                                      uint16_t* __restrict__ du,
+                                      int64_t rows, int I8,
+                                      int64_t g_stride, int64_t u_stride) {
+  const int64_t total = rows * (int64_t)I8;
+  for (int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       idx < total; idx += gstride()) {
+    const int64_t r = idx / I8;
+    const int64_t i = (idx - r * I8) * 8;
+    const ushortx8 gv = *reinterpret_cast<const ushortx8*>(
+        g + r * g_stride + i);
+    const ushortx8 uv = *reinterpret_cast<const ushortx8*>(
+        u + r * u_stride + i);
+    const ushortx8 dyv = *reinterpret_cast<const ushortx8*>(dy + idx * 8);
+    ushortx8 dgv, duv;
+    #pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      const float gf = bf16_to_f32(gv[j]);
+      const float dyf = bf16_to_f32(dyv[j]);
+      const float sig = 1.0f / (1.0f + __expf(-gf));
+      const float silu = gf * sig;
+      const float dsilu = sig * (1.0f + gf * (1.0f - sig));
+      dgv[j] = f32_to_bf16(dyf * bf16_to_f32(uv[j]) * dsilu);
+      duv[j] = f32_to_bf16(dyf * silu);
+    }
+    *reinterpret_cast<ushortx8*>(dg + r * g_stride + i) = dgv;
+    *reinterpret_cast<ushortx8*>(du + r * u_stride + i) = duv;
+  }
+}
+
 // y = silu(g) * u ; both halves of one fused gate_up output may be strided:
 // gate at row*2I + i, up at row*2I + I + i. We take separate pointers and a
 // row stride so the caller can pass either layout.
@@ -55,10 +116,24 @@ __global__ void swiglu_bwd_kernel(const typename E::storage* __restrict__ dy,
 
 extern "C" {
 
+static inline bool vec8_ok(const void* g, const void* u, int I,
+                           int64_t gs, int64_t us) {
+  return I % 8 == 0 && gs % 8 == 0 && us % 8 == 0
+      && ((uintptr_t)g & 15) == 0 && ((uintptr_t)u & 15) == 0;
+}
+
 hipError_t lumina_swiglu_fwd_bf16(const void* g, const void* u, void* y,
                                   int64_t rows, int I, int64_t gs, int64_t us,
                                   hipStream_t st) {
   const int block = 256;
+  if (vec8_ok(g, u, I, gs, us)) {
+    const int grid = elementwise_grid(rows * (int64_t)(I / 8), block, 4);
+    swiglu_fwd_vec_kernel<<<grid, block, 0, st>>>(
+        (const uint16_t*)g, (const uint16_t*)u, (uint16_t*)y, rows, I / 8,
+        gs, us);
+    HIP_CHECK_LAST();
+    return hipSuccess;
+  }
   const int grid = elementwise_grid(rows * (int64_t)I, block, 8);
   swiglu_fwd_kernel<BF16Elem><<<grid, block, 0, st>>>(
       (const uint16_t*)g, (const uint16_t*)u, (uint16_t*)y, rows, I, gs, us);
@@ -81,6 +156,14 @@ hipError_t lumina_swiglu_bwd_bf16(const void* dy, const void* g, const void* u,
                                   void* dg, void* du, int64_t rows, int I,
                                   int64_t gs, int64_t us, hipStream_t st) {
   const int block = 256;
+  if (vec8_ok(g, u, I, gs, us)) {
+    const int grid = elementwise_grid(rows * (int64_t)(I / 8), block, 4);
+    swiglu_bwd_vec_kernel<<<grid, block, 0, st>>>(
+        (const uint16_t*)dy, (const uint16_t*)g, (const uint16_t*)u,
+        (uint16_t*)dg, (uint16_t*)du, rows, I / 8, gs, us);
+    HIP_CHECK_LAST();
+    return hipSuccess;
+  }
   const int grid = elementwise_grid(rows * (int64_t)I, block, 8);
   swiglu_bwd_kernel<BF16Elem><<<grid, block, 0, st>>>(
       (const uint16_t*)dy, (const uint16_t*)g, (const uint16_t*)u,
