@@ -60,7 +60,7 @@ void launch_grouped_gemm_nt_v4(const void*, const void*, void*, int, int, int, i
 hipError_t lumina_gemv(const void*, const void*, void*, int, int64_t, int, hipStream_t);
 hipError_t lumina_dec_gemv(const void*, const void*, const void*, const void*, void*, int, int, float, int, hipStream_t);
 hipError_t lumina_dec_rope_cache(const void*, void*, void*, void*, const float*, const float*, const int*, int, int, int, hipStream_t);
-hipError_t lumina_dec_attn(const void*, const void*, const void*, void*, const int*, int, int, int, float, hipStream_t);
+hipError_t lumina_dec_attn(const void*, const void*, const void*, void*, const int*, int, int, int, int, float, hipStream_t);
 hipError_t lumina_dec_advance(int*, hipStream_t);
 hipError_t lumina_moe_gather_rows(const void*, const int64_t*, const bool*, void*, int64_t, int, int, hipStream_t);
 hipError_t lumina_moe_dispatch_bwd(const void*, const int64_t*, void*, int64_t, int, int, int64_t, int, hipStream_t);
@@ -596,9 +596,11 @@ void dec_rope_cache(const at::Tensor& qkv, at::Tensor& q_out, at::Tensor& kc,
 void dec_attn(const at::Tensor& q, const at::Tensor& kc, const at::Tensor& vc,
               at::Tensor& out, const at::Tensor& pos_dev, int64_t H,
               int64_t HKV, int64_t D, double scale) {
+  const int cap = (int)kc.size(0);
   check_hip(lumina_dec_attn(q.data_ptr(), kc.data_ptr(), vc.data_ptr(),
-                            out.data_ptr(), pos_dev.data_ptr<int>(), (int)H,
-                            (int)HKV, (int)D, (float)scale, cur_stream()),
+                            out.data_ptr(), pos_dev.data_ptr<int>(), cap,
+                            (int)H, (int)HKV, (int)D, (float)scale,
+                            cur_stream()),
             "dec_attn");
 }
 

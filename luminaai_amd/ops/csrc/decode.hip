@@ -144,9 +144,10 @@ __global__ void dec_advance_kernel(int* pos_dev) {
 
 // ---------------------------------------------------------------------------
 // single-token GQA attention over the cache: out[h][d] =
-// softmax(q_h . k_kv(s) * scale) @ v.  Block per q-head, 4 waves split the
-// sequence, online softmax per wave, LDS merge.  len = *pos_dev + 1 (the
-// new token was appended at *pos_dev).
+// softmax(q_h . k_kv(s) * scale) @ v.  Block per q-head; lane-per-s QK
+// scoring (each lane streams one K row), block softmax over an LDS score
+// array, then a d-parallel PV pass with coalesced V rows.  len =
+// *pos_dev + 1 (the new token was appended at *pos_dev).
 __global__ __launch_bounds__(256)
 void dec_attn_kernel(const uint16_t* __restrict__ q,
                      const uint16_t* __restrict__ kc,
@@ -154,79 +155,56 @@ void dec_attn_kernel(const uint16_t* __restrict__ q,
                      uint16_t* __restrict__ out,
                      const int* __restrict__ pos_dev,
                      int H, int HKV, int D, float scale) {
-  __shared__ float sm[4], sl[4];
-  __shared__ float sacc[4][192];
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  float* red = reinterpret_cast<float*>(smem);            // [16]
+  float* qs = red + 16;                                   // [D]
+  float* sc = qs + ((D + 3) & ~3);                        // [len]
 
   const int h = blockIdx.x;
   const int kvh = h / (H / HKV);
   const int len = *pos_dev + 1;
-  const int lane = threadIdx.x & 63;
-  const int wid = threadIdx.x >> 6;
+  const int t = threadIdx.x;
   const int KVS = HKV * D;
 
-  // q fragment: this lane's dims (lane, lane+64, lane+128)
-  float qv[3] = {};
-  #pragma unroll
-  for (int i = 0; i < 3; ++i) {
-    const int d = lane + i * 64;
-    if (d < D) qv[i] = bf16_to_f32(q[h * D + d]);
-  }
-
-  const int chunk = (len + 3) / 4;
-  const int s0 = wid * chunk;
-  const int s1 = s0 + chunk < len ? s0 + chunk : len;
-
-  float m = -1e30f, l = 0.0f, acc[3] = {};
-  for (int sp = s0; sp < s1; ++sp) {
-    const uint16_t* kr = kc + (int64_t)sp * KVS + kvh * D;
-    float d0 = 0.0f;
-    #pragma unroll
-    for (int i = 0; i < 3; ++i) {
-      const int d = lane + i * 64;
-      if (d < D) d0 = fmaf(qv[i], bf16_to_f32(kr[d]), d0);
-    }
-    d0 = wave_reduce_sum(d0);
-    d0 = __shfl(d0, 0, 64) * scale;
-    const float nm = fmaxf(m, d0);
-    const float f = __expf(m - nm);
-    const float p = __expf(d0 - nm);
-    m = nm;
-    l = l * f + p;
-    const uint16_t* vr = vc + (int64_t)sp * KVS + kvh * D;
-    #pragma unroll
-    for (int i = 0; i < 3; ++i) {
-      const int d = lane + i * 64;
-      if (d < D) acc[i] = acc[i] * f + p * bf16_to_f32(vr[d]);
-    }
-  }
-  // merge the 4 waves
-  if (lane == 0) { sm[wid] = s0 < s1 ? m : -1e30f; sl[wid] = l; }
-  #pragma unroll
-  for (int i = 0; i < 3; ++i) {
-    const int d = lane + i * 64;
-    if (d < D) sacc[wid][d] = acc[i];
-  }
+  for (int d = t; d < D; d += 256)
+    qs[d] = bf16_to_f32(q[h * D + d]);
   __syncthreads();
-  if (wid == 0) {
-    const float gm = fmaxf(fmaxf(sm[0], sm[1]), fmaxf(sm[2], sm[3]));
-    float gl = 0.0f;
-    float f[4];
-    #pragma unroll
-    for (int w = 0; w < 4; ++w) {
-      f[w] = __expf(sm[w] - gm);
-      gl += sl[w] * f[w];
+
+  // ---- pass 1: one K row per thread, vectorized along d
+  float pmax = -1e30f;
+  for (int sp = t; sp < len; sp += 256) {
+    const uint16_t* kr = kc + (int64_t)sp * KVS + kvh * D;
+    float dot = 0.0f;
+    int d = 0;
+    for (; d + 8 <= D; d += 8) {
+      const ushortx8 kv8 = *reinterpret_cast<const ushortx8*>(kr + d);
+      #pragma unroll
+      for (int j = 0; j < 8; ++j)
+        dot = fmaf(qs[d + j], bf16_to_f32(kv8[j]), dot);
     }
-    const float inv = 1.0f / fmaxf(gl, 1e-30f);
-    #pragma unroll
-    for (int i = 0; i < 3; ++i) {
-      const int d = lane + i * 64;
-      if (d < D) {
-        float o = 0.0f;
-        #pragma unroll
-        for (int w = 0; w < 4; ++w) o += sacc[w][d] * f[w];
-        out[h * D + d] = f32_to_bf16(o * inv);
-      }
-    }
+    for (; d < D; ++d)
+      dot = fmaf(qs[d], bf16_to_f32(kr[d]), dot);
+    dot *= scale;
+    sc[sp] = dot;
+    pmax = fmaxf(pmax, dot);
+  }
+  const float m = block_reduce_max(pmax, red);
+  float psum = 0.0f;
+  for (int sp = t; sp < len; sp += 256) {
+    const float p = __expf(sc[sp] - m);
+    sc[sp] = p;
+    psum += p;
+  }
+  const float inv = 1.0f / fmaxf(block_reduce_sum(psum, red), 1e-30f);
+  __syncthreads();
+
+  // ---- pass 2: d-parallel weighted V accumulation (coalesced rows)
+  for (int d = t; d < D; d += 256) {
+    float acc = 0.0f;
+    for (int sp = 0; sp < len; ++sp)
+      acc = fmaf(sc[sp], bf16_to_f32(vc[(int64_t)sp * KVS + kvh * D + d]),
+                 acc);
+    out[h * D + d] = f32_to_bf16(acc * inv);
   }
 }
 
@@ -285,11 +263,14 @@ extern "C" hipError_t lumina_dec_rope_cache(const void* qkv, void* q_out,
 
 extern "C" hipError_t lumina_dec_attn(const void* q, const void* kc,
                                       const void* vc, void* out,
-                                      const int* pos_dev, int H, int HKV,
+                                      const int* pos_dev, int cap,
+                                      int H, int HKV,
                                       int D, float scale,
                                       hipStream_t stream) {
   if (D > 192) return hipErrorInvalidValue;
-  hipLaunchKernelGGL(dec_attn_kernel, dim3(H), dim3(256), 0, stream,
+  const int lds = (16 + ((D + 3) & ~3)) * 4 + cap * 4;
+  if (lds > 160 * 1024) return hipErrorInvalidValue;
+  hipLaunchKernelGGL(dec_attn_kernel, dim3(H), dim3(256), lds, stream,
                      (const uint16_t*)q, (const uint16_t*)kc,
                      (const uint16_t*)vc, (uint16_t*)out, pos_dev,
                      H, HKV, D, scale);
