@@ -8,9 +8,15 @@ SwiGLU -- in the launches-baseline shape from the hardware guide, and
 hipGraph-captures the whole token step (all shapes static, cursor on
 device).
 
-Scope: dense and MoD models (MoD layers run dense at decode: the block's
-run_mod gate requires S > 1).  MoE models fall back to the standard
-engine path.
+Scope: dense, MoD (MoD layers run dense at decode: the block's run_mod
+gate requires S > 1) and single-GPU MoE models.  MoE layers decode as:
+rmsnorm -> router GEMV -> on-device top-k (dec_topk) -> per selected
+expert an indirect SWIGLU GEMV + scaled-residual down GEMV, the expert
+index read from device memory so the step still hipGraph-captures.
+Router softmax runs over bf16 logits (training gating is fp32): top-k
+selection can differ from eager only on near-exact routing ties.
+MoE with EP/TP shards, a load-balancer placement, or fp8/int8 expert
+storage falls back to the standard engine path.
 """
 
 from __future__ import annotations
@@ -24,6 +30,7 @@ from .. import ops
 _NORM = 1
 _RESID = 2
 _SWIGLU = 4
+_SCALE = 8
 
 
 def can_fuse_decode(model) -> bool:
@@ -34,7 +41,13 @@ def can_fuse_decode(model) -> bool:
         return False
     for layer in model.layers:
         if getattr(layer, "is_moe", False):
-            return False
+            f = layer.ffn
+            if (f.ep_size > 1 or getattr(f, "tp_group", None) is not None
+                    or getattr(f, "placement", None) is not None
+                    or getattr(f, "use_fp8", False)
+                    or getattr(f, "use_int8_weights", False)
+                    or f.num_experts > 64):
+                return False
         a = layer.attention
         if a.head_dim > 192 or a.qkv_proj.weight.dtype != torch.bfloat16:
             return False
@@ -70,8 +83,9 @@ class FusedDecoder:
         self.qkv = torch.zeros(qs + 2 * kvs, dtype=dt, device=dev)
         self.q = torch.zeros(qs, dtype=dt, device=dev)
         self.attn_out = torch.zeros(qs, dtype=dt, device=dev)
-        self.act = torch.zeros(model.layers[0].ffn.intermediate_size,
-                               dtype=dt, device=dev)
+        imax = max(l.ffn.intermediate_size for l in model.layers)
+        self.act = torch.zeros(imax, dtype=dt, device=dev)
+        self.xhat = torch.zeros(h, dtype=dt, device=dev)
         self.x2 = torch.zeros(h, dtype=dt, device=dev)
         V = model.lm_head.weight.shape[0]
         self.logits = torch.zeros(V, dtype=dt, device=dev)
@@ -88,17 +102,36 @@ class FusedDecoder:
                                device=dev) for _ in model.layers]
         self.vc = [torch.zeros_like(self.kc[0]) for _ in model.layers]
 
-        # contiguous bf16 weights views
+        # contiguous bf16 weights views; MoE experts pre-transposed into
+        # GEMV row-major layout ([E, 2I, h] / [E, h, I]) once at init
         self.Wqkv, self.Wo, self.Wgu, self.Wdn = [], [], [], []
         self.wn_in, self.wn_post = [], []
+        self.moe = []          # per-layer: None or (Wg, WguT, WdnT, k, temp)
         for layer in model.layers:
             at = layer.attention
             self.Wqkv.append(at.qkv_proj.weight.data.contiguous())
             self.Wo.append(at.o_proj.weight.data.contiguous())
-            self.Wgu.append(layer.ffn.gate_up_proj.weight.data.contiguous())
-            self.Wdn.append(layer.ffn.down_proj.weight.data.contiguous())
             self.wn_in.append(layer.input_norm.weight.data.contiguous())
             self.wn_post.append(layer.post_attn_norm.weight.data.contiguous())
+            if getattr(layer, "is_moe", False):
+                f = layer.ffn
+                self.Wgu.append(None)
+                self.Wdn.append(None)
+                self.moe.append((
+                    f.gate.weight.data.to(dt).contiguous(),
+                    f.w_gate_up.data.to(dt).permute(0, 2, 1).contiguous(),
+                    f.w_down.data.to(dt).permute(0, 2, 1).contiguous(),
+                    f.top_k, f.routing_temperature, f.intermediate_size))
+            else:
+                self.Wgu.append(
+                    layer.ffn.gate_up_proj.weight.data.contiguous())
+                self.Wdn.append(layer.ffn.down_proj.weight.data.contiguous())
+                self.moe.append(None)
+        kmax = max((m[3] for m in self.moe if m), default=0)
+        E = max((m[0].shape[0] for m in self.moe if m), default=1)
+        self.rlogits = torch.zeros(E, dtype=dt, device=dev)
+        self.eidx = torch.zeros(max(kmax, 1), dtype=torch.int32, device=dev)
+        self.ew = torch.zeros(max(kmax, 1), dtype=torch.float32, device=dev)
         self.wn_final = model.final_norm.weight.data.contiguous()
         self.Wlm = model.lm_head.weight.data.contiguous()
         self.embed = model.embed_tokens.weight.data
@@ -142,10 +175,25 @@ class FusedDecoder:
                        self.pos_dev, self.H, self.HKV, self.D, self.scale)
             e.dec_gemv(self.Wo[i], self.attn_out, None, x, self.x2,
                        self.eps, _RESID)
-            e.dec_gemv(self.Wgu[i], self.x2, self.wn_post[i], None, self.act,
-                       self.eps, _NORM | _SWIGLU)
-            e.dec_gemv(self.Wdn[i], self.act, None, self.x2, x,
-                       self.eps, _RESID)
+            if self.moe[i] is None:
+                e.dec_gemv(self.Wgu[i], self.x2, self.wn_post[i], None,
+                           self.act, self.eps, _NORM | _SWIGLU)
+                e.dec_gemv(self.Wdn[i], self.act, None, self.x2, x,
+                           self.eps, _RESID)
+            else:
+                Wg, WguT, WdnT, k, temp, I = self.moe[i]
+                e.dec_rmsnorm(self.x2, self.wn_post[i], self.xhat, self.eps)
+                rl = self.rlogits[:Wg.shape[0]]
+                e.dec_gemv(Wg, self.xhat, None, None, rl, self.eps, 0)
+                e.dec_topk(rl, self.eidx, self.ew, k, temp)
+                act = self.act[:I]
+                for slot in range(k):
+                    e.dec_gemv_moe(WguT, self.xhat, None, act,
+                                   self.eidx, self.ew, slot, _SWIGLU)
+                    e.dec_gemv_moe(WdnT, act,
+                                   self.x2 if slot == 0 else x, x,
+                                   self.eidx, self.ew, slot,
+                                   _RESID | _SCALE)
         e.dec_gemv(self.Wlm, x, self.wn_final, None, self.logits,
                    self.eps, _NORM)
         e.dec_advance(self.pos_dev)

@@ -62,6 +62,9 @@ hipError_t lumina_dec_gemv(const void*, const void*, const void*, const void*, v
 hipError_t lumina_dec_rope_cache(const void*, void*, void*, void*, const float*, const float*, const int*, int, int, int, hipStream_t);
 hipError_t lumina_dec_attn(const void*, const void*, const void*, void*, const int*, int, int, int, int, float, hipStream_t);
 hipError_t lumina_dec_advance(int*, hipStream_t);
+hipError_t lumina_dec_rmsnorm(const void*, const void*, void*, int, float, hipStream_t);
+hipError_t lumina_dec_topk(const void*, int*, float*, int, int, float, hipStream_t);
+hipError_t lumina_dec_gemv_moe(const void*, const void*, const void*, void*, const int*, const float*, int, int64_t, int, int, int, hipStream_t);
 hipError_t lumina_moe_gather_rows(const void*, const int64_t*, const bool*, void*, int64_t, int, int, hipStream_t);
 hipError_t lumina_moe_dispatch_bwd(const void*, const int64_t*, void*, int64_t, int, int, int64_t, int, hipStream_t);
 hipError_t lumina_moe_combine_fwd(const void*, const float*, const int64_t*, void*, int64_t, int, int, int64_t, int, hipStream_t);
@@ -609,6 +612,37 @@ void dec_advance(const at::Tensor& pos_dev) {
             "dec_advance");
 }
 
+void dec_rmsnorm(const at::Tensor& x, const at::Tensor& wn, at::Tensor& out,
+                 double eps) {
+  check_hip(lumina_dec_rmsnorm(x.data_ptr(), wn.data_ptr(), out.data_ptr(),
+                               (int)x.numel(), (float)eps, cur_stream()),
+            "dec_rmsnorm");
+}
+
+void dec_topk(const at::Tensor& logits, at::Tensor& eidx, at::Tensor& ew,
+              int64_t k, double temp) {
+  check_hip(lumina_dec_topk(logits.data_ptr(), eidx.data_ptr<int>(),
+                            ew.data_ptr<float>(), (int)logits.numel(),
+                            (int)k, (float)temp, cur_stream()),
+            "dec_topk");
+}
+
+void dec_gemv_moe(const at::Tensor& W, const at::Tensor& x,
+                  const std::optional<at::Tensor>& resid, at::Tensor& y,
+                  const at::Tensor& eidx, const at::Tensor& ew,
+                  int64_t slot, int64_t flags) {
+  // W: [E, N(or 2N), K] contiguous; rows per expert from y/x sizes
+  const int K = (int)x.numel();
+  const int N = (int)y.numel();
+  const int64_t estride = W.size(1) * W.size(2);
+  check_hip(lumina_dec_gemv_moe(W.data_ptr(), x.data_ptr(),
+                                resid ? resid->data_ptr() : nullptr,
+                                y.data_ptr(), eidx.data_ptr<int>(),
+                                ew.data_ptr<float>(), (int)slot, estride,
+                                N, K, (int)flags, cur_stream()),
+            "dec_gemv_moe");
+}
+
 at::Tensor gemv(const at::Tensor& x, const at::Tensor& w) {
   // y[N] = w[N,K] @ x[K]
   TORCH_CHECK(x.is_contiguous() && w.is_contiguous());
@@ -634,6 +668,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
           "decode RoPE + KV-cache append at the device cursor");
   mod.def("dec_attn", &dec_attn, "single-token GQA attention over the cache");
   mod.def("dec_advance", &dec_advance, "advance the decode cursor");
+  mod.def("dec_rmsnorm", &dec_rmsnorm, "decode rmsnorm -> xhat buffer");
+  mod.def("dec_topk", &dec_topk,
+          "decode router: softmax/temp top-k, renormalized, on device");
+  mod.def("dec_gemv_moe", &dec_gemv_moe,
+          "expert-indirect decode GEMV (weight = W + eidx[slot]*estride)");
   mod.def("mx_quant_rows", &mx_quant_rows,
           "rowwise e8m0 fp8 quantization, K zero-padded (gfx950)");
   mod.def("mx_quant_cols", &mx_quant_cols,

@@ -25,6 +25,7 @@ typedef const __attribute__((address_space(1))) void* dgas;
 #define DEC_NORM   1
 #define DEC_RESID  2
 #define DEC_SWIGLU 4
+#define DEC_SCALE  8
 
 DEV_INLINE float silu(float v) { return v / (1.0f + __expf(-v)); }
 
@@ -209,6 +210,114 @@ void dec_attn_kernel(const uint16_t* __restrict__ q,
 }
 
 // ---------------------------------------------------------------------------
+// MoE decode support: xhat = rmsnorm(x)*wn once (router and both experts
+// read it), a top-k router kernel, and an expert-indirect GEMV whose
+// weight pointer is W + eidx[slot]*estride -- the expert index stays on
+// device so the whole MoE token step hipGraph-captures.
+
+__global__ __launch_bounds__(256)
+void dec_rmsnorm_kernel(const uint16_t* __restrict__ x,
+                        const uint16_t* __restrict__ wn,
+                        uint16_t* __restrict__ out, int K, float eps) {
+  __shared__ float red[16];
+  float ss = 0.0f;
+  for (int k = threadIdx.x; k < K; k += 256) {
+    const float v = bf16_to_f32(x[k]);
+    ss = fmaf(v, v, ss);
+  }
+  const float inv = rsqrtf(block_reduce_sum(ss, red) / K + eps);
+  for (int k = threadIdx.x; k < K; k += 256)
+    out[k] = f32_to_bf16(bf16_to_f32(x[k]) * inv * bf16_to_f32(wn[k]));
+}
+
+// softmax(logits/temp) -> top-k (renormalized over the k), one wave.
+// Matches ops/reference.py topk_gating inference semantics (no noise).
+__global__ void dec_topk_kernel(const uint16_t* __restrict__ logits,
+                                int* __restrict__ eidx,
+                                float* __restrict__ ew,
+                                int E, int k, float temp) {
+  const int lane = threadIdx.x;
+  float v = lane < E ? bf16_to_f32(logits[lane]) / temp : -1e30f;
+  const float m = wave_reduce_max(v);
+  float p = lane < E ? __expf(v - m) : 0.0f;
+  const float Z = wave_reduce_sum(p);
+  p /= Z;
+  float wsum = 0.0f;
+  float pk = p;
+  for (int j = 0; j < k; ++j) {
+    const float mj = wave_reduce_max(pk);
+    // first lane holding the max claims the slot (stable tie-break: min id)
+    const uint64_t hit = __ballot(pk == mj);
+    const int who = __ffsll((unsigned long long)hit) - 1;
+    if (lane == who) {
+      eidx[j] = lane;
+      pk = -1.0f;
+    }
+    if (lane == 0) ew[j] = mj;
+    wsum += mj;
+  }
+  if (lane < k) ew[lane] = ew[lane] / fmaxf(wsum, 1e-9f);
+}
+
+// y[N] = W_e[N(or 2N), K] @ x[K] with W_e = W + eidx[slot]*estride.
+// SCALE multiplies the dot by ew[slot] before the residual add.
+template <int FLAGS>
+__global__ __launch_bounds__(256)
+void dec_gemv_moe_kernel(const uint16_t* __restrict__ W,
+                         const uint16_t* __restrict__ x,
+                         const uint16_t* __restrict__ resid,
+                         uint16_t* __restrict__ y,
+                         const int* __restrict__ eidx,
+                         const float* __restrict__ ew,
+                         int slot, int64_t estride, int N, int K) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  uint16_t* xs = reinterpret_cast<uint16_t*>(smem);
+  for (int kk = threadIdx.x; kk < K; kk += 256)
+    xs[kk] = x[kk];
+  __syncthreads();
+
+  const int row = blockIdx.x * 4 + (threadIdx.x >> 6);
+  if (row >= N) return;
+  const int lane = threadIdx.x & 63;
+  const uint16_t* We = W + eidx[slot] * estride;
+  const uint16_t* w0 = We + (int64_t)row * K;
+  const uint16_t* w1 = (FLAGS & DEC_SWIGLU)
+      ? We + (int64_t)(N + row) * K : nullptr;
+
+  float acc0 = 0.0f, acc1 = 0.0f;
+  int k = lane * 8;
+  for (; k + 8 <= K; k += 64 * 8) {
+    const ushortx8 xv = *reinterpret_cast<const ushortx8*>(xs + k);
+    const ushortx8 wv = __builtin_nontemporal_load(
+        reinterpret_cast<const ushortx8*>(w0 + k));
+    #pragma unroll
+    for (int i = 0; i < 8; ++i)
+      acc0 = fmaf(bf16_to_f32(wv[i]), bf16_to_f32(xv[i]), acc0);
+    if constexpr (FLAGS & DEC_SWIGLU) {
+      const ushortx8 uv = __builtin_nontemporal_load(
+          reinterpret_cast<const ushortx8*>(w1 + k));
+      #pragma unroll
+      for (int i = 0; i < 8; ++i)
+        acc1 = fmaf(bf16_to_f32(uv[i]), bf16_to_f32(xv[i]), acc1);
+    }
+  }
+  for (int kk = k; kk < K && kk < k + 8; ++kk) {
+    acc0 = fmaf(bf16_to_f32(w0[kk]), bf16_to_f32(xs[kk]), acc0);
+    if constexpr (FLAGS & DEC_SWIGLU)
+      acc1 = fmaf(bf16_to_f32(w1[kk]), bf16_to_f32(xs[kk]), acc1);
+  }
+  acc0 = wave_reduce_sum(acc0);
+  if constexpr (FLAGS & DEC_SWIGLU) acc1 = wave_reduce_sum(acc1);
+  if (lane == 0) {
+    float v = acc0;
+    if constexpr (FLAGS & DEC_SWIGLU) v = silu(acc0) * acc1;
+    if constexpr (FLAGS & DEC_SCALE) v *= ew[slot];
+    if constexpr (FLAGS & DEC_RESID) v += bf16_to_f32(resid[row]);
+    y[row] = f32_to_bf16(v);
+  }
+}
+
+// ---------------------------------------------------------------------------
 extern "C" hipError_t lumina_dec_gemv(const void* W, const void* x,
                                       const void* wn, const void* resid,
                                       void* y, int N, int K, float eps,
@@ -280,5 +389,50 @@ extern "C" hipError_t lumina_dec_attn(const void* q, const void* kc,
 extern "C" hipError_t lumina_dec_advance(int* pos_dev, hipStream_t stream) {
   hipLaunchKernelGGL(dec_advance_kernel, dim3(1), dim3(64), 0, stream,
                      pos_dev);
+  return hipGetLastError();
+}
+
+extern "C" hipError_t lumina_dec_rmsnorm(const void* x, const void* wn,
+                                         void* out, int K, float eps,
+                                         hipStream_t stream) {
+  hipLaunchKernelGGL(dec_rmsnorm_kernel, dim3(1), dim3(256), 0, stream,
+                     (const uint16_t*)x, (const uint16_t*)wn,
+                     (uint16_t*)out, K, eps);
+  return hipGetLastError();
+}
+
+extern "C" hipError_t lumina_dec_topk(const void* logits, int* eidx,
+                                      float* ew, int E, int k, float temp,
+                                      hipStream_t stream) {
+  if (E > 64 || k > E) return hipErrorInvalidValue;
+  hipLaunchKernelGGL(dec_topk_kernel, dim3(1), dim3(64), 0, stream,
+                     (const uint16_t*)logits, eidx, ew, E, k, temp);
+  return hipGetLastError();
+}
+
+extern "C" hipError_t lumina_dec_gemv_moe(const void* W, const void* x,
+                                          const void* resid, void* y,
+                                          const int* eidx, const float* ew,
+                                          int slot, int64_t estride,
+                                          int N, int K, int flags,
+                                          hipStream_t stream) {
+  const int lds = (K * 2 + 15) & ~15;
+  dim3 grid((N + 3) / 4), block(256);
+  switch (flags) {
+    case DEC_SWIGLU:
+      hipLaunchKernelGGL(dec_gemv_moe_kernel<DEC_SWIGLU>, grid, block, lds,
+                         stream, (const uint16_t*)W, (const uint16_t*)x,
+                         (const uint16_t*)resid, (uint16_t*)y, eidx, ew,
+                         slot, estride, N, K);
+      break;
+    case DEC_RESID | DEC_SCALE:
+      hipLaunchKernelGGL((dec_gemv_moe_kernel<DEC_RESID | DEC_SCALE>), grid,
+                         block, lds, stream, (const uint16_t*)W,
+                         (const uint16_t*)x, (const uint16_t*)resid,
+                         (uint16_t*)y, eidx, ew, slot, estride, N, K);
+      break;
+    default:
+      return hipErrorInvalidValue;
+  }
   return hipGetLastError();
 }

@@ -7,13 +7,14 @@ from luminaai_amd.config import ConfigPresets
 from luminaai_amd.models import DeepSeekTransformer, config_to_deepseek_config
 from luminaai_amd.inference.fused_decode import FusedDecoder, can_fuse_decode
 
-cfg = ConfigPresets.get("b1")
+preset = sys.argv[1] if len(sys.argv) > 1 else "b1"
+cfg = ConfigPresets.get(preset)
 mcfg = config_to_deepseek_config(cfg)
 torch.manual_seed(0)
 with torch.device("cuda"):
     model = DeepSeekTransformer(mcfg)
 model = model.to(torch.bfloat16).eval()
-print("can_fuse:", can_fuse_decode(model))
+print(preset, "can_fuse:", can_fuse_decode(model))
 
 dec = FusedDecoder(model, 2048)
 prompt = torch.randint(1, 1000, (1, 64), device="cuda")

@@ -723,6 +723,60 @@ def test_fused_decoder_matches_eager():
     assert got2 == got, (got2, got)
 
 
+def test_fused_decoder_moe_matches_eager():
+    """MoE fused decode (on-device router top-k + expert-indirect GEMVs)
+    matches the eager KV-cached MoE model's greedy stream.  Router runs
+    on bf16 logits (eager gating is fp32): allow a couple of near-tie
+    divergences."""
+    from luminaai_amd.models.transformer import (DeepSeekConfig,
+                                                 DeepSeekTransformer, KVCache)
+    from luminaai_amd.inference.fused_decode import (FusedDecoder,
+                                                     can_fuse_decode)
+    cfg = DeepSeekConfig(vocab_size=512, hidden_size=318, num_layers=3,
+                         num_heads=2, num_kv_heads=1, intermediate_size=256,
+                         use_moe=True, num_experts=4, moe_top_k=2,
+                         seq_length=128)
+    torch.manual_seed(3)
+    model = DeepSeekTransformer(cfg).to(_dev()).to(torch.bfloat16).eval()
+    assert can_fuse_decode(model)
+
+    prompt = torch.randint(1, 512, (1, 24), device=_dev())
+    n_new = 24
+
+    caches = [KVCache(max_len=128) for _ in model.layers]
+    with torch.no_grad():
+        logits, _, _ = model(prompt, kv_caches=caches)
+        ref_toks = []
+        tok = logits[:, -1].argmax(-1, keepdim=True)
+        for _ in range(n_new):
+            ref_toks.append(int(tok))
+            logits, _, _ = model(tok, kv_caches=caches)
+            tok = logits[:, -1].argmax(-1, keepdim=True)
+
+    dec = FusedDecoder(model, 128)
+    logits = dec.prefill(prompt)
+    tok = logits.argmax(-1).view(1)
+    got = []
+    for _ in range(n_new):
+        got.append(int(tok))
+        lg = dec.step(tok)
+        tok = lg.float().argmax().view(1)
+    agree = sum(a == b for a, b in zip(got, ref_toks))
+    assert agree >= n_new - 3, (got, ref_toks)
+
+    # captured graph path replays the exact same stream
+    dec.reset()
+    logits = dec.prefill(prompt)
+    dec.capture()
+    tok = logits.argmax(-1).view(1)
+    got2 = []
+    for _ in range(n_new):
+        got2.append(int(tok))
+        lg = dec.step(tok)
+        tok = lg.float().argmax().view(1)
+    assert got2 == got, (got2, got)
+
+
 def test_expert_bmm_fp32_tight():
     """fp32-path expert grouped GEMM with TIGHT tolerances (round-2: the
     bf16 test's loose rtol could hide a sign/magnitude bug in a tail)."""
