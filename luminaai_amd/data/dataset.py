@@ -104,9 +104,14 @@ class StreamingBaseTrainingDataset(IterableDataset):
                 and dist.get_world_size() > 1:
             from ..parallel.mesh import get_mesh
             mesh = get_mesh()
-            sp = mesh.sp_size if mesh is not None else 1
-            dp_world = dist.get_world_size() // max(sp, 1)
-            dp_rank = dist.get_rank() // max(sp, 1)
+            # SP ranks slice one batch along the sequence; TP ranks compute
+            # on one batch with sharded weights -> both count as ONE data
+            # rank (both are innermost-contiguous in the mesh layout)
+            shared = 1
+            if mesh is not None:
+                shared = max(mesh.sp_size, 1) * max(mesh.tp_size, 1)
+            dp_world = dist.get_world_size() // shared
+            dp_rank = dist.get_rank() // shared
         else:
             dp_rank, dp_world = 0, 1
         return dp_rank * nw + wid, dp_world * nw
@@ -236,9 +241,11 @@ def create_dataloader(dataset: Dataset, config, shuffle: bool = True) -> DataLoa
             and dist.get_world_size() > 1:
         from ..parallel.mesh import get_mesh
         mesh = get_mesh()
-        sp = mesh.sp_size if mesh is not None else 1
-        world = dist.get_world_size() // max(sp, 1)
-        rank = dist.get_rank() // max(sp, 1)
+        shared = 1
+        if mesh is not None:      # SP slices / TP shares one batch
+            shared = max(mesh.sp_size, 1) * max(mesh.tp_size, 1)
+        world = dist.get_world_size() // shared
+        rank = dist.get_rank() // shared
         from torch.utils.data.distributed import DistributedSampler
         kwargs["sampler"] = DistributedSampler(
             dataset, num_replicas=world, rank=rank, shuffle=shuffle,

@@ -69,7 +69,7 @@ class ZeroEngine:
         return [g for g in self.opt.groups if g.comm == "dp"]
 
     def _expert_groups(self):
-        return [g for g in self.opt.groups if g.comm == "expert"]
+        return [g for g in self.opt.groups if g.comm in ("expert", "tp")]
 
     # ------------------------------------------------------------- hooks
     def _install_hooks(self):
@@ -167,9 +167,11 @@ class ZeroEngine:
         # sharded-param grads (EP expert shards / TP weight shards):
         # replicas of the same shard live across the replica group (no-op
         # when every replica set has one member)
-        if self.mesh is not None and self.mesh.shard_replica_size > 1:
+        if self.mesh is not None:
             for g in self._expert_groups():
-                dist.all_reduce(g.flat_g, group=self.mesh.shard_replica_group)
+                if self.mesh.replica_size_for(g.comm) > 1:
+                    dist.all_reduce(
+                        g.flat_g, group=self.mesh.replica_group_for(g.comm))
         if self.mesh is not None and self.mesh.tp_size > 1:
             # TP ranks SHARE their batch: dense grads sum tp duplicates and
             # the uniform 1/world grad_scale averages them out, but a weight
@@ -192,17 +194,15 @@ class ZeroEngine:
             dist.all_reduce(ns, group=self.pg)
         # stages 0/1: dp grads are already globally reduced (summed); the norm
         # of the summed grad is what clipping applies to (after grad_scale).
-        exp = self._expert_groups()
-        if exp:
-            ns_e = None
-            for g in exp:
-                n = K.l2norm_sq(g.flat_g)
-                ns_e = n if ns_e is None else ns_e + n
-            if self.mesh is not None and self.mesh.shard_exchange_size > 1:
+        for g in self._expert_groups():
+            ns_e = K.l2norm_sq(g.flat_g)
+            if self.mesh is not None \
+                    and self.mesh.exchange_size_for(g.comm) > 1:
                 # each rank holds 1/N of the sharded params (post replica
                 # reduce) -> summing over ONE exchange group covers every
                 # shard exactly once
-                dist.all_reduce(ns_e, group=self.mesh.shard_exchange_group)
+                dist.all_reduce(ns_e,
+                                group=self.mesh.exchange_group_for(g.comm))
             ns = ns_e if ns is None else ns + ns_e
         return ns
 
@@ -249,10 +249,12 @@ class ZeroEngine:
             if not g._master_is_params:
                 g.master.copy_(
                     g.weight_view()[g.shard_lo:g.shard_hi].float())
-        if self.mesh is not None and self.mesh.shard_replica_size > 1:
-            pg = self.mesh.shard_replica_group
-            src = dist.get_process_group_ranks(pg)[0]
+        if self.mesh is not None:
             for g in self._expert_groups():
+                if self.mesh.replica_size_for(g.comm) <= 1:
+                    continue
+                pg = self.mesh.replica_group_for(g.comm)
+                src = dist.get_process_group_ranks(pg)[0]
                 dist.broadcast(g.weight_view(), src=src, group=pg)
                 if not g._master_is_params:
                     g.master.copy_(g.weight_view().float())

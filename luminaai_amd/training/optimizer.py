@@ -42,13 +42,17 @@ def split_decay_groups(model: nn.Module, ep_active: bool = False) -> List[Dict]:
     (comm='expert'): their grads are NOT replicated across the EP group and
     must not join the global DP all-reduce."""
     from ..parallel.expert_parallel import is_expert_param
-    decay, no_decay, expert = [], [], []
+    decay, no_decay, expert, tp_shard = [], [], [], []
     for name, p in model.named_parameters():
         if not p.requires_grad:
             continue
-        if (ep_active and is_expert_param(name)) \
-                or getattr(p, "_shard_parallel", False):
+        if ep_active and is_expert_param(name):
+            # EP-sharded (under TP x EP also TP-sharded): replicas across
+            # dp only -> own comm class
             expert.append(p)
+        elif getattr(p, "_shard_parallel", False):
+            # TP-sharded non-expert weights: replicas across dp x ep
+            tp_shard.append(p)
         elif any(k in name for k in NO_DECAY_KEYWORDS) or p.dim() <= 1:
             no_decay.append(p)
         else:
@@ -61,6 +65,9 @@ def split_decay_groups(model: nn.Module, ep_active: bool = False) -> List[Dict]:
     if expert:
         groups.append({"params": expert, "weight_decay": None,
                        "comm": "expert"})
+    if tp_shard:
+        groups.append({"params": tp_shard, "weight_decay": None,
+                       "comm": "tp"})
     return groups
 
 
@@ -73,7 +80,7 @@ class _FlatGroup:
         self.params = params
         self.lr = lr
         self.weight_decay = weight_decay
-        self.comm = comm                  # "dp" (replicated) | "expert" (EP-sharded)
+        self.comm = comm          # "dp" (replicated) | "expert" (EP) | "tp" (TP)
         self.shard_rank = shard_rank
         self.shard_world = shard_world
         self.numel = sum(p.numel() for p in params)
@@ -190,8 +197,10 @@ class FlatAdamW:
             _FlatGroup(list(g["params"]), lr=g.get("lr", lr),
                        weight_decay=(weight_decay if g.get("weight_decay") is None
                                      else g["weight_decay"]),
-                       shard_rank=0 if g.get("comm") == "expert" else shard_rank,
-                       shard_world=1 if g.get("comm") == "expert" else shard_world,
+                       shard_rank=0 if g.get("comm") in ("expert", "tp")
+                       else shard_rank,
+                       shard_world=1 if g.get("comm") in ("expert", "tp")
+                       else shard_world,
                        comm=g.get("comm", "dp"), offload=offload)
             for g in groups
         ]
@@ -333,9 +342,10 @@ class FlatAdamW:
         concatenating the saved shards in rank order reconstructs the full
         padded-at-world-A buffer; the first `numel` elements are re-padded
         for world B and re-sliced."""
-        if any(g.comm == "expert" for g in self.groups):
-            raise ValueError("elastic resharding with expert parallelism is "
-                             "not supported (expert placement changes)")
+        if any(g.comm in ("expert", "tp") for g in self.groups):
+            raise ValueError("elastic resharding with expert/tensor "
+                             "parallelism is not supported (shard placement "
+                             "changes)")
         saved_world = shard_sds[0].get("shard_world", 1)
         if len(shard_sds) != saved_world:
             raise ValueError(f"need all {saved_world} shards, got "
@@ -379,7 +389,7 @@ class FlatAdamW:
         new_groups = []
         for gd in groups:
             wd = gd.get("weight_decay")
-            expert = gd.get("comm") == "expert"
+            expert = gd.get("comm") in ("expert", "tp")
             fg = _FlatGroup(list(gd["params"]), lr=lr,
                             weight_decay=(self.defaults["weight_decay"]
                                           if wd is None else wd),

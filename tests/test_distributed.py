@@ -1025,6 +1025,137 @@ def test_ep_with_checkpointing():
 
 
 # ---- expert add/prune under EP (round-2: SURVEY build plan 7.6) ------------
+def tpep_worker(rank, world):
+    """TP(2) x EP(2) composed mesh (world 4, dp 1): full Trainer steps.
+    Ranks (e, t) hold expert shard e sliced by tp rank t; tp peers share a
+    batch shard, ep peers exchange tokens. Must match single-process
+    training on the concatenated global batch."""
+    from luminaai_amd.config import Config
+    from luminaai_amd.data.tokenizer import ConversationTokenizer
+    from luminaai_amd.models import (DeepSeekTransformer,
+                                     config_to_deepseek_config)
+    from luminaai_amd.parallel.mesh import init_mesh, reset_mesh
+    from luminaai_amd.parallel.tensor_parallel import convert_to_tensor_parallel
+    from luminaai_amd.training import Trainer
+    cfg = Config(vocab_size=512, hidden_size=64, num_layers=2, num_heads=4,
+                 num_kv_heads=2, seq_length=32, intermediate_size=128,
+                 micro_batch_size=2, gradient_accumulation_steps=1,
+                 num_workers=0, use_moe=True, num_experts=4, moe_top_k=2,
+                 routing_noise_std=0.0, use_mod=False,
+                 zero_stage=0, precision="fp32",
+                 experiment_name=f"tpep_{rank}",
+                 eval_every_n_batches=0, save_every_n_batches=0)
+    # build the FULL model first (identical RNG stream to the reference:
+    # EP-sharded construction consumes fewer randoms and diverges all
+    # later params), then copy dense params + this rank's expert slices
+    mcfg = config_to_deepseek_config(cfg)
+    torch.manual_seed(1234)
+    fullm = DeepSeekTransformer(mcfg)
+    for l in fullm.layers:                  # decisive routing: TP float-sum
+        l.ffn.gate.weight.data.mul_(50.0)   # reordering must not flip top-k
+    mesh = init_mesh(ep_size=2, tp_size=2)
+    assert (mesh.dp_size, mesh.ep_rank, mesh.tp_rank) == \
+        (1, rank // 2, rank % 2)
+    model = DeepSeekTransformer(mcfg)
+    EL = 4 // 2
+    e0 = mesh.ep_rank * EL
+    with torch.no_grad():
+        for (n, p), (n2, q) in zip(model.named_parameters(),
+                                   fullm.named_parameters()):
+            assert n == n2
+            p.copy_(q if p.shape == q.shape else q[e0:e0 + EL])
+    convert_to_tensor_parallel(model, mesh)
+    t = Trainer(model, ConversationTokenizer(), cfg)
+    t._setup_scheduler(10)
+    drank = rank // 2                       # tp peers share a batch shard
+    gens = [torch.Generator().manual_seed(640 + d) for d in range(2)]
+    for _ in range(2):
+        ids = torch.randint(1, cfg.vocab_size, (2, cfg.seq_length + 1),
+                            generator=gens[drank])
+        t.engine.set_sync(True)
+        t.train_step({"input_ids": ids[:, :-1], "labels": ids[:, 1:]})
+        t.optimizer_step()
+    emb = float(t.model.embed_tokens.weight.detach().sum())
+    gu = t.model.layers[0].ffn.w_gate_up.detach()
+    gn = t.optimizer.last_grad_norm()
+    reset_mesh()
+    return {"emb": emb, "gu_sum": float(gu.sum()),
+            "gu_shape": tuple(gu.shape), "grad_norm": gn, "rank": rank}
+
+
+def test_tpep_matches_single_process():
+    ctx = mp.get_context("spawn")
+    q = ctx.SimpleQueue()
+    port = _free_port()
+    procs = [ctx.Process(target=_run, args=(r, 4, port, "tpep_worker", q))
+             for r in range(4)]
+    for p in procs:
+        p.start()
+    res = {}
+    for _ in range(4):
+        rank, status, payload = q.get()
+        assert status == "ok", f"rank {rank} failed:\n{payload}"
+        res[rank] = payload
+    for p in procs:
+        p.join(timeout=180)
+
+    # single-process reference on the concatenated global batch
+    import torch as th
+    from luminaai_amd.config import Config
+    from luminaai_amd.data.tokenizer import ConversationTokenizer
+    from luminaai_amd.models import (DeepSeekTransformer,
+                                     config_to_deepseek_config)
+    from luminaai_amd.training import Trainer
+    os.environ.pop("WORLD_SIZE", None)
+    cfg = Config(vocab_size=512, hidden_size=64, num_layers=2, num_heads=4,
+                 num_kv_heads=2, seq_length=32, intermediate_size=128,
+                 micro_batch_size=4, gradient_accumulation_steps=1,
+                 num_workers=0, use_moe=True, num_experts=4, moe_top_k=2,
+                 routing_noise_std=0.0, use_mod=False,
+                 zero_stage=0, precision="fp32", experiment_name="tpep_ref",
+                 eval_every_n_batches=0, save_every_n_batches=0)
+    th.manual_seed(1234)
+    model = DeepSeekTransformer(config_to_deepseek_config(cfg))
+    for l in model.layers:
+        l.ffn.gate.weight.data.mul_(50.0)
+    t = Trainer(model, ConversationTokenizer(), cfg)
+    t._setup_scheduler(10)
+    gens = [th.Generator().manual_seed(640 + d) for d in range(2)]
+    for _ in range(2):
+        rows = []
+        for g in gens:
+            rows.append(th.randint(1, cfg.vocab_size,
+                                   (2, cfg.seq_length + 1), generator=g))
+        ids = th.cat(rows)
+        t.train_step({"input_ids": ids[:, :-1], "labels": ids[:, 1:]})
+        t.optimizer_step()
+    emb_ref = float(t.model.embed_tokens.weight.detach().sum())
+    gn_ref = t.optimizer.last_grad_norm()
+    gu = t.model.layers[0].ffn.w_gate_up.detach()      # [4, h, 2I]
+    I = cfg.intermediate_size
+    li = I // 2
+    for r in range(4):
+        e, tp = r // 2, r % 2
+        assert res[r]["emb"] == pytest.approx(emb_ref, abs=1e-3), \
+            (r, res[r]["emb"], emb_ref)
+        # the GLOBAL grad norm is the cross-class scaling check (sums the
+        # dense/"tp"/"expert" comm classes over their exchange groups);
+        # Adam updates are scale-invariant, weight sums alone can't see it.
+        # last_grad_norm is the RAW (pre-grad_scale) norm: distributed
+        # grads are world-summed, so raw norm = world x single-proc norm
+        assert res[r]["grad_norm"] / 4 == pytest.approx(gn_ref, rel=2e-3), \
+            (r, res[r]["grad_norm"], gn_ref)
+        ref = th.cat([gu[2 * e:2 * e + 2, :, tp * li:(tp + 1) * li],
+                      gu[2 * e:2 * e + 2, :, I + tp * li:I + (tp + 1) * li]],
+                     dim=2)
+        assert res[r]["gu_shape"] == tuple(ref.shape), \
+            (res[r]["gu_shape"], tuple(ref.shape))
+        # expert weights drift by Adam-amplified low-bit TP reorder noise
+        assert res[r]["gu_sum"] == pytest.approx(float(ref.sum()),
+                                                 abs=2e-2), \
+            (r, res[r]["gu_sum"], float(ref.sum()))
+
+
 def ep_add_prune_worker(rank, world):
     """add_expert/prune_expert under ep=2: shards stay even, every rank ends
     with an identical full expert stack, and the resharded model's forward
