@@ -71,7 +71,7 @@ class _Zero3Segment:
 
     def __init__(self, params: List[torch.Tensor], lr: float,
                  weight_decay: float, rank: int, world: int, pg=None,
-                 comm_kind: str = "dp"):
+                 comm_kind: str = "dp", offload: bool = False):
         self.params = params
         self.lr = lr
         self.weight_decay = weight_decay
@@ -100,12 +100,33 @@ class _Zero3Segment:
             self.offsets.append((off, n))
             off += n
 
-        # persistent shard state
+        # persistent shard state.  offload=True keeps the fp32 master and
+        # the Adam moments in host (pinned) memory and runs the update on
+        # the CPU (the ZeRO-3 analogue of FlatAdamW's cpu_offload_optimizer
+        # path; reference backend_deepspeed.py:129-165 offload_optimizer):
+        # only the bf16 w_shard/g_shard stay resident, cutting per-rank
+        # optimizer HBM from 12 to 4 bytes/param.
+        self.offload = bool(offload)
+        pin = device.type == "cuda"
+        state_dev = torch.device("cpu") if self.offload else device
         self.w_shard = self.flat_w[lo:hi].clone()
-        self.master = self.w_shard.float().clone()
+        mast = self.w_shard.float()
+        self.master = (mast.cpu().pin_memory() if self.offload and pin
+                       else mast.cpu() if self.offload else mast.clone())
         self.g_shard = torch.zeros_like(self.w_shard)
-        self.m = torch.zeros(self.shard_size, device=device, dtype=torch.float32)
+        self.m = torch.zeros(self.shard_size, device=state_dev,
+                             dtype=torch.float32)
         self.v = torch.zeros_like(self.m)
+        if self.offload and pin:
+            self._g_stage = torch.zeros(self.shard_size, dtype=self.dtype,
+                                        pin_memory=True)
+            self._w_stage = torch.zeros(self.shard_size, dtype=self.dtype,
+                                        pin_memory=True)
+        elif self.offload:
+            # CPU device (tests): the staging buffers alias the shards so
+            # the same two-phase step path runs without extra copies
+            self._g_stage = self.g_shard
+            self._w_stage = self.w_shard
 
     # -------------------------------------------------- storage management
     def weights_alloc(self) -> bool:
@@ -155,9 +176,25 @@ class _Zero3Segment:
 
     def step(self, step_count: int, norm_sq, max_norm: float,
              grad_scale: float):
+        assert not self.offload, "offloaded segments step via step_host()"
         K.adamw_step(self.master, self.g_shard, self.m, self.v, self.w_shard,
                      self.lr, 0.9, 0.95, 1e-8, self.weight_decay, step_count,
                      norm_sq, max_norm, grad_scale)
+
+    def stage_grad_d2h(self):
+        if self._g_stage is not self.g_shard:
+            self._g_stage.copy_(self.g_shard, non_blocking=True)
+
+    def step_host(self, step_count: int, norm_cpu, max_norm: float,
+                  grad_scale: float):
+        """CPU AdamW on the host-resident state (after stage_grad_d2h +
+        a device sync); async H2D of the refreshed weight shard."""
+        K.adamw_step(self.master, self._g_stage, self.m, self.v,
+                     self._w_stage, self.lr, 0.9, 0.95, 1e-8,
+                     self.weight_decay, step_count, norm_cpu, max_norm,
+                     grad_scale)
+        if self._w_stage is not self.w_shard:
+            self.w_shard.copy_(self._w_stage, non_blocking=True)
 
     def refresh_from_shard(self):
         """w_shard -> flat_w (persistent units after an optimizer step)."""
@@ -177,7 +214,7 @@ class _Zero3Segment:
         """flat_w -> w_shard/master (after load_state_dict wrote params)."""
         lo = self.rank * self.shard_size
         self.w_shard.copy_(self.flat_w[lo:lo + self.shard_size])
-        self.master.copy_(self.w_shard.float())
+        self.master.copy_(self.w_shard.float().to(self.master.device))
 
 
 class _Zero3Unit:
@@ -185,7 +222,8 @@ class _Zero3Unit:
 
     def __init__(self, name: str, module: nn.Module, lr: float,
                  weight_decay: float, rank: int, world: int,
-                 persistent: bool = False, pg=None, mesh=None):
+                 persistent: bool = False, pg=None, mesh=None,
+                 offload: bool = False):
         self.name = name
         self.module = module
         self.persistent = persistent
@@ -202,10 +240,11 @@ class _Zero3Unit:
         self.segments: List[_Zero3Segment] = []
         if decay:
             self.segments.append(_Zero3Segment(decay, lr, weight_decay,
-                                               rank, world, pg=pg))
+                                               rank, world, pg=pg,
+                                               offload=offload))
         if no_decay:
             self.segments.append(_Zero3Segment(no_decay, lr, 0.0, rank, world,
-                                               pg=pg))
+                                               pg=pg, offload=offload))
         if expert:
             # EP-sharded expert weights: shard only across the ranks that
             # hold the SAME experts (the expert replica group); with full EP
@@ -216,7 +255,8 @@ class _Zero3Unit:
                 dist.get_rank()) if r_pg is not None else 0)
             self.segments.append(_Zero3Segment(
                 [p for _, p in expert], lr, weight_decay,
-                r_rank, r_world, pg=r_pg, comm_kind="expert"))
+                r_rank, r_world, pg=r_pg, comm_kind="expert",
+                offload=offload))
         self.n_params = sum(len(s.params) for s in self.segments)
         self._grads_pending = self.n_params
 
@@ -291,8 +331,27 @@ class Zero3Optimizer:
         if norm_sq is None and self.max_grad_norm > 0:
             norm_sq = self.local_grad_norm_sq()
         self._last_norm_sq = norm_sq
+        offl = [g for g in self.groups if g.offload]
         for g in self.groups:
-            g.step(self.step_count, norm_sq, self.max_grad_norm, grad_scale)
+            if not g.offload:
+                g.step(self.step_count, norm_sq, self.max_grad_norm,
+                       grad_scale)
+        if offl:
+            # device-resident segments' kernels are in flight; overlap the
+            # D2H grad staging with them, sync once, then host math
+            for g in offl:
+                g.stage_grad_d2h()
+            if torch.cuda.is_available() and any(
+                    g.w_shard.is_cuda for g in offl):
+                torch.cuda.synchronize()
+            norm_cpu = (norm_sq.sum().cpu() if norm_sq is not None else None)
+            if norm_cpu is not None:
+                gn = float(norm_cpu) ** 0.5 * grad_scale
+                if not math.isfinite(gn):
+                    return      # NaN/Inf grads: skip the host update
+            for g in offl:
+                g.step_host(self.step_count, norm_cpu, self.max_grad_norm,
+                            grad_scale)
 
     def last_grad_norm(self) -> float:
         if self._last_norm_sq is None:
@@ -379,6 +438,8 @@ class Zero3Engine:
         self.overlap = getattr(config, "overlap_comm", True)
         lr = config.learning_rate
         wd = config.weight_decay
+        offload = bool(getattr(config, "cpu_offload_optimizer", False)
+                       or getattr(config, "aggressive_cpu_offload", False))
 
         blocks = list(getattr(model, "layers", []))
         # root unit: everything not inside a block (embed/final-norm/head)
@@ -386,11 +447,13 @@ class Zero3Engine:
         root_mod = _RootShell(model, blocks)
         self.units.append(_Zero3Unit("root", root_mod, lr, wd,
                                      self.rank, self.world, persistent=True,
-                                     pg=self.pg, mesh=mesh))
+                                     pg=self.pg, mesh=mesh,
+                                     offload=offload))
         for i, b in enumerate(blocks):
             self.units.append(_Zero3Unit(f"block{i}", b, lr, wd,
                                          self.rank, self.world,
-                                         pg=self.pg, mesh=mesh))
+                                         pg=self.pg, mesh=mesh,
+                                         offload=offload))
         self.optimizer = Zero3Optimizer(self)
         self._unit_of_param: Dict[int, _Zero3Unit] = {}
         self._hooks = []

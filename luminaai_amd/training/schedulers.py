@@ -43,6 +43,16 @@ class WarmupScheduler:
         self.step_num += 1
         self._apply()
 
+    def rebind(self, optimizer):
+        """Point the schedule at a REBUILT optimizer (expert add/prune under
+        ZeRO-3 replaces the optimizer object): keep the step position, take
+        the new group structure, re-apply the current multiplier."""
+        base = self.base_lrs[0] if self.base_lrs else None
+        self.optimizer = optimizer
+        self.base_lrs = [base if base is not None else g.lr
+                         for g in optimizer.groups]
+        self._apply()
+
     def get_last_lr(self):
         return [g.lr for g in self.optimizer.groups]
 

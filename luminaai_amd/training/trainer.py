@@ -654,10 +654,13 @@ class Trainer:
 
     def add_expert(self) -> bool:
         layers = self._moe_layers()
-        if not layers or self.engine.stage >= 3:
-            return False  # ZeRO-3 flat shards cannot grow in place
+        if not layers:
+            return False
         # under EP each layer grows by ep_size experts (one per shard) and
         # reshards; the fresh optimizer below re-flattens the new shapes
+        if self.engine.stage >= 3:
+            return self._mutate_experts_zero3(
+                lambda l: l.add_expert())
         for l in layers:
             l.add_expert()
         self.optimizer.rebuild(self.model)
@@ -666,17 +669,52 @@ class Trainer:
                                  overlap_comm=self.engine.overlap)
         return True
 
+    def _mutate_experts_zero3(self, fn) -> bool:
+        """Expert add/prune under ZeRO-3: the flat shards alias the param
+        storage and cannot grow in place, so materialise full weights, drop
+        the engine, mutate the layers, and build a fresh Zero3Engine over
+        the new shapes (fresh Adam moments for everything, like the
+        load-balance rebuild; step_count carries over for bias
+        correction)."""
+        from ..parallel.zero3 import Zero3Engine
+        old_engine = self.engine
+        step_count = self.optimizer.step_count
+        for u in old_engine.units:
+            u.gather()                      # p.data -> full weights
+        old_engine.remove_hooks()
+        # detach params from the old flat buffers so the new engine
+        # flattens from standalone storage
+        for p in self.model.parameters():
+            p.data = p.data.clone()
+            p.grad = None
+        for l in self._moe_layers():
+            fn(l)
+        self.engine = Zero3Engine(self.model, self.config, mesh=self.mesh)
+        self.optimizer = self.engine.optimizer
+        self.optimizer.step_count = step_count
+        if not self.engine._finalized:
+            self.engine._finalize_init()
+        if self.scheduler is not None:
+            self.scheduler.rebind(self.optimizer)
+        return True
+
     def prune_expert(self, expert_idx: Optional[int] = None) -> bool:
         layers = self._moe_layers()
         ep = self.mesh.ep_size if self.mesh is not None else 1
         if not layers or layers[0].num_experts - max(1, ep) < \
-                max(layers[0].top_k, 2) or self.engine.stage >= 3:
+                max(layers[0].top_k, 2):
             return False
-        for l in layers:
+
+        def _prune(l):
             idx = expert_idx
             if idx is None:
                 idx = int(l._usage_counts.argmin())
             l.prune_expert(idx)
+
+        if self.engine.stage >= 3:
+            return self._mutate_experts_zero3(_prune)
+        for l in layers:
+            _prune(l)
         self.optimizer.rebuild(self.model)
         self.engine = ZeroEngine(self.optimizer, stage=self.engine.stage,
                                  bucket_bytes=self.engine.bucket_bytes,
