@@ -234,6 +234,16 @@ class Config:
                 self.zero_stage = 1
 
     # ------------------------------------------------------------------
+    def _n_moe_layers(self) -> int:
+        """Layers that are actually MoE under moe_pattern (models/
+        transformer.py moe_layer_selector)."""
+        if not self.use_moe:
+            return 0
+        from .models.transformer import moe_layer_selector
+        return sum(moe_layer_selector(i, self.num_layers,
+                                      getattr(self, "moe_pattern", "all"))
+                   for i in range(self.num_layers))
+
     def estimate_active_params(self) -> int:
         h, L, V = self.hidden_size, self.num_layers, self.vocab_size
         inter = self.intermediate_size
@@ -241,10 +251,10 @@ class Config:
         head_dim = h // self.num_heads
         attn = h * h + 2 * h * kv * head_dim + h * h  # q,k,v,o
         ffn = 3 * h * inter  # gate, up, down
-        if self.use_moe:
-            ffn = ffn * self.moe_top_k
+        n_moe = self._n_moe_layers()
+        body = L * attn + n_moe * ffn * self.moe_top_k             + (L - n_moe) * ffn
         emb = V * h * (1 if self.tie_word_embeddings else 2)
-        return int(L * (attn + ffn) + emb)
+        return int(body + emb)
 
     def estimate_total_params(self) -> int:
         h, L, V = self.hidden_size, self.num_layers, self.vocab_size
@@ -253,10 +263,10 @@ class Config:
         head_dim = h // self.num_heads
         attn = h * h + 2 * h * kv * head_dim + h * h
         ffn = 3 * h * inter
-        if self.use_moe:
-            ffn = ffn * self.num_experts
+        n_moe = self._n_moe_layers()
+        body = L * attn + n_moe * (ffn * self.num_experts + h * self.num_experts)             + (L - n_moe) * ffn
         emb = V * h * (1 if self.tie_word_embeddings else 2)
-        return int(L * (attn + ffn) + emb)
+        return int(body + emb)
 
     def estimate_memory_gb(self) -> float:
         """Per-GPU training memory estimate (weights+grads+optimizer, bf16 compute)."""

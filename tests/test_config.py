@@ -78,3 +78,19 @@ def test_largest_divisor_helper():
     assert largest_divisor_leq(8, 5) == 4
     assert largest_divisor_leq(12, 5) == 4
     assert largest_divisor_leq(7, 3) == 1
+
+
+def test_estimator_respects_moe_pattern():
+    """hybrid_70b (MoE every 2nd layer): the estimator must count only
+    the MoE layers' expert stacks (round-2 fix: it assumed pattern=all
+    and over-counted hybrid_70b ~2x)."""
+    import torch
+    from luminaai_amd.config import ConfigPresets
+    from luminaai_amd.models import (DeepSeekTransformer,
+                                     config_to_deepseek_config)
+    cfg = ConfigPresets.hybrid_70b()
+    with torch.device("meta"):
+        m = DeepSeekTransformer(config_to_deepseek_config(cfg))
+    real = sum(p.numel() for p in m.parameters())
+    est = cfg.estimate_total_params()
+    assert abs(est - real) / real < 0.01, (est, real)
