@@ -122,13 +122,13 @@ DEV_INLINE float shfl_xor32(float v) { return __shfl_xor(v, 32, 64); }
 template <int DP, int NCH, int NT = ATTN_THREADS>
 struct Stage {
   ushortx8 v[NCH];
-  // per-thread chunk coordinates and image offsets are threadIdx-only:
-  // computed once, reused every step (the div-by-DP/8 and image swizzles
-  // were a measurable VALU cost when recomputed per tile)
-  int rowoff[NCH];     // row * DP  (global element offset of the row)
+  // per-thread image offsets are threadIdx-only: computed once, reused
+  // every step (the image swizzles were a measurable VALU cost when
+  // recomputed per tile).  row/rowoff are re-derived in load() from
+  // roff's row field to keep the struct at 2 ints/chunk (the kernels
+  // sit within a register of the 2-waves/SIMD budget).
   int roff[NCH];       // row-image byte offset
   int boff[NCH];       // blocked-image element offset
-  int row_[NCH];
   DEV_INLINE void init() {
     const int t = threadIdx.x;
     #pragma unroll
@@ -136,8 +136,6 @@ struct Stage {
       const int c = t + i * NT;
       const int row = c < TS * DP / 8 ? c / (DP / 8) : 0;
       const int col8 = c - row * (DP / 8);
-      row_[i] = row;
-      rowoff[i] = row * DP + col8 * 8;
       roff[i] = rimg<DP>(row, col8 * 16);
       boff[i] = blk_off<DP>(row, col8 * 8);
     }
@@ -147,8 +145,11 @@ struct Stage {
     for (int i = 0; i < NCH; ++i) {
       const int c = threadIdx.x + i * NT;
       if (c < TS * DP / 8) {
-        const int over = row0 + row_[i] - (S - 1);   // clamp row to S-1
-        const int off = rowoff[i] - (over > 0 ? over * DP : 0);
+        const int row = roff[i] / (DP * 2);          // rimg row field
+        const int col8 = (c - row * (DP / 8)) * 8;
+        const int over = row0 + row - (S - 1);       // clamp row to S-1
+        const int off = row * DP + col8
+                        - (over > 0 ? over * DP : 0);
         v[i] = *reinterpret_cast<const ushortx8*>(base + (int64_t)row0 * DP
                                                   + off);
       }
@@ -616,6 +617,217 @@ void attn_bwd_dkdv_kernel(const uint16_t* __restrict__ Q,
 }
 
 // ===========================================================================
+// dK/dV, 8-wave split-duty variant: 512 threads at 2 waves/SIMD.  The
+// 4-wave kernel above holds dK AND dV accumulators (160 VGPRs) plus
+// resident K/V fragments per wave -> full 512-register file, occupancy 1,
+// ~13% MFMA busy (profiles/r02_SUMMARY.md).  Here wave pair (g, g+4)
+// shares kv rows [g*TS, g*TS+TS): wave g accumulates dV only, wave g+4
+// dK only (80 accumulator VGPRs each), and K/V live in LDS blocked
+// images read per-MFMA with brfrag.  The S tile is computed by BOTH
+// waves of a pair (+25% MFMA issue), paid for by 2x occupancy on a pipe
+// that was 87% idle (guide: Two waves per SIMD, item 1).
+// ===========================================================================
+template <int DP>
+__global__ __launch_bounds__(512, 2)
+void attn_bwd_dkdv8_kernel(const uint16_t* __restrict__ Q,
+                           const uint16_t* __restrict__ K,
+                           const uint16_t* __restrict__ V,
+                           const uint16_t* __restrict__ dO,
+                           const float* __restrict__ LSE2,
+                           const float* __restrict__ Delta,
+                           uint16_t* __restrict__ dK,
+                           uint16_t* __restrict__ dV,
+                           int B, int H, int HKV, int S, int SP,
+                           float scale) {
+  constexpr int NT = 512;
+  constexpr int NCH = (TS * DP / 8 + NT - 1) / NT;
+  constexpr int DT = DP / 32;
+  constexpr int KS = DP / 16;
+  __shared__ uint16_t lsQt[2][BLK_ELEMS(DP)];
+  __shared__ uint16_t lsOt[2][BLK_ELEMS(DP)];
+  __shared__ uint16_t lsK[4][BLK_ELEMS(DP)];   // resident kv block
+  __shared__ uint16_t lsV[4][BLK_ELEMS(DP)];
+  __shared__ float    lsL[2][TS];
+  __shared__ float    lsD[2][TS];
+
+  const int hkv  = blockIdx.y;
+  const int b    = blockIdx.z;
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int g    = wave & 3;                // kv row group of the pair
+  const bool kduty = wave >= 4;             // dK duty; waves 0-3 do dV
+  const int G    = H / HKV;
+  const int nkb  = (S + KVB - 1) / KVB;
+  #pragma unroll 1
+  for (int pass = 0; pass < 2; ++pass) {
+  const int kb = pass == 0 ? blockIdx.x : nkb - 1 - blockIdx.x;
+  if (pass == 1 && kb <= (int)blockIdx.x) break;
+  if (pass == 1) __syncthreads();           // LDS reuse fence
+  const int kv0w = kb * KVB + g * TS;
+
+  const uint16_t* Kp = K + ((int64_t)b * HKV + hkv) * S * DP;
+  const uint16_t* Vp = V + ((int64_t)b * HKV + hkv) * S * DP;
+  uint16_t* dKp = dK + ((int64_t)b * HKV + hkv) * S * DP;
+  uint16_t* dVp = dV + ((int64_t)b * HKV + hkv) * S * DP;
+
+  const float c2 = scale * LOG2E;
+
+  // ---- stage the kv block's K/V into blocked LDS images (whole block)
+  {
+    constexpr int CH = KVB * DP / 8;
+    #pragma unroll
+    for (int i = 0; i < (CH + NT - 1) / NT; ++i) {
+      const int c = threadIdx.x + i * NT;
+      if (c < CH) {
+        const int row = c / (DP / 8);
+        const int col8 = (c - row * (DP / 8)) * 8;
+        const int gr = kb * KVB + row;
+        const int64_t gc = gr >= S ? S - 1 : gr;
+        const int off = blk_off<DP>(row & 31, col8);
+        *reinterpret_cast<ushortx8*>(lsK[row >> 5] + off) =
+            *reinterpret_cast<const ushortx8*>(Kp + gc * DP + col8);
+        *reinterpret_cast<ushortx8*>(lsV[row >> 5] + off) =
+            *reinterpret_cast<const ushortx8*>(Vp + gc * DP + col8);
+      }
+    }
+  }
+
+  f32x16 acc[DT] = {};                      // dV (waves 0-3) or dK (4-7)
+
+  const int tq0 = kb * (KVB / TS);
+  const int ntq = (S + TS - 1) / TS - tq0;
+  const int total = ntq * G;
+
+  auto stage_load = [&](Stage<DP, NCH, NT>& sq, Stage<DP, NCH, NT>& so,
+                        float lse[1], float del[1], int step) {
+    const int gi = step / ntq;
+    const int qt = tq0 + (step - gi * ntq);
+    const int h = hkv * G + gi;
+    const uint16_t* Qp = Q + ((int64_t)b * H + h) * S * DP;
+    const uint16_t* Op = dO + ((int64_t)b * H + h) * S * DP;
+    sq.load(Qp, qt * TS, S);
+    so.load(Op, qt * TS, S);
+    if (threadIdx.x < TS) {
+      const float* Lp = LSE2 + ((int64_t)b * H + h) * SP + qt * TS;
+      lse[0] = Lp[threadIdx.x];
+    } else if (threadIdx.x < 2 * TS) {
+      const float* Dp = Delta + ((int64_t)b * H + h) * SP + qt * TS;
+      del[0] = Dp[threadIdx.x - TS];
+    }
+  };
+  Stage<DP, NCH, NT> sq, so;
+  sq.init();
+  so.init();
+  float lse1[1], del1[1];
+  stage_load(sq, so, lse1, del1, 0);
+  sq.write_blk(lsQt[0]);
+  so.write_blk(lsOt[0]);
+  if (threadIdx.x < TS) lsL[0][threadIdx.x] = lse1[0];
+  else if (threadIdx.x < 2 * TS) lsD[0][threadIdx.x - TS] = del1[0];
+  __syncthreads();
+
+  for (int step = 0; step < total; ++step) {
+    const int buf = step & 1;
+    const int gi = step / ntq;
+    const int qt = tq0 + (step - gi * ntq);
+    if (step + 1 < total)
+      stage_load(sq, so, lse1, del1, step + 1);
+
+    const bool active = qt * TS + TS - 1 >= kv0w;
+    if (active) {
+      // ---- S tile (both duty waves): rows(regs) = q, cols(lanes) = kv
+      f32x16 st = {};
+      __builtin_amdgcn_s_setprio(1);
+      #pragma unroll
+      for (int ks = 0; ks < KS; ++ks)
+        st = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+            brfrag<DP>(lsQt[buf], lane, ks), brfrag<DP>(lsK[g], lane, ks),
+            st, 0, 0, 0);
+      __builtin_amdgcn_s_setprio(0);
+
+      const int kvg = kv0w + (lane & 31);
+      const bool interior = (qt * TS >= kv0w + TS) && (qt * TS + TS <= S);
+      float p[16];
+      #pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const float lse = lsL[buf][crow(r, lane >> 5)];
+        float pr = __builtin_exp2f(fmaf(st[r], c2, -lse));
+        if (!interior) {
+          const int qg = qt * TS + crow(r, lane >> 5);
+          pr = (kvg <= qg && qg < S) ? pr : 0.0f;
+        }
+        p[r] = pr;
+      }
+
+      if (!kduty) {
+        // ---- dV += P^T dO
+        bf16x8 pf[2];
+        pack_frags(p, pf);
+        __builtin_amdgcn_s_setprio(1);
+        #pragma unroll
+        for (int n = 0; n < DT; ++n)
+          #pragma unroll
+          for (int sfrag = 0; sfrag < 2; ++sfrag)
+            acc[n] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+                pf[sfrag], tfrag<DP>(lsOt[buf], lane, n, sfrag), acc[n],
+                0, 0, 0);
+        __builtin_amdgcn_s_setprio(0);
+      } else {
+        // ---- dP tile, then dK += dS^T Q
+        f32x16 dp = {};
+        __builtin_amdgcn_s_setprio(1);
+        #pragma unroll
+        for (int ks = 0; ks < KS; ++ks)
+          dp = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+              brfrag<DP>(lsOt[buf], lane, ks), brfrag<DP>(lsV[g], lane, ks),
+              dp, 0, 0, 0);
+        __builtin_amdgcn_s_setprio(0);
+        float ds[16];
+        #pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          const float dlt = lsD[buf][crow(r, lane >> 5)];
+          ds[r] = p[r] * (dp[r] - dlt) * scale;
+        }
+        bf16x8 df[2];
+        pack_frags(ds, df);
+        __builtin_amdgcn_s_setprio(1);
+        #pragma unroll
+        for (int n = 0; n < DT; ++n)
+          #pragma unroll
+          for (int sfrag = 0; sfrag < 2; ++sfrag)
+            acc[n] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+                df[sfrag], tfrag<DP>(lsQt[buf], lane, n, sfrag), acc[n],
+                0, 0, 0);
+        __builtin_amdgcn_s_setprio(0);
+      }
+    }
+
+    if (step + 1 < total) {
+      const int bi = buf ^ 1;
+      sq.write_blk(lsQt[bi]);
+      so.write_blk(lsOt[bi]);
+      if (threadIdx.x < TS) lsL[bi][threadIdx.x] = lse1[0];
+      else if (threadIdx.x < 2 * TS) lsD[bi][threadIdx.x - TS] = del1[0];
+    }
+    __syncthreads();
+  }
+
+  // ---- epilogue: C rows = kv (crow regs), cols = d (lanes)
+  uint16_t* outp = kduty ? dKp : dVp;
+  #pragma unroll
+  for (int r = 0; r < 16; ++r) {
+    const int row = kv0w + crow(r, lane >> 5);
+    if (row < S) {
+      #pragma unroll
+      for (int n = 0; n < DT; ++n)
+        outp[(int64_t)row * DP + n * 32 + (lane & 31)] =
+            f32_to_bf16(acc[n][r]);
+    }
+  }
+  }  // pass loop
+}
+
+// ===========================================================================
 // dQ: 4 waves, wave owns 32 q rows of a 128-row q block; iterates kv tiles.
 // S^T orientation (lse/delta lane-local).
 // ===========================================================================
@@ -652,7 +864,10 @@ void attn_bwd_dq_kernel(const uint16_t* __restrict__ Q,
   const float c2 = scale * LOG2E;
 
   // resident B-fragments: Q^T (pre-scaled by c2) and dO^T; lane-local
-  // lse2/delta (q = l&31 of this wave's rows)
+  // lse2/delta (q = l&31 of this wave's rows).  (An LDS-resident variant
+  // measured 211 vs 217 TF: the 80 freed VGPRs don't buy occupancy here
+  // -- the q-block images push LDS past the 2-block budget -- so the
+  // per-MFMA LDS re-reads are pure cost.  Kept register-resident.)
   bf16x8 qtf[KS], otf[KS];
   float lse, dlt;
   {
@@ -806,13 +1021,22 @@ extern "C" hipError_t lumina_attn_delta(const void* dO, const void* O,
 
 #define INSTANT_DP(DPV)                                                      \
   if (DP == DPV) {                                                           \
-    hipLaunchKernelGGL(attn_bwd_dkdv_kernel<DPV>, gridkv,                    \
-                       dim3(ATTN_THREADS), 0, stream,                        \
-                       (const uint16_t*)Q, (const uint16_t*)K,               \
-                       (const uint16_t*)V, (const uint16_t*)dO,              \
-                       (const float*)LSE2, (const float*)Delta,              \
-                       (uint16_t*)dK, (uint16_t*)dV, B, H, HKV, S, SP,       \
-                       scale);                                               \
+    if (dkdv4)                                                               \
+      hipLaunchKernelGGL(attn_bwd_dkdv_kernel<DPV>, gridkv,                  \
+                         dim3(ATTN_THREADS), 0, stream,                      \
+                         (const uint16_t*)Q, (const uint16_t*)K,             \
+                         (const uint16_t*)V, (const uint16_t*)dO,            \
+                         (const float*)LSE2, (const float*)Delta,            \
+                         (uint16_t*)dK, (uint16_t*)dV, B, H, HKV, S, SP,     \
+                         scale);                                             \
+    else                                                                     \
+      hipLaunchKernelGGL(attn_bwd_dkdv8_kernel<DPV>, gridkv,                 \
+                         dim3(512), 0, stream,                               \
+                         (const uint16_t*)Q, (const uint16_t*)K,             \
+                         (const uint16_t*)V, (const uint16_t*)dO,            \
+                         (const float*)LSE2, (const float*)Delta,            \
+                         (uint16_t*)dK, (uint16_t*)dV, B, H, HKV, S, SP,     \
+                         scale);                                             \
     hipError_t e = hipGetLastError();                                        \
     if (e != hipSuccess) return e;                                           \
     hipLaunchKernelGGL(attn_bwd_dq_kernel<DPV>, gridq, dim3(ATTN_THREADS),   \
@@ -832,6 +1056,8 @@ extern "C" hipError_t lumina_attn_bwd(const void* Q, const void* K,
                                       hipStream_t stream) {
   const int nkb = (S + KVB - 1) / KVB;
   dim3 gridkv((nkb + 1) / 2, HKV, B);
+  // LUMINA_DKDV4=1 forces the 4-wave kernel (A/B knob)
+  static const bool dkdv4 = getenv("LUMINA_DKDV4") != nullptr;
   dim3 gridq((S + QB - 1) / QB, H, B);
   INSTANT_DP(64)
   INSTANT_DP(128)
