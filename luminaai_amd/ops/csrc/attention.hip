@@ -832,7 +832,7 @@ void attn_bwd_dkdv8_kernel(const uint16_t* __restrict__ Q,
 // S^T orientation (lse/delta lane-local).
 // ===========================================================================
 template <int DP>
-__global__ __launch_bounds__(ATTN_THREADS, 1)
+__global__ __launch_bounds__(ATTN_THREADS, 2)
 void attn_bwd_dq_kernel(const uint16_t* __restrict__ Q,
                         const uint16_t* __restrict__ K,
                         const uint16_t* __restrict__ V,
@@ -846,6 +846,15 @@ void attn_bwd_dq_kernel(const uint16_t* __restrict__ Q,
   constexpr int KS = DP / 16;
   __shared__ uint16_t lsKt[2][BLK_ELEMS(DP)];  // K blocked image (raw)
   __shared__ uint16_t lsV[2][TS * DP];         // V row image
+  // upper-half dO^T fragments live in LDS for wide heads: the fully
+  // register-resident kernel sits at 268 regs (occupancy 1); parking
+  // KS-OKS fragments here brings it under the 2-waves/SIMD budget
+  constexpr int OKS = DP >= 128 ? KS / 2 : KS;   // register-resident count
+  // image stride padded to a 32-element multiple: rimg's 4-slot swizzle
+  // windows must not straddle rows (DP/2 = 80 would overflow slot 8..9
+  // into the next row -- the round-2 dq numerics bug)
+  constexpr int DPH = ((DP / 2) + 31) & ~31;
+  __shared__ uint16_t lsOh[4][OKS < KS ? TS * DPH : 1];
 
   const int qb   = gridDim.x - 1 - blockIdx.x;
   const int h    = blockIdx.y;
@@ -868,7 +877,7 @@ void attn_bwd_dq_kernel(const uint16_t* __restrict__ Q,
   // measured 211 vs 217 TF: the 80 freed VGPRs don't buy occupancy here
   // -- the q-block images push LDS past the 2-block budget -- so the
   // per-MFMA LDS re-reads are pure cost.  Kept register-resident.)
-  bf16x8 qtf[KS], otf[KS];
+  bf16x8 qtf[KS], otf[OKS];
   float lse, dlt;
   {
     const int qg0 = q0w + (lane & 31);
@@ -884,8 +893,16 @@ void attn_bwd_dq_kernel(const uint16_t* __restrict__ Q,
         f[j] = __builtin_bit_cast(__bf16, u);
       }
       qtf[ks] = f;
-      otf[ks] = *reinterpret_cast<const bf16x8*>(
+      const ushortx8 ov = *reinterpret_cast<const ushortx8*>(
           Op + (int64_t)qg * DP + 16 * ks + 8 * (lane >> 5));
+      if (ks < OKS) {
+        otf[ks] = __builtin_bit_cast(bf16x8, ov);
+      } else {
+        *reinterpret_cast<ushortx8*>(
+            reinterpret_cast<char*>(lsOh[wave])
+            + rimg<DPH>(lane & 31, (ks - OKS) * 32 + 16 * (lane >> 5))) =
+            ov;
+      }
     }
     const int64_t bh = (int64_t)blockIdx.z * H + h;
     lse = LSE2[bh * SP + (q0w + (lane & 31) < SP ? q0w + (lane & 31) : SP - 1)];
@@ -926,7 +943,10 @@ void attn_bwd_dq_kernel(const uint16_t* __restrict__ Q,
       #pragma unroll
       for (int ks = 0; ks < KS; ++ks)
         dp = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
-            rfrag<DP>(lsV[buf], lane, ks), otf[ks], dp, 0, 0, 0);
+            rfrag<DP>(lsV[buf], lane, ks),
+            ks < OKS ? otf[ks < OKS ? ks : 0]
+                     : rfrag<DPH>(lsOh[wave], lane, ks - OKS),
+            dp, 0, 0, 0);
       __builtin_amdgcn_s_setprio(0);
 
       const int qg = q0w + (lane & 31);
