@@ -246,6 +246,17 @@ class BatchedLinearFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, w):
         ctx.save_for_backward(x, w)
+        import os as _os
+        # A/B knob only: the NN-form 256^2 kernel measured 36.3k vs
+        # hipBLASLt's 39.2k tok/s on the b1 step (the tr16-read B path
+        # loses to the library's NN form at these shapes) -- default off
+        if _os.environ.get("LUMINA_GG8P_FWD") and use_hip(x) \
+                and x.dtype == torch.bfloat16 and w.shape[2] % 8 == 0 \
+                and x.shape[2] == w.shape[1]:
+            K = x.shape[2]
+            xp = x if K % 32 == 0 else \
+                torch.nn.functional.pad(x, (0, 32 - K % 32))
+            return get_ext().gg8p_nn(xp.contiguous(), w.contiguous())
         return torch.bmm(x, w)
 
     @staticmethod
