@@ -475,10 +475,9 @@ class SwiGLUExpert(nn.Module):
             x = tp_copy(x, self.tp_group)
         gu = _linear(self.gate_up_proj, x)
         shp = gu.shape[:-1]
-        gu2 = gu.view(-1, 2 * self.intermediate_size)
-        gate = gu2.narrow(1, 0, self.intermediate_size)
-        up = gu2.narrow(1, self.intermediate_size, self.intermediate_size)
-        act = ops.swiglu(gate, up).view(*shp, self.intermediate_size)
+        act = ops.swiglu_fused(
+            gu.view(-1, 2 * self.intermediate_size)
+        ).view(*shp, self.intermediate_size)
         y = _linear(self.down_proj, act)
         if self.tp_group is not None:
             from ..parallel.tensor_parallel import tp_reduce
@@ -599,8 +598,7 @@ class MoEFFNLayer(nn.Module):
 
         def _mlp(z):
             gu = _ebmm(z, w_gu)
-            gu2 = gu.reshape(-1, 2 * I)
-            act = ops.swiglu(gu2.narrow(1, 0, I), gu2.narrow(1, I, I))
+            act = ops.swiglu_fused(gu.reshape(-1, 2 * I))
             out = _ebmm(act.view(EL, -1, I), w_dn)
             if tp_group is not None:
                 # TP over experts: w_gate_up column-sharded, w_down

@@ -12,12 +12,12 @@ from .interface import (
     l2norm_sq,
     rmsnorm,
     rope,
-    swiglu,
+    swiglu, swiglu_fused,
 )
 from .reference import rope_cache
 
 __all__ = [
     "adamw_step", "can_flash_attention", "flash_attention",
     "fused_cross_entropy", "get_ext", "has_ext", "l2norm_sq",
-    "reference", "rmsnorm", "rope", "rope_cache", "swiglu",
+    "reference", "rmsnorm", "rope", "rope_cache", "swiglu", "swiglu_fused",
 ]

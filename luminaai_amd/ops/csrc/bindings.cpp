@@ -203,6 +203,27 @@ std::tuple<at::Tensor, at::Tensor> swiglu_bwd(const at::Tensor& dy,
   return {dg, du};
 }
 
+at::Tensor swiglu_bwd_fused(const at::Tensor& dy, const at::Tensor& gu) {
+  // dy [rows, I], gu [rows, 2I] contiguous -> dgu [rows, 2I].  Taking the
+  // FUSED projection output directly keeps autograd away from narrow()
+  // slices: the two narrow-backwards cost a zero-fill + two slice copies
+  // + a buffer add on [rows, 2I] (measured 6% of the b1 training step).
+  TORCH_CHECK(dy.is_contiguous() && gu.is_contiguous());
+  TORCH_CHECK(gu.dim() == 2 && dy.dim() == 2);
+  const int64_t rows = gu.size(0);
+  const int I = (int)(gu.size(1) / 2);
+  TORCH_CHECK(dy.size(0) == rows && dy.size(1) == I);
+  auto dgu = at::empty_like(gu);
+  auto fn = is_bf16(gu) ? lumina_swiglu_bwd_bf16 : lumina_swiglu_bwd_f32;
+  const char* g = (const char*)gu.data_ptr();
+  char* dg = (char*)dgu.data_ptr();
+  const int64_t ib = (int64_t)I * gu.element_size();
+  check_hip(fn(dy.data_ptr(), g, g + ib, dg, dg + ib, rows, I,
+               2 * (int64_t)I, 2 * (int64_t)I, cur_stream()),
+            "swiglu_bwd_fused");
+  return dgu;
+}
+
 // ---- fused CE ------------------------------------------------------------
 std::tuple<at::Tensor, at::Tensor> ce_fwd(const at::Tensor& logits,
                                           const at::Tensor& labels,
@@ -668,6 +689,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
           "decode RoPE + KV-cache append at the device cursor");
   mod.def("dec_attn", &dec_attn, "single-token GQA attention over the cache");
   mod.def("dec_advance", &dec_advance, "advance the decode cursor");
+  mod.def("swiglu_bwd_fused", &swiglu_bwd_fused,
+          "swiglu backward into one fused [rows, 2I] grad buffer");
   mod.def("dec_rmsnorm", &dec_rmsnorm, "decode rmsnorm -> xhat buffer");
   mod.def("dec_topk", &dec_topk,
           "decode router: softmax/temp top-k, renormalized, on device");

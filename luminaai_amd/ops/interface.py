@@ -178,6 +178,36 @@ class SwiGLUFn(torch.autograd.Function):
         return dg, du
 
 
+class SwiGLUFusedFn(torch.autograd.Function):
+    """y = silu(gu[:, :I]) * gu[:, I:] on the FUSED [M, 2I] projection
+    output.  Compared with swiglu() on two narrow() views, autograd never
+    sees the slices: backward writes one [M, 2I] grad buffer directly
+    instead of narrow-backward's zero-fill + slice copies + add (measured
+    ~6% of the b1 training step in eager glue kernels)."""
+
+    @staticmethod
+    def forward(ctx, gu2):
+        I = gu2.shape[1] // 2
+        ctx.save_for_backward(gu2)
+        return get_ext().swiglu_fwd(gu2.narrow(1, 0, I), gu2.narrow(1, I, I))
+
+    @staticmethod
+    def backward(ctx, gy):
+        (gu2,) = ctx.saved_tensors
+        return get_ext().swiglu_bwd_fused(gy.contiguous(), gu2)
+
+
+def swiglu_fused(gu2: torch.Tensor) -> torch.Tensor:
+    """SwiGLU over the fused gate_up output [..., 2I] -> [..., I]."""
+    I = gu2.shape[-1] // 2
+    flat = gu2.reshape(-1, 2 * I)
+    if use_hip(flat) and flat.is_contiguous():
+        y = SwiGLUFusedFn.apply(flat)
+    else:
+        y = SwiGLUFn.apply(flat.narrow(1, 0, I), flat.narrow(1, I, I))
+    return y.view(*gu2.shape[:-1], I)
+
+
 def swiglu(gate: torch.Tensor, up: torch.Tensor) -> torch.Tensor:
     shape = gate.shape
     g2 = gate.reshape(-1, shape[-1]) if gate.dim() != 2 else gate

@@ -723,6 +723,26 @@ def test_fused_decoder_matches_eager():
     assert got2 == got, (got2, got)
 
 
+def test_swiglu_fused_matches_autograd():
+    """Fused-input SwiGLU (one [M,2I] tensor) vs fp32 autograd reference."""
+    torch.manual_seed(5)
+    M, I = 512, 256
+    gu = (torch.randn(M, 2 * I, device=_dev(), dtype=torch.bfloat16)
+          .requires_grad_())
+    from luminaai_amd.ops.interface import swiglu_fused
+    y = swiglu_fused(gu)
+    gy = torch.randn_like(y)
+    y.backward(gy)
+
+    ref = gu.detach().float().clone().requires_grad_()
+    g, u = ref[:, :I], ref[:, I:]
+    yr = torch.nn.functional.silu(g) * u
+    yr.backward(gy.float())
+    torch.testing.assert_close(y.float(), yr, rtol=2e-2, atol=2e-2)
+    torch.testing.assert_close(gu.grad.float(), ref.grad, rtol=3e-2,
+                               atol=3e-2)
+
+
 def test_fused_decoder_moe_matches_eager():
     """MoE fused decode (on-device router top-k + expert-indirect GEMVs)
     matches the eager KV-cached MoE model's greedy stream.  Router runs
