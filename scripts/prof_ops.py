@@ -14,6 +14,7 @@ cfg = ConfigPresets.get("b1_moe")
 cfg.micro_batch_size = 16
 cfg.gradient_accumulation_steps = 2
 cfg.num_workers = 0
+cfg.gradient_checkpointing = False   # match bench.py (preset default is True)
 cfg.eval_every_n_batches = 0
 cfg.save_every_n_batches = 0
 device = torch.device("cuda")
