@@ -21,7 +21,8 @@ ARCH = os.environ.get("PYTORCH_ROCM_ARCH", "gfx950")
 
 KERNELS = ["rmsnorm.hip", "rope.hip", "swiglu.hip", "ce_loss.hip", "optim.hip",
            "grouped_gemm.hip", "moe.hip", "gemv.hip", "attention.hip",
-           "gemm8p.hip", "mxfp8.hip", "decode.hip"]
+           "gemm8p.hip",
+    "gemm8t.hip", "mxfp8.hip", "decode.hip"]
 
 
 def _newer(a: Path, b: Path) -> bool:

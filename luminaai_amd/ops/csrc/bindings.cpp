@@ -62,6 +62,7 @@ hipError_t lumina_dec_gemv(const void*, const void*, const void*, const void*, v
 hipError_t lumina_dec_rope_cache(const void*, void*, void*, void*, const float*, const float*, const int*, int, int, int, hipStream_t);
 hipError_t lumina_dec_attn(const void*, const void*, const void*, void*, const int*, int, int, int, int, float, hipStream_t);
 hipError_t lumina_dec_advance(int*, hipStream_t);
+hipError_t lumina_gg8t_nt(const void*, const void*, void*, int, int, int, int, int64_t, int64_t, int64_t, int, hipStream_t);
 hipError_t lumina_dec_rmsnorm(const void*, const void*, void*, int, float, hipStream_t);
 hipError_t lumina_dec_topk(const void*, int*, float*, int, int, float, hipStream_t);
 hipError_t lumina_dec_gemv_moe(const void*, const void*, const void*, void*, const int*, const float*, int, int64_t, int, int, int, hipStream_t);
@@ -496,6 +497,21 @@ std::tuple<at::Tensor, at::Tensor, at::Tensor> attn_bwd(
 // nt: out[e] = A[e] @ B[e]^T, A [E,M,K], B [E,N,K]; K % 32 == 0
 // nn: out[e] = A[e] @ B[e],   A [E,M,K], B [E,Kb,N]; K % 32, Kb <= K
 //     (A zero-padded along K up to a multiple of 32; B rows clamped)
+at::Tensor gg8t_nt(const at::Tensor& A, const at::Tensor& B, int64_t pf2) {
+  TORCH_CHECK(A.is_contiguous() && B.is_contiguous());
+  TORCH_CHECK(is_bf16(A) && is_bf16(B));
+  TORCH_CHECK(A.dim() == 3 && B.dim() == 3 && A.size(2) == B.size(2));
+  TORCH_CHECK(A.size(2) % 64 == 0, "gg8t requires K % 64 == 0");
+  const int E = (int)A.size(0), M = (int)A.size(1);
+  const int K = (int)A.size(2), N = (int)B.size(1);
+  auto O = at::empty({E, M, N}, A.options());
+  check_hip(lumina_gg8t_nt(A.data_ptr(), B.data_ptr(), O.data_ptr(), E, M,
+                           N, K, (int64_t)M * K, (int64_t)N * K,
+                           (int64_t)M * N, (int)pf2, cur_stream()),
+            "gg8t_nt");
+  return O;
+}
+
 at::Tensor gg8p_nt(const at::Tensor& A, const at::Tensor& B) {
   TORCH_CHECK(A.is_contiguous() && B.is_contiguous());
   TORCH_CHECK(is_bf16(A) && is_bf16(B));
@@ -702,6 +718,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
           "columnwise e8m0 fp8 quantization + transpose (gfx950)");
   mod.def("gg_mx_nt", &gg_mx_nt,
           "grouped MX-fp8 GEMM A.B^T at the ~5PF fp8 MFMA rate (gfx950)");
+  mod.def("gg8t_nt", &gg8t_nt,
+          "8-phase-template grouped NT GEMM (256^2, BK64, 16x16x32)");
   mod.def("gg8p_nt", &gg8p_nt,
           "256^2 pipelined grouped GEMM, A.B^T (gfx950)");
   mod.def("gg8p_nn", &gg8p_nn,
