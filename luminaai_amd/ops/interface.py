@@ -371,6 +371,33 @@ def flash_attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
     return o.transpose(1, 2)
 
 
+_PAD_CACHE: dict = {}
+
+
+def invalidate_pad_cache():
+    """Drop cached padded weights (call after in-place weight updates:
+    custom HIP optimizer kernels don't bump torch version counters)."""
+    _PAD_CACHE.clear()
+
+
+def _pad_k_cached(w: torch.Tensor, pad: int) -> torch.Tensor:
+    """Zero-pad the trailing (K) dim, cached per (id, _version): grad_x
+    re-pads the SAME weight tensor every micro-batch (62 calls/step on b1
+    -- each a fill + full copy of a ~150 MB tensor). The version counter
+    invalidates after the optimizer's in-place update; a weakref guards
+    id recycling (same pattern as ops/fp8.py's weight cache)."""
+    import weakref
+    key = id(w)
+    ent = _PAD_CACHE.get(key)
+    if ent is not None and ent[0]() is w and ent[1] == w._version:
+        return ent[2]
+    out = torch.nn.functional.pad(w, (0, pad))
+    if len(_PAD_CACHE) > 256:
+        _PAD_CACHE.clear()
+    _PAD_CACHE[key] = (weakref.ref(w), w._version, out)
+    return out
+
+
 def grouped_gemm_nt(a: torch.Tensor, b: torch.Tensor) -> torch.Tensor:
     """out[e] = a[e] @ b[e]^T with a [E,M,K], b [E,N,K] both row-major
     (contraction over the trailing dim). HIP MFMA kernel on GPU bf16;
@@ -385,7 +412,7 @@ def grouped_gemm_nt(a: torch.Tensor, b: torch.Tensor) -> torch.Tensor:
         if K % 64:
             pad = 64 - K % 64
             a = torch.nn.functional.pad(a, (0, pad))
-            b = torch.nn.functional.pad(b, (0, pad))
+            b = _pad_k_cached(b, pad)
         import os as _os
         if _os.environ.get("LUMINA_GG128"):     # A/B knob: 128^2 kernel
             return get_ext().grouped_gemm_nt(a.contiguous(), b.contiguous())

@@ -211,6 +211,8 @@ class Trainer:
         if self.precision.spec.name == "fp8":
             from ..ops import fp8 as _fp8
             _fp8.invalidate_weight_cache()
+        from ..ops import interface as _ops_if
+        _ops_if.invalidate_pad_cache()   # padded-K grad_x weight cache
         self._micro_in_cycle = 0
         self.global_step += 1
         self._maybe_expire_lr_override()
