@@ -103,6 +103,7 @@ class Config:
     use_deepspeed: bool = False
     zero_stage: int = 0
     cpu_offload: bool = False
+    fp8_alltoall: bool = False   # e4m3/e5m2 EP token exchange (halves a2a bytes)
     cpu_offload_optimizer: bool = False
     cpu_offload_parameters: bool = False
     aggressive_cpu_offload: bool = False

@@ -65,6 +65,7 @@ class DeepSeekConfig:
     load_balancing_weight: float = 0.01
     routing_temperature: float = 1.0
     routing_noise_std: float = 0.1
+    fp8_alltoall: bool = False     # e4m3/e5m2 EP token exchange
     moe_pattern: Union[str, Callable] = "all"
     dense_start_layers: int = 2
 
@@ -129,6 +130,7 @@ def config_to_deepseek_config(config) -> DeepSeekConfig:
         load_balancing_weight=config.load_balancing_weight,
         routing_temperature=getattr(config, "routing_temperature", 1.0),
         routing_noise_std=getattr(config, "routing_noise_std", 0.1),
+        fp8_alltoall=getattr(config, "fp8_alltoall", False),
         moe_pattern=getattr(config, "moe_pattern", "all"),
         use_mod=config.use_mod,
         mod_capacity_factor=getattr(config, "mod_capacity_factor", 0.5),
@@ -513,6 +515,7 @@ class MoEFFNLayer(nn.Module):
         self.load_balancing_weight = config.load_balancing_weight
         self.routing_temperature = config.routing_temperature
         self.routing_noise_std = config.routing_noise_std
+        self.fp8_alltoall = getattr(config, "fp8_alltoall", False)
         self.expert_dropout = 0.0
 
         # expert parallelism: local shard of the expert weights, token
@@ -617,7 +620,8 @@ class MoEFFNLayer(nn.Module):
             nch = 2 if C >= 2 and getattr(self, "overlap_alltoall", True) \
                 else 1
             y = expert_pipeline(bufv, _mlp, self.ep_group, self.ep_size,
-                                n_chunks=nch)
+                                n_chunks=nch,
+                                fp8=getattr(self, "fp8_alltoall", False))
         else:
             y = _mlp(bufv)
 

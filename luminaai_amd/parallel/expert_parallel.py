@@ -113,8 +113,76 @@ class _A2AWaitFn(torch.autograd.Function):
         return out, None, None
 
 
+# ---- fp8 token exchange (round-2 roadmap 4b) ------------------------------
+# Quantize the a2a payload to OCP fp8 with a per-token (row) fp32 scale:
+# activations travel as e4m3, gradients as e5m2 (the wider-exponent grad
+# format), halving the bytes on the per-link-bound xGMI rings.  The pair
+# mirrors _A2AStartFn/_A2AWaitFn's token registry; the tensor linking
+# Start->Wait in the autograd graph is a shape/dtype carrier only (its
+# values are never read -- the real payload+scale works live in the
+# registry).
+
+E4M3_MAX = 448.0
+E5M2_MAX = 57344.0
+
+
+def _quant_rows_comm(x, dtype, maxv):
+    amax = x.float().abs().amax(-1, keepdim=True).clamp_min(1e-12)
+    s = maxv / amax
+    q = (x.float() * s).clamp(-maxv, maxv).to(dtype)
+    return q, s.squeeze(-1).contiguous()
+
+
+def _dequant_rows_comm(q, s, out_dtype):
+    return (q.float() / s.unsqueeze(-1)).to(out_dtype)
+
+
+def _launch_fp8_a2a(x, group, dtype, maxv):
+    q, s = _quant_rows_comm(x, dtype, maxv)
+    po, pw = _launch_a2a(q.view(torch.uint8), group)
+    so, sw = _launch_a2a(s, group)
+    return (pw, sw, po, so, dtype)
+
+
+def _join_fp8_a2a(ent, out_dtype):
+    pw, sw, po, so, dtype = ent
+    pw.wait()
+    sw.wait()
+    return _dequant_rows_comm(po.view(dtype), so, out_dtype)
+
+
+class _Fp8A2AStartFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, group, token):
+        ctx.group = group
+        ctx.token = token
+        _A2A_WORKS[token] = _launch_fp8_a2a(x, group, torch.float8_e4m3fn,
+                                            E4M3_MAX)
+        return x.new_empty(x.shape)        # shape carrier (values unused)
+
+    @staticmethod
+    def backward(ctx, gy):
+        ent = _A2A_WORKS.pop(("g", ctx.token))
+        return _join_fp8_a2a(ent, gy.dtype), None, None
+
+
+class _Fp8A2AWaitFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, group, token):
+        ctx.group = group
+        ctx.token = token
+        ent = _A2A_WORKS.pop(token)
+        return _join_fp8_a2a(ent, x.dtype)
+
+    @staticmethod
+    def backward(ctx, gy):
+        _A2A_WORKS[("g", ctx.token)] = _launch_fp8_a2a(
+            gy, ctx.group, torch.float8_e5m2, E5M2_MAX)
+        return gy.new_empty(gy.shape), None, None
+
+
 def expert_pipeline(bufv: torch.Tensor, mlp_fn, group, ep: int,
-                    n_chunks: int = 2):
+                    n_chunks: int = 2, fp8: bool = False):
     """Dispatch-exchange -> expert MLP -> return-exchange with the capacity
     dim split into `n_chunks` so exchange and compute overlap.
 
@@ -131,20 +199,22 @@ def expert_pipeline(bufv: torch.Tensor, mlp_fn, group, ep: int,
         t = _A2A_NEXT[0]
         _A2A_NEXT[0] += 1
         tokens.append(t)
-    started = [_A2AStartFn.apply(c.contiguous(), group, t)
+    Start = _Fp8A2AStartFn if fp8 else _A2AStartFn
+    Wait = _Fp8A2AWaitFn if fp8 else _A2AWaitFn
+    started = [Start.apply(c.contiguous(), group, t)
                for c, t in zip(chunks, tokens)]
     rets, rtokens = [], []
     for s, t, c in zip(started, tokens, chunks):
         cc = c.shape[1]
-        z = _A2AWaitFn.apply(s, group, t)
+        z = Wait.apply(s, group, t)
         z = z.view(ep, EL, cc, h).transpose(0, 1).reshape(EL, ep * cc, h)
         y = mlp_fn(z)
         y = y.view(EL, ep, cc, h).transpose(0, 1).reshape(ep * EL, cc, h)
         rt = _A2A_NEXT[0]
         _A2A_NEXT[0] += 1
         rtokens.append(rt)
-        rets.append(_A2AStartFn.apply(y.contiguous(), group, rt))
-    outs = [_A2AWaitFn.apply(r, group, t) for r, t in zip(rets, rtokens)]
+        rets.append(Start.apply(y.contiguous(), group, rt))
+    outs = [Wait.apply(r, group, t) for r, t in zip(rets, rtokens)]
     return torch.cat(outs, dim=1)
 
 

@@ -1025,6 +1025,46 @@ def test_ep_with_checkpointing():
 
 
 # ---- expert add/prune under EP (round-2: SURVEY build plan 7.6) ------------
+def ep_fp8_a2a_worker(rank, world):
+    """EP(2) with fp8_alltoall: forward/backward equal the full-expert
+    model within e4m3 quantization tolerance (payload e4m3, grads e5m2)."""
+    from luminaai_amd.models import DeepSeekTransformer
+    from luminaai_amd.parallel.mesh import init_mesh, reset_mesh
+
+    mcfg = _moe_model_cfg()
+    reset_mesh()
+    torch.manual_seed(1234)
+    full = DeepSeekTransformer(mcfg)
+    init_mesh(world)
+    torch.manual_seed(1234)
+    ep = DeepSeekTransformer(mcfg)
+    with torch.no_grad():
+        for (n, p), (n2, q) in zip(ep.named_parameters(),
+                                   full.named_parameters()):
+            p.copy_(q if p.shape == q.shape
+                    else q[rank * 2:(rank + 1) * 2])
+    for l in ep.layers:
+        l.ffn.fp8_alltoall = True
+    torch.manual_seed(77)
+    ids_all = torch.randint(1, mcfg.vocab_size, (2, 2, mcfg.seq_length))
+    lf, _, _ = full(ids_all.reshape(4, mcfg.seq_length))
+    le, _, _ = ep(ids_all[rank])
+    ref = lf.view(2, 2, mcfg.seq_length, -1)[rank]
+    rel = (le - ref).abs().max() / ref.abs().max().clamp_min(1e-6)
+    # grads flow (e5m2 path) without error
+    le.float().pow(2).mean().backward()
+    g = ep.layers[0].ffn.w_gate_up.grad
+    reset_mesh()
+    return {"rel": float(rel), "grad_finite": bool(torch.isfinite(g).all())}
+
+
+def test_ep_fp8_alltoall():
+    res = _spawn("ep_fp8_a2a_worker")
+    for r in range(WORLD):
+        assert res[r]["rel"] < 0.05, res       # e4m3 rowwise quant band
+        assert res[r]["grad_finite"], res
+
+
 def tpep_worker(rank, world):
     """TP(2) x EP(2) composed mesh (world 4, dp 1): full Trainer steps.
     Ranks (e, t) hold expert shard e sliced by tp rank t; tp peers share a
