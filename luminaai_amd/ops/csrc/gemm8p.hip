@@ -28,6 +28,7 @@
 // M and N tails are handled by clamped staging + guarded epilogue.
 
 #include "common.h"
+#include <stdlib.h>
 
 typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
 typedef __attribute__((ext_vector_type(4))) float  f32x4;
@@ -86,7 +87,7 @@ void gg8p_kernel(const uint16_t* __restrict__ Aall,
                  const uint16_t* __restrict__ Ball,
                  uint16_t* __restrict__ Oall,
                  int M, int N, int K, int Kb,
-                 int64_t sA, int64_t sB, int64_t sO) {
+                 int64_t sA, int64_t sB, int64_t sO, int noremap) {
   // ONE __shared__ array only: a second __shared__ object makes hipcc emit
   // s_waitcnt vmcnt(0) before the first ds_read of every k-step of a glds
   // pipeline (guide Sec.5 'Three .s-level traps', a) -- that full drain was
@@ -110,8 +111,9 @@ void gg8p_kernel(const uint16_t* __restrict__ Aall,
   const int nwg = gridDim.x * gridDim.y;
   const int orig = blockIdx.x + blockIdx.y * gridDim.x;
   const int q = nwg >> 3, r = nwg & 7, xcd = orig & 7;
-  const int wgid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q)
-                   + (orig >> 3);
+  int wgid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q)
+             + (orig >> 3);
+  if (noremap) wgid = orig;        // A/B knob (8192^3 regression diag)
   const int tileM = (wgid % gridDim.x) * G8_BM;
   const int tileN = (wgid / gridDim.x) * G8_BN;
   const int t = threadIdx.x;
@@ -305,12 +307,18 @@ extern "C" void launch_gg8p(const void* A, const void* B, void* O,
                             int64_t sA, int64_t sB, int64_t sO,
                             int nn_form, hipStream_t stream) {
   dim3 grid((M + G8_BM - 1) / G8_BM, (N + G8_BN - 1) / G8_BN, E);
+  // round-2 diag: the XCD-contiguous remap LOSES at every probed size
+  // (8192^3: 743 vs 1040 TF; b1 grad_x shapes: 971 vs 1012) -- at these
+  // grids an XCD's contiguous chunk walks whole M-columns, thrashing its
+  // L2 on the A panel.  Linear order is the default; LUMINA_GG8P_REMAP=1
+  // re-enables the remap for A/B.
+  static const int noremap = getenv("LUMINA_GG8P_REMAP") == nullptr;
   if (nn_form)
     hipLaunchKernelGGL(gg8p_kernel<true>, grid, dim3(G8_THREADS), 0, stream,
                        (const uint16_t*)A, (const uint16_t*)B, (uint16_t*)O,
-                       M, N, K, Kb, sA, sB, sO);
+                       M, N, K, Kb, sA, sB, sO, noremap);
   else
     hipLaunchKernelGGL(gg8p_kernel<false>, grid, dim3(G8_THREADS), 0, stream,
                        (const uint16_t*)A, (const uint16_t*)B, (uint16_t*)O,
-                       M, N, K, Kb, sA, sB, sO);
+                       M, N, K, Kb, sA, sB, sO, noremap);
 }
