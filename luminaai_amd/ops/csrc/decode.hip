@@ -61,42 +61,73 @@ void dec_gemv_kernel(const uint16_t* __restrict__ W,
   }
   __syncthreads();
 
-  const int row = blockIdx.x * 4 + (threadIdx.x >> 6);
+  // two rows per wave, interleaved in one k-loop (2x memory-level
+  // parallelism per wave; measured NEUTRAL vs one row/wave at b1 decode
+  // shapes -- the streams are bandwidth-saturated -- kept for the halved
+  // launch grid)
+  const int row = blockIdx.x * 8 + (threadIdx.x >> 6) * 2;
   if (row >= N) return;
   const int lane = threadIdx.x & 63;
+  const bool two = row + 1 < N;
   const uint16_t* w0 = W + (int64_t)row * K;
+  const uint16_t* w0b = w0 + (two ? K : 0);
   const uint16_t* w1 = (FLAGS & DEC_SWIGLU)
       ? W + (int64_t)(N + row) * K : nullptr;
+  const uint16_t* w1b = (FLAGS & DEC_SWIGLU) ? w1 + (two ? K : 0) : nullptr;
 
-  float acc0 = 0.0f, acc1 = 0.0f;
+  float acc0 = 0.0f, acc1 = 0.0f, acc0b = 0.0f, acc1b = 0.0f;
   int k = lane * 8;
   for (; k + 8 <= K; k += 64 * 8) {
     const ushortx8 xv = *reinterpret_cast<const ushortx8*>(xs + k);
     const ushortx8 wv = __builtin_nontemporal_load(
         reinterpret_cast<const ushortx8*>(w0 + k));
+    const ushortx8 wvb = __builtin_nontemporal_load(
+        reinterpret_cast<const ushortx8*>(w0b + k));
     #pragma unroll
-    for (int i = 0; i < 8; ++i)
-      acc0 = fmaf(bf16_to_f32(wv[i]), bf16_to_f32(xv[i]), acc0);
+    for (int i = 0; i < 8; ++i) {
+      const float xf = bf16_to_f32(xv[i]);
+      acc0 = fmaf(bf16_to_f32(wv[i]), xf, acc0);
+      acc0b = fmaf(bf16_to_f32(wvb[i]), xf, acc0b);
+    }
     if constexpr (FLAGS & DEC_SWIGLU) {
       const ushortx8 uv = __builtin_nontemporal_load(
           reinterpret_cast<const ushortx8*>(w1 + k));
+      const ushortx8 uvb = __builtin_nontemporal_load(
+          reinterpret_cast<const ushortx8*>(w1b + k));
       #pragma unroll
-      for (int i = 0; i < 8; ++i)
-        acc1 = fmaf(bf16_to_f32(uv[i]), bf16_to_f32(xv[i]), acc1);
+      for (int i = 0; i < 8; ++i) {
+        const float xf = bf16_to_f32(xv[i]);
+        acc1 = fmaf(bf16_to_f32(uv[i]), xf, acc1);
+        acc1b = fmaf(bf16_to_f32(uvb[i]), xf, acc1b);
+      }
     }
   }
   for (int kk = k; kk < K && kk < k + 8; ++kk) {
-    acc0 = fmaf(bf16_to_f32(w0[kk]), bf16_to_f32(xs[kk]), acc0);
-    if constexpr (FLAGS & DEC_SWIGLU)
-      acc1 = fmaf(bf16_to_f32(w1[kk]), bf16_to_f32(xs[kk]), acc1);
+    const float xf = bf16_to_f32(xs[kk]);
+    acc0 = fmaf(bf16_to_f32(w0[kk]), xf, acc0);
+    acc0b = fmaf(bf16_to_f32(w0b[kk]), xf, acc0b);
+    if constexpr (FLAGS & DEC_SWIGLU) {
+      acc1 = fmaf(bf16_to_f32(w1[kk]), xf, acc1);
+      acc1b = fmaf(bf16_to_f32(w1b[kk]), xf, acc1b);
+    }
   }
   acc0 = wave_reduce_sum(acc0);
-  if constexpr (FLAGS & DEC_SWIGLU) acc1 = wave_reduce_sum(acc1);
+  acc0b = wave_reduce_sum(acc0b);
+  if constexpr (FLAGS & DEC_SWIGLU) {
+    acc1 = wave_reduce_sum(acc1);
+    acc1b = wave_reduce_sum(acc1b);
+  }
   if (lane == 0) {
     float v = acc0;
     if constexpr (FLAGS & DEC_SWIGLU) v = silu(acc0) * acc1;
     if constexpr (FLAGS & DEC_RESID) v += bf16_to_f32(resid[row]);
     y[row] = f32_to_bf16(v);
+    if (two) {
+      float vb = acc0b;
+      if constexpr (FLAGS & DEC_SWIGLU) vb = silu(acc0b) * acc1b;
+      if constexpr (FLAGS & DEC_RESID) vb += bf16_to_f32(resid[row + 1]);
+      y[row + 1] = f32_to_bf16(vb);
+    }
   }
 }
 
@@ -323,7 +354,7 @@ extern "C" hipError_t lumina_dec_gemv(const void* W, const void* x,
                                       void* y, int N, int K, float eps,
                                       int flags, hipStream_t stream) {
   const int lds = 64 + ((K * 2 + 15) & ~15);
-  dim3 grid((N + 3) / 4), block(256);
+  dim3 grid((N + 7) / 8), block(256);
   switch (flags) {
     case 0:
       hipLaunchKernelGGL(dec_gemv_kernel<0>, grid, block, lds, stream,
