@@ -13,6 +13,7 @@ recipe — wgrad/dgrad keep full bf16 fidelity).
 
 from __future__ import annotations
 
+import os
 import weakref
 
 import torch
@@ -219,9 +220,32 @@ def _mx_quantized_weight(w: torch.Tensor):
     return qT, sc
 
 
+def _mx_quantized_weight_rows(w: torch.Tensor):
+    """w [E, K, N] bf16 -> rowwise-quantized NATURAL layout (q [E, K, Np]
+    u8, s [E, K] u8), cached per optimizer epoch.  This is the B operand
+    of the MX dgrad GEMM: grad_x = go @ w^T contracts over N, and w's
+    rows are already N-contiguous -- no transpose needed."""
+    ent = _WCACHE.get(("mxr", id(w)))
+    if ent is not None and ent[3] == _EPOCH and ent[0]() is w:
+        return ent[1], ent[2]
+    from .interface import get_ext
+    q, sc = get_ext().mx_quant_rows(w.contiguous(), 0)
+    _WCACHE[("mxr", id(w))] = (weakref.ref(w), q, sc, _EPOCH)
+    return q, sc
+
+
+# MX dgrad measured SLOWER end-to-end than bf16 dgrad (40.7k vs 41.5k
+# tok/s on the b1 fp8 step): the per-call rowwise quantization of go
+# costs more than the 1.3-1.5 PF GEMM saves.  bf16 dgrad is the default;
+# LUMINA_FP8_MX_DGRAD=1 opts in (capability + memory-bound regimes).
+_MX_DGRAD = bool(os.environ.get("LUMINA_FP8_MX_DGRAD"))
+
+
 class _MXExpertBmmFn(torch.autograd.Function):
-    """Grouped expert GEMM: MX-fp8 forward, bf16 backward (NT MFMA kernel
-    for grad_x, hipBLASLt for grad_w)."""
+    """Grouped expert GEMM: MX-fp8 forward (+ optional MX dgrad, roadmap
+    item 4a: grad_x contracts w's natural rows so the same NT kernel
+    serves it), hipBLASLt bf16 for wgrad (needs a TN form the MX kernel
+    doesn't have)."""
 
     @staticmethod
     def forward(ctx, x, w):
@@ -235,12 +259,18 @@ class _MXExpertBmmFn(torch.autograd.Function):
 
     @staticmethod
     def backward(ctx, go):
-        from .interface import grouped_gemm_nt
+        from .interface import get_ext, grouped_gemm_nt
         x, w = ctx.saved_tensors
         go = go.contiguous()
         gx = gw = None
         if ctx.needs_input_grad[0]:
-            gx = grouped_gemm_nt(go, w)
+            if _MX_DGRAD:
+                ext = get_ext()
+                qg, sg = ext.mx_quant_rows(go, 0)
+                wq, ws = _mx_quantized_weight_rows(w)
+                gx = ext.gg_mx_nt(qg, sg, wq, ws)
+            else:
+                gx = grouped_gemm_nt(go, w)
         if ctx.needs_input_grad[1]:
             gw = torch.bmm(x.transpose(1, 2), go)
         return gx, gw

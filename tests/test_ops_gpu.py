@@ -363,8 +363,19 @@ def test_fp8_expert_bmm_matches_bf16():
     ref = torch.matmul(x.float(), w.float())
     err = (y.float() - ref).abs().max() / ref.abs().max()
     assert float(err) < 0.1, f"fp8 expert bmm relerr {float(err)}"
-    y.sum().backward()
-    assert torch.isfinite(x.grad.float()).all()
+    import luminaai_amd.ops.fp8 as _f8
+    _f8._MX_DGRAD = True                 # exercise the MX dgrad path
+    try:
+        gy = torch.randn_like(y)
+        y.backward(gy)
+    finally:
+        _f8._MX_DGRAD = False
+    # MX dgrad (grad_x through the fp8 NT kernel): within the e4m3
+    # rowwise quant band of the bf16 reference
+    gx_ref = torch.matmul(gy.float(), w.detach().float().transpose(1, 2))
+    relg = (x.grad.float() - gx_ref).abs().max() \
+        / gx_ref.abs().max().clamp_min(1e-3)
+    assert relg < 0.08, float(relg)
     assert torch.isfinite(w.grad.float()).all()
 
 
@@ -664,8 +675,19 @@ def test_mx_expert_bmm_autograd():
     ref = torch.matmul(x.detach().float(), w.detach().float())
     rel = (y.float() - ref).abs().max() / ref.abs().max().clamp_min(1e-3)
     assert rel < 0.06, float(rel)
-    y.sum().backward()
-    assert torch.isfinite(x.grad.float()).all()
+    import luminaai_amd.ops.fp8 as _f8
+    _f8._MX_DGRAD = True                 # exercise the MX dgrad path
+    try:
+        gy = torch.randn_like(y)
+        y.backward(gy)
+    finally:
+        _f8._MX_DGRAD = False
+    # MX dgrad (grad_x through the fp8 NT kernel): within the e4m3
+    # rowwise quant band of the bf16 reference
+    gx_ref = torch.matmul(gy.float(), w.detach().float().transpose(1, 2))
+    relg = (x.grad.float() - gx_ref).abs().max() \
+        / gx_ref.abs().max().clamp_min(1e-3)
+    assert relg < 0.08, float(relg)
     assert torch.isfinite(w.grad.float()).all()
 
 
