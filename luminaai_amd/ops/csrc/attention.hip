@@ -649,6 +649,7 @@ void attn_bwd_dkdv8_kernel(const uint16_t* __restrict__ Q,
   __shared__ uint16_t lsV[4][BLK_ELEMS(DP)];
   __shared__ float    lsL[2][TS];
   __shared__ float    lsD[2][TS];
+  __shared__ float    lsP[4][TS * TS];         // P handoff dV-wave -> dK-wave
 
   const int hkv  = blockIdx.y;
   const int b    = blockIdx.z;
@@ -734,33 +735,34 @@ void attn_bwd_dkdv8_kernel(const uint16_t* __restrict__ Q,
       stage_load(sq, so, lse1, del1, step + 1);
 
     const bool active = qt * TS + TS - 1 >= kv0w;
+    // phase 1: dV wave computes S -> P (writes the P slab + its own dV
+    // MFMAs); dK wave computes ONLY dP.  The S tile is no longer
+    // duplicated: the dK wave reads P from the slab after the mid-step
+    // barrier (pair MFMAs/step 50 -> 40).
     if (active) {
-      // ---- S tile (both duty waves): rows(regs) = q, cols(lanes) = kv
-      f32x16 st = {};
-      __builtin_amdgcn_s_setprio(1);
-      #pragma unroll
-      for (int ks = 0; ks < KS; ++ks)
-        st = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
-            brfrag<DP>(lsQt[buf], lane, ks), brfrag<DP>(lsK[g], lane, ks),
-            st, 0, 0, 0);
-      __builtin_amdgcn_s_setprio(0);
-
-      const int kvg = kv0w + (lane & 31);
-      const bool interior = (qt * TS >= kv0w + TS) && (qt * TS + TS <= S);
-      float p[16];
-      #pragma unroll
-      for (int r = 0; r < 16; ++r) {
-        const float lse = lsL[buf][crow(r, lane >> 5)];
-        float pr = __builtin_exp2f(fmaf(st[r], c2, -lse));
-        if (!interior) {
-          const int qg = qt * TS + crow(r, lane >> 5);
-          pr = (kvg <= qg && qg < S) ? pr : 0.0f;
-        }
-        p[r] = pr;
-      }
-
       if (!kduty) {
-        // ---- dV += P^T dO
+        f32x16 st = {};
+        __builtin_amdgcn_s_setprio(1);
+        #pragma unroll
+        for (int ks = 0; ks < KS; ++ks)
+          st = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+              brfrag<DP>(lsQt[buf], lane, ks), brfrag<DP>(lsK[g], lane, ks),
+              st, 0, 0, 0);
+        __builtin_amdgcn_s_setprio(0);
+        const int kvg = kv0w + (lane & 31);
+        const bool interior = (qt * TS >= kv0w + TS) && (qt * TS + TS <= S);
+        float p[16];
+        #pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          const float lse = lsL[buf][crow(r, lane >> 5)];
+          float pr = __builtin_exp2f(fmaf(st[r], c2, -lse));
+          if (!interior) {
+            const int qg = qt * TS + crow(r, lane >> 5);
+            pr = (kvg <= qg && qg < S) ? pr : 0.0f;
+          }
+          p[r] = pr;
+          lsP[g][crow(r, lane >> 5) * TS + (lane & 31)] = pr;
+        }
         bf16x8 pf[2];
         pack_frags(p, pf);
         __builtin_amdgcn_s_setprio(1);
@@ -773,7 +775,6 @@ void attn_bwd_dkdv8_kernel(const uint16_t* __restrict__ Q,
                 0, 0, 0);
         __builtin_amdgcn_s_setprio(0);
       } else {
-        // ---- dP tile, then dK += dS^T Q
         f32x16 dp = {};
         __builtin_amdgcn_s_setprio(1);
         #pragma unroll
@@ -782,11 +783,13 @@ void attn_bwd_dkdv8_kernel(const uint16_t* __restrict__ Q,
               brfrag<DP>(lsOt[buf], lane, ks), brfrag<DP>(lsV[g], lane, ks),
               dp, 0, 0, 0);
         __builtin_amdgcn_s_setprio(0);
+        __syncthreads();                     // P slab ready
         float ds[16];
         #pragma unroll
         for (int r = 0; r < 16; ++r) {
           const float dlt = lsD[buf][crow(r, lane >> 5)];
-          ds[r] = p[r] * (dp[r] - dlt) * scale;
+          const float pr = lsP[g][crow(r, lane >> 5) * TS + (lane & 31)];
+          ds[r] = pr * (dp[r] - dlt) * scale;
         }
         bf16x8 df[2];
         pack_frags(ds, df);
@@ -801,6 +804,8 @@ void attn_bwd_dkdv8_kernel(const uint16_t* __restrict__ Q,
         __builtin_amdgcn_s_setprio(0);
       }
     }
+    if (!active || !kduty)
+      __syncthreads();                       // pairs with the dK-wave wait
 
     if (step + 1 < total) {
       const int bi = buf ^ 1;

@@ -47,11 +47,13 @@ __global__ void swiglu_bwd_vec_kernel(const uint16_t* __restrict__ dy,
        idx < total; idx += gstride()) {
     const int64_t r = idx / I8;
     const int64_t i = (idx - r * I8) * 8;
-    const ushortx8 gv = *reinterpret_cast<const ushortx8*>(
-        g + r * g_stride + i);
-    const ushortx8 uv = *reinterpret_cast<const ushortx8*>(
-        u + r * u_stride + i);
-    const ushortx8 dyv = *reinterpret_cast<const ushortx8*>(dy + idx * 8);
+    // nontemporal: strictly streaming, never re-read (bypass L2 churn)
+    const ushortx8 gv = __builtin_nontemporal_load(
+        reinterpret_cast<const ushortx8*>(g + r * g_stride + i));
+    const ushortx8 uv = __builtin_nontemporal_load(
+        reinterpret_cast<const ushortx8*>(u + r * u_stride + i));
+    const ushortx8 dyv = __builtin_nontemporal_load(
+        reinterpret_cast<const ushortx8*>(dy + idx * 8));
     ushortx8 dgv, duv;
     #pragma unroll
     for (int j = 0; j < 8; ++j) {
@@ -63,8 +65,10 @@ __global__ void swiglu_bwd_vec_kernel(const uint16_t* __restrict__ dy,
       dgv[j] = f32_to_bf16(dyf * bf16_to_f32(uv[j]) * dsilu);
       duv[j] = f32_to_bf16(dyf * silu);
     }
-    *reinterpret_cast<ushortx8*>(dg + r * g_stride + i) = dgv;
-    *reinterpret_cast<ushortx8*>(du + r * u_stride + i) = duv;
+    __builtin_nontemporal_store(
+        dgv, reinterpret_cast<ushortx8*>(dg + r * g_stride + i));
+    __builtin_nontemporal_store(
+        duv, reinterpret_cast<ushortx8*>(du + r * u_stride + i));
   }
 }
 
