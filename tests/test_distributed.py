@@ -1123,6 +1123,111 @@ def tpep_worker(rank, world):
             "gu_shape": tuple(gu.shape), "grad_norm": gn, "rank": rank}
 
 
+def dpep_worker(rank, world):
+    """DP(2) x EP(2) (world 4): the expert-replica grad all-reduce
+    (expert_dp_group, stride ep) finally has >1 member -- dense params
+    sync over the world group, expert shards over their replica pair.
+    Must match single-process training on the concatenated global batch."""
+    from luminaai_amd.config import Config
+    from luminaai_amd.data.tokenizer import ConversationTokenizer
+    from luminaai_amd.models import (DeepSeekTransformer,
+                                     config_to_deepseek_config)
+    from luminaai_amd.parallel.mesh import init_mesh, reset_mesh
+    from luminaai_amd.training import Trainer
+    cfg = Config(vocab_size=512, hidden_size=64, num_layers=2, num_heads=4,
+                 num_kv_heads=2, seq_length=32, intermediate_size=128,
+                 micro_batch_size=2, gradient_accumulation_steps=1,
+                 num_workers=0, use_moe=True, num_experts=4, moe_top_k=2,
+                 routing_noise_std=0.0, use_mod=False,
+                 zero_stage=0, precision="fp32",
+                 experiment_name=f"dpep_{rank}",
+                 eval_every_n_batches=0, save_every_n_batches=0)
+    mcfg = config_to_deepseek_config(cfg)
+    torch.manual_seed(1234)
+    fullm = DeepSeekTransformer(mcfg)
+    mesh = init_mesh(ep_size=2)
+    assert (mesh.dp_size, mesh.ep_rank, mesh.dp_rank) == \
+        (2, rank % 2, rank // 2)
+    model = DeepSeekTransformer(mcfg)
+    e0 = mesh.ep_rank * 2
+    with torch.no_grad():
+        for (n, p), (n2, q) in zip(model.named_parameters(),
+                                   fullm.named_parameters()):
+            p.copy_(q if p.shape == q.shape else q[e0:e0 + 2])
+    t = Trainer(model, ConversationTokenizer(), cfg)
+    t._setup_scheduler(10)
+    # data rank = dp_rank (ep peers of one replica share... no: EP ranks
+    # have DISTINCT batch shards; 4 ranks = 4 shards)
+    gens = [torch.Generator().manual_seed(820 + r) for r in range(4)]
+    for _ in range(2):
+        ids = torch.randint(1, cfg.vocab_size, (2, cfg.seq_length + 1),
+                            generator=gens[rank])
+        t.engine.set_sync(True)
+        t.train_step({"input_ids": ids[:, :-1], "labels": ids[:, 1:]})
+        t.optimizer_step()
+    emb = float(t.model.embed_tokens.weight.detach().sum())
+    gu = t.model.layers[0].ffn.w_gate_up.detach()
+    gn = t.optimizer.last_grad_norm()
+    reset_mesh()
+    return {"emb": emb, "gu_sum": float(gu.sum()), "grad_norm": gn,
+            "rank": rank}
+
+
+def test_dpep_matches_single_process():
+    ctx = mp.get_context("spawn")
+    q = ctx.SimpleQueue()
+    port = _free_port()
+    procs = [ctx.Process(target=_run, args=(r, 4, port, "dpep_worker", q))
+             for r in range(4)]
+    for p in procs:
+        p.start()
+    res = {}
+    for _ in range(4):
+        rank, status, payload = q.get()
+        assert status == "ok", f"rank {rank} failed:\n{payload}"
+        res[rank] = payload
+    for p in procs:
+        p.join(timeout=180)
+
+    import torch as th
+    from luminaai_amd.config import Config
+    from luminaai_amd.data.tokenizer import ConversationTokenizer
+    from luminaai_amd.models import (DeepSeekTransformer,
+                                     config_to_deepseek_config)
+    from luminaai_amd.training import Trainer
+    os.environ.pop("WORLD_SIZE", None)
+    cfg = Config(vocab_size=512, hidden_size=64, num_layers=2, num_heads=4,
+                 num_kv_heads=2, seq_length=32, intermediate_size=128,
+                 micro_batch_size=8, gradient_accumulation_steps=1,
+                 num_workers=0, use_moe=True, num_experts=4, moe_top_k=2,
+                 routing_noise_std=0.0, use_mod=False,
+                 zero_stage=0, precision="fp32", experiment_name="dpep_ref",
+                 eval_every_n_batches=0, save_every_n_batches=0)
+    th.manual_seed(1234)
+    model = DeepSeekTransformer(config_to_deepseek_config(cfg))
+    t = Trainer(model, ConversationTokenizer(), cfg)
+    t._setup_scheduler(10)
+    gens = [th.Generator().manual_seed(820 + r) for r in range(4)]
+    for _ in range(2):
+        rows = [th.randint(1, cfg.vocab_size, (2, cfg.seq_length + 1),
+                           generator=g) for g in gens]
+        ids = th.cat(rows)
+        t.train_step({"input_ids": ids[:, :-1], "labels": ids[:, 1:]})
+        t.optimizer_step()
+    emb_ref = float(t.model.embed_tokens.weight.detach().sum())
+    gn_ref = t.optimizer.last_grad_norm()
+    gu = t.model.layers[0].ffn.w_gate_up.detach()
+    for r in range(4):
+        e = r % 2
+        assert res[r]["emb"] == pytest.approx(emb_ref, abs=1e-3), \
+            (r, res[r]["emb"], emb_ref)
+        assert res[r]["grad_norm"] / 4 == pytest.approx(gn_ref, rel=2e-3), \
+            (r, res[r]["grad_norm"], gn_ref)
+        ref = float(gu[2 * e:2 * e + 2].sum())
+        assert res[r]["gu_sum"] == pytest.approx(ref, abs=1e-2), \
+            (r, res[r]["gu_sum"], ref)
+
+
 def test_tpep_matches_single_process():
     ctx = mp.get_context("spawn")
     q = ctx.SimpleQueue()
