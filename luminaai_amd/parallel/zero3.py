@@ -432,6 +432,11 @@ class Zero3Engine:
         self.model = model
         self.pg = process_group
         self.mesh = mesh
+        if mesh is not None and getattr(mesh, "tp_size", 1) > 1:
+            raise ValueError(
+                "ZeRO-3 does not compose with TP (the dense-segment "
+                "collectives assume the world group is pure DP x EP); "
+                "use ZeRO-0/1/2 with tensor parallelism")
         self.world = comm.get_world_size()
         self.rank = comm.get_rank()
         self.bucket_bytes = getattr(config, "reduce_bucket_size", 50_000_000)

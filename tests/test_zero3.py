@@ -362,6 +362,25 @@ def test_zero3_expert_add_prune():
     t.engine.remove_hooks()
 
 
+def test_zero3_rejects_tp_mesh():
+    """ZeRO-3 + TP is not a supported composition (the dense-segment
+    collectives assume the world group is pure DP); it must raise a
+    clear error instead of deadlocking mid-training."""
+    from luminaai_amd.models import (DeepSeekTransformer,
+                                     config_to_deepseek_config)
+    from luminaai_amd.parallel.zero3 import Zero3Engine
+    from luminaai_amd.parallel import mesh as mesh_mod
+    cfg = _cfg(3)
+    model = DeepSeekTransformer(config_to_deepseek_config(cfg))
+
+    class _FakeMesh:
+        tp_size = 2
+        ep_size = 1
+
+    with pytest.raises(ValueError, match="ZeRO-3.*TP"):
+        Zero3Engine(model, cfg, mesh=_FakeMesh())
+
+
 def z3_elastic_save_worker(rank, world):
     """Train ZeRO-3 at world=2 with SAME data on both ranks and save."""
     from luminaai_amd.training import CheckpointManager
