@@ -89,6 +89,56 @@ __global__ void adamw_step_kernel(float* __restrict__ master,
   }
 }
 
+// 4-wide variant (n % 4 == 0 and 16B-aligned pointers -- the flat-buffer
+// design pads every group to 256 elements, so this is the standard path):
+// float4 moments/master and 4-element grad/weight vectors cut the scalar
+// version's 7 memory ops/element to 7 per 4 elements.
+template <typename G, typename W>
+__global__ void adamw_step_vec_kernel(
+    float* __restrict__ master, const typename G::storage* __restrict__ grad,
+    float* __restrict__ m, float* __restrict__ v,
+    typename W::storage* __restrict__ w_out,
+    int64_t n4, float lr, float beta1, float beta2,
+    float eps, float wd, float bias1, float bias2,
+    const float* __restrict__ gnorm_sq, float max_norm, float grad_scale) {
+  float clip = 1.0f;
+  if (gnorm_sq) {
+    const float gn = sqrtf(*gnorm_sq) * grad_scale;
+    if (!isfinite(gn)) return;
+    if (max_norm > 0.f && gn > max_norm) clip = max_norm / (gn + 1e-6f);
+  }
+  const float gs = grad_scale * clip;
+  typedef __attribute__((ext_vector_type(4))) float f4;
+  typedef __attribute__((ext_vector_type(4))) typename G::storage g4;
+  typedef __attribute__((ext_vector_type(4))) typename W::storage w4;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n4;
+       i += gstride()) {
+    const g4 gv = __builtin_nontemporal_load(
+        reinterpret_cast<const g4*>(grad) + i);
+    f4 mi = *(reinterpret_cast<f4*>(m) + i);
+    f4 vi = *(reinterpret_cast<f4*>(v) + i);
+    f4 p = *(reinterpret_cast<f4*>(master) + i);
+    w4 wv;
+    #pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      const typename G::storage ge = gv[j];
+      const float g = G::load(&ge) * gs;
+      mi[j] = beta1 * mi[j] + (1.f - beta1) * g;
+      vi[j] = beta2 * vi[j] + (1.f - beta2) * g * g;
+      p[j] -= lr * (mi[j] * bias1 / (sqrtf(vi[j] * bias2) + eps)
+                    + wd * p[j]);
+      typename W::storage we;
+      W::store(&we, p[j]);
+      wv[j] = we;
+    }
+    *(reinterpret_cast<f4*>(m) + i) = mi;
+    *(reinterpret_cast<f4*>(v) + i) = vi;
+    *(reinterpret_cast<f4*>(master) + i) = p;
+    if (w_out)
+      __builtin_nontemporal_store(wv, reinterpret_cast<w4*>(w_out) + i);
+  }
+}
+
 extern "C" {
 
 hipError_t lumina_l2norm_sq_bf16(const void* x, int64_t n, float* out,
@@ -116,6 +166,36 @@ hipError_t lumina_adamw_step(float* master, const void* grad, int grad_is_bf16,
                              const float* gnorm_sq, float max_norm,
                              float grad_scale, hipStream_t s) {
   const int block = 256;
+  const bool vec4 = (n % 4 == 0)
+      && ((uintptr_t)grad & 7) == 0 && ((uintptr_t)master & 15) == 0
+      && ((uintptr_t)m & 15) == 0 && ((uintptr_t)v & 15) == 0
+      && (!w_out || ((uintptr_t)w_out & 7) == 0);
+  if (vec4) {
+    const int64_t n4 = n / 4;
+    const int grid4 = elementwise_grid(n4, block, 4);
+    if (grad_is_bf16 && wout_is_bf16)
+      adamw_step_vec_kernel<BF16Elem, BF16Elem><<<grid4, block, 0, s>>>(
+          master, (const uint16_t*)grad, m, v, (uint16_t*)w_out, n4, lr,
+          beta1, beta2, eps, wd, bias1, bias2, gnorm_sq, max_norm,
+          grad_scale);
+    else if (grad_is_bf16)
+      adamw_step_vec_kernel<BF16Elem, F32Elem><<<grid4, block, 0, s>>>(
+          master, (const uint16_t*)grad, m, v, (float*)w_out, n4, lr,
+          beta1, beta2, eps, wd, bias1, bias2, gnorm_sq, max_norm,
+          grad_scale);
+    else if (wout_is_bf16)
+      adamw_step_vec_kernel<F32Elem, BF16Elem><<<grid4, block, 0, s>>>(
+          master, (const float*)grad, m, v, (uint16_t*)w_out, n4, lr,
+          beta1, beta2, eps, wd, bias1, bias2, gnorm_sq, max_norm,
+          grad_scale);
+    else
+      adamw_step_vec_kernel<F32Elem, F32Elem><<<grid4, block, 0, s>>>(
+          master, (const float*)grad, m, v, (float*)w_out, n4, lr,
+          beta1, beta2, eps, wd, bias1, bias2, gnorm_sq, max_norm,
+          grad_scale);
+    HIP_CHECK_LAST();
+    return hipSuccess;
+  }
   const int grid = elementwise_grid(n, block, 4);
   if (grad_is_bf16 && wout_is_bf16)
     adamw_step_kernel<BF16Elem, BF16Elem><<<grid, block, 0, s>>>(

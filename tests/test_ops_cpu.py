@@ -196,3 +196,27 @@ def test_flat_adamw_offload_fp32_unsharded_falls_back():
     opt.zero_grad()
     m(torch.randn(2, 8)).sum().backward()
     opt.step()  # still steps fine
+
+
+def test_pad_k_cache_invalidation():
+    """_pad_k_cached serves stale data only until invalidate_pad_cache();
+    the Trainer calls it every optimizer step (grad_x correctness)."""
+    import torch
+    from luminaai_amd.ops.interface import (_pad_k_cached,
+                                            invalidate_pad_cache)
+    invalidate_pad_cache()
+    w = torch.randn(2, 4, 6)
+    p1 = _pad_k_cached(w, 2)
+    assert p1.shape == (2, 4, 8)
+    assert torch.equal(p1[..., :6], w) and p1[..., 6:].abs().sum() == 0
+    assert _pad_k_cached(w, 2) is p1          # cache hit
+    with torch.no_grad():
+        w.mul_(2.0)                           # in-place update...
+    invalidate_pad_cache()                    # ...trainer invalidates
+    p2 = _pad_k_cached(w, 2)
+    assert torch.equal(p2[..., :6], w)
+    # id-recycling safety: a NEW tensor at a recycled id must miss
+    del w
+    w2 = torch.randn(2, 4, 6)
+    p3 = _pad_k_cached(w2, 2)
+    assert torch.equal(p3[..., :6], w2)
