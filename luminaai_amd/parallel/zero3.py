@@ -327,6 +327,8 @@ class Zero3Optimizer:
     @torch.no_grad()
     def step(self, grad_scale: float = 1.0, closure=None,
              norm_sq: Optional[torch.Tensor] = None, shard_only: bool = True):
+        from ..ops.interface import invalidate_pad_cache
+        invalidate_pad_cache()    # weights change: padded-K grad_x cache
         self.step_count += 1
         if norm_sq is None and self.max_grad_norm > 0:
             norm_sq = self.local_grad_norm_sq()

@@ -237,6 +237,8 @@ class FlatAdamW:
         all-reduced value); default computes it locally.
         shard_only: grads already reduced into this rank's shard (ZeRO-2).
         """
+        from ..ops.interface import invalidate_pad_cache
+        invalidate_pad_cache()    # weights change: padded-K grad_x cache
         self.step_count += 1
         if norm_sq is None and self.max_grad_norm > 0:
             norm_sq = self.local_grad_norm_sq(shard_only=shard_only)
