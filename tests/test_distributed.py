@@ -1025,6 +1025,66 @@ def test_ep_with_checkpointing():
 
 
 # ---- expert add/prune under EP (round-2: SURVEY build plan 7.6) ------------
+def ep_zero1_worker(rank, world):
+    """EP(2) + ZeRO-1 -- the exact default combo bench.py uses multi-GPU
+    (zero_stage 1, ep = world). Must train identically to EP + ZeRO-0
+    on the same data (sharded-optimizer bookkeeping must not disturb
+    the expert comm groups)."""
+    from luminaai_amd.config import Config
+    from luminaai_amd.data.tokenizer import ConversationTokenizer
+    from luminaai_amd.models import (DeepSeekTransformer,
+                                     config_to_deepseek_config)
+    from luminaai_amd.parallel.mesh import init_mesh, reset_mesh
+    from luminaai_amd.training import Trainer
+
+    def run(zstage):
+        reset_mesh()
+        cfg = Config(vocab_size=512, hidden_size=64, num_layers=2,
+                     num_heads=4, num_kv_heads=2, seq_length=32,
+                     intermediate_size=128, micro_batch_size=2,
+                     gradient_accumulation_steps=1, num_workers=0,
+                     use_moe=True, num_experts=4, moe_top_k=2,
+                     routing_noise_std=0.0, use_mod=False,
+                     zero_stage=zstage, precision="fp32",
+                     experiment_name=f"epz{zstage}_{rank}",
+                     eval_every_n_batches=0, save_every_n_batches=0)
+        init_mesh(world)
+        torch.manual_seed(1234)
+        model = DeepSeekTransformer(config_to_deepseek_config(cfg))
+        t = Trainer(model, ConversationTokenizer(), cfg)
+        t._setup_scheduler(10)
+        torch.manual_seed(640 + rank)
+        for _ in range(2):
+            ids = torch.randint(1, cfg.vocab_size, (2, cfg.seq_length + 1))
+            t.engine.set_sync(True)
+            t.train_step({"input_ids": ids[:, :-1], "labels": ids[:, 1:]})
+            t.optimizer_step()
+        emb = t.model.embed_tokens.weight.detach().clone()
+        gu = t.model.layers[0].ffn.w_gate_up.detach().clone()
+        gn = t.optimizer.last_grad_norm()
+        t.engine.remove_hooks()
+        reset_mesh()
+        return emb, gu, gn
+
+    emb1, gu1, gn1 = run(1)
+    emb2, gu2, gn2 = run(2)
+    emb0, gu0, gn0 = run(0)
+    return {"demb1": float((emb1 - emb0).abs().max()),
+            "dgu1": float((gu1 - gu0).abs().max()),
+            "demb2": float((emb2 - emb0).abs().max()),
+            "dgu2": float((gu2 - gu0).abs().max()),
+            "gn1": gn1, "gn2": gn2, "gn0": gn0, "rank": rank}
+
+
+def test_ep_with_zero1_zero2():
+    res = _spawn("ep_zero1_worker")
+    for r in range(WORLD):
+        assert res[r]["demb1"] < 1e-5 and res[r]["dgu1"] < 1e-5, res
+        assert res[r]["demb2"] < 1e-5 and res[r]["dgu2"] < 1e-5, res
+        assert res[r]["gn1"] == pytest.approx(res[r]["gn0"], rel=1e-5), res
+        assert res[r]["gn2"] == pytest.approx(res[r]["gn0"], rel=1e-5), res
+
+
 def ep_fp8_a2a_worker(rank, world):
     """EP(2) with fp8_alltoall: forward/backward equal the full-expert
     model within e4m3 quantization tolerance (payload e4m3, grads e5m2)."""
