@@ -182,10 +182,8 @@ class FusedDecoder:
                            self.eps, _RESID)
             else:
                 Wg, WguT, WdnT, k, temp, I = self.moe[i]
-                e.dec_rmsnorm(self.x2, self.wn_post[i], self.xhat, self.eps)
-                rl = self.rlogits[:Wg.shape[0]]
-                e.dec_gemv(Wg, self.xhat, None, None, rl, self.eps, 0)
-                e.dec_topk(rl, self.eidx, self.ew, k, temp)
+                e.dec_router(self.x2, self.wn_post[i], Wg, self.xhat,
+                             self.eidx, self.ew, k, temp, self.eps)
                 act = self.act[:I]
                 for slot in range(k):
                     e.dec_gemv_moe(WguT, self.xhat, None, act,

@@ -65,6 +65,7 @@ hipError_t lumina_dec_advance(int*, hipStream_t);
 hipError_t lumina_gg8t_nt(const void*, const void*, void*, int, int, int, int, int64_t, int64_t, int64_t, int, hipStream_t);
 hipError_t lumina_dec_rmsnorm(const void*, const void*, void*, int, float, hipStream_t);
 hipError_t lumina_dec_topk(const void*, int*, float*, int, int, float, hipStream_t);
+hipError_t lumina_dec_router(const void*, const void*, const void*, void*, int*, float*, int, int, int, float, float, hipStream_t);
 hipError_t lumina_dec_gemv_moe(const void*, const void*, const void*, void*, const int*, const float*, int, int64_t, int, int, int, hipStream_t);
 hipError_t lumina_moe_gather_rows(const void*, const int64_t*, const bool*, void*, int64_t, int, int, hipStream_t);
 hipError_t lumina_moe_dispatch_bwd(const void*, const int64_t*, void*, int64_t, int, int, int64_t, int, hipStream_t);
@@ -656,6 +657,17 @@ void dec_rmsnorm(const at::Tensor& x, const at::Tensor& wn, at::Tensor& out,
             "dec_rmsnorm");
 }
 
+void dec_router(const at::Tensor& x, const at::Tensor& wn,
+                const at::Tensor& Wg, at::Tensor& xhat, at::Tensor& eidx,
+                at::Tensor& ew, int64_t k, double temp, double eps) {
+  check_hip(lumina_dec_router(x.data_ptr(), wn.data_ptr(), Wg.data_ptr(),
+                              xhat.data_ptr(), eidx.data_ptr<int>(),
+                              ew.data_ptr<float>(), (int)x.numel(),
+                              (int)Wg.size(0), (int)k, (float)temp,
+                              (float)eps, cur_stream()),
+            "dec_router");
+}
+
 void dec_topk(const at::Tensor& logits, at::Tensor& eidx, at::Tensor& ew,
               int64_t k, double temp) {
   check_hip(lumina_dec_topk(logits.data_ptr(), eidx.data_ptr<int>(),
@@ -708,6 +720,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("swiglu_bwd_fused", &swiglu_bwd_fused,
           "swiglu backward into one fused [rows, 2I] grad buffer");
   mod.def("dec_rmsnorm", &dec_rmsnorm, "decode rmsnorm -> xhat buffer");
+  mod.def("dec_router", &dec_router,
+          "fused decode MoE router: rmsnorm + gate GEMV + top-k");
   mod.def("dec_topk", &dec_topk,
           "decode router: softmax/temp top-k, renormalized, on device");
   mod.def("dec_gemv_moe", &dec_gemv_moe,
