@@ -10,7 +10,7 @@ device).
 
 Scope: dense, MoD (MoD layers run dense at decode: the block's run_mod
 gate requires S > 1) and single-GPU MoE models.  MoE layers decode as:
-rmsnorm -> router GEMV -> on-device top-k (dec_topk) -> per selected
+fused router kernel (rmsnorm + gate GEMV + on-device top-k) -> per selected
 expert an indirect SWIGLU GEMV + scaled-residual down GEMV, the expert
 index read from device memory so the step still hipGraph-captures.
 Router softmax runs over bf16 logits (training gating is fp32): top-k
@@ -128,8 +128,6 @@ class FusedDecoder:
                 self.Wdn.append(layer.ffn.down_proj.weight.data.contiguous())
                 self.moe.append(None)
         kmax = max((m[3] for m in self.moe if m), default=0)
-        E = max((m[0].shape[0] for m in self.moe if m), default=1)
-        self.rlogits = torch.zeros(E, dtype=dt, device=dev)
         self.eidx = torch.zeros(max(kmax, 1), dtype=torch.int32, device=dev)
         self.ew = torch.zeros(max(kmax, 1), dtype=torch.float32, device=dev)
         self.wn_final = model.final_norm.weight.data.contiguous()
