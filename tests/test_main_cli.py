@@ -199,3 +199,55 @@ def test_main_ring_sp(tmp_path):
     for p in procs:
         p.join(timeout=120)
     assert results[0]["global_step"] >= 2
+
+
+def _tp_cli_worker(rank, world, port, q, tmp):
+    import traceback
+    os.environ.update({
+        "MASTER_ADDR": "127.0.0.1", "MASTER_PORT": str(port),
+        "RANK": str(rank), "LOCAL_RANK": str(rank), "WORLD_SIZE": str(world),
+    })
+    os.chdir(tmp)
+    try:
+        from luminaai_amd.main import main
+        result = main([
+            "--preset", "debug", "--synthetic-steps", "2",
+            "--experiment-name", "cli_tp", "--precision", "fp32",
+            "--micro-batch", "2", "--accum", "1", "--seq-len", "64",
+            "--tp", "2",
+            "--set", "hidden_size=64", "--set", "num_layers=2",
+            "--set", "num_heads=4", "--set", "num_kv_heads=2",
+            "--set", "vocab_size=512", "--set", "intermediate_size=128",
+            "--set", "num_workers=0", "--set", "use_moe=false",
+            "--set", "use_mod=false", "--set", "gradient_checkpointing=false",
+            "--set", "eval_every_n_batches=0", "--set", "save_every_n_batches=0",
+        ])
+        q.put((rank, "ok", result))
+    except Exception:  # noqa: BLE001
+        q.put((rank, "err", traceback.format_exc()))
+
+
+def test_main_tensor_parallel(tmp_path):
+    """--tp 2 trains end-to-end through the CLI (gloo x2)."""
+    import socket
+
+    import torch.multiprocessing as mp
+    ctx = mp.get_context("spawn")
+    q = ctx.SimpleQueue()
+    with socket.socket() as s:
+        s.bind(("127.0.0.1", 0))
+        port = s.getsockname()[1]
+    procs = [ctx.Process(target=_tp_cli_worker,
+                         args=(r, 2, port, q, str(tmp_path)))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    results = {}
+    for _ in range(2):
+        rank, status, payload = q.get()
+        assert status == "ok", f"rank {rank} failed:\n{payload}"
+        results[rank] = payload
+    for p in procs:
+        p.join(timeout=120)
+    for r in range(2):
+        assert results[r]["global_step"] >= 2
