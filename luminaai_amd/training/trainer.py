@@ -728,11 +728,11 @@ class Trainer:
         parallel/load_balance.py — ColossalAI LoadBalancer counterpart).
         Rebuilds the optimizer (fresh moments for expert weights)."""
         layers = self._moe_layers()
-        if not layers or self.engine.stage >= 3:
+        if not layers:
             return False
         from ..parallel.load_balance import (apply_placement, imbalance,
                                              plan_placement)
-        changed = False
+        orders = {}
         for l in layers:
             counts = l._usage_counts.clone()
             if comm.is_distributed():
@@ -749,7 +749,19 @@ class Trainer:
             ident = list(range(l.num_experts))
             if imbalance(loads, order, ep) < imbalance(loads, ident, ep) - 1e-6 \
                     or getattr(l, "placement", None) is not None:
-                apply_placement(l, order)
+                orders[id(l)] = order
+        if not orders:
+            return False
+        if self.engine.stage >= 3:
+            # ZeRO-3: materialise weights, re-place, rebuild the engine
+            # (fresh moments -- same machinery as expert add/prune)
+            return self._mutate_experts_zero3(
+                lambda l: apply_placement(l, orders[id(l)])
+                if id(l) in orders else None)
+        changed = False
+        for l in layers:
+            if id(l) in orders:
+                apply_placement(l, orders[id(l)])
                 changed = True
         if changed:
             # weights moved IN PLACE through the flat buffers; the Adam
