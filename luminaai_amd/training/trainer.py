@@ -407,6 +407,7 @@ class Trainer:
                 if os.path.exists(fixed):
                     which = fixed
         payload = self.checkpoints.load_checkpoint(which, map_location=self.device)
+        payload = self._maybe_reshard_ep(which, payload, load_optimizer)
         if self.engine.stage >= 3:
             with self.engine.gathered_weights():
                 self.model.load_state_dict(payload["model_state_dict"])
@@ -451,6 +452,129 @@ class Trainer:
             self.scheduler.load_state_dict(payload["scheduler_state_dict"])
         self.global_step = payload.get("global_step", 0)
         self.epoch = payload.get("epoch", 0)
+        return payload
+
+    def _maybe_reshard_ep(self, which, payload, load_optimizer):
+        """Elastic EP resume: the run was saved at a different ep_size
+        (detected by the expert tensors' local-expert dim).  Model expert
+        weights and the optimizer's expert flat group are re-assembled
+        from the per-ep-rank checkpoint files by slicing along the expert
+        dim.  Scope: zero_stage 0 for the optimizer part (dense state is
+        replicated; ZeRO-1/2 dense shards reshard separately via
+        load_resharded and are not combined with an EP change)."""
+        msd = payload.get("model_state_dict")
+        if not msd:
+            return payload
+        cur = self.model.state_dict()
+        ek = next((k for k in msd
+                   if k.endswith(".w_gate_up") and k in cur), None)
+        if ek is None or msd[ek].shape[0] == cur[ek].shape[0]:
+            return payload
+        el_a, el_b = msd[ek].shape[0], cur[ek].shape[0]
+        ep_b = self.mesh.ep_size if self.mesh is not None else 1
+        E = el_b * max(ep_b, 1)
+        if E % el_a:
+            raise ValueError(f"elastic EP resume: saved local experts "
+                             f"{el_a} do not divide the global count {E}")
+        ep_a = E // el_a
+        ep_rank_b = self.mesh.ep_rank if self.mesh is not None else 0
+        # gather the ep_a saved payloads (rank order)
+        import re as _re
+        base = str(self.checkpoints.resolve(which) or which)
+        payloads = []
+        for e in range(ep_a):
+            if ep_a == 1:
+                payloads.append(payload)
+                break
+            p = _re.sub(r"_ep_rank_\d+", f"_ep_rank_{e}", base)
+            if not os.path.exists(p):
+                raise FileNotFoundError(
+                    f"elastic EP resume: missing saved expert shard {p}")
+            payloads.append(torch.load(p, map_location=self.device,
+                                       weights_only=False))
+        # ---- model weights: slice-merge the expert dims
+        new_msd = {}
+        for k, v in payloads[0]["model_state_dict"].items():
+            tgt = cur.get(k)
+            if tgt is None or tuple(tgt.shape) == tuple(v.shape):
+                new_msd[k] = v
+                continue
+            rows = []
+            for j in range(el_b):
+                e = ep_rank_b * el_b + j
+                src = payloads[e // el_a]["model_state_dict"][k]
+                rows.append(src[e % el_a])
+            new_msd[k] = torch.stack(rows)
+        payload = dict(payload)
+        payload["model_state_dict"] = new_msd
+        if not load_optimizer or not payload.get("optimizer_state_dict"):
+            return payload
+        if self.config.zero_stage != 0:
+            raise NotImplementedError(
+                "elastic EP resume of optimizer state requires "
+                "zero_stage=0 (pass load_optimizer=False to resume "
+                "weights only)")
+        # ---- optimizer: rebuild EVERY group param-by-param.  The group
+        # STRUCTURE differs across ep sizes (ep-active runs keep expert
+        # params in their own comm group; ep=1 folds them into decay), so
+        # reconstruct the saved layout by re-running the deterministic
+        # classification with the SAVED ep semantics, then source each
+        # new param's state slice by name.
+        from ..parallel.expert_parallel import is_expert_param
+        from .optimizer import NO_DECAY_KEYWORDS
+        named = [(n, p) for n, p in self.model.named_parameters()
+                 if p.requires_grad]
+        saved_layout = {}                   # name -> (saved_gi, off, old_n)
+        offs = [0, 0, 0]
+        for n, p in named:
+            exp = is_expert_param(n)
+            # expert tensors were saved at el_a local experts regardless
+            # of which group they sat in
+            old_n = p.numel() // p.shape[0] * el_a if exp else p.numel()
+            if ep_a > 1 and exp:
+                gi = 2
+            elif any(k in n for k in NO_DECAY_KEYWORDS) or p.dim() <= 1:
+                gi = 1
+            else:
+                gi = 0
+            saved_layout[n] = (gi, offs[gi], old_n)
+            offs[gi] += old_n
+        saved_groups = payloads[0]["optimizer_state_dict"]["groups"]
+        for gi, total in enumerate(offs):
+            if total and saved_groups[gi]["numel"] != total:
+                raise ValueError(
+                    f"elastic EP resume: reconstructed saved group {gi} "
+                    f"size {total} != checkpoint {saved_groups[gi]['numel']}")
+        name_of = {id(p): n for n, p in named}
+
+        opt = dict(payloads[0]["optimizer_state_dict"])
+        groups = []
+        for g in self.optimizer.groups:
+            tmpl = {"master": g.master, "m": g.m, "v": g.v}
+            new_g = {"lr": g.lr, "weight_decay": g.weight_decay,
+                     "numel": g.numel}
+            for key, t in tmpl.items():
+                buf = torch.zeros(t.shape, dtype=torch.float32)
+                for p, (off, n) in zip(g.params, g.offsets):
+                    nme = name_of[id(p)]
+                    sgi, soff, old_n = saved_layout[nme]
+                    if is_expert_param(nme):
+                        pe = n // p.shape[0]
+                        for j in range(p.shape[0]):
+                            e = ep_rank_b * el_b + j
+                            src = payloads[e // el_a][
+                                "optimizer_state_dict"]["groups"][sgi][key]
+                            so = soff + (e % el_a) * pe
+                            buf[off + j * pe: off + (j + 1) * pe] = \
+                                src[so: so + pe].float()
+                    else:
+                        src = payloads[0][
+                            "optimizer_state_dict"]["groups"][sgi][key]
+                        buf[off: off + n] = src[soff: soff + old_n].float()
+                new_g[key] = buf
+            groups.append(new_g)
+        opt["groups"] = groups
+        payload["optimizer_state_dict"] = opt
         return payload
 
     def _gather_saved_optim_shards(self, which, payload):

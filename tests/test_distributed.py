@@ -1125,6 +1125,189 @@ def test_ep_fp8_alltoall():
         assert res[r]["grad_finite"], res
 
 
+def ep_elastic_save_worker(rank, world):
+    """Train EP(2) 2 steps, save, then ONE more step (the reference
+    continuation for the elastic single-proc resume)."""
+    from luminaai_amd.config import Config
+    from luminaai_amd.data.tokenizer import ConversationTokenizer
+    from luminaai_amd.models import (DeepSeekTransformer,
+                                     config_to_deepseek_config)
+    from luminaai_amd.parallel.mesh import init_mesh, reset_mesh
+    from luminaai_amd.training import CheckpointManager, Trainer
+    tmp = os.environ["EP_ELASTIC_TMP"]
+    os.chdir(tmp)
+    cfg = Config(vocab_size=512, hidden_size=64, num_layers=2, num_heads=4,
+                 num_kv_heads=2, seq_length=32, intermediate_size=128,
+                 micro_batch_size=2, gradient_accumulation_steps=1,
+                 num_workers=0, use_moe=True, num_experts=4, moe_top_k=2,
+                 routing_noise_std=0.0, use_mod=False,
+                 zero_stage=0, precision="fp32",
+                 experiment_name=f"epel_{rank}",
+                 eval_every_n_batches=0, save_every_n_batches=0)
+    init_mesh(world)
+    torch.manual_seed(1234)
+    model = DeepSeekTransformer(config_to_deepseek_config(cfg))
+    t = Trainer(model, ConversationTokenizer(), cfg)
+    t._setup_scheduler(10)
+    t.checkpoints = CheckpointManager(os.path.join(tmp, "ep_ckpts"))
+    gens = [torch.Generator().manual_seed(870 + r) for r in range(world)]
+    for _ in range(2):
+        ids = torch.randint(1, cfg.vocab_size, (2, cfg.seq_length + 1),
+                            generator=gens[rank])
+        t.engine.set_sync(True)
+        t.train_step({"input_ids": ids[:, :-1], "labels": ids[:, 1:]})
+        t.optimizer_step()
+    t.save_checkpoint(tag="elastic_test")
+    import torch.distributed as dist
+    dist.barrier()
+    # reference continuation step (same per-rank shards)
+    ids = torch.randint(1, cfg.vocab_size, (2, cfg.seq_length + 1),
+                        generator=gens[rank])
+    t.engine.set_sync(True)
+    t.train_step({"input_ids": ids[:, :-1], "labels": ids[:, 1:]})
+    t.optimizer_step()
+    emb = float(t.model.embed_tokens.weight.detach().sum())
+    gu = t.model.layers[0].ffn.w_gate_up.detach().clone()
+    reset_mesh()
+    return {"emb": emb, "gu_sum": float(gu.sum()), "rank": rank}
+
+
+def test_ep_elastic_resume_world1(tmp_path):
+    """Elastic EP resume: a world-2/EP-2 run's checkpoint loads into a
+    single process (all 4 experts + optimizer moments re-assembled from
+    the two _ep_rank files) and continues training identically."""
+    os.environ["EP_ELASTIC_TMP"] = str(tmp_path)
+    res = _spawn("ep_elastic_save_worker")
+
+    import torch as th
+    from luminaai_amd.config import Config
+    from luminaai_amd.data.tokenizer import ConversationTokenizer
+    from luminaai_amd.models import (DeepSeekTransformer,
+                                     config_to_deepseek_config)
+    from luminaai_amd.training import CheckpointManager, Trainer
+    os.environ.pop("WORLD_SIZE", None)
+    cfg = Config(vocab_size=512, hidden_size=64, num_layers=2, num_heads=4,
+                 num_kv_heads=2, seq_length=32, intermediate_size=128,
+                 micro_batch_size=4, gradient_accumulation_steps=1,
+                 num_workers=0, use_moe=True, num_experts=4, moe_top_k=2,
+                 routing_noise_std=0.0, use_mod=False,
+                 zero_stage=0, precision="fp32", experiment_name="epel_r",
+                 eval_every_n_batches=0, save_every_n_batches=0)
+    th.manual_seed(1)
+    model = DeepSeekTransformer(config_to_deepseek_config(cfg))
+    t = Trainer(model, ConversationTokenizer(), cfg)
+    t._setup_scheduler(10)
+    t.checkpoints = CheckpointManager(str(tmp_path / "ep_ckpts"))
+    t.load_checkpoint("latest")
+    # continuation on the CONCATENATED global batch of the ref step
+    gens = [th.Generator().manual_seed(870 + r) for r in range(2)]
+    for g in gens:                      # advance past the 2 pre-save draws
+        for _ in range(2):
+            th.randint(1, cfg.vocab_size, (2, cfg.seq_length + 1),
+                       generator=g)
+    rows = [th.randint(1, cfg.vocab_size, (2, cfg.seq_length + 1),
+                       generator=g) for g in gens]
+    ids = th.cat(rows)
+    t.train_step({"input_ids": ids[:, :-1], "labels": ids[:, 1:]})
+    t.optimizer_step()
+    emb = float(t.model.embed_tokens.weight.detach().sum())
+    gu = t.model.layers[0].ffn.w_gate_up.detach()     # [4, h, 2I]
+    assert emb == pytest.approx(res[0]["emb"], abs=1e-3), \
+        (emb, res[0]["emb"])
+    for r in range(2):
+        ref = float(gu[2 * r:2 * r + 2].sum())
+        assert res[r]["gu_sum"] == pytest.approx(ref, abs=1e-3), \
+            (r, res[r]["gu_sum"], ref)
+
+
+def ep_elastic_grow_worker(rank, world):
+    """Resume a SINGLE-process MoE checkpoint at EP(2): elastic grow."""
+    from luminaai_amd.config import Config
+    from luminaai_amd.data.tokenizer import ConversationTokenizer
+    from luminaai_amd.models import (DeepSeekTransformer,
+                                     config_to_deepseek_config)
+    from luminaai_amd.parallel.mesh import init_mesh, reset_mesh
+    from luminaai_amd.training import CheckpointManager, Trainer
+    tmp = os.environ["EP_ELASTIC_TMP"]
+    os.chdir(tmp)
+    cfg = Config(vocab_size=512, hidden_size=64, num_layers=2, num_heads=4,
+                 num_kv_heads=2, seq_length=32, intermediate_size=128,
+                 micro_batch_size=2, gradient_accumulation_steps=1,
+                 num_workers=0, use_moe=True, num_experts=4, moe_top_k=2,
+                 routing_noise_std=0.0, use_mod=False,
+                 zero_stage=0, precision="fp32",
+                 experiment_name=f"epgrow_{rank}",
+                 eval_every_n_batches=0, save_every_n_batches=0)
+    init_mesh(world)
+    torch.manual_seed(7)
+    model = DeepSeekTransformer(config_to_deepseek_config(cfg))
+    t = Trainer(model, ConversationTokenizer(), cfg)
+    t._setup_scheduler(10)
+    t.checkpoints = CheckpointManager(os.path.join(tmp, "grow_ckpts"))
+    t.load_checkpoint("latest")
+    gens = [torch.Generator().manual_seed(880 + r) for r in range(world)]
+    ids = torch.randint(1, cfg.vocab_size, (2, cfg.seq_length + 1),
+                        generator=gens[rank])
+    t.engine.set_sync(True)
+    t.train_step({"input_ids": ids[:, :-1], "labels": ids[:, 1:]})
+    t.optimizer_step()
+    emb = float(t.model.embed_tokens.weight.detach().sum())
+    gu = t.model.layers[0].ffn.w_gate_up.detach().clone()
+    reset_mesh()
+    return {"emb": emb, "gu_sum": float(gu.sum()), "rank": rank}
+
+
+def test_ep_elastic_grow_world2(tmp_path):
+    """Single-process MoE run saved, resumed under EP(2): expert weights
+    AND moments slice across the new ranks; continuation matches the
+    single-process continuation."""
+    import torch as th
+    from luminaai_amd.config import Config
+    from luminaai_amd.data.tokenizer import ConversationTokenizer
+    from luminaai_amd.models import (DeepSeekTransformer,
+                                     config_to_deepseek_config)
+    from luminaai_amd.training import CheckpointManager, Trainer
+    os.environ["EP_ELASTIC_TMP"] = str(tmp_path)
+    os.environ.pop("WORLD_SIZE", None)
+    cfg = Config(vocab_size=512, hidden_size=64, num_layers=2, num_heads=4,
+                 num_kv_heads=2, seq_length=32, intermediate_size=128,
+                 micro_batch_size=4, gradient_accumulation_steps=1,
+                 num_workers=0, use_moe=True, num_experts=4, moe_top_k=2,
+                 routing_noise_std=0.0, use_mod=False,
+                 zero_stage=0, precision="fp32", experiment_name="epg_r",
+                 eval_every_n_batches=0, save_every_n_batches=0)
+    th.manual_seed(7)
+    model = DeepSeekTransformer(config_to_deepseek_config(cfg))
+    t = Trainer(model, ConversationTokenizer(), cfg)
+    t._setup_scheduler(10)
+    t.checkpoints = CheckpointManager(str(tmp_path / "grow_ckpts"))
+    gens = [th.Generator().manual_seed(860 + r) for r in range(2)]
+    for _ in range(2):
+        rows = [th.randint(1, cfg.vocab_size, (2, cfg.seq_length + 1),
+                           generator=g) for g in gens]
+        ids = th.cat(rows)
+        t.train_step({"input_ids": ids[:, :-1], "labels": ids[:, 1:]})
+        t.optimizer_step()
+    t.save_checkpoint(tag="grow_test")
+    # single-process continuation (the reference)
+    gens2 = [th.Generator().manual_seed(880 + r) for r in range(2)]
+    rows = [th.randint(1, cfg.vocab_size, (2, cfg.seq_length + 1),
+                       generator=g) for g in gens2]
+    ids = th.cat(rows)
+    t.train_step({"input_ids": ids[:, :-1], "labels": ids[:, 1:]})
+    t.optimizer_step()
+    emb_ref = float(t.model.embed_tokens.weight.detach().sum())
+    gu = t.model.layers[0].ffn.w_gate_up.detach()
+
+    res = _spawn("ep_elastic_grow_worker")
+    for r in range(2):
+        assert res[r]["emb"] == pytest.approx(emb_ref, abs=1e-3), \
+            (r, res[r]["emb"], emb_ref)
+        ref = float(gu[2 * r:2 * r + 2].sum())
+        assert res[r]["gu_sum"] == pytest.approx(ref, abs=1e-3), \
+            (r, res[r]["gu_sum"], ref)
+
+
 def tpep_worker(rank, world):
     """TP(2) x EP(2) composed mesh (world 4, dp 1): full Trainer steps.
     Ranks (e, t) hold expert shard e sliced by tp rank t; tp peers share a
