@@ -252,3 +252,46 @@ def test_source_processors_clean_each_format(tmp_path):
     stats = a.prepare_text_corpus(str(p), str(out), source="arxiv")
     assert stats["kept"] == 3
     assert out.read_text().count("Title: Paper") == 3
+
+
+def test_dataloader_rank_tp_sp_share_batches(monkeypatch):
+    """create_dataloader's data-rank math: SP and TP peers count as ONE
+    data rank (both innermost in the mesh layout); EP/DP ranks get
+    distinct shards."""
+    import torch.distributed as dist
+    from luminaai_amd.data import dataset as ds_mod
+
+    class _Mesh:
+        def __init__(self, sp, tp):
+            self.sp_size, self.tp_size = sp, tp
+
+    calls = {}
+
+    class _Sampler:
+        def __init__(self, dataset, num_replicas, rank, **kw):
+            calls["world"] = num_replicas
+            calls["rank"] = rank
+
+    monkeypatch.setattr(dist, "is_available", lambda: True)
+    monkeypatch.setattr(dist, "is_initialized", lambda: True)
+    monkeypatch.setattr(dist, "get_world_size", lambda: 8)
+    monkeypatch.setattr(dist, "get_rank", lambda: 5)
+    import torch.utils.data.distributed as tdd
+    monkeypatch.setattr(tdd, "DistributedSampler", _Sampler)
+    from luminaai_amd.config import Config
+    from luminaai_amd.parallel import mesh as mesh_mod
+    cfg = Config(vocab_size=512, hidden_size=64, num_layers=1, num_heads=4,
+                 num_kv_heads=2, seq_length=32, intermediate_size=128,
+                 micro_batch_size=1, num_workers=0)
+
+    class _DS(list):
+        pass
+
+    data = _DS([{"input_ids": None}] * 16)
+    for sp, tp, want_world, want_rank in ((1, 1, 8, 5), (1, 2, 4, 2),
+                                          (2, 1, 4, 2), (1, 4, 2, 1)):
+        monkeypatch.setattr(mesh_mod, "_MESH", _Mesh(sp, tp))
+        ds_mod.create_dataloader(data, cfg, shuffle=False)
+        assert (calls["world"], calls["rank"]) == (want_world, want_rank), \
+            (sp, tp, calls)
+    monkeypatch.setattr(mesh_mod, "_MESH", None)
